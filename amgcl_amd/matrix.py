@@ -1,0 +1,100 @@
+"""Host CSR matrix (fp64 values, int32 indices) and setup-phase algebra.
+
+The canonical host container mirroring the reference's backend::crs
+(amgcl/backend/builtin.hpp:61): raw ptr/col/val arrays, with transpose /
+product / diagonal implemented by the OpenMP C++ engine (_core).
+The AMG hierarchy is always assembled in this format on the host and then
+moved to a compute backend — the reference's core architectural invariant
+(docs/design.rst:20-38).
+"""
+import numpy as np
+
+from . import _core
+
+
+class CSR:
+    __slots__ = ("nrows", "ncols", "ptr", "col", "val")
+
+    def __init__(self, nrows, ncols, ptr, col, val):
+        self.nrows = int(nrows)
+        self.ncols = int(ncols)
+        self.ptr = np.ascontiguousarray(ptr, dtype=np.int32)
+        self.col = np.ascontiguousarray(col, dtype=np.int32)
+        self.val = np.ascontiguousarray(val, dtype=np.float64)
+
+    @property
+    def nnz(self):
+        return int(self.col.size)
+
+    @property
+    def shape(self):
+        return (self.nrows, self.ncols)
+
+    def bytes(self):
+        return self.ptr.nbytes + self.col.nbytes + self.val.nbytes
+
+    def diagonal(self):
+        return _core.diagonal(self.nrows, self.ptr, self.col, self.val)
+
+    def transpose(self):
+        tp, tc, tv = _core.transpose(self.nrows, self.ncols, self.ptr, self.col, self.val)
+        return CSR(self.ncols, self.nrows, tp, tc, tv)
+
+    def __matmul__(self, other):
+        if isinstance(other, CSR):
+            if self.ncols != other.nrows:
+                raise ValueError("dimension mismatch in CSR product")
+            cp, cc, cv = _core.spgemm(
+                self.nrows, self.ncols, other.ncols,
+                self.ptr, self.col, self.val,
+                other.ptr, other.col, other.val,
+            )
+            return CSR(self.nrows, other.ncols, cp, cc, cv)
+        x = np.ascontiguousarray(other, dtype=np.float64)
+        y = np.empty(self.nrows, dtype=np.float64)
+        _core.spmv(1.0, self.nrows, self.ptr, self.col, self.val, x, 0.0, y)
+        return y
+
+    def spmv(self, alpha, x, beta, y):
+        _core.spmv(alpha, self.nrows, self.ptr, self.col, self.val, x, beta, y)
+
+    def residual(self, b, x, r):
+        _core.residual(self.nrows, self.ptr, self.col, self.val, b, x, r)
+
+    def to_scipy(self):
+        import scipy.sparse as sp
+
+        return sp.csr_matrix((self.val, self.col, self.ptr), shape=self.shape)
+
+    @staticmethod
+    def from_scipy(m):
+        m = m.tocsr()
+        return CSR(m.shape[0], m.shape[1], m.indptr, m.indices, m.data)
+
+    @staticmethod
+    def from_dense(a):
+        a = np.asarray(a, dtype=np.float64)
+        nrows, ncols = a.shape
+        mask = a != 0.0
+        counts = mask.sum(axis=1)
+        ptr = np.zeros(nrows + 1, dtype=np.int64)
+        np.cumsum(counts, out=ptr[1:])
+        col = np.nonzero(mask)[1].astype(np.int32)
+        val = a[mask]
+        return CSR(nrows, ncols, ptr, col, val)
+
+    def to_dense(self):
+        a = np.zeros(self.shape, dtype=np.float64)
+        for i in range(self.nrows):
+            for j in range(self.ptr[i], self.ptr[i + 1]):
+                a[i, self.col[j]] = self.val[j]
+        return a
+
+    def __repr__(self):
+        return f"CSR({self.nrows}x{self.ncols}, nnz={self.nnz})"
+
+
+def galerkin(R, A, P):
+    """Coarse operator Ac = R*(A*P) via two SpGEMMs
+    (parity: amgcl/coarsening/detail/galerkin.hpp:42)."""
+    return R @ (A @ P)

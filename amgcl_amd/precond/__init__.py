@@ -1,0 +1,24 @@
+"""Preconditioners.
+
+Preconditioner concept (parity: amgcl/amg.hpp:289-306): constructed from the
+host system matrix + params + backend; .apply(rhs, x) approximately solves
+A x = rhs; .system_matrix() returns the backend matrix the Krylov solver
+iterates with.
+"""
+from .amg import AMG
+from .dummy import Dummy
+from .relaxation_precond import RelaxationPrecond
+
+REGISTRY = {
+    "amg": AMG,
+    "relaxation": RelaxationPrecond,
+    "dummy": Dummy,
+}
+
+
+def make_preconditioner(A, prm=None, backend=None):
+    prm = dict(prm or {})
+    kind = prm.pop("class", "amg")
+    if kind not in REGISTRY:
+        raise ValueError(f"unknown preconditioner class '{kind}'")
+    return REGISTRY[kind](A, prm, backend)

@@ -1,0 +1,170 @@
+"""The AMG hierarchy preconditioner.
+
+Parity: amgcl/amg.hpp:68-602 — levels built on the host (coarsening +
+Galerkin) until coarse_enough, each level's operators moved to the backend;
+V/W-cycle recursion with npre/npost smoothing, residual restriction, coarse
+correction prolongation, and a direct solve at the coarsest level.
+
+MI355X-native deltas from the reference:
+  - coarsest level solved by a precomputed dense inverse applied as a
+    device-resident GEMV (no D<->H round-trip; cf. backend/hip.hpp:73-96);
+  - the whole cycle is expressible as a fixed kernel sequence, enabling
+    hipGraph capture on the HIP backend (small-level launch overhead).
+"""
+import numpy as np
+
+from ..matrix import CSR
+from ..params import merge_params
+from ..coarsening import make_coarsening
+from ..relaxation import make_relaxation_factory
+
+
+class Level:
+    __slots__ = ("A", "P", "R", "f", "u", "t", "relax", "rows", "nnz", "A_host")
+
+    def __init__(self):
+        self.A = self.P = self.R = None
+        self.f = self.u = self.t = None
+        self.relax = None
+        self.rows = self.nnz = 0
+        self.A_host = None
+
+
+class AMG:
+    @staticmethod
+    def defaults():
+        return {
+            "coarsening": {"type": "smoothed_aggregation"},
+            "relax": {"type": "spai0"},
+            "coarse_enough": 3000,
+            "max_levels": 100,
+            "npre": 1,
+            "npost": 1,
+            "ncycle": 1,      # 1 = V-cycle, 2 = W-cycle
+            "pre_cycles": 1,
+            "direct_coarse": True,
+            "keep_host_matrices": False,
+        }
+
+    def __init__(self, A, prm=None, backend=None):
+        if backend is None:
+            from ..backend import make_backend
+
+            backend = make_backend("cpu")
+        self.backend = backend
+        self.prm = merge_params(self.defaults(), prm)
+        self.levels = []
+        self.coarse_solve = None
+        self._build(A)
+
+    # --- setup (host) ------------------------------------------------------
+    def _build(self, A: CSR):
+        prm = self.prm
+        coarsening = make_coarsening(prm["coarsening"])
+        relax_factory = make_relaxation_factory(prm["relax"])
+        backend = self.backend
+
+        coarse_enough = int(prm["coarse_enough"])
+        A_host = A
+
+        while True:
+            lvl = Level()
+            lvl.rows, lvl.nnz = A_host.nrows, A_host.nnz
+            if prm["keep_host_matrices"] or backend.name == "cpu":
+                lvl.A_host = A_host
+            lvl.A = backend.matrix(A_host)
+            last = (
+                A_host.nrows <= coarse_enough
+                or len(self.levels) + 1 >= int(prm["max_levels"])
+            )
+            if not last:
+                lvl.relax = relax_factory(A_host, backend)
+                lvl.f = backend.vector(A_host.nrows) if self.levels else None
+                lvl.u = backend.vector(A_host.nrows) if self.levels else None
+                lvl.t = backend.vector(A_host.nrows)
+                try:
+                    P, R = coarsening.transfer_operators(A_host)
+                except RuntimeError:
+                    # empty level (all nodes removed): stop coarsening here
+                    last = True
+                if not last:
+                    if P.ncols == 0:
+                        last = True
+            if last:
+                # coarsest level
+                lvl.f = backend.vector(A_host.nrows) if self.levels else None
+                lvl.u = backend.vector(A_host.nrows) if self.levels else None
+                lvl.t = backend.vector(A_host.nrows)
+                if prm["direct_coarse"]:
+                    self.coarse_solve = backend.coarse_solver(A_host)
+                else:
+                    lvl.relax = relax_factory(A_host, backend)
+                self.levels.append(lvl)
+                break
+
+            lvl.P = backend.matrix(P)
+            lvl.R = backend.matrix(R)
+            self.levels.append(lvl)
+            A_host = coarsening.coarse_operator(A_host, P, R)
+
+    # --- solve-phase -------------------------------------------------------
+    def system_matrix(self):
+        return self.levels[0].A
+
+    def cycle(self, i, f, u):
+        """One multigrid cycle at level i with rhs f and iterate u
+        (parity: amgcl/amg.hpp:514-553)."""
+        b = self.backend
+        lvl = self.levels[i]
+        prm = self.prm
+
+        if i + 1 == len(self.levels):
+            if self.coarse_solve is not None:
+                self.coarse_solve(f, u)
+            else:
+                for _ in range(int(prm["npre"])):
+                    lvl.relax.apply_pre(lvl.A, f, u, lvl.t)
+                for _ in range(int(prm["npost"])):
+                    lvl.relax.apply_post(lvl.A, f, u, lvl.t)
+            return
+
+        nxt = self.levels[i + 1]
+        for _ in range(int(prm["npre"])):
+            lvl.relax.apply_pre(lvl.A, f, u, lvl.t)
+        b.residual(f, lvl.A, u, lvl.t)
+        b.spmv(1.0, lvl.R, lvl.t, 0.0, nxt.f)
+        b.clear(nxt.u)
+        for _ in range(int(prm["ncycle"])):
+            self.cycle(i + 1, nxt.f, nxt.u)
+        b.spmv(1.0, lvl.P, nxt.u, 1.0, u)
+        for _ in range(int(prm["npost"])):
+            lvl.relax.apply_post(lvl.A, f, u, lvl.t)
+
+    def apply(self, rhs, x):
+        """x = M^-1 rhs via pre_cycles cycles from a zero initial guess
+        (parity: amgcl/amg.hpp:289-297)."""
+        b = self.backend
+        b.clear(x)
+        for _ in range(int(self.prm["pre_cycles"])):
+            self.cycle(0, rhs, x)
+
+    # --- observability (parity: amgcl/amg.hpp:561-598) ---------------------
+    def __str__(self):
+        lines = ["level     unknowns       nonzeros"]
+        total_rows = sum(l.rows for l in self.levels)
+        total_nnz = sum(l.nnz for l in self.levels)
+        for i, l in enumerate(self.levels):
+            pct = 100.0 * l.nnz / total_nnz if total_nnz else 0
+            lines.append(f"{i:5d} {l.rows:12d} {l.nnz:14d} ({pct:5.2f}%)")
+        oc = total_nnz / self.levels[0].nnz if self.levels[0].nnz else 0
+        gc = total_rows / self.levels[0].rows if self.levels[0].rows else 0
+        lines.append(f"operator complexity: {oc:.2f}")
+        lines.append(f"grid complexity:     {gc:.2f}")
+        return "\n".join(lines)
+
+    def bytes(self):
+        tot = 0
+        for l in self.levels:
+            if l.A_host is not None:
+                tot += l.A_host.bytes()
+        return tot

@@ -1,0 +1,110 @@
+"""Unit tests for the host setup engine (_core) against scipy references."""
+import numpy as np
+import pytest
+import scipy.sparse as sp
+
+import amgcl_amd as am
+from amgcl_amd import _core
+from amgcl_amd.matrix import CSR
+
+
+def rand_csr(rng, n, m, density=0.05):
+    a = sp.random(n, m, density=density, random_state=rng, format="csr")
+    a.data = rng.standard_normal(a.nnz)
+    return CSR(n, m, a.indptr, a.indices, a.data), a
+
+
+def test_poisson3d_matches_kron_construction():
+    n = 8
+    A, b = am.poisson3d(n)
+    # scipy reference: 3D Laplacian via kron sums, h=1
+    one = sp.identity(n)
+    t = sp.diags([-1, 2, -1], [-1, 0, 1], shape=(n, n))
+    ref = (
+        sp.kron(sp.kron(one, one), t)
+        + sp.kron(sp.kron(one, t), one)
+        + sp.kron(sp.kron(t, one), one)
+    ).tocsr()
+    ours = A.to_scipy()
+    assert (abs(ours - ref)).max() < 1e-14
+    assert b.shape == (n**3,)
+
+
+def test_transpose_matches_scipy():
+    rng = np.random.default_rng(0)
+    A, a = rand_csr(rng, 60, 37)
+    T = A.transpose()
+    diff = abs(T.to_scipy() - a.T.tocsr())
+    assert diff.nnz == 0 or diff.max() < 1e-14
+
+
+def test_spgemm_matches_scipy():
+    rng = np.random.default_rng(1)
+    A, a = rand_csr(rng, 40, 55)
+    B, b = rand_csr(rng, 55, 33)
+    C = A @ B
+    diff = abs(C.to_scipy() - (a @ b).tocsr())
+    assert diff.nnz == 0 or diff.max() < 1e-12
+    # rows sorted
+    for i in range(C.nrows):
+        row = C.col[C.ptr[i] : C.ptr[i + 1]]
+        assert np.all(np.diff(row) > 0)
+
+
+def test_spmv_residual_match_numpy():
+    rng = np.random.default_rng(2)
+    A, a = rand_csr(rng, 70, 70, 0.1)
+    x = rng.standard_normal(70)
+    y = rng.standard_normal(70)
+    y2 = y.copy()
+    A.spmv(1.5, x, 0.25, y2)
+    assert np.allclose(y2, 1.5 * (a @ x) + 0.25 * y)
+    r = np.empty(70)
+    A.residual(y, x, r)
+    assert np.allclose(r, y - a @ x)
+
+
+def test_aggregates_cover_all_connected_nodes():
+    A, _ = am.poisson3d(8)
+    naggr, ids, strong = _core.aggregates(A.nrows, A.ptr, A.col, A.val, 0.08)
+    ids = np.asarray(ids)
+    assert naggr > 0
+    assert ids.max() == naggr - 1
+    # Poisson has no Dirichlet-isolated rows -> every node aggregated
+    assert (ids >= 0).all()
+    # every aggregate non-empty
+    assert len(np.unique(ids)) == naggr
+
+
+def test_galerkin_matches_scipy_triple_product():
+    A, _ = am.poisson3d(6)
+    naggr, ids, strong = _core.aggregates(A.nrows, A.ptr, A.col, A.val, 0.08)
+    pp, pc, pv = _core.smoothed_prolongation(
+        A.nrows, A.ptr, A.col, A.val, strong, ids, naggr, 2.0 / 3.0
+    )
+    P = CSR(A.nrows, naggr, pp, pc, pv)
+    R = P.transpose()
+    Ac = am.galerkin(R, A, P)
+    ref = (P.to_scipy().T @ A.to_scipy() @ P.to_scipy()).tocsr()
+    diff = abs(Ac.to_scipy() - ref)
+    assert diff.nnz == 0 or diff.max() < 1e-12
+
+
+def test_ilu0_factorization_reproduces_dense_ilu():
+    A, _ = am.poisson3d(4)
+    lu, dia = _core.ilu0_factor(A.nrows, A.ptr, A.col, A.val)
+    # solve with it: must reduce residual substantially for diagonally dominant A
+    rng = np.random.default_rng(3)
+    b = rng.standard_normal(A.nrows)
+    z = b.copy()
+    _core.ilu0_solve(A.nrows, A.ptr, A.col, lu, dia, z)
+    r = b - A @ z
+    assert np.linalg.norm(r) < 0.5 * np.linalg.norm(b)
+
+
+def test_unknown_param_raises():
+    A, b = am.poisson3d(8)
+    with pytest.raises(Exception, match="unknown"):
+        am.make_solver(A, {"precond": {"class": "amg", "coarsening": {"type": "smoothed_aggregation", "bogus": 1}}})
+    with pytest.raises(Exception, match="unknown"):
+        am.make_solver(A, {"bogus": {}})

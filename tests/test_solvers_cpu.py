@@ -1,0 +1,116 @@
+"""Combinatorial convergence matrix on the CPU backend.
+
+Mirrors the reference's test strategy (tests/test_solver.hpp:111-210): loop
+coarsenings x smoothers x Krylov solvers through the runtime interface on a
+3D Poisson problem and assert the relative residual. Sizes are kept small so
+the suite runs in seconds on the CPU-only container.
+"""
+import numpy as np
+import pytest
+
+import amgcl_amd as am
+
+COARSENING = ["smoothed_aggregation", "aggregation"]
+RELAX = ["spai0", "damped_jacobi", "chebyshev", "gauss_seidel", "ilu0"]
+SOLVERS = ["cg", "bicgstab", "gmres", "richardson"]
+
+N = 16  # 4096 unknowns
+
+
+@pytest.fixture(scope="module")
+def problem():
+    A, b = am.poisson3d(N, rhs="random")
+    return A, b
+
+
+@pytest.mark.parametrize("coarsening", COARSENING)
+@pytest.mark.parametrize("relax", RELAX)
+@pytest.mark.parametrize("solver", SOLVERS)
+def test_convergence_matrix(problem, coarsening, relax, solver):
+    A, b = problem
+    maxiter = 300 if solver == "richardson" else 100
+    s = am.make_solver(
+        A,
+        {
+            "precond": {
+                "class": "amg",
+                "coarsening": {"type": coarsening},
+                "relax": {"type": relax},
+                "coarse_enough": 500,
+            },
+            "solver": {"type": solver, "tol": 1e-8, "maxiter": maxiter},
+        },
+    )
+    x, iters, resid = s(b)
+    assert resid < 1e-6, f"{coarsening}/{relax}/{solver}: resid={resid} iters={iters}"
+    r = b - A @ x
+    assert np.linalg.norm(r) / np.linalg.norm(b) < 1e-6
+    assert iters < maxiter
+
+
+@pytest.mark.parametrize("relax", RELAX)
+def test_smoother_as_preconditioner(problem, relax):
+    """Each smoother standalone (parity: tests/test_solver.hpp test_rap)."""
+    A, b = problem
+    s = am.make_solver(
+        A,
+        {
+            "precond": {"class": "relaxation", "type": relax},
+            "solver": {"type": "bicgstab", "tol": 1e-8, "maxiter": 1000},
+        },
+    )
+    x, iters, resid = s(b)
+    assert resid < 1e-6
+
+
+def test_cg_iteration_count_parity(problem):
+    """SA + SPAI0 + CG on Poisson must stay in the reference's iteration
+    class (reference: 12-24 its on comparable Poisson problems)."""
+    A, b = problem
+    s = am.make_solver(A, {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}})
+    x, iters, resid = s(b)
+    assert iters <= 30
+    assert resid < 1e-8
+
+
+def test_dummy_preconditioner(problem):
+    A, b = problem
+    s = am.make_solver(
+        A,
+        {
+            "precond": {"class": "dummy"},
+            "solver": {"type": "cg", "tol": 1e-8, "maxiter": 2000},
+        },
+    )
+    x, iters, resid = s(b)
+    assert resid < 1e-7
+
+
+def test_w_cycle_and_pre_post_counts(problem):
+    A, b = problem
+    s = am.make_solver(
+        A,
+        {
+            "precond": {"class": "amg", "ncycle": 2, "npre": 2, "npost": 2},
+            "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100},
+        },
+    )
+    x, iters, resid = s(b)
+    assert resid < 1e-8
+    assert iters <= 20
+
+
+def test_zero_rhs(problem):
+    A, _ = problem
+    s = am.make_solver(A, {"solver": {"type": "cg"}})
+    x, iters, resid = s(np.zeros(A.nrows))
+    assert iters == 0
+    assert np.all(np.asarray(x) == 0)
+
+
+def test_preonly_nested():
+    A, b = am.poisson3d(12, rhs="random")
+    s = am.make_solver(A, {"solver": {"type": "preonly"}})
+    x, iters, resid = s(b)
+    assert iters == 1
+    assert resid < 1.0
