@@ -1,0 +1,68 @@
+"""ctypes bindings to the hand-written gfx950 kernel library (libamghip.so).
+
+The library is pure HIP (no torch dependency); tensors cross the boundary as
+raw device pointers + the current torch HIP stream. Fails loudly if the
+library is missing on a GPU machine — GPU ops must never fall back silently
+to eager PyTorch.
+"""
+import ctypes
+import os
+
+_LIB = None
+
+_SIGS = {
+    "amg_spmv_f64": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_void_p, ctypes.c_void_p, ctypes.c_double, ctypes.c_double,
+                     ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+    "amg_residual_f64": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+                         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                         ctypes.c_int, ctypes.c_void_p],
+    "amg_relax_diag_f64": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+                           ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                           ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+    "amg_axpby_f64": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_double,
+                      ctypes.c_void_p, ctypes.c_void_p],
+    "amg_axpbypcz_f64": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_double,
+                         ctypes.c_void_p, ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p],
+    "amg_vmul_f64": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p],
+    "amg_fill_f64": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p],
+    "amg_dot_f64": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                    ctypes.c_void_p],
+    "amg_dot2_f64": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
+    "amg_gather_f64": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                       ctypes.c_void_p],
+    "amg_scatter_f64": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                        ctypes.c_void_p],
+    "amg_gemv_f64": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_void_p],
+}
+
+
+def lib():
+    global _LIB
+    if _LIB is None:
+        path = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                            "_hip", "libamghip.so")
+        if not os.path.exists(path):
+            # try to build (works even without a GPU: hipcc cross-compiles)
+            from ..build import build_hip_lib
+
+            build_hip_lib()
+        if not os.path.exists(path):
+            raise RuntimeError(
+                f"libamghip.so not found at {path}; run `python setup.py build_ext "
+                "--inplace` — GPU ops do not fall back to eager torch"
+            )
+        _LIB = ctypes.CDLL(path)
+        for name, argtypes in _SIGS.items():
+            fn = getattr(_LIB, name)
+            fn.argtypes = argtypes
+            fn.restype = ctypes.c_int
+    return _LIB
+
+
+def check(rc, what):
+    if rc != 0:
+        raise RuntimeError(f"HIP kernel '{what}' failed with hipError_t={rc}")

@@ -1,0 +1,201 @@
+#!/usr/bin/env python3
+"""Flagship benchmark (driver contract).
+
+Metric (BASELINE.json): setup+solve time (s) and iterations to rel. 1e-6 on a
+synthetic 7-point 3D Poisson system (512^3 default) with a random RHS,
+CG + smoothed-aggregation AMG + SPAI0, fp64, at 1/2/4/8 MI355X GPUs.
+
+One "step" = one full setup (host hierarchy assembly) + solve (GPU) cycle.
+Strong scaling: the 512^3 problem is partitioned in row blocks across ranks.
+
+Single GPU:  python bench.py [--size 512 --steps 1 --warmup 1]
+Multi GPU :  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                 --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+BASELINE_TOTAL_S = 2.03  # MN4 6144 cores: 0.68 setup + 1.35 solve
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=int(os.environ.get("WORLD_SIZE", "1")))
+    p.add_argument("--steps", type=int, default=1)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--size", type=int, default=512, help="grid edge n (n^3 unknowns)")
+    p.add_argument("--tol", type=float, default=1e-6)
+    p.add_argument("--solver", default="cg")
+    p.add_argument("--maxiter", type=int, default=300)
+    p.add_argument("--backend", default=None, choices=[None, "hip", "cpu"])
+    p.add_argument("--precond", default=None, help="override precond JSON")
+    return p.parse_args()
+
+
+def one_step(args, dist_ctx):
+    """Full setup+solve; returns (elapsed_s, setup_s, solve_s, iters, resid, true_rel)."""
+    import numpy as np
+
+    import amgcl_amd as am
+
+    t0 = time.perf_counter()
+    if dist_ctx is None:
+        A, b = am.poisson3d(args.size, rhs="random")
+        backend = args.backend or ("hip" if _has_gpu() else "cpu")
+        prm = {
+            "precond": json.loads(args.precond) if args.precond else {"class": "amg"},
+            "solver": {"type": args.solver, "tol": args.tol, "maxiter": args.maxiter},
+        }
+        solve = am.make_solver(A, prm, backend=backend)
+        t1 = time.perf_counter()
+        solve.backend.synchronize()
+        x, iters, resid = solve(b)
+        solve.backend.synchronize()
+        t2 = time.perf_counter()
+        xh = solve.backend.to_host(x)
+        true_rel = float(np.linalg.norm(b - A @ xh) / np.linalg.norm(b))
+    else:
+        t1, t2, iters, resid, true_rel = run_distributed(args, dist_ctx, t0)
+    return t2 - t0, t1 - t0, t2 - t1, iters, resid, true_rel
+
+
+def run_distributed(args, dist_ctx, t0):
+    import numpy as np
+    import torch
+    import torch.distributed as dist
+
+    import amgcl_amd as am
+    from amgcl_amd.parallel import DistBackend, make_dist_solver
+
+    rank, world = dist_ctx
+    A_strip, _, row_beg, row_end = am.poisson3d_strip(args.size, rank, world, rhs=None)
+    rng = np.random.default_rng(42)
+    b_global = rng.standard_normal(args.size**3)
+    b = b_global[row_beg:row_end]
+
+    backend_name = args.backend or ("hip" if _has_gpu() else "cpu")
+    solve = make_dist_solver(
+        A_strip,
+        {
+            "precond": json.loads(args.precond) if args.precond else {"class": "amg"},
+            "solver": {"type": args.solver, "tol": args.tol, "maxiter": args.maxiter},
+        },
+        backend=backend_name,
+    )
+    t1 = time.perf_counter()
+    solve.backend.synchronize()
+    dist.barrier()
+    x, iters, resid = solve(b)
+    solve.backend.synchronize()
+    dist.barrier()
+    t2 = time.perf_counter()
+    # true residual check on the strip
+    xh = solve.gather_solution(x)
+    if xh is not None:
+        A_full, _ = am.poisson3d(args.size)
+        true_rel = float(
+            np.linalg.norm(b_global - A_full @ xh) / np.linalg.norm(b_global)
+        )
+    else:
+        true_rel = -1.0
+    return t1, t2, iters, resid, true_rel
+
+
+def _has_gpu():
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    dist_ctx = None
+    if world > 1:
+        import torch
+        import torch.distributed as dist
+
+        rank = int(os.environ["RANK"])
+        local_rank = int(os.environ.get("LOCAL_RANK", rank))
+        if _has_gpu():
+            torch.cuda.set_device(local_rank)
+            dist.init_process_group("nccl")
+        else:
+            dist.init_process_group("gloo")
+        dist_ctx = (rank, world)
+
+    rank = dist_ctx[0] if dist_ctx else 0
+
+    for _ in range(args.warmup):
+        one_step(args, dist_ctx)
+
+    if dist_ctx:
+        import torch.distributed as dist
+
+        dist.barrier()
+    times, setups, solves, iters_l, resid, true_rel = [], [], [], [], 0.0, 0.0
+    t_start = time.perf_counter()
+    for _ in range(args.steps):
+        el, setup_s, solve_s, iters, resid, true_rel = one_step(args, dist_ctx)
+        times.append(el)
+        setups.append(setup_s)
+        solves.append(solve_s)
+        iters_l.append(iters)
+    total = time.perf_counter() - t_start
+
+    if dist_ctx:
+        import torch
+        import torch.distributed as dist
+
+        t = torch.tensor([total], dtype=torch.float64,
+                         device="cuda" if _has_gpu() else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        total = float(t.item())
+
+    if rank == 0:
+        value = total / args.steps
+        out = {
+            "metric": f"setup+solve time (s) & iterations to 1e-6, 3D Poisson {args.size}^3",
+            "value": value,
+            "unit": "s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": value * 1000.0,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": value / BASELINE_TOTAL_S,
+            "dtype": "fp64",
+            "data": "synthetic 7-pt Poisson, random RHS",
+            "config": {
+                "model": f"3D Poisson {args.size}^3 (7-point stencil)",
+                "unknowns": args.size**3,
+                "solver": args.solver,
+                "precond": "smoothed_aggregation AMG + SPAI0",
+                "tol": args.tol,
+                "iters": iters_l[-1],
+                "setup_s": setups[-1],
+                "solve_s": solves[-1],
+                "final_rel_resid": resid,
+                "true_rel_resid": true_rel,
+                "parallelism": f"row-block dd x{world}" if world > 1 else "single GPU",
+            },
+        }
+        print(json.dumps(out))
+
+    if dist_ctx:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
