@@ -28,9 +28,14 @@ class SmoothedAggregation:
         self.eps_strong = float(self.prm["eps_strong"])
 
     def transfer_operators(self, A: CSR):
-        naggr, aggr_id, strong = _core.aggregates(
-            A.nrows, A.ptr, A.col, A.val, self.eps_strong
-        )
+        from ..profiler import prof
+
+        with prof.scope("aggregates"):
+            # large levels: deterministic parallel MIS(2) aggregation
+            # (mirrors the reference's own distributed pmis design);
+            # small levels: the exact greedy reference pass.
+            agg = _core.aggregates_parallel if A.nrows > 100_000 else _core.aggregates
+            naggr, aggr_id, strong = agg(A.nrows, A.ptr, A.col, A.val, self.eps_strong)
         self.eps_strong *= 0.5  # reference halves eps per level (s_a.hpp:140)
 
         omega = float(self.prm["relax"])
@@ -39,11 +44,13 @@ class SmoothedAggregation:
         else:
             omega *= 2.0 / 3.0
 
-        pp, pc, pv = _core.smoothed_prolongation(
-            A.nrows, A.ptr, A.col, A.val, strong, aggr_id, naggr, omega
-        )
-        P = CSR(A.nrows, naggr, pp, pc, pv)
-        R = P.transpose()
+        with prof.scope("smooth_P"):
+            pp, pc, pv = _core.smoothed_prolongation(
+                A.nrows, A.ptr, A.col, A.val, strong, aggr_id, naggr, omega
+            )
+            P = CSR(A.nrows, naggr, pp, pc, pv)
+        with prof.scope("transpose_R"):
+            R = P.transpose()
         return P, R
 
     def coarse_operator(self, A, P, R):

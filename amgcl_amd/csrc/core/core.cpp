@@ -43,6 +43,8 @@ struct CsrView {
     const double *val;
 };
 
+static void scan_i32(i32 *a, i64 n);
+
 static CsrView view(i64 nrows, i64 ncols, const arr<i32> &ptr, const arr<i32> &col,
                     const arr<double> &val) {
     if (ptr.size() != nrows + 1) throw std::runtime_error("bad ptr size");
@@ -70,8 +72,7 @@ static py::tuple poisson3d(i64 n, double anisotropy) {
         int cnt = 1 + (k > 0) + (j > 0) + (i > 0) + (i + 1 < n) + (j + 1 < n) + (k + 1 < n);
         P[idx + 1] = cnt;
     }
-    // exclusive scan (serial; 134M adds is fast)
-    for (i64 i = 0; i < n3; ++i) P[i + 1] += P[i];
+    scan_i32(P + 1, n3);
     const i64 nnz = P[n3];
     if (nnz >= (i64(1) << 31)) throw std::runtime_error("nnz exceeds int32");
 
@@ -128,7 +129,7 @@ static py::tuple transpose(i64 nrows, i64 ncols, arr<i32> ptr, arr<i32> col, arr
 #pragma omp atomic
         TP[A.col[j] + 1]++;
     }
-    for (i64 c = 0; c < ncols; ++c) TP[c + 1] += TP[c];
+    scan_i32(TP + 1, ncols);
 
     arr<i32> tcol(nnz);
     arr<double> tval(nnz);
@@ -193,9 +194,13 @@ static py::tuple spgemm(i64 an, i64 am, i64 bm,
             CP[i + 1] = cnt;
         }
     }
-    i64 total = 0;
-    for (i64 i = 0; i < an; ++i) { total += CP[i + 1]; CP[i + 1] = (i32)total; }
-    if (total >= (i64(1) << 31)) throw std::runtime_error("spgemm: nnz exceeds int32");
+    {
+        i64 chk = 0;
+        for (i64 i = 0; i < an; ++i) chk += CP[i + 1];
+        if (chk >= (i64(1) << 31)) throw std::runtime_error("spgemm: nnz exceeds int32");
+    }
+    scan_i32(CP + 1, an);
+    const i64 total = CP[an];
 
     arr<i32> ccol(total);
     arr<double> cval(total);
@@ -235,6 +240,233 @@ static py::tuple spgemm(i64 an, i64 am, i64 bm,
         }
     }
     return py::make_tuple(cptr, ccol, cval);
+}
+
+// parallel inclusive scan over an i32 array (blocked 3-phase)
+static void scan_i32(i32 *a, i64 n) {
+    if (n < (1 << 16)) {
+        for (i64 i = 1; i < n; ++i) a[i] += a[i - 1];
+        return;
+    }
+    int nt = omp_get_max_threads();
+    std::vector<i64> sums(nt + 1, 0);
+#pragma omp parallel num_threads(nt)
+    {
+        int t = omp_get_thread_num();
+        i64 b = n * t / nt, e = n * (t + 1) / nt;
+        i64 s = 0;
+        for (i64 i = b; i < e; ++i) { s += a[i]; }
+        sums[t + 1] = s;
+#pragma omp barrier
+#pragma omp single
+        for (int k = 1; k <= nt; ++k) sums[k] += sums[k - 1];
+        i64 off = sums[t];
+        for (i64 i = b; i < e; ++i) { off += a[i]; a[i] = (i32)off; }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Parallel aggregation: distance-2 maximal-independent-set roots with 1-ring
+// claiming and leftover adoption. Parallel counterpart of the reference's
+// greedy pass, mirroring its own distributed PMIS design
+// (amgcl/mpi/coarsening/pmis.hpp:50 — deterministic random tie-breaking).
+// Deterministic regardless of thread count (hash-keyed competition).
+// ---------------------------------------------------------------------------
+static inline uint64_t agg_key(i32 i) {
+    uint32_t x = (uint32_t)i;
+    x ^= x >> 16; x *= 0x7feb352dU; x ^= x >> 15; x *= 0x846ca68bU; x ^= x >> 16;
+    return ((uint64_t)x << 32) | (uint32_t)i;
+}
+
+static inline void atomic_max_u64(std::atomic<uint64_t> &a, uint64_t v) {
+    uint64_t cur = a.load(std::memory_order_relaxed);
+    while (cur < v && !a.compare_exchange_weak(cur, v, std::memory_order_relaxed)) {
+    }
+}
+
+static py::tuple aggregates_parallel(i64 nrows, arr<i32> ptr, arr<i32> col,
+                                     arr<double> val, double eps_strong) {
+    auto A = view(nrows, nrows, ptr, col, val);
+    const i64 nnz = (i64)col.size();
+    const double eps2 = eps_strong * eps_strong;
+
+    arr<double> dia_a = diagonal(nrows, ptr, col, val);
+    const double *D = dia_a.data();
+
+    arr<uint8_t> strong(nnz);
+    uint8_t *S = strong.mutable_data();
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < nrows; ++i) {
+        double edi = eps2 * D[i];
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i32 c = A.col[j];
+            double v = A.val[j];
+            S[j] = (c != (i32)i) && (edi * D[c] < v * v);
+        }
+    }
+
+    // States mirror the greedy reference: UNDEF competes to seed; FIRM nodes
+    // are roots or 1-ring members; PROVISIONAL nodes (2nd ring, adjacent to
+    // an aggregate) hold an adoptive id, never seed, and may be re-claimed
+    // by a later root's 1-ring — exactly the greedy pass's semantics, made
+    // parallel via distance-2 MIS root selection with hashed keys.
+    constexpr i32 UNDEF = -1, REMOVED = -2;
+    arr<i32> id_a(nrows);
+    i32 *id = id_a.mutable_data();  // UNDEF / REMOVED / root-node index
+    std::vector<uint8_t> prov(nrows, 0);
+    std::vector<std::atomic<uint64_t>> m1(nrows);
+
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < nrows; ++i) {
+        m1[i].store(0, std::memory_order_relaxed);
+        i32 st = REMOVED;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j)
+            if (S[j]) { st = UNDEF; break; }
+        id[i] = st;
+    }
+
+    std::vector<i32> active;
+    active.reserve(nrows);
+    for (i64 i = 0; i < nrows; ++i)
+        if (id[i] == UNDEF) active.push_back((i32)i);
+
+    std::vector<i32> touched, next_active;
+    touched.reserve(active.size());
+
+    bool m1_dirty_all = false;
+    while (!active.empty()) {
+        const i64 na = (i64)active.size();
+        if (na > nrows / 16) {
+            // pass 1, dense gather (no atomics): one sweep over all rows
+#pragma omp parallel for schedule(static)
+            for (i64 i = 0; i < nrows; ++i) {
+                uint64_t m = (id[i] == UNDEF) ? agg_key((i32)i) : 0;
+                for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+                    if (!S[j]) continue;
+                    i32 c = A.col[j];
+                    if (id[c] == UNDEF) {
+                        uint64_t k = agg_key(c);
+                        if (k > m) m = k;
+                    }
+                }
+                m1[i].store(m, std::memory_order_relaxed);
+            }
+            m1_dirty_all = true;
+            touched.clear();
+        } else {
+            // clear stale m1, then sparse scatter from the active set
+            if (m1_dirty_all) {
+#pragma omp parallel for schedule(static)
+                for (i64 i = 0; i < nrows; ++i)
+                    m1[i].store(0, std::memory_order_relaxed);
+                m1_dirty_all = false;
+                touched.clear();
+            } else {
+#pragma omp parallel for schedule(static)
+                for (i64 t = 0; t < (i64)touched.size(); ++t)
+                    m1[touched[t]].store(0, std::memory_order_relaxed);
+                touched.clear();
+            }
+#pragma omp parallel
+            {
+                std::vector<i32> local_touched;
+#pragma omp for schedule(static) nowait
+                for (i64 t = 0; t < na; ++t) {
+                    i32 u = active[t];
+                    uint64_t k = agg_key(u);
+                    atomic_max_u64(m1[u], k);
+                    local_touched.push_back(u);
+                    for (i32 j = A.ptr[u]; j < A.ptr[u + 1]; ++j) {
+                        if (!S[j]) continue;
+                        i32 c = A.col[j];
+                        atomic_max_u64(m1[c], k);
+                        local_touched.push_back(c);
+                    }
+                }
+#pragma omp critical
+                touched.insert(touched.end(), local_touched.begin(), local_touched.end());
+            }
+        }
+
+        // pass 2: u is a root iff its key is the max within distance 2 of
+        // the UNDEF subgraph (propagated through any intermediate node).
+#pragma omp parallel for schedule(static)
+        for (i64 t = 0; t < na; ++t) {
+            i32 u = active[t];
+            uint64_t key = agg_key(u);
+            uint64_t m2 = m1[u].load(std::memory_order_relaxed);
+            for (i32 j = A.ptr[u]; j < A.ptr[u + 1] && m2 <= key; ++j) {
+                if (!S[j]) continue;
+                uint64_t v = m1[A.col[j]].load(std::memory_order_relaxed);
+                if (v > m2) m2 = v;
+            }
+            if (m2 == key) id[u] = u;  // new root (FIRM)
+        }
+
+        // pass 3: new roots claim their strong 1-ring FIRM, overwriting
+        // provisional members (greedy: "later claimed by other aggregates").
+        // No conflicts: two new roots are never within distance 2.
+#pragma omp parallel for schedule(static)
+        for (i64 t = 0; t < na; ++t) {
+            i32 u = active[t];
+            if (id[u] != u) continue;
+            for (i32 j = A.ptr[u]; j < A.ptr[u + 1]; ++j) {
+                i32 c = A.col[j];
+                if (!S[j]) continue;
+                if (id[c] == UNDEF || prov[c]) {
+                    id[c] = u;
+                    prov[c] = 0;
+                }
+            }
+        }
+
+        // pass 4: remaining UNDEF nodes adjacent to a FIRM member become
+        // provisional members of that aggregate (greedy 2nd-ring claim);
+        // deterministic choice: the strong neighbor with max root key wins.
+        next_active.clear();
+#pragma omp parallel
+        {
+            std::vector<i32> local_next;
+#pragma omp for schedule(static) nowait
+            for (i64 t = 0; t < na; ++t) {
+                i32 u = active[t];
+                if (id[u] != UNDEF) continue;
+                uint64_t best = 0;
+                i32 root = -1;
+                for (i32 j = A.ptr[u]; j < A.ptr[u + 1]; ++j) {
+                    i32 c = A.col[j];
+                    if (!S[j] || c == u) continue;
+                    if (id[c] >= 0 && !prov[c]) {
+                        uint64_t k = agg_key(id[c]);
+                        if (k > best) { best = k; root = id[c]; }
+                    }
+                }
+                if (root >= 0) {
+                    id[u] = root;
+                    prov[u] = 1;
+                } else {
+                    local_next.push_back(u);
+                }
+            }
+#pragma omp critical
+            next_active.insert(next_active.end(), local_next.begin(), local_next.end());
+        }
+        std::swap(active, next_active);
+    }
+
+    // renumber root nodes to compact aggregate ids (deterministic: node order)
+    std::vector<i32> mark(nrows, 0);
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < nrows; ++i)
+        if (id[i] == (i32)i) mark[i] = 1;
+    scan_i32(mark.data(), nrows);  // inclusive
+    i64 count = mark[nrows - 1];
+    if (!count) throw std::runtime_error("empty level in aggregation");
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < nrows; ++i)
+        if (id[i] >= 0) id[i] = mark[id[i]] - 1;
+
+    return py::make_tuple((i64)count, id_a, strong);
 }
 
 // ---------------------------------------------------------------------------
@@ -544,6 +776,7 @@ PYBIND11_MODULE(_core, m) {
     m.def("transpose", &transpose);
     m.def("spgemm", &spgemm);
     m.def("aggregates", &aggregates);
+    m.def("aggregates_parallel", &aggregates_parallel);
     m.def("smoothed_prolongation", &smoothed_prolongation);
     m.def("tentative_prolongation", &tentative_prolongation);
     m.def("spai0", &spai0);

@@ -232,18 +232,24 @@ extern "C" int amg_fill_f64(int64_t n, double v, double *x, hipStream_t stream) 
 // ---------------------------------------------------------------------------
 __global__ void dot_k(int64_t n, const double *__restrict__ x, const double *__restrict__ y,
                       double *__restrict__ out) {
+    __shared__ double lds[4];  // 256 threads = 4 waves
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     double s = 0.0;
     for (; i < n; i += stride) s += x[i] * y[i];
 #pragma unroll
     for (int off = WAVE / 2; off > 0; off >>= 1) s += __shfl_down(s, off, WAVE);
-    if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, s);
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
+    if (lane == 0) lds[wid] = s;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd(out, lds[0] + lds[1] + lds[2] + lds[3]);
 }
 
 __global__ void dot2_k(int64_t n, const double *__restrict__ x1, const double *__restrict__ y1,
                        const double *__restrict__ x2, const double *__restrict__ y2,
                        double *__restrict__ out) {
+    __shared__ double lds[8];
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     double s1 = 0.0, s2 = 0.0;
@@ -256,16 +262,22 @@ __global__ void dot2_k(int64_t n, const double *__restrict__ x1, const double *_
         s1 += __shfl_down(s1, off, WAVE);
         s2 += __shfl_down(s2, off, WAVE);
     }
-    if ((threadIdx.x & (WAVE - 1)) == 0) {
-        atomicAdd(out, s1);
-        atomicAdd(out + 1, s2);
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
+    if (lane == 0) {
+        lds[wid] = s1;
+        lds[wid + 4] = s2;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        atomicAdd(out, lds[0] + lds[1] + lds[2] + lds[3]);
+        atomicAdd(out + 1, lds[4] + lds[5] + lds[6] + lds[7]);
     }
 }
 
 extern "C" int amg_dot_f64(int64_t n, const double *x, const double *y, double *out,
                            hipStream_t stream) {
     fill_k<<<1, 64, 0, stream>>>(1, 0.0, out);
-    dot_k<<<nblocks(n), 256, 0, stream>>>(n, x, y, out);
+    dot_k<<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x, y, out);
     return (int)hipGetLastError();
 }
 
@@ -273,7 +285,7 @@ extern "C" int amg_dot2_f64(int64_t n, const double *x1, const double *y1,
                             const double *x2, const double *y2, double *out,
                             hipStream_t stream) {
     fill_k<<<1, 64, 0, stream>>>(2, 0.0, out);
-    dot2_k<<<nblocks(n), 256, 0, stream>>>(n, x1, y1, x2, y2, out);
+    dot2_k<<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x1, y1, x2, y2, out);
     return (int)hipGetLastError();
 }
 

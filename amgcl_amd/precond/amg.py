@@ -66,24 +66,28 @@ class AMG:
 
         coarse_enough = int(prm["coarse_enough"])
         A_host = A
+        from ..profiler import prof
 
         while True:
             lvl = Level()
             lvl.rows, lvl.nnz = A_host.nrows, A_host.nnz
             if prm["keep_host_matrices"] or backend.name == "cpu":
                 lvl.A_host = A_host
-            lvl.A = backend.matrix(A_host)
+            with prof.scope("move_to_backend"):
+                lvl.A = backend.matrix(A_host)
             last = (
                 A_host.nrows <= coarse_enough
                 or len(self.levels) + 1 >= int(prm["max_levels"])
             )
             if not last:
-                lvl.relax = relax_factory(A_host, backend)
+                with prof.scope("relax_setup"):
+                    lvl.relax = relax_factory(A_host, backend)
                 lvl.f = backend.vector(A_host.nrows) if self.levels else None
                 lvl.u = backend.vector(A_host.nrows) if self.levels else None
                 lvl.t = backend.vector(A_host.nrows)
                 try:
-                    P, R = coarsening.transfer_operators(A_host)
+                    with prof.scope("transfer_operators"):
+                        P, R = coarsening.transfer_operators(A_host)
                 except RuntimeError:
                     # empty level (all nodes removed): stop coarsening here
                     last = True
@@ -96,16 +100,19 @@ class AMG:
                 lvl.u = backend.vector(A_host.nrows) if self.levels else None
                 lvl.t = backend.vector(A_host.nrows)
                 if prm["direct_coarse"]:
-                    self.coarse_solve = backend.coarse_solver(A_host)
+                    with prof.scope("coarse_solver"):
+                        self.coarse_solve = backend.coarse_solver(A_host)
                 else:
                     lvl.relax = relax_factory(A_host, backend)
                 self.levels.append(lvl)
                 break
 
-            lvl.P = backend.matrix(P)
-            lvl.R = backend.matrix(R)
+            with prof.scope("move_to_backend"):
+                lvl.P = backend.matrix(P)
+                lvl.R = backend.matrix(R)
             self.levels.append(lvl)
-            A_host = coarsening.coarse_operator(A_host, P, R)
+            with prof.scope("galerkin"):
+                A_host = coarsening.coarse_operator(A_host, P, R)
 
     # --- solve-phase -------------------------------------------------------
     def system_matrix(self):
