@@ -168,6 +168,69 @@ static py::tuple transpose(i64 nrows, i64 ncols, arr<i32> ptr, arr<i32> col, arr
 // Two passes: symbolic row sizes, then numeric fill; per-thread marker array.
 // Output rows sorted.
 // ---------------------------------------------------------------------------
+// Per-thread open-addressing accumulator for one sparse row at a time.
+// Fixed KB-scale footprint (L2-resident) instead of the classic O(ncols)
+// marker array — the marker costs 256 threads x ncols x 4B of first-touch
+// page faults on a 2-socket EPYC and was the dominant setup cost at 512^3.
+struct RowHash {
+    std::vector<i32> keys;
+    std::vector<double> vals;
+    std::vector<i32> used;
+    uint32_t mask;
+
+    explicit RowHash(int cap_pow2 = 1 << 11)
+        : keys(cap_pow2, -1), vals(cap_pow2, 0.0), mask(cap_pow2 - 1) {
+        used.reserve(cap_pow2);
+    }
+
+    void ensure(i64 row_ub) {
+        // keep load factor <= 1/2
+        if ((i64)(mask + 1) < 2 * row_ub) {
+            i64 cap = mask + 1;
+            while (cap < 2 * row_ub) cap *= 2;
+            keys.assign(cap, -1);
+            vals.assign(cap, 0.0);
+            mask = (uint32_t)cap - 1;
+            used.clear();
+        }
+    }
+
+    inline uint32_t slot(i32 key) const {
+        uint32_t h = ((uint32_t)key * 2654435761u) & mask;
+        while (keys[h] != key && keys[h] != -1) h = (h + 1) & mask;
+        return h;
+    }
+
+    inline void add(i32 key, double v) {
+        uint32_t h = slot(key);
+        if (keys[h] == -1) {
+            keys[h] = key;
+            vals[h] = v;
+            used.push_back((i32)h);
+        } else {
+            vals[h] += v;
+        }
+    }
+
+    inline int count_add(i32 key) {  // symbolic: returns 1 if new
+        uint32_t h = slot(key);
+        if (keys[h] == -1) {
+            keys[h] = key;
+            used.push_back((i32)h);
+            return 1;
+        }
+        return 0;
+    }
+
+    inline void reset() {
+        for (i32 h : used) {
+            keys[h] = -1;
+            vals[h] = 0.0;
+        }
+        used.clear();
+    }
+};
+
 static py::tuple spgemm(i64 an, i64 am, i64 bm,
                         arr<i32> aptr, arr<i32> acol, arr<double> aval,
                         arr<i32> bptr, arr<i32> bcol, arr<double> bval) {
@@ -178,18 +241,40 @@ static py::tuple spgemm(i64 an, i64 am, i64 bm,
     i32 *CP = cptr.mutable_data();
     CP[0] = 0;
 
+    // dense per-thread markers are faster but cost nthreads*bm*4B of
+    // first-touch pages; switch to per-row hash accumulators when that
+    // footprint would thrash (the 256-core EPYC regime).
+    const bool use_marker =
+        (i64)omp_get_max_threads() * bm * 4 < (i64(1) << 29);
+
 #pragma omp parallel
     {
-        std::vector<i32> marker(bm, -1);
+        RowHash hash;
+        std::vector<i32> marker(use_marker ? bm : 0, -1);
 #pragma omp for schedule(dynamic, 4096)
         for (i64 i = 0; i < an; ++i) {
             i32 cnt = 0;
-            for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
-                i32 ca = A.col[ja];
-                for (i32 jb = B.ptr[ca]; jb < B.ptr[ca + 1]; ++jb) {
-                    i32 cb = B.col[jb];
-                    if (marker[cb] != (i32)i) { marker[cb] = (i32)i; ++cnt; }
+            if (use_marker) {
+                for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
+                    i32 ca = A.col[ja];
+                    for (i32 jb = B.ptr[ca]; jb < B.ptr[ca + 1]; ++jb) {
+                        i32 cb = B.col[jb];
+                        if (marker[cb] != (i32)i) { marker[cb] = (i32)i; ++cnt; }
+                    }
                 }
+            } else {
+                i64 ub = 0;
+                for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
+                    i32 ca = A.col[ja];
+                    ub += B.ptr[ca + 1] - B.ptr[ca];
+                }
+                hash.ensure(ub);
+                for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
+                    i32 ca = A.col[ja];
+                    for (i32 jb = B.ptr[ca]; jb < B.ptr[ca + 1]; ++jb)
+                        cnt += hash.count_add(B.col[jb]);
+                }
+                hash.reset();
             }
             CP[i + 1] = cnt;
         }
@@ -209,26 +294,50 @@ static py::tuple spgemm(i64 an, i64 am, i64 bm,
 
 #pragma omp parallel
     {
-        std::vector<i32> marker(bm, -1);
+        RowHash hash;
+        std::vector<i32> marker(use_marker ? bm : 0, -1);
+        std::vector<i32> marker_row(use_marker ? bm : 0, -1);
 #pragma omp for schedule(dynamic, 4096)
         for (i64 i = 0; i < an; ++i) {
             const i32 row_beg = CP[i];
             i32 row_end = row_beg;
-            for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
-                i32 ca = A.col[ja];
-                double va = A.val[ja];
-                for (i32 jb = B.ptr[ca]; jb < B.ptr[ca + 1]; ++jb) {
-                    i32 cb = B.col[jb];
-                    double vb = B.val[jb];
-                    if (marker[cb] < row_beg) {
-                        marker[cb] = row_end;
-                        CC[row_end] = cb;
-                        CV[row_end] = va * vb;
-                        ++row_end;
-                    } else {
-                        CV[marker[cb]] += va * vb;
+            if (use_marker) {
+                for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
+                    i32 ca = A.col[ja];
+                    double va = A.val[ja];
+                    for (i32 jb = B.ptr[ca]; jb < B.ptr[ca + 1]; ++jb) {
+                        i32 cb = B.col[jb];
+                        double vb = B.val[jb];
+                        if (marker_row[cb] != (i32)i) {
+                            marker_row[cb] = (i32)i;
+                            marker[cb] = row_end;
+                            CC[row_end] = cb;
+                            CV[row_end] = va * vb;
+                            ++row_end;
+                        } else {
+                            CV[marker[cb]] += va * vb;
+                        }
                     }
                 }
+            } else {
+                i64 ub = 0;
+                for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
+                    i32 ca = A.col[ja];
+                    ub += B.ptr[ca + 1] - B.ptr[ca];
+                }
+                hash.ensure(ub);
+                for (i32 ja = A.ptr[i]; ja < A.ptr[i + 1]; ++ja) {
+                    i32 ca = A.col[ja];
+                    double va = A.val[ja];
+                    for (i32 jb = B.ptr[ca]; jb < B.ptr[ca + 1]; ++jb)
+                        hash.add(B.col[jb], va * B.val[jb]);
+                }
+                for (i32 h : hash.used) {
+                    CC[row_end] = hash.keys[h];
+                    CV[row_end] = hash.vals[h];
+                    ++row_end;
+                }
+                hash.reset();
             }
             // sort the (short) row
             for (i32 k = row_beg + 1; k < row_end; ++k) {
@@ -563,21 +672,24 @@ static py::tuple smoothed_prolongation(i64 nrows, arr<i32> ptr, arr<i32> col, ar
 
 #pragma omp parallel
     {
-        std::vector<i32> marker(naggr, -1);
+        RowHash hash(256);
 #pragma omp for schedule(static)
         for (i64 i = 0; i < nrows; ++i) {
+            i64 row_len = A.ptr[i + 1] - A.ptr[i];
+            hash.ensure(row_len);
             i32 cnt = 0;
             for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
                 i32 c = A.col[j];
                 if (c != (i32)i && !S[j]) continue;  // weak off-diagonal: skipped
                 i32 a = id[c];
-                if (a >= 0 && marker[a] != (i32)i) { marker[a] = (i32)i; ++cnt; }
+                if (a >= 0) cnt += hash.count_add(a);
             }
+            hash.reset();
             PP[i + 1] = cnt;
         }
     }
-    i64 total = 0;
-    for (i64 i = 0; i < nrows; ++i) { total += PP[i + 1]; PP[i + 1] = (i32)total; }
+    scan_i32(PP + 1, nrows);
+    const i64 total = PP[nrows];
 
     arr<i32> pcol(total);
     arr<double> pval(total);
@@ -586,7 +698,7 @@ static py::tuple smoothed_prolongation(i64 nrows, arr<i32> ptr, arr<i32> col, ar
 
 #pragma omp parallel
     {
-        std::vector<i32> marker(naggr, -1);
+        RowHash hash(256);
 #pragma omp for schedule(static)
         for (i64 i = 0; i < nrows; ++i) {
             // filtered diagonal: original diagonal plus weak off-diagonal values
@@ -595,23 +707,23 @@ static py::tuple smoothed_prolongation(i64 nrows, arr<i32> ptr, arr<i32> col, ar
                 if (A.col[j] == (i32)i || !S[j]) dia += A.val[j];
             if (dia != 0.0) dia = -omega / dia;
 
-            const i32 row_beg = PP[i];
-            i32 row_end = row_beg;
+            hash.ensure(A.ptr[i + 1] - A.ptr[i]);
             for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
                 i32 c = A.col[j];
                 if (c != (i32)i && !S[j]) continue;
                 i32 a = id[c];
                 if (a < 0) continue;
                 double v = (c == (i32)i) ? (1.0 - omega) : dia * A.val[j];
-                if (marker[a] < row_beg) {
-                    marker[a] = row_end;
-                    PC[row_end] = a;
-                    PV[row_end] = v;
-                    ++row_end;
-                } else {
-                    PV[marker[a]] += v;
-                }
+                hash.add(a, v);
             }
+            const i32 row_beg = PP[i];
+            i32 row_end = row_beg;
+            for (i32 h : hash.used) {
+                PC[row_end] = hash.keys[h];
+                PV[row_end] = hash.vals[h];
+                ++row_end;
+            }
+            hash.reset();
             for (i32 k = row_beg + 1; k < row_end; ++k) {
                 i32 ck = PC[k]; double vk = PV[k];
                 i32 m = k;

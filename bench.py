@@ -56,6 +56,11 @@ def one_step(args, dist_ctx):
         solve = am.make_solver(A, prm, backend=backend)
         solve.backend.synchronize()
         t1 = time.perf_counter()
+        if os.environ.get("AMGCL_PROFILE"):
+            from amgcl_amd.profiler import prof
+
+            print(prof.report(), file=sys.stderr)
+            print(solve.P, file=sys.stderr)
         x, iters, resid = solve(b)
         solve.backend.synchronize()
         t2 = time.perf_counter()
