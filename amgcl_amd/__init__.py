@@ -19,6 +19,23 @@ import os as _os
 # passive waiting is the right default for this call pattern.
 _os.environ.setdefault("OMP_WAIT_POLICY", "PASSIVE")
 
+
+def _effective_cpus():
+    """nproc capped by the cgroup CPU quota (GPU boxes expose 256 CPUs but
+    cap the container at a much smaller cpu.max; oversubscribed OpenMP
+    threads then thrash against CFS throttling)."""
+    n = _os.cpu_count() or 1
+    try:
+        quota, period = open("/sys/fs/cgroup/cpu.max").read().split()
+        if quota != "max":
+            n = min(n, max(1, int(int(quota) / int(period))))
+    except (OSError, ValueError):
+        pass
+    return n
+
+
+_os.environ.setdefault("OMP_NUM_THREADS", str(_effective_cpus()))
+
 from . import build as _build_mod
 
 try:

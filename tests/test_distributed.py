@@ -1,0 +1,118 @@
+"""Multi-process distributed tests (gloo backend, CPU, world_size 2/4).
+
+Replaces the reference's mpirun-based example checks
+(examples/mpi/check_direct.cpp, runtime_sdd.cpp) with spawn-based
+torch.distributed tests that run on the CPU-only container.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch.multiprocessing as mp
+
+
+def _run_dist(rank, world, fn, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        out = fn(rank, world)
+        results[rank] = out
+    finally:
+        dist.destroy_process_group()
+
+
+def spawn(world, fn, port):
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        results = mgr.dict()
+        procs = [
+            ctx.Process(target=_run_dist, args=(r, world, fn, port, results))
+            for r in range(world)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(300)
+        for p in procs:
+            assert p.exitcode == 0, f"rank process failed: {p.exitcode}"
+        return dict(results)
+
+
+def _solve_poisson(rank, world):
+    import amgcl_amd as am
+    from amgcl_amd.parallel import make_dist_solver
+
+    n = 20
+    strip, b, row_beg, row_end = am.poisson3d_strip(n, rank, world, rhs="ones")
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "amg", "coarse_enough": 200},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 200}},
+        backend="cpu",
+    )
+    x, iters, resid = solve(b)
+    xg = solve.gather_solution(x)
+    return iters, resid, None if xg is None else xg.tolist()
+
+
+def _spmv_check(rank, world):
+    import amgcl_amd as am
+    from amgcl_amd.parallel import DistBackend
+    from amgcl_amd.backend import make_backend
+
+    n = 12
+    strip, _, row_beg, row_end = am.poisson3d_strip(n, rank, world)
+    backend = DistBackend(make_backend("cpu"))
+    A = backend.matrix(strip)
+    rng = np.random.default_rng(5)
+    x_global = rng.standard_normal(n**3)
+    x = backend.from_host(x_global[row_beg:row_end])
+    y = backend.vector(row_end - row_beg)
+    backend.spmv(1.0, A, x, 0.0, y)
+    return row_beg, row_end, backend.to_host(y).tolist()
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_dist_spmv_matches_serial(world):
+    results = spawn(world, _spmv_check, 29511 + world)
+    import amgcl_amd as am
+
+    n = 12
+    A, _ = am.poisson3d(n)
+    rng = np.random.default_rng(5)
+    x = rng.standard_normal(n**3)
+    ref = A @ x
+    for rank, (rb, re_, y) in results.items():
+        np.testing.assert_allclose(np.array(y), ref[rb:re_], rtol=1e-13, atol=1e-13)
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_solve_converges_and_matches(world):
+    results = spawn(world, _solve_poisson, 29611 + world)
+    import amgcl_amd as am
+
+    n = 20
+    A, b = am.poisson3d(n, rhs="ones")
+    iters, resid, xg = results[0]
+    assert resid < 1e-8
+    assert iters < 120
+    xg = np.array(xg)
+    r = b - A @ xg
+    assert np.linalg.norm(r) / np.linalg.norm(b) < 1e-7
+
+
+def test_dist_inner_product_single_rank_passthrough():
+    # world_size=1 path: DistMatrix with no neighbors must behave like local
+    # (exercised via bench --gpus 1; here just the mask split logic)
+    import amgcl_amd as am
+    from amgcl_amd.matrix import CSR
+
+    n = 8
+    strip, b, rb, re_ = am.poisson3d_strip(n, 0, 2)
+    assert strip.nrows == n**3 // 2
+    assert strip.ncols == n**3
