@@ -37,6 +37,15 @@ class DeviceCSR:
         self.val = torch.from_numpy(np.asarray(csr.val)).to(device)
         self.subw = subw  # 0 = auto by mean row length
 
+    @classmethod
+    def from_tensors(cls, nrows, ncols, ptr, col, val, subw=0):
+        self = cls.__new__(cls)
+        self.nrows = int(nrows)
+        self.ncols = int(ncols)
+        self.ptr, self.col, self.val = ptr, col, val
+        self.subw = subw
+        return self
+
     @property
     def nnz(self):
         return self.col.numel()
@@ -61,7 +70,9 @@ class HipBackend:
         self._dothost = torch.zeros(2, dtype=torch.float64, pin_memory=True)
 
     # --- containers -------------------------------------------------------
-    def matrix(self, csr: CSR):
+    def matrix(self, csr):
+        if isinstance(csr, DeviceCSR):
+            return csr
         return DeviceCSR(csr, self.device)
 
     def vector(self, n):
@@ -132,7 +143,11 @@ class HipBackend:
                                     x.data_ptr(), _stream()), "scatter")
 
     # --- coarse direct solver ---------------------------------------------
-    def coarse_solver(self, csr: CSR):
+    def coarse_solver(self, csr):
+        if isinstance(csr, DeviceCSR):
+            from . import hip_setup
+
+            csr = hip_setup.download(csr)
         return DeviceDenseSolver(csr, self)
 
     def synchronize(self):

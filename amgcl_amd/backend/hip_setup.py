@@ -1,0 +1,217 @@
+"""Device-side AMG setup orchestration.
+
+Python driver for the gfx950 setup kernels (csrc/hip/setup.hip): the whole
+smoothed-aggregation hierarchy — strong connections, parallel aggregation,
+prolongation smoothing, R = P^T, and the Galerkin triple product — is built
+in device memory, eliminating the reference's host-assembly + per-level H2D
+upload design (amgcl/amg.hpp:467-512, backend/hip.hpp:382-409), which is the
+dominant cost on CPU-quota'd GPU nodes.
+
+The algorithms are the exact device twins of the host engine
+(csrc/core/core.cpp), with identical deterministic aggregation keys, so the
+device-built hierarchy matches the host-built one level for level.
+"""
+import numpy as np
+
+from ..matrix import CSR
+from ._hiplib import check, lib
+
+
+def _stream():
+    import torch
+
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _torch():
+    import torch
+
+    return torch
+
+
+class _Dev:
+    """Lazily-allocated workspace shared across levels."""
+
+    def __init__(self, device):
+        self.device = device
+        t = _torch()
+        self.flag = t.zeros(2, dtype=t.int32, device=device)  # overflow/remaining
+
+
+def device_csr(nrows, ncols, ptr, col, val, subw=0):
+    from .hip import DeviceCSR
+
+    return DeviceCSR.from_tensors(nrows, ncols, ptr, col, val, subw)
+
+
+def _new_ptr(n, device):
+    t = _torch()
+    return t.zeros(n + 1, dtype=t.int32, device=device)
+
+
+def _scan_ptr(ptr):
+    """Inclusive scan of counts stored in ptr[1:]; ptr[0] stays 0."""
+    n = ptr.numel() - 1
+    check(lib().amg_scan_i32(ptr[1:].data_ptr(), n, _stream()), "scan")
+    return ptr
+
+
+def diagonal(A):
+    t = _torch()
+    d = t.empty(A.nrows, dtype=t.float64, device=A.val.device)
+    check(lib().amg_setup_diag(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
+                               A.val.data_ptr(), d.data_ptr(), _stream()), "diag")
+    return d
+
+
+def spai0(A):
+    t = _torch()
+    m = t.empty(A.nrows, dtype=t.float64, device=A.val.device)
+    check(lib().amg_setup_spai0(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
+                                A.val.data_ptr(), m.data_ptr(), _stream()), "spai0")
+    return m
+
+
+def aggregates(A, eps_strong):
+    """Device twin of _core.aggregates_parallel: returns (naggr, id, strong)."""
+    t = _torch()
+    dev = A.val.device
+    n = A.nrows
+    d = diagonal(A)
+    S = t.empty(A.nnz, dtype=t.uint8, device=dev)
+    check(lib().amg_setup_strong(n, A.ptr.data_ptr(), A.col.data_ptr(), A.val.data_ptr(),
+                                 d.data_ptr(), float(eps_strong) ** 2, S.data_ptr(),
+                                 _stream()), "strong")
+    ids = t.empty(n, dtype=t.int32, device=dev)
+    check(lib().amg_agg_init(n, A.ptr.data_ptr(), S.data_ptr(), ids.data_ptr(),
+                             _stream()), "agg_init")
+    prov = t.zeros(n, dtype=t.uint8, device=dev)
+    m1 = t.empty(n, dtype=t.int64, device=dev)
+    newroot = t.empty(n, dtype=t.uint8, device=dev)
+    remaining = t.zeros(1, dtype=t.int32, device=dev)
+    for _round in range(64):
+        remaining.zero_()
+        check(lib().amg_agg_round(n, A.ptr.data_ptr(), A.col.data_ptr(), S.data_ptr(),
+                                  ids.data_ptr(), prov.data_ptr(), m1.data_ptr(),
+                                  newroot.data_ptr(), remaining.data_ptr(), _stream()),
+              "agg_round")
+        if int(remaining.item()) == 0:
+            break
+    else:
+        raise RuntimeError("device aggregation did not converge")
+    mark = t.empty(n, dtype=t.int32, device=dev)
+    check(lib().amg_agg_renumber(n, ids.data_ptr(), mark.data_ptr(), _stream()),
+          "agg_renumber")
+    naggr = int(mark[-1].item())
+    if naggr == 0:
+        raise RuntimeError("empty level in aggregation")
+    return naggr, ids, S
+
+
+def smoothed_prolongation(A, S, ids, naggr, omega):
+    t = _torch()
+    dev = A.val.device
+    n = A.nrows
+    pptr = _new_ptr(n, dev)
+    overflow = t.zeros(1, dtype=t.int32, device=dev)
+    check(lib().amg_psmooth_count(n, A.ptr.data_ptr(), A.col.data_ptr(), S.data_ptr(),
+                                  ids.data_ptr(), pptr[1:].data_ptr(),
+                                  overflow.data_ptr(), _stream()), "psmooth_count")
+    if int(overflow.item()) != 0:
+        raise OverflowError("P row exceeds device buffer; use host setup")
+    _scan_ptr(pptr)
+    nnz = int(pptr[-1].item())
+    pcol = t.empty(nnz, dtype=t.int32, device=dev)
+    pval = t.empty(nnz, dtype=t.float64, device=dev)
+    check(lib().amg_psmooth_fill(n, A.ptr.data_ptr(), A.col.data_ptr(), A.val.data_ptr(),
+                                 S.data_ptr(), ids.data_ptr(), float(omega),
+                                 pptr[1:].data_ptr(), pcol.data_ptr(), pval.data_ptr(),
+                                 _stream()), "psmooth_fill")
+    return device_csr(n, naggr, pptr, pcol, pval)
+
+
+def tentative_prolongation(A, ids, naggr):
+    t = _torch()
+    dev = A.val.device
+    n = A.nrows
+    pptr = _new_ptr(n, dev)
+    check(lib().amg_ptent_count(n, ids.data_ptr(), pptr[1:].data_ptr(), _stream()),
+          "ptent_count")
+    _scan_ptr(pptr)
+    nnz = int(pptr[-1].item())
+    pcol = t.empty(nnz, dtype=t.int32, device=dev)
+    pval = t.empty(nnz, dtype=t.float64, device=dev)
+    check(lib().amg_ptent_fill(n, ids.data_ptr(), pptr[1:].data_ptr(), pcol.data_ptr(),
+                               pval.data_ptr(), _stream()), "ptent_fill")
+    return device_csr(n, naggr, pptr, pcol, pval)
+
+
+def transpose(A):
+    t = _torch()
+    dev = A.val.device
+    tptr = _new_ptr(A.ncols, dev)
+    check(lib().amg_transpose_count(A.nnz, A.col.data_ptr(), tptr[1:].data_ptr(),
+                                    _stream()), "tcount")
+    _scan_ptr(tptr)
+    cursor = tptr[:-1].clone()
+    tcol = t.empty(A.nnz, dtype=t.int32, device=dev)
+    tval = t.empty(A.nnz, dtype=t.float64, device=dev)
+    check(lib().amg_transpose_scatter(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
+                                      A.val.data_ptr(), cursor.data_ptr(),
+                                      tcol.data_ptr(), tval.data_ptr(), _stream()),
+          "tscatter")
+    check(lib().amg_sort_rows(A.ncols, tptr[1:].data_ptr(), tcol.data_ptr(),
+                              tval.data_ptr(), _stream()), "sort_rows")
+    return device_csr(A.ncols, A.nrows, tptr, tcol, tval)
+
+
+def spgemm(A, B):
+    t = _torch()
+    dev = A.val.device
+    cptr = _new_ptr(A.nrows, dev)
+    overflow = t.zeros(1, dtype=t.int32, device=dev)
+    check(lib().amg_spgemm_count(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
+                                 B.ptr.data_ptr(), B.col.data_ptr(),
+                                 cptr[1:].data_ptr(), overflow.data_ptr(), _stream()),
+          "spgemm_count")
+    if int(overflow.item()) != 0:
+        raise OverflowError("spgemm row exceeds LDS hash; use host setup")
+    _scan_ptr(cptr)
+    nnz = int(cptr[-1].item())
+    ccol = t.empty(nnz, dtype=t.int32, device=dev)
+    cval = t.empty(nnz, dtype=t.float64, device=dev)
+    check(lib().amg_spgemm_fill(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
+                                A.val.data_ptr(), B.ptr.data_ptr(), B.col.data_ptr(),
+                                B.val.data_ptr(), cptr[1:].data_ptr(), ccol.data_ptr(),
+                                cval.data_ptr(), _stream()), "spgemm_fill")
+    return device_csr(A.nrows, B.ncols, cptr, ccol, cval)
+
+
+def download(A):
+    """Device CSR -> host CSR (sorted rows)."""
+    ptr = A.ptr.cpu().numpy()
+    col = A.col.cpu().numpy()
+    val = A.val.cpu().numpy()
+    csr = CSR(A.nrows, A.ncols, ptr, col, val)
+    # rows may be unsorted (spgemm extraction order): sort on host
+    for i in range(csr.nrows):
+        lo, hi = csr.ptr[i], csr.ptr[i + 1]
+        order = np.argsort(csr.col[lo:hi], kind="stable")
+        csr.col[lo:hi] = csr.col[lo:hi][order]
+        csr.val[lo:hi] = csr.val[lo:hi][order]
+    return csr
+
+
+def poisson3d_device(n, device="cuda"):
+    """7-point Poisson fixture generated directly on the GPU."""
+    t = _torch()
+    n3 = n**3
+    ptr = _new_ptr(n3, device)
+    check(lib().amg_poisson_cnt(n, ptr[1:].data_ptr(), _stream()), "poisson_cnt")
+    _scan_ptr(ptr)
+    nnz = int(ptr[-1].item())
+    col = t.empty(nnz, dtype=t.int32, device=device)
+    val = t.empty(nnz, dtype=t.float64, device=device)
+    check(lib().amg_poisson_fill(n, ptr[1:].data_ptr(), col.data_ptr(), val.data_ptr(),
+                                 _stream()), "poisson_fill")
+    return device_csr(n3, n3, ptr, col, val)

@@ -18,7 +18,14 @@ class Aggregation:
         self.prm = merge_params(self.defaults(), prm)
         self.eps_strong = float(self.prm["eps_strong"])
 
-    def transfer_operators(self, A: CSR):
+    def transfer_operators(self, A):
+        if not isinstance(A, CSR):
+            from ..backend import hip_setup
+
+            naggr, ids, _strong = hip_setup.aggregates(A, self.eps_strong)
+            self.eps_strong *= 0.5
+            P = hip_setup.tentative_prolongation(A, ids, naggr)
+            return P, hip_setup.transpose(P)
         naggr, aggr_id, _strong = _core.aggregates(
             A.nrows, A.ptr, A.col, A.val, self.eps_strong
         )

@@ -27,9 +27,11 @@ class SmoothedAggregation:
         self.prm = merge_params(self.defaults(), prm)
         self.eps_strong = float(self.prm["eps_strong"])
 
-    def transfer_operators(self, A: CSR):
+    def transfer_operators(self, A):
         from ..profiler import prof
 
+        if not isinstance(A, CSR):
+            return self._transfer_operators_device(A)
         with prof.scope("aggregates"):
             # large levels: deterministic parallel MIS(2) aggregation
             # (mirrors the reference's own distributed pmis design);
@@ -51,6 +53,21 @@ class SmoothedAggregation:
             P = CSR(A.nrows, naggr, pp, pc, pv)
         with prof.scope("transpose_R"):
             R = P.transpose()
+        return P, R
+
+    def _transfer_operators_device(self, A):
+        """Device twin (backend/hip_setup.py): same algorithm, same keys."""
+        from ..backend import hip_setup
+        from ..profiler import prof
+
+        with prof.scope("aggregates(dev)"):
+            naggr, ids, strong = hip_setup.aggregates(A, self.eps_strong)
+        self.eps_strong *= 0.5
+        omega = float(self.prm["relax"]) * (2.0 / 3.0)
+        with prof.scope("smooth_P(dev)"):
+            P = hip_setup.smoothed_prolongation(A, strong, ids, naggr, omega)
+        with prof.scope("transpose_R(dev)"):
+            R = hip_setup.transpose(P)
         return P, R
 
     def coarse_operator(self, A, P, R):

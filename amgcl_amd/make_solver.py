@@ -68,7 +68,9 @@ def make_solver(A, prm=None, backend="cpu", **backend_kwargs):
     A may be a CSR, a scipy sparse matrix, or a (ptr, col, val) tuple
     (the crs_tuple adapter, cf. amgcl/adapter/crs_tuple.hpp:83).
     """
-    if not isinstance(A, CSR):
+    if type(A).__name__ == "DeviceCSR":
+        pass  # device-resident input: setup runs on the GPU
+    elif not isinstance(A, CSR):
         if isinstance(A, tuple) and len(A) in (3, 4):
             if len(A) == 4:
                 n, ptr, col, val = A

@@ -96,5 +96,13 @@ class CSR:
 
 def galerkin(R, A, P):
     """Coarse operator Ac = R*(A*P) via two SpGEMMs
-    (parity: amgcl/coarsening/detail/galerkin.hpp:42)."""
+    (parity: amgcl/coarsening/detail/galerkin.hpp:42).
+    Runs on the device (backend/hip_setup.py spgemm) when the operands are
+    device-resident."""
+    if not isinstance(A, CSR):
+        from .backend import hip_setup
+        from .profiler import prof
+
+        with prof.scope("galerkin(dev)"):
+            return hip_setup.spgemm(R, hip_setup.spgemm(A, P))
     return R @ (A @ P)

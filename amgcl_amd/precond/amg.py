@@ -113,6 +113,13 @@ class AMG:
             self.levels.append(lvl)
             with prof.scope("galerkin"):
                 A_host = coarsening.coarse_operator(A_host, P, R)
+            # device-setup path: hand small coarse levels back to the host
+            # engine (tiny kernels are launch-bound; the host path also
+            # supports every smoother/coarse solver)
+            if not isinstance(A_host, CSR) and A_host.nrows <= 20000:
+                from ..backend import hip_setup
+
+                A_host = hip_setup.download(A_host)
 
     # --- solve-phase -------------------------------------------------------
     def system_matrix(self):

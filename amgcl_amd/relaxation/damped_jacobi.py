@@ -16,7 +16,13 @@ class DampedJacobi(DiagonalSmootherBase):
         return {"damping": 0.72}
 
     def __init__(self, A, prm, backend):
+        from ..matrix import CSR
+
         p = merge_params(self.defaults(), prm)
-        d = A.diagonal()
-        m = float(p["damping"]) / np.asarray(d)
+        if isinstance(A, CSR):
+            m = float(p["damping"]) / np.asarray(A.diagonal())
+        else:
+            from ..backend import hip_setup
+
+            m = float(p["damping"]) / hip_setup.diagonal(A)
         self._setup_m(m, backend)

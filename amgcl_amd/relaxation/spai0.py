@@ -17,7 +17,9 @@ class DiagonalSmootherBase:
 
     def _setup_m(self, m_host, backend):
         self.backend = backend
-        self.M = backend.from_host(m_host)
+        import numpy as np
+
+        self.M = backend.from_host(m_host) if isinstance(m_host, np.ndarray) else m_host
         self._fused = hasattr(backend, "relax_diag")
 
     def _step(self, A, rhs, x, tmp):
@@ -47,6 +49,13 @@ class Spai0(DiagonalSmootherBase):
         return {}
 
     def __init__(self, A, prm, backend):
+        from ..matrix import CSR
+
         merge_params(self.defaults(), prm)
-        m = _core.spai0(A.nrows, A.ptr, A.col, A.val)
+        if isinstance(A, CSR):
+            m = _core.spai0(A.nrows, A.ptr, A.col, A.val)
+        else:  # device-resident level matrix: compute weights on device
+            from ..backend import hip_setup
+
+            m = hip_setup.spai0(A)
         self._setup_m(m, backend)

@@ -1,0 +1,147 @@
+"""Device-side setup engine vs the host engine (equivalence tests).
+
+The device kernels implement the exact same deterministic algorithms as the
+host engine (same aggregation hash keys), so hierarchies must match
+structurally; SpGEMM values match to reduction-order roundoff.
+"""
+import numpy as np
+import pytest
+
+import amgcl_amd as am
+from amgcl_amd import _core
+from amgcl_amd.matrix import CSR
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def hip():
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from amgcl_amd.backend import make_backend
+
+    return make_backend("hip")
+
+
+def to_scipy_dev(D):
+    import scipy.sparse as sp
+
+    m = sp.csr_matrix(
+        (D.val.cpu().numpy(), D.col.cpu().numpy(), D.ptr.cpu().numpy()),
+        shape=(D.nrows, D.ncols),
+    )
+    m.sort_indices()
+    return m
+
+
+def test_poisson_device_matches_host(hip):
+    from amgcl_amd.backend.hip_setup import poisson3d_device
+
+    n = 12
+    Ad = poisson3d_device(n)
+    A, _ = am.poisson3d(n)
+    diff = abs(to_scipy_dev(Ad) - A.to_scipy())
+    assert diff.nnz == 0 or diff.max() < 1e-14
+
+
+def test_device_aggregates_match_host_parallel(hip):
+    from amgcl_amd.backend import hip_setup
+    from amgcl_amd.backend.hip import DeviceCSR
+
+    A, _ = am.poisson3d(16)
+    naggr_h, id_h, strong_h = _core.aggregates_parallel(
+        A.nrows, A.ptr, A.col, A.val, 0.08
+    )
+    Ad = DeviceCSR(A, hip.device)
+    naggr_d, id_d, strong_d = hip_setup.aggregates(Ad, 0.08)
+    assert naggr_d == naggr_h
+    np.testing.assert_array_equal(id_d.cpu().numpy(), np.asarray(id_h))
+    np.testing.assert_array_equal(strong_d.cpu().numpy(), np.asarray(strong_h))
+
+
+def test_device_prolongation_and_galerkin_match_host(hip):
+    from amgcl_amd.backend import hip_setup
+    from amgcl_amd.backend.hip import DeviceCSR
+    from amgcl_amd.matrix import galerkin
+
+    A, _ = am.poisson3d(12)
+    Ad = DeviceCSR(A, hip.device)
+    naggr, ids, strong = hip_setup.aggregates(Ad, 0.08)
+    P = hip_setup.smoothed_prolongation(Ad, strong, ids, naggr, 2.0 / 3.0)
+    ph, pc, pv = _core.smoothed_prolongation(
+        A.nrows, A.ptr, A.col, A.val,
+        strong.cpu().numpy(), ids.cpu().numpy().astype(np.int32), naggr, 2.0 / 3.0,
+    )
+    P_h = CSR(A.nrows, naggr, ph, pc, pv)
+    diff = abs(to_scipy_dev(P) - P_h.to_scipy())
+    assert diff.nnz == 0 or diff.max() < 1e-13
+
+    R = hip_setup.transpose(P)
+    diff = abs(to_scipy_dev(R) - P_h.transpose().to_scipy())
+    assert diff.nnz == 0 or diff.max() < 1e-13
+
+    Ac = galerkin(R, Ad, P)
+    Ac_h = P_h.transpose() @ (A @ P_h)
+    d = to_scipy_dev(Ac) - Ac_h.to_scipy()
+    assert abs(d).max() < 1e-11
+
+
+def test_device_spgemm_random(hip):
+    import scipy.sparse as sp
+
+    from amgcl_amd.backend import hip_setup
+    from amgcl_amd.backend.hip import DeviceCSR
+
+    rng = np.random.default_rng(3)
+    a = sp.random(300, 200, density=0.05, random_state=rng, format="csr")
+    a.data = rng.standard_normal(a.nnz)
+    b = sp.random(200, 250, density=0.05, random_state=rng, format="csr")
+    b.data = rng.standard_normal(b.nnz)
+    Ad = DeviceCSR(CSR.from_scipy(a), hip.device)
+    Bd = DeviceCSR(CSR.from_scipy(b), hip.device)
+    Cd = hip_setup.spgemm(Ad, Bd)
+    diff = abs(to_scipy_dev(Cd) - (a @ b).tocsr())
+    assert diff.nnz == 0 or diff.max() < 1e-12
+
+
+def test_full_device_setup_solve(hip):
+    """End-to-end: device-generated A, device setup, device solve."""
+    import torch
+
+    from amgcl_amd.backend.hip_setup import poisson3d_device
+
+    n = 48
+    A = poisson3d_device(n)
+    g = torch.Generator(device="cuda").manual_seed(7)
+    b = torch.randn(n**3, dtype=torch.float64, device="cuda", generator=g)
+    solve = am.make_solver(
+        A, {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}}, backend=hip
+    )
+    x, iters, resid = solve(b)
+    assert resid < 1e-8
+    assert iters <= 30
+    # true residual via tested kernels
+    r = hip.vector(n**3)
+    hip.residual(b, solve.system_matrix(), x, r)
+    import math
+
+    rel = math.sqrt(hip.dot(r, r)) / math.sqrt(hip.dot(b, b))
+    assert rel < 1e-7
+
+
+def test_device_setup_iteration_parity_with_host(hip):
+    """Device-built and host-built hierarchies give identical iterations."""
+    A, b = am.poisson3d(32, rhs="random")
+    prm = {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}}
+    s_host = am.make_solver(A, prm, backend=hip)
+    x1, it_host, _ = s_host(b)
+
+    from amgcl_amd.backend.hip import DeviceCSR
+
+    Ad = DeviceCSR(A, hip.device)
+    s_dev = am.make_solver(Ad, prm, backend=hip)
+    x2, it_dev, _ = s_dev(b)
+    # host greedy vs device MIS aggregation may differ slightly
+    assert abs(it_dev - it_host) <= 3
