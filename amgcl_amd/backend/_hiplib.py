@@ -44,7 +44,7 @@ _SIGS = {
     "amg_setup_spai0": [ctypes.c_int64] + [ctypes.c_void_p] * 5,
     "amg_scan_i32": [ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p],
     "amg_agg_init": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
-    "amg_agg_round": [ctypes.c_int64] + [ctypes.c_void_p] * 9,
+    "amg_agg_round": [ctypes.c_int64] + [ctypes.c_void_p] * 10,
     "amg_agg_renumber": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     "amg_psmooth_count": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
     "amg_psmooth_fill": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
@@ -55,8 +55,8 @@ _SIGS = {
     "amg_transpose_count": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     "amg_transpose_scatter": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
     "amg_sort_rows": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
-    "amg_spgemm_count": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
-    "amg_spgemm_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 10,
+    "amg_spgemm_count": [ctypes.c_int64] + [ctypes.c_void_p] * 8,
+    "amg_spgemm_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 11,
     "amg_poisson_cnt": [ctypes.c_int64] + [ctypes.c_void_p] * 2,
     "amg_poisson_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
 }

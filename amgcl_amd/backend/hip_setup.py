@@ -88,12 +88,14 @@ def aggregates(A, eps_strong):
     prov = t.zeros(n, dtype=t.uint8, device=dev)
     m1 = t.empty(n, dtype=t.int64, device=dev)
     newroot = t.empty(n, dtype=t.uint8, device=dev)
+    near = t.empty(n, dtype=t.uint8, device=dev)
     remaining = t.zeros(1, dtype=t.int32, device=dev)
     for _round in range(64):
         remaining.zero_()
         check(lib().amg_agg_round(n, A.ptr.data_ptr(), A.col.data_ptr(), S.data_ptr(),
                                   ids.data_ptr(), prov.data_ptr(), m1.data_ptr(),
-                                  newroot.data_ptr(), remaining.data_ptr(), _stream()),
+                                  newroot.data_ptr(), near.data_ptr(),
+                                  remaining.data_ptr(), _stream()),
               "agg_round")
         if int(remaining.item()) == 0:
             break
@@ -169,9 +171,10 @@ def spgemm(A, B):
     t = _torch()
     dev = A.val.device
     cptr = _new_ptr(A.nrows, dev)
+    ub = t.empty(A.nrows, dtype=t.int32, device=dev)
     overflow = t.zeros(1, dtype=t.int32, device=dev)
     check(lib().amg_spgemm_count(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
-                                 B.ptr.data_ptr(), B.col.data_ptr(),
+                                 B.ptr.data_ptr(), B.col.data_ptr(), ub.data_ptr(),
                                  cptr[1:].data_ptr(), overflow.data_ptr(), _stream()),
           "spgemm_count")
     if int(overflow.item()) != 0:
@@ -182,24 +185,22 @@ def spgemm(A, B):
     cval = t.empty(nnz, dtype=t.float64, device=dev)
     check(lib().amg_spgemm_fill(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
                                 A.val.data_ptr(), B.ptr.data_ptr(), B.col.data_ptr(),
-                                B.val.data_ptr(), cptr[1:].data_ptr(), ccol.data_ptr(),
-                                cval.data_ptr(), _stream()), "spgemm_fill")
+                                B.val.data_ptr(), ub.data_ptr(), cptr[1:].data_ptr(),
+                                ccol.data_ptr(), cval.data_ptr(), _stream()),
+          "spgemm_fill")
     return device_csr(A.nrows, B.ncols, cptr, ccol, cval)
 
 
 def download(A):
     """Device CSR -> host CSR (sorted rows)."""
-    ptr = A.ptr.cpu().numpy()
-    col = A.col.cpu().numpy()
-    val = A.val.cpu().numpy()
-    csr = CSR(A.nrows, A.ncols, ptr, col, val)
-    # rows may be unsorted (spgemm extraction order): sort on host
-    for i in range(csr.nrows):
-        lo, hi = csr.ptr[i], csr.ptr[i + 1]
-        order = np.argsort(csr.col[lo:hi], kind="stable")
-        csr.col[lo:hi] = csr.col[lo:hi][order]
-        csr.val[lo:hi] = csr.val[lo:hi][order]
-    return csr
+    import scipy.sparse as sp
+
+    m = sp.csr_matrix(
+        (A.val.cpu().numpy(), A.col.cpu().numpy(), A.ptr.cpu().numpy()),
+        shape=(A.nrows, A.ncols),
+    )
+    m.sort_indices()
+    return CSR(A.nrows, A.ncols, m.indptr, m.indices, m.data)
 
 
 def poisson3d_device(n, device="cuda"):

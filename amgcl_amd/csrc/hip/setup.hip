@@ -165,12 +165,19 @@ __global__ void agg_init_k(int64_t n, const int *__restrict__ ptr, const uint8_t
     }
 }
 
+// m1 needs to be valid for every node within distance 1 of an UNDEF node.
+// `near` marks those nodes (set during the previous round's passes or the
+// init pass); everything else keeps m1 = 0 without re-reading its row.
 __global__ void agg_m1_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
                          const uint8_t *__restrict__ S, const int *__restrict__ id,
-                         uint64_t *__restrict__ m1) {
+                         const uint8_t *__restrict__ near, uint64_t *__restrict__ m1) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
+        if (!near[i]) {
+            m1[i] = 0;
+            continue;
+        }
         uint64_t m = (id[i] == -1) ? agg_key_d((int)i) : 0;
         for (int j = ptr[i]; j < ptr[i + 1]; ++j) {
             if (!S[j]) continue;
@@ -181,6 +188,21 @@ __global__ void agg_m1_k(int64_t n, const int *__restrict__ ptr, const int *__re
             }
         }
         m1[i] = m;
+    }
+}
+
+// near[i] = i is UNDEF or has an UNDEF strong neighbor
+__global__ void agg_near_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
+                           const uint8_t *__restrict__ S, const int *__restrict__ id,
+                           uint8_t *__restrict__ near) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        if (id[i] == -1) {
+            near[i] = 1;
+            for (int j = ptr[i]; j < ptr[i + 1]; ++j)
+                if (S[j]) near[col[j]] = 1;
+        }
     }
 }
 
@@ -269,8 +291,10 @@ extern "C" int amg_agg_init(int64_t n, const int *ptr, const uint8_t *S, int *id
 }
 extern "C" int amg_agg_round(int64_t n, const int *ptr, const int *col, const uint8_t *S,
                              int *id, uint8_t *prov, uint64_t *m1, uint8_t *newroot,
-                             int *remaining, hipStream_t s) {
-    agg_m1_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, m1);
+                             uint8_t *near, int *remaining, hipStream_t s) {
+    hipMemsetAsync(near, 0, n, s);
+    agg_near_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, near);
+    agg_m1_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, near, m1);
     agg_roots_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, m1, newroot);
     agg_claim_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, prov, newroot);
     agg_adopt_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, prov, remaining);
@@ -485,13 +509,130 @@ extern "C" int amg_sort_rows(int64_t n, const int *ptr_scanned, int *col, double
 // ---------------------------------------------------------------------------
 #define HSLOTS 512
 #define HMASK (HSLOTS - 1)
+// small-row bin: 8 lanes per row, 128-slot tables (rows with ub <= SGSMALL)
+#define SGRP 8
+#define SSLOTS 128
+#define SSMASK (SSLOTS - 1)
+#define SGSMALL 96
+
+// per-row product upper bound (for binning)
+__global__ void spgemm_ub_k(int64_t an, const int *__restrict__ aptr,
+                            const int *__restrict__ acol, const int *__restrict__ bptr,
+                            int *__restrict__ ub) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < an; i += stride) {
+        int u = 0;
+        for (int ja = aptr[i]; ja < aptr[i + 1]; ++ja) {
+            int ca = acol[ja];
+            u += bptr[ca + 1] - bptr[ca];
+        }
+        ub[i] = u;
+    }
+}
+
+__global__ void spgemm_count_small_k(int64_t an, const int *__restrict__ aptr,
+                                     const int *__restrict__ acol,
+                                     const int *__restrict__ bptr,
+                                     const int *__restrict__ bcol,
+                                     const int *__restrict__ ub, int *__restrict__ cnt) {
+    __shared__ int keys[32][SSLOTS];  // 32 groups of 8 lanes (256 threads)
+    int gid = threadIdx.x / SGRP;
+    int lane = threadIdx.x & (SGRP - 1);
+    int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / SGRP;
+    int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SGRP;
+    int *tk = keys[gid];
+    for (; row < an; row += stride) {
+        if (ub[row] > SGSMALL) continue;
+        for (int t = lane; t < SSLOTS; t += SGRP) tk[t] = -1;
+        __builtin_amdgcn_s_waitcnt(0);
+        __builtin_amdgcn_wave_barrier();
+        int inserted = 0;
+        for (int ja = aptr[row]; ja < aptr[row + 1]; ++ja) {
+            int ca = acol[ja];
+            for (int jb = bptr[ca] + lane; jb < bptr[ca + 1]; jb += SGRP) {
+                int key = bcol[jb];
+                uint32_t h = ((uint32_t)key * 2654435761u) & SSMASK;
+                while (true) {
+                    int old = atomicCAS(&tk[h], -1, key);
+                    if (old == -1) { ++inserted; break; }
+                    if (old == key) break;
+                    h = (h + 1) & SSMASK;
+                }
+            }
+        }
+#pragma unroll
+        for (int off = SGRP / 2; off > 0; off >>= 1)
+            inserted += __shfl_down(inserted, off, SGRP);
+        if (lane == 0) cnt[row] = inserted;
+        __builtin_amdgcn_wave_barrier();
+    }
+}
+
+__global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
+                                    const int *__restrict__ acol,
+                                    const double *__restrict__ aval,
+                                    const int *__restrict__ bptr,
+                                    const int *__restrict__ bcol,
+                                    const double *__restrict__ bval,
+                                    const int *__restrict__ ub,
+                                    const int *__restrict__ cptr_scanned,
+                                    int *__restrict__ ccol, double *__restrict__ cval) {
+    __shared__ int keys[32][SSLOTS];
+    __shared__ double vals[32][SSLOTS];
+    __shared__ int cur[32];
+    int gid = threadIdx.x / SGRP;
+    int lane = threadIdx.x & (SGRP - 1);
+    int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / SGRP;
+    int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SGRP;
+    int *tk = keys[gid];
+    double *tv = vals[gid];
+    for (; row < an; row += stride) {
+        if (ub[row] > SGSMALL) continue;
+        for (int t = lane; t < SSLOTS; t += SGRP) {
+            tk[t] = -1;
+            tv[t] = 0.0;
+        }
+        if (lane == 0) cur[gid] = 0;
+        __builtin_amdgcn_s_waitcnt(0);
+        __builtin_amdgcn_wave_barrier();
+        for (int ja = aptr[row]; ja < aptr[row + 1]; ++ja) {
+            int ca = acol[ja];
+            double va = aval[ja];
+            for (int jb = bptr[ca] + lane; jb < bptr[ca + 1]; jb += SGRP) {
+                int key = bcol[jb];
+                double v = va * bval[jb];
+                uint32_t h = ((uint32_t)key * 2654435761u) & SSMASK;
+                while (true) {
+                    int old = atomicCAS(&tk[h], -1, key);
+                    if (old == -1 || old == key) {
+                        atomicAdd(&tv[h], v);
+                        break;
+                    }
+                    h = (h + 1) & SSMASK;
+                }
+            }
+        }
+        __builtin_amdgcn_s_waitcnt(0);
+        __builtin_amdgcn_wave_barrier();
+        int base = (row == 0) ? 0 : cptr_scanned[row - 1];
+        for (int t = lane; t < SSLOTS; t += SGRP) {
+            int key = tk[t];
+            if (key != -1) {
+                int pos = base + atomicAdd(&cur[gid], 1);
+                ccol[pos] = key;
+                cval[pos] = tv[t];
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+    }
+}
 
 __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
                                const int *__restrict__ acol, const int *__restrict__ bptr,
-                               const int *__restrict__ bcol, int *__restrict__ cnt,
-                               int *__restrict__ overflow) {
+                               const int *__restrict__ bcol, const int *__restrict__ ub,
+                               int *__restrict__ cnt, int *__restrict__ overflow) {
     __shared__ int keys[4][HSLOTS];
-    __shared__ int rowcnt[4];
     int wid = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -499,6 +640,7 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
     int *tk = keys[wid];
 
     for (; row < an; row += stride) {
+        if (ub[row] <= SGSMALL) continue;
         for (int t = lane; t < HSLOTS; t += WAVE) tk[t] = -1;
         if (lane == 0) rowcnt[wid] = 0;
         __builtin_amdgcn_s_waitcnt(0);
@@ -535,7 +677,7 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
 __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
                               const int *__restrict__ acol, const double *__restrict__ aval,
                               const int *__restrict__ bptr, const int *__restrict__ bcol,
-                              const double *__restrict__ bval,
+                              const double *__restrict__ bval, const int *__restrict__ ub,
                               const int *__restrict__ cptr_scanned, int *__restrict__ ccol,
                               double *__restrict__ cval) {
     __shared__ int keys[4][HSLOTS];
@@ -549,6 +691,7 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
     double *tv = vals[wid];
 
     for (; row < an; row += stride) {
+        if (ub[row] <= SGSMALL) continue;
         for (int t = lane; t < HSLOTS; t += WAVE) {
             tk[t] = -1;
             tv[t] = 0.0;
@@ -632,15 +775,21 @@ extern "C" int amg_poisson_fill(int64_t n, const int *ptr_scanned, int *col, dou
 }
 
 extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol, const int *bptr,
-                                const int *bcol, int *cnt, int *overflow, hipStream_t s) {
-    spgemm_count_k<<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, bptr, bcol, cnt, overflow);
+                                const int *bcol, int *ub, int *cnt, int *overflow,
+                                hipStream_t s) {
+    spgemm_ub_k<<<nblk(an), 256, 0, s>>>(an, aptr, acol, bptr, ub);
+    spgemm_count_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub, cnt);
+    spgemm_count_k<<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub, cnt,
+                                                   overflow);
     return (int)hipGetLastError();
 }
 extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol, const double *aval,
                                const int *bptr, const int *bcol, const double *bval,
-                               const int *cptr_scanned, int *ccol, double *cval,
-                               hipStream_t s) {
+                               const int *ub, const int *cptr_scanned, int *ccol,
+                               double *cval, hipStream_t s) {
+    spgemm_fill_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol,
+                                                        bval, ub, cptr_scanned, ccol, cval);
     spgemm_fill_k<<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol, bval,
-                                                  cptr_scanned, ccol, cval);
+                                                  ub, cptr_scanned, ccol, cval);
     return (int)hipGetLastError();
 }
