@@ -642,7 +642,6 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
     for (; row < an; row += stride) {
         if (ub[row] <= SGSMALL) continue;
         for (int t = lane; t < HSLOTS; t += WAVE) tk[t] = -1;
-        if (lane == 0) rowcnt[wid] = 0;
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
         int inserted = 0;
