@@ -162,8 +162,9 @@ def transpose(A):
                                       A.val.data_ptr(), cursor.data_ptr(),
                                       tcol.data_ptr(), tval.data_ptr(), _stream()),
           "tscatter")
-    check(lib().amg_sort_rows(A.ncols, tptr[1:].data_ptr(), tcol.data_ptr(),
-                              tval.data_ptr(), _stream()), "sort_rows")
+    # rows are left unsorted: no solve/setup kernel depends on intra-row
+    # order (gather bandwidth depends on the address SET, not order), and
+    # sorting 40+-entry rows in global memory costs more than everything else
     return device_csr(A.ncols, A.nrows, tptr, tcol, tval)
 
 

@@ -251,6 +251,7 @@ __global__ void agg_adopt_k(int64_t n, const int *__restrict__ ptr, const int *_
                             uint8_t *__restrict__ prov, int *__restrict__ remaining) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int my_remaining = 0;
     for (; i < n; i += stride) {
         if (id[i] != -1) continue;
         uint64_t best = 0;
@@ -267,9 +268,15 @@ __global__ void agg_adopt_k(int64_t n, const int *__restrict__ ptr, const int *_
             id[i] = root;
             prov[i] = 1;
         } else {
-            atomicAdd(remaining, 1);
+            ++my_remaining;
         }
     }
+    // one atomic per wave, not per undecided node (contention)
+#pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        my_remaining += __shfl_down(my_remaining, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && my_remaining)
+        atomicAdd(remaining, my_remaining);
 }
 
 __global__ void agg_mark_roots_k(int64_t n, const int *__restrict__ id, int *__restrict__ mark) {
@@ -628,11 +635,30 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
     }
 }
 
+// Big rows: wave per row, with the (ja, jb) product space FLATTENED across
+// the 64 lanes via an LDS prefix of B-row lengths (short B rows would leave
+// most lanes idle under the naive per-A-entry split). A rows longer than
+// BIGROW raise `overflow` -> host fallback.
+#define BIGROW 128
+
+__device__ __forceinline__ int pfx_find(const int *pfx, int len, int t) {
+    // largest ja with pfx[ja] <= t  (pfx[0] = 0, ascending)
+    int lo = 0, hi = len;  // invariant: pfx[lo] <= t < pfx[hi+? ]
+    while (hi - lo > 1) {
+        int mid = (lo + hi) >> 1;
+        if (pfx[mid] <= t) lo = mid;
+        else hi = mid;
+    }
+    return lo;
+}
+
 __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
                                const int *__restrict__ acol, const int *__restrict__ bptr,
                                const int *__restrict__ bcol, const int *__restrict__ ub,
                                int *__restrict__ cnt, int *__restrict__ overflow) {
     __shared__ int keys[4][HSLOTS];
+    __shared__ int pfx[4][BIGROW + 1];
+    __shared__ int bbeg[4][BIGROW];
     int wid = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -640,31 +666,43 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
     int *tk = keys[wid];
 
     for (; row < an; row += stride) {
-        if (ub[row] <= SGSMALL) continue;
+        int total = ub[row];
+        if (total <= SGSMALL) continue;
+        int ab = aptr[row], alen = aptr[row + 1] - ab;
+        if (alen > BIGROW || total > 4 * HSLOTS) {
+            if (lane == 0) atomicAdd(overflow, 1);
+            continue;
+        }
         for (int t = lane; t < HSLOTS; t += WAVE) tk[t] = -1;
+        // build the product-space prefix (serial scan by lane 0: alen <= 128)
+        if (lane == 0) {
+            int acc = 0;
+            for (int j = 0; j < alen; ++j) {
+                pfx[wid][j] = acc;
+                int ca = acol[ab + j];
+                bbeg[wid][j] = bptr[ca];
+                acc += bptr[ca + 1] - bptr[ca];
+            }
+            pfx[wid][alen] = acc;
+        }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
         int inserted = 0;
         bool ovf = false;
-        for (int ja = aptr[row]; ja < aptr[row + 1]; ++ja) {
-            int ca = acol[ja];
-            for (int jb = bptr[ca] + lane; jb < bptr[ca + 1]; jb += WAVE) {
-                int key = bcol[jb];
-                uint32_t h = ((uint32_t)key * 2654435761u) & HMASK;
-                int probes = 0;
-                while (true) {
-                    int old = atomicCAS(&tk[h], -1, key);
-                    if (old == -1) { ++inserted; break; }
-                    if (old == key) break;
-                    h = (h + 1) & HMASK;
-                    if (++probes >= HSLOTS) { ovf = true; break; }
-                }
-                if (ovf) break;
+        for (int t = lane; t < total && !ovf; t += WAVE) {
+            int ja = pfx_find(pfx[wid], alen, t);
+            int key = bcol[bbeg[wid][ja] + (t - pfx[wid][ja])];
+            uint32_t h = ((uint32_t)key * 2654435761u) & HMASK;
+            int probes = 0;
+            while (true) {
+                int old = atomicCAS(&tk[h], -1, key);
+                if (old == -1) { ++inserted; break; }
+                if (old == key) break;
+                h = (h + 1) & HMASK;
+                if (++probes >= HSLOTS) { ovf = true; break; }
             }
-            if (ovf) break;
         }
-        if (ovf) atomicAdd(overflow, 1);
-        // wave-reduce inserted
+        if (ovf && lane == 0) atomicAdd(overflow, 1);
 #pragma unroll
         for (int off = WAVE / 2; off > 0; off >>= 1)
             inserted += __shfl_down(inserted, off, WAVE);
@@ -681,6 +719,9 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
                               double *__restrict__ cval) {
     __shared__ int keys[4][HSLOTS];
     __shared__ double vals[4][HSLOTS];
+    __shared__ int pfx[4][BIGROW + 1];
+    __shared__ int bbeg[4][BIGROW];
+    __shared__ double av[4][BIGROW];
     __shared__ int wcur[4];
     int wid = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
@@ -690,34 +731,45 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
     double *tv = vals[wid];
 
     for (; row < an; row += stride) {
-        if (ub[row] <= SGSMALL) continue;
+        int total = ub[row];
+        if (total <= SGSMALL) continue;
+        int ab = aptr[row], alen = aptr[row + 1] - ab;
+        if (alen > BIGROW || total > 4 * HSLOTS) continue;  // host fallback
         for (int t = lane; t < HSLOTS; t += WAVE) {
             tk[t] = -1;
             tv[t] = 0.0;
         }
-        if (lane == 0) wcur[wid] = 0;
+        for (int j = lane; j < alen; j += WAVE) av[wid][j] = aval[ab + j];
+        if (lane == 0) {
+            wcur[wid] = 0;
+            int acc = 0;
+            for (int j = 0; j < alen; ++j) {
+                pfx[wid][j] = acc;
+                int ca = acol[ab + j];
+                bbeg[wid][j] = bptr[ca];
+                acc += bptr[ca + 1] - bptr[ca];
+            }
+            pfx[wid][alen] = acc;
+        }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
-        for (int ja = aptr[row]; ja < aptr[row + 1]; ++ja) {
-            int ca = acol[ja];
-            double va = aval[ja];
-            for (int jb = bptr[ca] + lane; jb < bptr[ca + 1]; jb += WAVE) {
-                int key = bcol[jb];
-                double v = va * bval[jb];
-                uint32_t h = ((uint32_t)key * 2654435761u) & HMASK;
-                while (true) {
-                    int old = atomicCAS(&tk[h], -1, key);
-                    if (old == -1 || old == key) {
-                        atomicAdd(&tv[h], v);
-                        break;
-                    }
-                    h = (h + 1) & HMASK;
+        for (int t = lane; t < total; t += WAVE) {
+            int ja = pfx_find(pfx[wid], alen, t);
+            int jb = bbeg[wid][ja] + (t - pfx[wid][ja]);
+            int key = bcol[jb];
+            double v = av[wid][ja] * bval[jb];
+            uint32_t h = ((uint32_t)key * 2654435761u) & HMASK;
+            while (true) {
+                int old = atomicCAS(&tk[h], -1, key);
+                if (old == -1 || old == key) {
+                    atomicAdd(&tv[h], v);
+                    break;
                 }
+                h = (h + 1) & HMASK;
             }
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
-        // extract used slots
         int base = (row == 0) ? 0 : cptr_scanned[row - 1];
         for (int t = lane; t < HSLOTS; t += WAVE) {
             int key = tk[t];

@@ -103,6 +103,12 @@ def galerkin(R, A, P):
         from .backend import hip_setup
         from .profiler import prof
 
-        with prof.scope("galerkin(dev)"):
-            return hip_setup.spgemm(R, hip_setup.spgemm(A, P))
+        try:
+            with prof.scope("galerkin(dev)"):
+                return hip_setup.spgemm(R, hip_setup.spgemm(A, P))
+        except OverflowError:
+            Rh = hip_setup.download(R)
+            Ah = hip_setup.download(A)
+            Ph = hip_setup.download(P)
+            return Rh @ (Ah @ Ph)
     return R @ (A @ P)

@@ -88,6 +88,15 @@ class AMG:
                 try:
                     with prof.scope("transfer_operators"):
                         P, R = coarsening.transfer_operators(A_host)
+                except OverflowError:
+                    # device setup hit a row denser than its LDS buffers:
+                    # fall back to the host engine for this and deeper levels
+                    from ..backend import hip_setup
+
+                    A_host = hip_setup.download(A_host)
+                    lvl.A_host = A_host if (prm["keep_host_matrices"] or backend.name == "cpu") else None
+                    with prof.scope("transfer_operators"):
+                        P, R = coarsening.transfer_operators(A_host)
                 except RuntimeError:
                     # empty level (all nodes removed): stop coarsening here
                     last = True
