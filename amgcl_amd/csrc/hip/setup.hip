@@ -639,7 +639,7 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
 // the 64 lanes via an LDS prefix of B-row lengths (short B rows would leave
 // most lanes idle under the naive per-A-entry split). A rows longer than
 // BIGROW raise `overflow` -> host fallback.
-#define BIGROW 128
+#define BIGROW 256
 
 __device__ __forceinline__ int pfx_find(const int *pfx, int len, int t) {
     // largest ja with pfx[ja] <= t  (pfx[0] = 0, ascending)
@@ -669,7 +669,7 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
         int total = ub[row];
         if (total <= SGSMALL) continue;
         int ab = aptr[row], alen = aptr[row + 1] - ab;
-        if (alen > BIGROW || total > 4 * HSLOTS) {
+        if (alen > BIGROW) {  // true table overflow is caught by the probes
             if (lane == 0) atomicAdd(overflow, 1);
             continue;
         }
@@ -734,7 +734,7 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
         int total = ub[row];
         if (total <= SGSMALL) continue;
         int ab = aptr[row], alen = aptr[row + 1] - ab;
-        if (alen > BIGROW || total > 4 * HSLOTS) continue;  // host fallback
+        if (alen > BIGROW) continue;  // host fallback
         for (int t = lane; t < HSLOTS; t += WAVE) {
             tk[t] = -1;
             tv[t] = 0.0;
