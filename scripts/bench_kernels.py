@@ -32,6 +32,7 @@ def timeit(fn, iters=20):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--size", type=int, default=384)
+    ap.add_argument("--sort", action="store_true", help="sort level rows first")
     args = ap.parse_args()
 
     hip = make_backend("hip")
@@ -39,6 +40,18 @@ def main():
     solve = am.make_solver(A, {"solver": {"type": "cg", "tol": 1e-6}}, backend=hip)
     amg = solve.P
     print(amg)
+
+    if args.sort:
+        from amgcl_amd.backend._hiplib import check, lib
+        from amgcl_amd.backend.hip_setup import _stream
+
+        for lvl in amg.levels:
+            Ad = lvl.A
+            check(lib().amg_sort_rows(Ad.nrows, Ad.ptr[1:].data_ptr(),
+                                      Ad.col.data_ptr(), Ad.val.data_ptr(),
+                                      _stream()), "sort")
+        torch.cuda.synchronize()
+        print("(rows sorted)")
 
     for li, lvl in enumerate(amg.levels):
         Ad = lvl.A
