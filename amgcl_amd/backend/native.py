@@ -1,0 +1,206 @@
+"""Native solve driver bindings.
+
+Hands the whole AMG-preconditioned CG/BiCGStab solve to the C++ driver
+(csrc/hip/driver.hip) — one ctypes call per solve, no Python in the hot
+path. Engaged automatically by make_solver for HIP-backend AMG with diagonal
+smoothers (SPAI0/damped Jacobi), dense-inverse coarse solve, and cg/bicgstab;
+anything else uses the generic Python orchestration.
+"""
+import ctypes
+
+from ._hiplib import lib as _lib
+
+
+class LevelDescC(ctypes.Structure):
+    _fields_ = [
+        ("nrows", ctypes.c_int64),
+        ("nnz", ctypes.c_int64),
+        ("ptr", ctypes.c_void_p),
+        ("col", ctypes.c_void_p),
+        ("val", ctypes.c_void_p),
+        ("subw", ctypes.c_int),
+        ("pnnz", ctypes.c_int64),
+        ("pptr", ctypes.c_void_p),
+        ("pcol", ctypes.c_void_p),
+        ("pval", ctypes.c_void_p),
+        ("psubw", ctypes.c_int),
+        ("rnnz", ctypes.c_int64),
+        ("rptr", ctypes.c_void_p),
+        ("rcol", ctypes.c_void_p),
+        ("rval", ctypes.c_void_p),
+        ("rsubw", ctypes.c_int),
+        ("M", ctypes.c_void_p),
+        ("f", ctypes.c_void_p),
+        ("u", ctypes.c_void_p),
+        ("t", ctypes.c_void_p),
+    ]
+
+
+_driver_bound = False
+
+
+def _bind():
+    global _driver_bound
+    L = _lib()
+    if not _driver_bound:
+        L.amg_driver_create.argtypes = [
+            ctypes.POINTER(LevelDescC), ctypes.c_int, ctypes.c_void_p, ctypes.c_int64,
+            ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+        ]
+        L.amg_driver_create.restype = ctypes.c_void_p
+        L.amg_driver_destroy.argtypes = [ctypes.c_void_p]
+        L.amg_driver_destroy.restype = None
+        L.amg_driver_cg.argtypes = (
+            [ctypes.c_void_p] + [ctypes.c_void_p] * 7
+            + [ctypes.c_double, ctypes.c_double, ctypes.c_int,
+               ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_double)]
+        )
+        L.amg_driver_cg.restype = ctypes.c_int
+        L.amg_driver_bicgstab.argtypes = (
+            [ctypes.c_void_p] + [ctypes.c_void_p] * 10
+            + [ctypes.c_double, ctypes.c_double, ctypes.c_int,
+               ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_double)]
+        )
+        L.amg_driver_bicgstab.restype = ctypes.c_int
+        _driver_bound = True
+    return L
+
+
+def _ptr(t):
+    return ctypes.c_void_p(0 if t is None else t.data_ptr())
+
+
+class NativeDriver:
+    """Owns the C driver handle plus references to every device tensor it
+    points at (torch would otherwise free them under the caching allocator)."""
+
+    def __init__(self, amg, backend, solver_kind, solver_prm):
+        import torch
+
+        from ..relaxation.spai0 import DiagonalSmootherBase
+        from .hip import DeviceCSR, DeviceDenseSolver
+
+        self.backend = backend
+        self.solver_kind = solver_kind
+        self.tol = float(solver_prm["tol"])
+        self.abstol = float(solver_prm["abstol"])
+        self.maxiter = int(solver_prm["maxiter"])
+
+        levels = amg.levels
+        if not all(isinstance(l.A, DeviceCSR) for l in levels):
+            raise TypeError("native driver needs device-resident levels")
+        for l in levels[:-1]:
+            if not isinstance(l.relax, DiagonalSmootherBase):
+                raise TypeError("native driver supports diagonal smoothers only")
+        if amg.coarse_solve is not None and not isinstance(
+            amg.coarse_solve, DeviceDenseSolver
+        ):
+            raise TypeError("native driver needs the dense coarse solver")
+        if amg.coarse_solve is None and not isinstance(
+            levels[-1].relax, DiagonalSmootherBase
+        ):
+            raise TypeError("native coarsest smoother must be diagonal")
+
+        self._keep = []  # tensor refs
+        descs = (LevelDescC * len(levels))()
+        n0 = levels[0].A.nrows
+        for i, l in enumerate(levels):
+            d = descs[i]
+            A = l.A
+            d.nrows, d.nnz = A.nrows, int(A.nnz)
+            d.ptr, d.col, d.val = _ptr(A.ptr), _ptr(A.col), _ptr(A.val)
+            d.subw = A.subw or _auto_subw(A)
+            if l.P is not None:
+                P, R = l.P, l.R
+                d.pnnz = int(P.nnz)
+                d.pptr, d.pcol, d.pval = _ptr(P.ptr), _ptr(P.col), _ptr(P.val)
+                d.psubw = P.subw or _auto_subw(P)
+                d.rnnz = int(R.nnz)
+                d.rptr, d.rcol, d.rval = _ptr(R.ptr), _ptr(R.col), _ptr(R.val)
+                d.rsubw = R.subw or _auto_subw(R)
+            relax = l.relax
+            d.M = _ptr(relax.M if relax is not None else None)
+            d.f = _ptr(l.f)
+            d.u = _ptr(l.u)
+            d.t = _ptr(l.t)
+            self._keep.extend([A.ptr, A.col, A.val, l.f, l.u, l.t])
+            if l.P is not None:
+                self._keep.extend([l.P.ptr, l.P.col, l.P.val, l.R.ptr, l.R.col, l.R.val])
+            if relax is not None:
+                self._keep.append(relax.M)
+
+        inv = amg.coarse_solve.inv if amg.coarse_solve is not None else None
+        if inv is not None:
+            self._keep.append(inv)
+        prm = amg.prm
+        L = _bind()
+        stream = torch.cuda.current_stream().cuda_stream
+        self.handle = L.amg_driver_create(
+            descs, len(levels), _ptr(inv),
+            amg.coarse_solve.n if amg.coarse_solve is not None else 0,
+            int(prm["npre"]), int(prm["npost"]), int(prm["ncycle"]),
+            int(prm["pre_cycles"]), ctypes.c_void_p(stream),
+        )
+        if not self.handle:
+            raise RuntimeError("amg_driver_create failed")
+        nwork = 5 if solver_kind == "cg" else 8
+        self._work = [torch.empty(n0, dtype=torch.float64, device=backend.device)
+                      for _ in range(nwork)]
+        self._descs = descs
+
+    def solve(self, rhs, x):
+        L = _bind()
+        iters = ctypes.c_int64(0)
+        resid = ctypes.c_double(0.0)
+        w = [_ptr(t) for t in self._work]
+        if self.solver_kind == "cg":
+            rc = L.amg_driver_cg(self.handle, _ptr(rhs), _ptr(x), *w, self.tol,
+                                 self.abstol, self.maxiter,
+                                 ctypes.byref(iters), ctypes.byref(resid))
+        else:
+            rc = L.amg_driver_bicgstab(self.handle, _ptr(rhs), _ptr(x), *w, self.tol,
+                                       self.abstol, self.maxiter,
+                                       ctypes.byref(iters), ctypes.byref(resid))
+        if rc != 0:
+            raise RuntimeError(f"native solve failed rc={rc}")
+        return int(iters.value), float(resid.value)
+
+    def __del__(self):
+        try:
+            _bind().amg_driver_destroy(self.handle)
+        except Exception:
+            pass
+
+
+def _auto_subw(A):
+    m = A.nnz / max(A.nrows, 1)
+    for lim, sw in ((4, 2), (10, 4), (24, 8), (128, 16)):
+        if m <= lim:
+            return sw
+    return 32
+
+
+def try_native(make_solver_obj):
+    """Attach a native driver to a MakeSolver when the configuration allows;
+    returns None otherwise."""
+    from ..precond.amg import AMG
+    from ..solver.bicgstab import BiCGStab
+    from ..solver.cg import CG
+
+    P, S, backend = make_solver_obj.P, make_solver_obj.S, make_solver_obj.backend
+    if getattr(backend, "name", "") != "hip":
+        return None
+    if not isinstance(P, AMG):
+        return None
+    if isinstance(S, CG):
+        kind = "cg"
+    elif isinstance(S, BiCGStab) and S.prm["pside"] == "right" and not S.prm["check_after"]:
+        kind = "bicgstab"
+    else:
+        return None
+    if S.prm.get("verbose") or S.prm.get("ns_search"):
+        return None
+    try:
+        return NativeDriver(P, backend, kind, S.prm)
+    except TypeError:
+        return None

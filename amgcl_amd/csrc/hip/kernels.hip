@@ -97,9 +97,8 @@ static inline int pick_subw(int64_t nrows, int64_t nnz) {
     if (m <= 4) return 2;
     if (m <= 10) return 4;
     if (m <= 24) return 8;
-    if (m <= 48) return 16;
-    if (m <= 96) return 32;
-    return 64;
+    if (m <= 128) return 16;
+    return 32;
 }
 
 extern "C" int amg_spmv_f64(int64_t nrows, int64_t nnz, const int *ptr, const int *col,

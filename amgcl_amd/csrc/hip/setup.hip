@@ -31,6 +31,33 @@ static inline int nblk(int64_t work, int block = 256, int cap = 4096) {
     return (int)(b < 1 ? 1 : (b > cap ? cap : b));
 }
 
+// Bitonic sort of an LDS (key,val) table with LANES cooperating threads.
+// Empty slots must hold key = INT_MAX (sorted to the tail). SLOTS power of 2.
+template <int LANES, int SLOTS>
+__device__ __forceinline__ void lds_bitonic(int *k, double *v, int lane) {
+    for (int size = 2; size <= SLOTS; size <<= 1) {
+        for (int strd = size >> 1; strd > 0; strd >>= 1) {
+            __builtin_amdgcn_s_waitcnt(0);
+            __builtin_amdgcn_wave_barrier();
+            for (int t = lane; t < SLOTS / 2; t += LANES) {
+                int i = ((t / strd) * strd * 2) + (t % strd);
+                int j = i + strd;
+                bool up = ((i & size) == 0);
+                int ki = k[i], kj = k[j];
+                if ((ki > kj) == up) {
+                    k[i] = kj;
+                    k[j] = ki;
+                    double tv = v[i];
+                    v[i] = v[j];
+                    v[j] = tv;
+                }
+            }
+        }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_wave_barrier();
+}
+
 __device__ __forceinline__ uint64_t agg_key_d(int i) {
     uint32_t x = (uint32_t)i;
     x ^= x >> 16; x *= 0x7feb352dU; x ^= x >> 15; x *= 0x846ca68bU; x ^= x >> 16;
@@ -587,7 +614,6 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
                                     int *__restrict__ ccol, double *__restrict__ cval) {
     __shared__ int keys[32][SSLOTS];
     __shared__ double vals[32][SSLOTS];
-    __shared__ int cur[32];
     int gid = threadIdx.x / SGRP;
     int lane = threadIdx.x & (SGRP - 1);
     int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / SGRP;
@@ -600,7 +626,6 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
             tk[t] = -1;
             tv[t] = 0.0;
         }
-        if (lane == 0) cur[gid] = 0;
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
         for (int ja = aptr[row]; ja < aptr[row + 1]; ++ja) {
@@ -622,14 +647,15 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
+        // sorted extraction: empties to +inf, bitonic, write the prefix
+        for (int t = lane; t < SSLOTS; t += SGRP)
+            if (tk[t] == -1) tk[t] = 0x7fffffff;
+        lds_bitonic<SGRP, SSLOTS>(tk, tv, lane);
         int base = (row == 0) ? 0 : cptr_scanned[row - 1];
-        for (int t = lane; t < SSLOTS; t += SGRP) {
-            int key = tk[t];
-            if (key != -1) {
-                int pos = base + atomicAdd(&cur[gid], 1);
-                ccol[pos] = key;
-                cval[pos] = tv[t];
-            }
+        int len = cptr_scanned[row] - base;
+        for (int t = lane; t < len; t += SGRP) {
+            ccol[base + t] = tk[t];
+            cval[base + t] = tv[t];
         }
         __builtin_amdgcn_wave_barrier();
     }
@@ -722,7 +748,6 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
     __shared__ int pfx[4][BIGROW + 1];
     __shared__ int bbeg[4][BIGROW];
     __shared__ double av[4][BIGROW];
-    __shared__ int wcur[4];
     int wid = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -741,7 +766,6 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
         }
         for (int j = lane; j < alen; j += WAVE) av[wid][j] = aval[ab + j];
         if (lane == 0) {
-            wcur[wid] = 0;
             int acc = 0;
             for (int j = 0; j < alen; ++j) {
                 pfx[wid][j] = acc;
@@ -770,14 +794,14 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
+        for (int t = lane; t < HSLOTS; t += WAVE)
+            if (tk[t] == -1) tk[t] = 0x7fffffff;
+        lds_bitonic<WAVE, HSLOTS>(tk, tv, lane);
         int base = (row == 0) ? 0 : cptr_scanned[row - 1];
-        for (int t = lane; t < HSLOTS; t += WAVE) {
-            int key = tk[t];
-            if (key != -1) {
-                int pos = base + atomicAdd(&wcur[wid], 1);
-                ccol[pos] = key;
-                cval[pos] = tv[t];
-            }
+        int len = cptr_scanned[row] - base;
+        for (int t = lane; t < len; t += WAVE) {
+            ccol[base + t] = tk[t];
+            cval[base + t] = tv[t];
         }
         __builtin_amdgcn_wave_barrier();
     }

@@ -37,6 +37,12 @@ class MakeSolver:
         self.S = make_solver_component(
             A.nrows, prm.get("solver"), backend, inner_product
         )
+        # native C++ solve driver (HIP + AMG + diagonal smoothers + cg/bicgstab)
+        self._native = None
+        if inner_product is None and getattr(backend, "name", "") == "hip":
+            from .backend.native import try_native
+
+            self._native = try_native(self)
 
     def __call__(self, rhs, x=None, A=None):
         """Solve A x = rhs. Returns (x, iters, relative residual)."""
@@ -46,7 +52,10 @@ class MakeSolver:
             x_dev = b.vector(self.A_host.nrows)
         else:
             x_dev = x if not isinstance(x, np.ndarray) or b.name == "cpu" else b.from_host(x)
-        iters, resid = self.S(self.P, rhs_dev, x_dev, A=A)
+        if self._native is not None and A is None:
+            iters, resid = self._native.solve(rhs_dev, x_dev)
+        else:
+            iters, resid = self.S(self.P, rhs_dev, x_dev, A=A)
         return x_dev, iters, resid
 
     def solve(self, rhs, x=None, A=None):
