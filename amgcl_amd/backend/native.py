@@ -62,6 +62,8 @@ def _bind():
                ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_double)]
         )
         L.amg_driver_bicgstab.restype = ctypes.c_int
+        L.amg_driver_precond.argtypes = [ctypes.c_void_p] * 4
+        L.amg_driver_precond.restype = ctypes.c_int
         _driver_bound = True
     return L
 
@@ -74,7 +76,7 @@ class NativeDriver:
     """Owns the C driver handle plus references to every device tensor it
     points at (torch would otherwise free them under the caching allocator)."""
 
-    def __init__(self, amg, backend, solver_kind, solver_prm):
+    def __init__(self, amg, backend, solver_kind="cg", solver_prm=None):
         import torch
 
         from ..relaxation.spai0 import DiagonalSmootherBase
@@ -82,6 +84,7 @@ class NativeDriver:
 
         self.backend = backend
         self.solver_kind = solver_kind
+        solver_prm = solver_prm or {"tol": 1e-8, "abstol": 0.0, "maxiter": 100}
         self.tol = float(solver_prm["tol"])
         self.abstol = float(solver_prm["abstol"])
         self.maxiter = int(solver_prm["maxiter"])
@@ -164,6 +167,13 @@ class NativeDriver:
         if rc != 0:
             raise RuntimeError(f"native solve failed rc={rc}")
         return int(iters.value), float(resid.value)
+
+    def precond_apply(self, rhs, x):
+        """x = M^-1 rhs (one native V-cycle stack application)."""
+        L = _bind()
+        rc = L.amg_driver_precond(self.handle, _ptr(rhs), _ptr(x), _ptr(self._work[0]))
+        if rc != 0:
+            raise RuntimeError(f"native precond failed rc={rc}")
 
     def __del__(self):
         try:

@@ -120,6 +120,43 @@ static py::tuple poisson3d(i64 n, double anisotropy) {
     return py::make_tuple(ptr, col, val);
 }
 
+// Row strip [row_beg, row_end) of the n^3 Poisson matrix with GLOBAL column
+// indices (for the distributed tests/benchmarks; parity:
+// examples/mpi/mpi_solver.cpp:47 assembles the local strip per rank).
+static py::tuple poisson3d_strip(i64 n, i64 row_beg, i64 row_end) {
+    const i64 nloc = row_end - row_beg;
+    const double dia = 6.0;
+    arr<i32> ptr(nloc + 1);
+    i32 *P = ptr.mutable_data();
+    P[0] = 0;
+#pragma omp parallel for schedule(static)
+    for (i64 r = 0; r < nloc; ++r) {
+        i64 idx = row_beg + r;
+        i64 i = idx % n, j = (idx / n) % n, k = idx / (n * n);
+        P[r + 1] = 1 + (k > 0) + (j > 0) + (i > 0) + (i + 1 < n) + (j + 1 < n) + (k + 1 < n);
+    }
+    scan_i32(P + 1, nloc);
+    const i64 nnz = P[nloc];
+    arr<i32> col = big_arr<i32>(nnz);
+    arr<double> val = big_arr<double>(nnz);
+    i32 *C = col.mutable_data();
+    double *V = val.mutable_data();
+#pragma omp parallel for schedule(static)
+    for (i64 r = 0; r < nloc; ++r) {
+        i64 idx = row_beg + r;
+        i64 i = idx % n, j = (idx / n) % n, k = idx / (n * n);
+        i64 h = P[r];
+        if (k > 0)     { C[h] = (i32)(idx - n * n); V[h] = -1.0; ++h; }
+        if (j > 0)     { C[h] = (i32)(idx - n);     V[h] = -1.0; ++h; }
+        if (i > 0)     { C[h] = (i32)(idx - 1);     V[h] = -1.0; ++h; }
+        C[h] = (i32)idx; V[h] = dia; ++h;
+        if (i + 1 < n) { C[h] = (i32)(idx + 1);     V[h] = -1.0; ++h; }
+        if (j + 1 < n) { C[h] = (i32)(idx + n);     V[h] = -1.0; ++h; }
+        if (k + 1 < n) { C[h] = (i32)(idx + n * n); V[h] = -1.0; ++h; }
+    }
+    return py::make_tuple(ptr, col, val);
+}
+
 // ---------------------------------------------------------------------------
 // diagonal
 // ---------------------------------------------------------------------------
@@ -911,6 +948,7 @@ static void ilu0_solve(i64 nrows, arr<i32> ptr, arr<i32> col, arr<double> lu,
 PYBIND11_MODULE(_core, m) {
     m.doc() = "amgcl_amd host setup engine (OpenMP)";
     m.def("poisson3d", &poisson3d, py::arg("n"), py::arg("anisotropy") = 1.0);
+    m.def("poisson3d_strip", &poisson3d_strip);
     m.def("diagonal", &diagonal);
     m.def("transpose", &transpose);
     m.def("spgemm", &spgemm);

@@ -225,6 +225,14 @@ extern "C" void amg_driver_destroy(void *h) {
     delete D;
 }
 
+// standalone preconditioner application (used by the distributed layer:
+// the Krylov loop stays in Python for halo exchange, but each local V-cycle
+// runs natively)
+extern "C" int amg_driver_precond(void *h, const double *rhs, double *x, double *x_swap) {
+    Driver *D = (Driver *)h;
+    return precond_apply(D, rhs, x, x_swap);
+}
+
 // Preconditioned CG (parity: amgcl/solver/cg.hpp:152-204).
 // Work vectors r,s,p,q + s_swap provided by the caller (device, n doubles).
 extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, double *s,

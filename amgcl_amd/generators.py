@@ -31,11 +31,8 @@ def poisson3d_strip(n, rank, nranks, rhs="ones"):
     n3 = n**3
     row_beg = (n3 * rank) // nranks
     row_end = (n3 * (rank + 1)) // nranks
-    ptr, col, val = _core.poisson3d(int(n), 1.0)
-    A = CSR(n3, n3, ptr, col, val)
-    lo, hi = A.ptr[row_beg], A.ptr[row_end]
-    ptr_s = (A.ptr[row_beg : row_end + 1] - lo).astype(np.int32)
-    strip = CSR(row_end - row_beg, n3, ptr_s, A.col[lo:hi], A.val[lo:hi])
+    ptr, col, val = _core.poisson3d_strip(int(n), int(row_beg), int(row_end))
+    strip = CSR(row_end - row_beg, n3, ptr, col, val)
     if rhs == "ones":
         b = np.ones(row_end - row_beg, dtype=np.float64)
     else:
