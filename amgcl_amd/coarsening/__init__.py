@@ -5,10 +5,12 @@ an object with .defaults(), .transfer_operators(A) -> (P, R) and
 .coarse_operator(A, P, R) -> Ac.
 """
 from .aggregation import Aggregation
+from .ruge_stuben import RugeStuben
 from .smoothed_aggregation import SmoothedAggregation
 
 REGISTRY = {
     "aggregation": Aggregation,
+    "ruge_stuben": RugeStuben,
     "smoothed_aggregation": SmoothedAggregation,
 }
 
@@ -16,12 +18,6 @@ REGISTRY = {
 def make_coarsening(prm=None):
     prm = dict(prm or {})
     kind = prm.pop("type", "smoothed_aggregation")
-    try:
-        from .ruge_stuben import RugeStuben  # optional, added later
-
-        REGISTRY.setdefault("ruge_stuben", RugeStuben)
-    except ImportError:
-        pass
     if kind not in REGISTRY:
         raise ValueError(f"unknown coarsening '{kind}'")
     return REGISTRY[kind](prm)

@@ -820,6 +820,212 @@ static py::tuple tentative_prolongation(i64 nrows, arr<i32> id_a, i64 naggr) {
 }
 
 // ---------------------------------------------------------------------------
+// Ruge-Stuben (classic) coarsening
+// (parity: amgcl/coarsening/ruge_stuben.hpp:54-458 — strong negative
+// couplings -a_ij >= eps*max|a_ik^-|, standard lambda-bucket C/F splitting,
+// direct interpolation with optional truncation+rescaling).
+// Returns (Pptr, Pcol, Pval, nc).
+// ---------------------------------------------------------------------------
+static py::tuple ruge_stuben(i64 n, arr<i32> ptr_a, arr<i32> col_a, arr<double> val_a,
+                             double eps_strong, bool do_trunc, double eps_trunc) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    const i64 nnz = (i64)col_a.size();
+    constexpr double tiny = 1e-300;
+
+    // strong connections: S[j] = (col != i) && (a_ij < eps * min_k a_ik < 0)
+    std::vector<uint8_t> S(nnz, 0);
+    std::vector<char> cf(n, 'U');
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        double amin = 0.0;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j)
+            if (A.col[j] != (i32)i && A.val[j] < amin) amin = A.val[j];
+        if (amin > -tiny) {
+            cf[i] = 'F';  // no negative couplings
+            continue;
+        }
+        amin *= eps_strong;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j)
+            S[j] = (A.col[j] != (i32)i) && (A.val[j] < amin);
+    }
+
+    // transpose of the strong graph (S^T: who strongly depends on i)
+    std::vector<i32> tptr(n + 1, 0), tcol;
+    for (i64 j = 0; j < nnz; ++j)
+        if (S[j]) ++tptr[A.col[j] + 1];
+    for (i64 i = 0; i < n; ++i) tptr[i + 1] += tptr[i];
+    tcol.resize(tptr[n]);
+    {
+        std::vector<i32> cur(tptr.begin(), tptr.end() - 1);
+        for (i64 i = 0; i < n; ++i)
+            for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j)
+                if (S[j]) tcol[cur[A.col[j]]++] = (i32)i;
+    }
+
+    // lambda-bucket C/F split (standard RS first pass)
+    std::vector<i64> lambda(n);
+    for (i64 i = 0; i < n; ++i) {
+        i64 t = 0;
+        for (i32 j = tptr[i]; j < tptr[i + 1]; ++j)
+            t += (cf[tcol[j]] == 'U') ? 1 : 2;
+        lambda[i] = t;
+    }
+    std::vector<i64> bptr(n + 2, 0), bcnt(n + 1, 0), i2n(n), n2i(n);
+    for (i64 i = 0; i < n; ++i) ++bptr[lambda[i] + 1];
+    for (i64 i = 0; i <= n; ++i) bptr[i + 1] += bptr[i];
+    for (i64 i = 0; i < n; ++i) {
+        i64 lam = lambda[i];
+        i64 idx = bptr[lam] + bcnt[lam]++;
+        i2n[idx] = i;
+        n2i[i] = idx;
+    }
+    auto bucket_move = [&](i64 node, i64 from, i64 to_pos) {
+        i64 old_pos = n2i[node];
+        n2i[i2n[old_pos]] = to_pos;
+        n2i[i2n[to_pos]] = old_pos;
+        std::swap(i2n[old_pos], i2n[to_pos]);
+        (void)from;
+    };
+    for (i64 top = n; top-- > 0;) {
+        i64 i = i2n[top];
+        i64 lam = lambda[i];
+        if (lam == 0) {
+            for (i64 k = 0; k < n; ++k)
+                if (cf[k] == 'U') cf[k] = 'C';
+            break;
+        }
+        --bcnt[lam];
+        if (cf[i] == 'F') continue;
+        cf[i] = 'C';
+        for (i32 j = tptr[i]; j < tptr[i + 1]; ++j) {
+            i32 c = tcol[j];
+            if (cf[c] != 'U') continue;
+            cf[c] = 'F';
+            // bump lambdas of the new F's strong neighbours
+            for (i32 aj = A.ptr[c]; aj < A.ptr[c + 1]; ++aj) {
+                if (!S[aj]) continue;
+                i32 ac = A.col[aj];
+                i64 lam_a = lambda[ac];
+                if (cf[ac] != 'U' || lam_a + 1 >= n) continue;
+                bucket_move(ac, lam_a, bptr[lam_a] + bcnt[lam_a] - 1);
+                --bcnt[lam_a];
+                ++bcnt[lam_a + 1];
+                bptr[lam_a + 1] = bptr[lam_a] + bcnt[lam_a];
+                lambda[ac] = lam_a + 1;
+            }
+        }
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            if (!S[j]) continue;
+            i32 c = A.col[j];
+            i64 lam_c = lambda[c];
+            if (cf[c] != 'U' || lam_c == 0) continue;
+            bucket_move(c, lam_c, bptr[lam_c]);
+            --bcnt[lam_c];
+            ++bcnt[lam_c - 1];
+            ++bptr[lam_c];
+            lambda[c] = lam_c - 1;
+        }
+    }
+
+    // coarse index
+    std::vector<i32> cidx(n, -1);
+    i64 nc = 0;
+    for (i64 i = 0; i < n; ++i)
+        if (cf[i] == 'C') cidx[i] = (i32)nc++;
+    if (!nc) throw std::runtime_error("empty level in ruge_stuben");
+
+    // direct interpolation with optional truncation
+    arr<i32> pptr(n + 1);
+    i32 *PP = pptr.mutable_data();
+    PP[0] = 0;
+    std::vector<double> Amin(do_trunc ? n : 0), Amax(do_trunc ? n : 0);
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        i32 cnt = 0;
+        if (cf[i] == 'C') {
+            cnt = 1;
+        } else if (do_trunc) {
+            double amin = 0, amax = 0;
+            for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+                if (!S[j] || cf[A.col[j]] != 'C') continue;
+                amin = std::min(amin, A.val[j]);
+                amax = std::max(amax, A.val[j]);
+            }
+            Amin[i] = amin * eps_trunc;
+            Amax[i] = amax * eps_trunc;
+            for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+                if (!S[j] || cf[A.col[j]] != 'C') continue;
+                if (A.val[j] < Amin[i] || A.val[j] > Amax[i]) ++cnt;
+            }
+        } else {
+            for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j)
+                if (S[j] && cf[A.col[j]] == 'C') ++cnt;
+        }
+        PP[i + 1] = cnt;
+    }
+    scan_i32(PP + 1, n);
+    const i64 pnz = PP[n];
+    arr<i32> pcol = big_arr<i32>(pnz);
+    arr<double> pval = big_arr<double>(pnz);
+    i32 *PC = pcol.mutable_data();
+    double *PV = pval.mutable_data();
+
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        i32 head = PP[i];
+        if (cf[i] == 'C') {
+            PC[head] = cidx[i];
+            PV[head] = 1.0;
+            continue;
+        }
+        double dia = 0, a_num = 0, a_den = 0, b_num = 0, b_den = 0, d_neg = 0, d_pos = 0;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i32 c = A.col[j];
+            double v = A.val[j];
+            if (c == (i32)i) {
+                dia = v;
+                continue;
+            }
+            if (v < 0) {
+                a_num += v;
+                if (S[j] && cf[c] == 'C') {
+                    a_den += v;
+                    if (do_trunc && Amin[i] < v) d_neg += v;
+                }
+            } else {
+                b_num += v;
+                if (S[j] && cf[c] == 'C') {
+                    b_den += v;
+                    if (do_trunc && v < Amax[i]) d_pos += v;
+                }
+            }
+        }
+        double cf_neg = 1, cf_pos = 1;
+        if (do_trunc) {
+            if (std::abs(a_den - d_neg) > tiny) cf_neg = std::abs(a_den) / std::abs(a_den - d_neg);
+            if (std::abs(b_den - d_pos) > tiny) cf_pos = std::abs(b_den) / std::abs(b_den - d_pos);
+        }
+        if (b_num > 0 && std::abs(b_den) < tiny) dia += b_num;
+        double alpha = std::abs(a_den) > tiny
+                           ? -cf_neg * std::abs(a_num) / (std::abs(dia) * std::abs(a_den))
+                           : 0.0;
+        double beta = std::abs(b_den) > tiny
+                          ? -cf_pos * std::abs(b_num) / (std::abs(dia) * std::abs(b_den))
+                          : 0.0;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i32 c = A.col[j];
+            double v = A.val[j];
+            if (!S[j] || cf[c] != 'C') continue;
+            if (do_trunc && Amin[i] <= v && v <= Amax[i]) continue;
+            PC[head] = cidx[c];
+            PV[head] = (v < 0 ? alpha : beta) * v;
+            ++head;
+        }
+    }
+    return py::make_tuple(pptr, pcol, pval, nc);
+}
+
+// ---------------------------------------------------------------------------
 // SPAI-0 weights (parity: amgcl/relaxation/spai0.hpp:66-77): m_i = a_ii / sum_j a_ij^2
 // ---------------------------------------------------------------------------
 static arr<double> spai0(i64 nrows, arr<i32> ptr, arr<i32> col, arr<double> val) {
@@ -957,6 +1163,7 @@ PYBIND11_MODULE(_core, m) {
     m.def("smoothed_prolongation", &smoothed_prolongation);
     m.def("tentative_prolongation", &tentative_prolongation);
     m.def("spai0", &spai0);
+    m.def("ruge_stuben", &ruge_stuben);
     m.def("spmv", &spmv);
     m.def("residual", &residual);
     m.def("gauss_seidel", &gauss_seidel);

@@ -10,9 +10,10 @@ import pytest
 
 import amgcl_amd as am
 
-COARSENING = ["smoothed_aggregation", "aggregation"]
+COARSENING = ["smoothed_aggregation", "aggregation", "ruge_stuben"]
 RELAX = ["spai0", "damped_jacobi", "chebyshev", "gauss_seidel", "ilu0"]
 SOLVERS = ["cg", "bicgstab", "gmres", "richardson"]
+SOLVERS_EXTRA = ["bicgstabl", "fgmres", "lgmres", "idrs"]
 
 N = 16  # 4096 unknowns
 
@@ -114,3 +115,28 @@ def test_preonly_nested():
     x, iters, resid = s(b)
     assert iters == 1
     assert resid < 1.0
+
+
+@pytest.mark.parametrize("solver", SOLVERS_EXTRA)
+def test_extra_solvers(problem, solver):
+    A, b = problem
+    s = am.make_solver(
+        A,
+        {"precond": {"class": "amg", "coarse_enough": 500},
+         "solver": {"type": solver, "tol": 1e-8, "maxiter": 200}},
+    )
+    x, iters, resid = s(b)
+    r = b - A @ x
+    assert np.linalg.norm(r) / np.linalg.norm(b) < 1e-6
+
+
+def test_ruge_stuben_iteration_class(problem):
+    A, b = problem
+    s = am.make_solver(
+        A,
+        {"precond": {"class": "amg", "coarsening": {"type": "ruge_stuben"},
+                     "coarse_enough": 500},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+    )
+    x, iters, resid = s(b)
+    assert resid < 1e-8 and iters <= 20
