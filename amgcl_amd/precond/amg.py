@@ -20,10 +20,12 @@ from ..relaxation import make_relaxation_factory
 
 
 class Level:
-    __slots__ = ("A", "P", "R", "f", "u", "t", "relax", "rows", "nnz", "A_host")
+    __slots__ = ("A", "P", "R", "P_build", "R_build", "f", "u", "t", "relax",
+                 "rows", "nnz", "A_host")
 
     def __init__(self):
         self.A = self.P = self.R = None
+        self.P_build = self.R_build = None
         self.f = self.u = self.t = None
         self.relax = None
         self.rows = self.nnz = 0
@@ -55,12 +57,37 @@ class AMG:
         self.prm = merge_params(self.defaults(), prm)
         self.levels = []
         self.coarse_solve = None
+        self._coarsening = None
         self._build(A)
+
+    def rebuild(self, A_new):
+        """Reuse the transfer operators for a matrix with changed coefficients
+        (time-dependent problems; parity: amgcl/amg.hpp:250-269 rebuild)."""
+        backend = self.backend
+        relax_factory = make_relaxation_factory(self.prm["relax"])
+        A = A_new
+        for lvl in self.levels[:-1]:
+            lvl.rows, lvl.nnz = A.nrows, A.nnz
+            if lvl.A_host is not None:
+                lvl.A_host = A
+            lvl.A = backend.matrix(A)
+            lvl.relax = relax_factory(A, backend)
+            A = self._coarsening.coarse_operator(A, lvl.P_build, lvl.R_build)
+        last = self.levels[-1]
+        last.rows, last.nnz = A.nrows, A.nnz
+        if last.A_host is not None:
+            last.A_host = A
+        last.A = backend.matrix(A)
+        if self.prm["direct_coarse"]:
+            self.coarse_solve = backend.coarse_solver(A)
+        else:
+            last.relax = relax_factory(A, backend)
 
     # --- setup (host) ------------------------------------------------------
     def _build(self, A: CSR):
         prm = self.prm
         coarsening = make_coarsening(prm["coarsening"])
+        self._coarsening = coarsening
         relax_factory = make_relaxation_factory(prm["relax"])
         backend = self.backend
 
@@ -119,6 +146,7 @@ class AMG:
             with prof.scope("move_to_backend"):
                 lvl.P = backend.matrix(P)
                 lvl.R = backend.matrix(R)
+            lvl.P_build, lvl.R_build = P, R
             self.levels.append(lvl)
             with prof.scope("galerkin"):
                 A_host = coarsening.coarse_operator(A_host, P, R)

@@ -1,0 +1,129 @@
+"""I/O round-trips, adapters, rebuild, CLI smoke."""
+import os
+
+import numpy as np
+import pytest
+import scipy.sparse as sp
+
+import amgcl_amd as am
+from amgcl_amd import io
+from amgcl_amd.adapter import Reordered, ScaledProblem, zero_copy
+from amgcl_amd.matrix import CSR
+
+
+def rand_csr(rng, n, density=0.1, spd=True):
+    a = sp.random(n, n, density=density, random_state=rng, format="csr")
+    a.data = rng.standard_normal(a.nnz)
+    if spd:
+        a = (a + a.T).tocsr()
+        a = (a + sp.identity(n) * (abs(a).sum(axis=1).max() + 1)).tocsr()
+    a.sort_indices()
+    return CSR(n, n, a.indptr, a.indices, a.data)
+
+
+def test_mm_roundtrip(tmp_path):
+    rng = np.random.default_rng(0)
+    A = rand_csr(rng, 40)
+    p = tmp_path / "a.mtx"
+    io.mm_write(str(p), A)
+    B = io.mm_read(str(p))
+    assert abs(B.to_scipy() - A.to_scipy()).max() < 1e-14
+    v = rng.standard_normal(40)
+    io.mm_write(str(tmp_path / "v.mtx"), v)
+    v2 = io.mm_read(str(tmp_path / "v.mtx")).ravel()
+    np.testing.assert_allclose(v2, v)
+
+
+def test_mm_symmetric_expansion(tmp_path):
+    p = tmp_path / "s.mtx"
+    p.write_text(
+        "%%MatrixMarket matrix coordinate real symmetric\n"
+        "3 3 4\n1 1 2.0\n2 1 -1.0\n2 2 2.0\n3 3 1.0\n"
+    )
+    A = io.mm_read(str(p))
+    d = A.to_dense()
+    assert d[0, 1] == -1.0 and d[1, 0] == -1.0
+
+
+def test_binary_roundtrip_and_strip(tmp_path):
+    rng = np.random.default_rng(1)
+    A = rand_csr(rng, 50)
+    p = str(tmp_path / "a.bin")
+    io.write_crs(p, A)
+    B = io.read_crs(p)
+    assert abs(B.to_scipy() - A.to_scipy()).max() < 1e-15
+    S = io.read_crs(p, row_beg=10, row_end=30)
+    assert S.nrows == 20
+    assert abs(S.to_scipy() - A.to_scipy()[10:30]).max() < 1e-15
+    x = rng.standard_normal(50)
+    io.write_dense(str(tmp_path / "x.bin"), x)
+    np.testing.assert_allclose(io.read_dense(str(tmp_path / "x.bin")), x)
+
+
+def test_zero_copy_no_copy():
+    A, _ = am.poisson3d(8)
+    Z = zero_copy(A.nrows, A.ptr, A.col, A.val)
+    assert Z.val is A.val  # same buffer
+    assert Z.nnz == A.nnz
+
+
+def test_reorder_solves_same_system():
+    A, b = am.poisson3d(10, rhs="random")
+    R = Reordered(A)
+    s = am.make_solver(R.A, {"solver": {"type": "cg", "tol": 1e-10, "maxiter": 100}})
+    y, iters, resid = s(R.forward(b))
+    x = R.inverse(y)
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-8
+    # bandwidth actually reduced
+    row_of = np.repeat(np.arange(R.A.nrows), np.diff(R.A.ptr))
+    bw_new = int(np.abs(row_of - R.A.col).max())
+    row_of0 = np.repeat(np.arange(A.nrows), np.diff(A.ptr))
+    bw_old = int(np.abs(row_of0 - A.col).max())
+    assert bw_new <= bw_old
+
+
+def test_scaled_problem():
+    A, b = am.poisson3d(10, rhs="random")
+    S = ScaledProblem(A)
+    s = am.make_solver(S.A, {"solver": {"type": "cg", "tol": 1e-10, "maxiter": 100}})
+    y, iters, resid = s(S.scale_rhs(b))
+    x = S.unscale_x(y)
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-8
+    assert np.allclose(np.asarray(S.A.diagonal()), 1.0)
+
+
+def test_rebuild_reuses_hierarchy():
+    A, b = am.poisson3d(12, rhs="random")
+    s = am.make_solver(A, {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}})
+    x, it0, _ = s(b)
+    # scale the matrix: same structure, new values
+    A2 = CSR(A.nrows, A.ncols, A.ptr, A.col, A.val * 2.0)
+    s.P.rebuild(A2)
+    x2 = s.backend.vector(A.nrows)
+    it2, resid2 = s.S(s.P, b, x2)
+    assert resid2 < 1e-8
+    assert np.linalg.norm(b - A2 @ x2) / np.linalg.norm(b) < 1e-7
+    assert abs(it2 - it0) <= 2
+
+
+def test_cli_smoke(tmp_path, capsys):
+    from amgcl_amd.cli import main
+
+    rc = main(["--poisson", "12", "-p", "solver.type=cg", "-p", "solver.tol=1e-8",
+               "-o", str(tmp_path / "x.bin")])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "iters:" in out
+    x = io.read_dense(str(tmp_path / "x.bin"))
+    assert x.shape[0] == 12**3
+
+
+def test_cli_mm_with_reorder_scale(tmp_path):
+    from amgcl_amd.cli import main
+
+    rng = np.random.default_rng(3)
+    A = rand_csr(rng, 60)
+    io.mm_write(str(tmp_path / "a.mtx"), A)
+    rc = main(["-A", str(tmp_path / "a.mtx"), "--reorder", "--scale",
+               "-p", "solver.type=bicgstab"])
+    assert rc == 0
