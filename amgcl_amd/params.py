@@ -13,8 +13,11 @@ class UnknownParameter(ValueError):
     pass
 
 
-def merge_params(defaults, prm, path=""):
-    """Merge user dict `prm` onto `defaults`, raising on unknown keys."""
+def merge_params(defaults, prm, path="", opaque=()):
+    """Merge user dict `prm` onto `defaults`, raising on unknown keys.
+    Keys in `opaque` (nested component configs, e.g. a sub-solver tree)
+    are taken verbatim without recursive validation — the sub-component
+    validates its own tree when constructed."""
     out = copy.deepcopy(defaults)
     if prm is None:
         return out
@@ -23,10 +26,10 @@ def merge_params(defaults, prm, path=""):
     for key, value in prm.items():
         if key not in out:
             raise UnknownParameter(f"unknown parameter '{path + key}'")
-        if isinstance(out[key], dict) and not key.endswith("_raw"):
-            out[key] = merge_params(out[key], value, path + key + ".")
-        else:
+        if key in opaque or key.endswith("_raw") or not isinstance(out[key], dict):
             out[key] = value
+        else:
+            out[key] = merge_params(out[key], value, path + key + ".")
     return out
 
 

@@ -16,9 +16,30 @@ REGISTRY = {
 }
 
 
+def _lazy(kind):
+    if kind == "schur_pressure_correction" and kind not in REGISTRY:
+        from .schur import SchurPressureCorrection
+
+        REGISTRY[kind] = SchurPressureCorrection
+    if kind in ("cpr", "cpr_drs") and kind not in REGISTRY:
+        from .cpr import CPR
+
+        REGISTRY["cpr"] = CPR
+
+        class CPRDRS(CPR):
+            @staticmethod
+            def defaults():
+                d = CPR.defaults()
+                d["drs"] = True
+                return d
+
+        REGISTRY["cpr_drs"] = CPRDRS
+
+
 def make_preconditioner(A, prm=None, backend=None):
     prm = dict(prm or {})
     kind = prm.pop("class", "amg")
+    _lazy(kind)
     if kind not in REGISTRY:
         raise ValueError(f"unknown preconditioner class '{kind}'")
     return REGISTRY[kind](A, prm, backend)
