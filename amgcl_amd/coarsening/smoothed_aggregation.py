@@ -21,11 +21,15 @@ class SmoothedAggregation:
             "relax": 1.0,
             "estimate_spectral_radius": False,
             "power_iters": 5,
+            "nullspace_raw": None,  # (n x k) near-nullspace block (e.g. RBM)
+            "block_size": 1,        # dofs per grid point (pointwise aggregation)
         }
 
     def __init__(self, prm=None):
         self.prm = merge_params(self.defaults(), prm)
         self.eps_strong = float(self.prm["eps_strong"])
+        ns = self.prm["nullspace_raw"]
+        self.B = None if ns is None else np.ascontiguousarray(ns, dtype=np.float64)
 
     def transfer_operators(self, A):
         from ..profiler import prof
@@ -36,8 +40,23 @@ class SmoothedAggregation:
             # large levels: deterministic parallel MIS(2) aggregation
             # (mirrors the reference's own distributed pmis design);
             # small levels: the exact greedy reference pass.
-            agg = _core.aggregates_parallel if A.nrows > 100_000 else _core.aggregates
-            naggr, aggr_id, strong = agg(A.nrows, A.ptr, A.col, A.val, self.eps_strong)
+            bsize = int(self.prm["block_size"])
+            if bsize > 1 and A.nrows % bsize == 0:
+                # pointwise (block) aggregation: condense to points, aggregate,
+                # expand (parity: coarsening/pointwise_aggregates.hpp:85)
+                pp, pc, pv = _core.pointwise_matrix(A.nrows, A.ptr, A.col, A.val, bsize)
+                np_ = A.nrows // bsize
+                agg = _core.aggregates_parallel if np_ > 100_000 else _core.aggregates
+                naggr, id_p, strong_p = agg(np_, pp, pc, pv, self.eps_strong)
+                aggr_id = np.where(
+                    np.repeat(np.asarray(id_p), bsize) >= 0,
+                    np.repeat(np.asarray(id_p), bsize), -2
+                ).astype(np.int32)
+                strong = _core.expand_strong(A.nrows, A.ptr, A.col, bsize,
+                                             pp, pc, strong_p)
+            else:
+                agg = _core.aggregates_parallel if A.nrows > 100_000 else _core.aggregates
+                naggr, aggr_id, strong = agg(A.nrows, A.ptr, A.col, A.val, self.eps_strong)
         self.eps_strong *= 0.5  # reference halves eps per level (s_a.hpp:140)
 
         omega = float(self.prm["relax"])
@@ -47,10 +66,29 @@ class SmoothedAggregation:
             omega *= 2.0 / 3.0
 
         with prof.scope("smooth_P"):
-            pp, pc, pv = _core.smoothed_prolongation(
-                A.nrows, A.ptr, A.col, A.val, strong, aggr_id, naggr, omega
-            )
-            P = CSR(A.nrows, naggr, pp, pc, pv)
+            if self.B is None:
+                pp, pc, pv = _core.smoothed_prolongation(
+                    A.nrows, A.ptr, A.col, A.val, strong, aggr_id, naggr, omega
+                )
+                P = CSR(A.nrows, naggr, pp, pc, pv)
+            else:
+                # nullspace path: per-aggregate QR tentative + explicit
+                # filtered-Jacobi smoother matrix (reference: tentative_
+                # prolongation.hpp:134-207 + smoothed_aggregation smoothing)
+                k = self.B.shape[1]
+                tp, tc, tv, Bnew = _core.tentative_nullspace(
+                    A.nrows, aggr_id, naggr, self.B, k
+                )
+                P_tent = CSR(A.nrows, naggr * k, tp, tc, tv)
+                sp_, sc_, sv_ = _core.filtered_smoother_matrix(
+                    A.nrows, A.ptr, A.col, A.val, strong, omega
+                )
+                S_F = CSR(A.nrows, A.nrows, sp_, sc_, sv_)
+                P = S_F @ P_tent
+                self.B = np.asarray(Bnew).reshape(naggr * k, k)
+                self.prm["block_size"] = 1  # coarse levels: k dofs per aggregate
+                if k > 1:
+                    self.prm["block_size"] = k  # aggregate coarse points (k dofs each)
         with prof.scope("transpose_R"):
             R = P.transpose()
         return P, R

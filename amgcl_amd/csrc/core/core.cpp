@@ -820,6 +820,261 @@ static py::tuple tentative_prolongation(i64 nrows, arr<i32> id_a, i64 naggr) {
 }
 
 // ---------------------------------------------------------------------------
+// Pointwise (block) condensation for vector problems
+// (parity: amgcl/backend/builtin.hpp:505 pointwise_matrix +
+// amgcl/coarsening/pointwise_aggregates.hpp:85): condense the scalar matrix
+// with node-major interleaved dofs (block_size unknowns per point) to a
+// point matrix whose values are Frobenius norms of the blocks; aggregation
+// then runs on points and ids/strength expand back to unknowns.
+// ---------------------------------------------------------------------------
+static py::tuple pointwise_matrix(i64 n, arr<i32> ptr_a, arr<i32> col_a,
+                                  arr<double> val_a, i64 b) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    if (n % b) throw std::runtime_error("pointwise: size not divisible by block");
+    const i64 np_ = n / b;
+    arr<i32> pptr(np_ + 1);
+    i32 *PP = pptr.mutable_data();
+    PP[0] = 0;
+#pragma omp parallel
+    {
+        RowHash hash(256);
+#pragma omp for schedule(static)
+        for (i64 p = 0; p < np_; ++p) {
+            i64 ub = 0;
+            for (i64 r = p * b; r < (p + 1) * b; ++r) ub += A.ptr[r + 1] - A.ptr[r];
+            hash.ensure(ub);
+            i32 cnt = 0;
+            for (i64 r = p * b; r < (p + 1) * b; ++r)
+                for (i32 j = A.ptr[r]; j < A.ptr[r + 1]; ++j)
+                    cnt += hash.count_add(A.col[j] / (i32)b);
+            hash.reset();
+            PP[p + 1] = cnt;
+        }
+    }
+    scan_i32(PP + 1, np_);
+    arr<i32> pcol = big_arr<i32>(PP[np_]);
+    arr<double> pval = big_arr<double>(PP[np_]);
+    i32 *PC = pcol.mutable_data();
+    double *PV = pval.mutable_data();
+#pragma omp parallel
+    {
+        RowHash hash(256);
+#pragma omp for schedule(static)
+        for (i64 p = 0; p < np_; ++p) {
+            i64 ub = 0;
+            for (i64 r = p * b; r < (p + 1) * b; ++r) ub += A.ptr[r + 1] - A.ptr[r];
+            hash.ensure(ub);
+            for (i64 r = p * b; r < (p + 1) * b; ++r)
+                for (i32 j = A.ptr[r]; j < A.ptr[r + 1]; ++j)
+                    hash.add(A.col[j] / (i32)b, A.val[j] * A.val[j]);
+            i32 head = PP[p];
+            for (i32 h : hash.used) {
+                PC[head] = hash.keys[h];
+                PV[head] = std::sqrt(hash.vals[h]);  // Frobenius norm of block
+                ++head;
+            }
+            hash.reset();
+            // sort row
+            i32 beg = PP[p];
+            for (i32 k2 = beg + 1; k2 < head; ++k2) {
+                i32 ck = PC[k2];
+                double vk = PV[k2];
+                i32 m = k2;
+                while (m > beg && PC[m - 1] > ck) {
+                    PC[m] = PC[m - 1];
+                    PV[m] = PV[m - 1];
+                    --m;
+                }
+                PC[m] = ck;
+                PV[m] = vk;
+            }
+        }
+    }
+    return py::make_tuple(pptr, pcol, pval);
+}
+
+// Expand a point-level strong mask to the scalar entries: scalar entry (i,j)
+// is strong iff the point pair (i/b, col/b) is strong in the point matrix.
+static arr<uint8_t> expand_strong(i64 n, arr<i32> ptr_a, arr<i32> col_a, i64 b,
+                                  arr<i32> pptr_a, arr<i32> pcol_a,
+                                  arr<uint8_t> pstrong_a) {
+    const i32 *ptr = ptr_a.data();
+    const i32 *col = col_a.data();
+    const i32 *pp = pptr_a.data();
+    const i32 *pc = pcol_a.data();
+    const uint8_t *ps = pstrong_a.data();
+    const i64 nnz = (i64)col_a.size();
+    arr<uint8_t> S(nnz);
+    uint8_t *Sv = S.mutable_data();
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        i32 p = (i32)(i / b);
+        for (i32 j = ptr[i]; j < ptr[i + 1]; ++j) {
+            i32 q = col[j] / (i32)b;
+            uint8_t st = 0;
+            if (q != p) {
+                // binary search q in point row p (sorted)
+                i32 lo = pp[p], hi = pp[p + 1] - 1;
+                while (lo <= hi) {
+                    i32 mid = (lo + hi) / 2;
+                    if (pc[mid] == q) { st = ps[mid]; break; }
+                    if (pc[mid] < q) lo = mid + 1;
+                    else hi = mid - 1;
+                }
+            }
+            Sv[j] = st;
+        }
+    }
+    return S;
+}
+
+// ---------------------------------------------------------------------------
+// Nullspace-aware tentative prolongation
+// (parity: amgcl/coarsening/tentative_prolongation.hpp:120-207 — per
+// aggregate, a thin QR of the near-nullspace block B gives the P block (Q)
+// and the coarse-level nullspace (R)). Modified Gram-Schmidt with one
+// re-orthogonalization pass (k <= ~6 columns).
+// Returns (Pptr, Pcol, Pval, Bnew) with P: n x (naggr*k).
+// ---------------------------------------------------------------------------
+static py::tuple tentative_nullspace(i64 n, arr<i32> id_a, i64 naggr,
+                                     arr<double> B_a, i64 k) {
+    const i32 *id = id_a.data();
+    const double *B = B_a.data();  // n x k row-major
+
+    // group rows by aggregate
+    std::vector<i32> acnt(naggr + 1, 0);
+    for (i64 i = 0; i < n; ++i)
+        if (id[i] >= 0) ++acnt[id[i] + 1];
+    for (i64 a = 0; a < naggr; ++a) acnt[a + 1] += acnt[a];
+    std::vector<i32> members(acnt[naggr]);
+    {
+        std::vector<i32> cur(acnt.begin(), acnt.end() - 1);
+        for (i64 i = 0; i < n; ++i)
+            if (id[i] >= 0) members[cur[id[i]]++] = (i32)i;
+    }
+
+    arr<i32> pptr(n + 1);
+    i32 *PP = pptr.mutable_data();
+    PP[0] = 0;
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) PP[i + 1] = (id[i] >= 0) ? (i32)k : 0;
+    scan_i32(PP + 1, n);
+    const i64 pnz = PP[n];
+    arr<i32> pcol = big_arr<i32>(pnz);
+    arr<double> pval = big_arr<double>(pnz);
+    i32 *PC = pcol.mutable_data();
+    double *PV = pval.mutable_data();
+    arr<double> Bnew_a(naggr * k * k);
+    double *Bnew = Bnew_a.mutable_data();
+
+#pragma omp parallel
+    {
+        std::vector<double> Q;  // d x k column-major
+#pragma omp for schedule(dynamic, 256)
+        for (i64 a = 0; a < naggr; ++a) {
+            i64 beg = acnt[a], end = acnt[a + 1];
+            i64 d = end - beg;
+            Q.assign(d * k, 0.0);
+            for (i64 r = 0; r < d; ++r)
+                for (i64 c = 0; c < (i64)k; ++c)
+                    Q[c * d + r] = B[(i64)members[beg + r] * k + c];
+            double *Rm = Bnew + a * k * k;  // k x k row-major
+            std::fill(Rm, Rm + k * k, 0.0);
+            // MGS with one re-orthogonalization
+            for (i64 c = 0; c < (i64)k; ++c) {
+                double *qc = &Q[c * d];
+                for (int pass = 0; pass < 2; ++pass) {
+                    for (i64 p = 0; p < c; ++p) {
+                        const double *qp = &Q[p * d];
+                        double h = 0;
+                        for (i64 r = 0; r < d; ++r) h += qp[r] * qc[r];
+                        for (i64 r = 0; r < d; ++r) qc[r] -= h * qp[r];
+                        Rm[p * k + c] += h;
+                    }
+                }
+                double nrm = 0;
+                for (i64 r = 0; r < d; ++r) nrm += qc[r] * qc[r];
+                nrm = std::sqrt(nrm);
+                Rm[c * k + c] = nrm;
+                if (nrm > 1e-300)
+                    for (i64 r = 0; r < d; ++r) qc[r] /= nrm;
+            }
+            // write P rows
+            for (i64 r = 0; r < d; ++r) {
+                i32 row = members[beg + r];
+                i32 head = PP[row];
+                for (i64 c = 0; c < (i64)k; ++c) {
+                    PC[head + c] = (i32)(a * k + c);
+                    PV[head + c] = Q[c * d + r];
+                }
+            }
+        }
+    }
+    return py::make_tuple(pptr, pcol, pval, Bnew_a);
+}
+
+// Filtered Jacobi smoother matrix S_F = I - omega * Df^-1 * Af as explicit
+// CSR (strong off-diagonals kept, weak ones folded into the diagonal) —
+// used to smooth a general (nullspace) tentative P via spgemm
+// (parity: the implicit smoothing pass of smoothed_aggregation.hpp:157-232).
+static py::tuple filtered_smoother_matrix(i64 n, arr<i32> ptr_a, arr<i32> col_a,
+                                          arr<double> val_a, arr<uint8_t> strong_a,
+                                          double omega) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    const uint8_t *S = strong_a.data();
+    arr<i32> sptr(n + 1);
+    i32 *SP = sptr.mutable_data();
+    SP[0] = 0;
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        i32 cnt = 0;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j)
+            if (A.col[j] == (i32)i || S[j]) ++cnt;
+        // ensure a diagonal entry even if A has none
+        bool has_dia = false;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j)
+            if (A.col[j] == (i32)i) { has_dia = true; break; }
+        if (!has_dia) ++cnt;
+        SP[i + 1] = cnt;
+    }
+    scan_i32(SP + 1, n);
+    arr<i32> scol = big_arr<i32>(SP[n]);
+    arr<double> sval = big_arr<double>(SP[n]);
+    i32 *SC = scol.mutable_data();
+    double *SV = sval.mutable_data();
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        double dia = 0;
+        bool has_dia = false;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            if (A.col[j] == (i32)i) has_dia = true;
+            if (A.col[j] == (i32)i || !S[j]) dia += A.val[j];
+        }
+        double w = (dia != 0.0) ? -omega / dia : 0.0;
+        i32 head = SP[i];
+        bool wrote_dia = false;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i32 c = A.col[j];
+            if (c == (i32)i) {
+                SC[head] = c;
+                SV[head] = 1.0 - omega;
+                wrote_dia = true;
+                ++head;
+            } else if (S[j]) {
+                SC[head] = c;
+                SV[head] = w * A.val[j];
+                ++head;
+            }
+        }
+        if (!has_dia && !wrote_dia) {
+            SC[head] = (i32)i;
+            SV[head] = 1.0 - omega;
+        }
+    }
+    return py::make_tuple(sptr, scol, sval);
+}
+
+// ---------------------------------------------------------------------------
 // Ruge-Stuben (classic) coarsening
 // (parity: amgcl/coarsening/ruge_stuben.hpp:54-458 — strong negative
 // couplings -a_ij >= eps*max|a_ik^-|, standard lambda-bucket C/F splitting,
@@ -1164,6 +1419,10 @@ PYBIND11_MODULE(_core, m) {
     m.def("tentative_prolongation", &tentative_prolongation);
     m.def("spai0", &spai0);
     m.def("ruge_stuben", &ruge_stuben);
+    m.def("tentative_nullspace", &tentative_nullspace);
+    m.def("pointwise_matrix", &pointwise_matrix);
+    m.def("expand_strong", &expand_strong);
+    m.def("filtered_smoother_matrix", &filtered_smoother_matrix);
     m.def("spmv", &spmv);
     m.def("residual", &residual);
     m.def("gauss_seidel", &gauss_seidel);

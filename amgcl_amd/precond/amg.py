@@ -54,7 +54,7 @@ class AMG:
 
             backend = make_backend("cpu")
         self.backend = backend
-        self.prm = merge_params(self.defaults(), prm)
+        self.prm = merge_params(self.defaults(), prm, opaque=("coarsening", "relax"))
         self.levels = []
         self.coarse_solve = None
         self._coarsening = None
