@@ -64,7 +64,14 @@ class DistBackend:
 
     # primitives
     def spmv(self, alpha, A, x, beta, y):
-        if isinstance(A, DistMatrix):
+        from .deflation import ProjectedDistMatrix
+
+        if isinstance(A, ProjectedDistMatrix):
+            if beta != 0.0:
+                raise ValueError("projected spmv supports beta=0 only")
+            self.spmv(alpha, A.inner, x, 0.0, y)
+            A.defl.project(y)
+        elif isinstance(A, DistMatrix):
             works = A.start_exchange(x)
             self.base.spmv(alpha, A.A_loc, x, beta, y)  # overlapped with halo
             A.finish_exchange(works)
@@ -74,7 +81,12 @@ class DistBackend:
             self.base.spmv(alpha, A, x, beta, y)
 
     def residual(self, b, A, x, r):
-        if isinstance(A, DistMatrix):
+        from .deflation import ProjectedDistMatrix
+
+        if isinstance(A, ProjectedDistMatrix):
+            self.spmv(1.0, A, x, 0.0, r)  # r = P A x
+            self.base.axpby(1.0, b, -1.0, r)  # r = b - P A x
+        elif isinstance(A, DistMatrix):
             works = A.start_exchange(x)
             self.base.residual(b, A.A_loc, x, r)
             A.finish_exchange(works)

@@ -16,7 +16,7 @@ class DistSolver:
         self.backend = DistBackend(backend, group)
         prm = dict(prm or {})
         for key in prm:
-            if key not in ("precond", "solver"):
+            if key not in ("precond", "solver", "deflation"):
                 raise ValueError(f"unknown parameter '{key}'")
         self.A = self.backend.matrix(strip)
         self.P = LocalBlockPrecond(self.A, prm.get("precond"), self.backend)
@@ -24,6 +24,19 @@ class DistSolver:
         self.S = make_solver_component(
             self.A.n_loc, prm.get("solver"), self.backend, self.inner
         )
+        self.defl = None
+        dprm = prm.get("deflation")
+        if dprm:
+            from .deflation import ProjectedDistMatrix, SubdomainDeflation
+
+            dprm = dict(dprm)
+            self.defl = SubdomainDeflation(
+                self.A, self.backend, dprm.pop("type", "constant"),
+                dprm.pop("coords_raw", None),
+            )
+            if dprm:
+                raise ValueError(f"unknown deflation parameters {list(dprm)}")
+            self._proj = ProjectedDistMatrix(self.A, self.defl)
 
     def __call__(self, rhs, x=None):
         b = self.backend
@@ -31,7 +44,25 @@ class DistSolver:
             rhs = b.from_host(rhs)
         if x is None:
             x = b.vector(self.A.n_loc)
-        iters, resid = self.S(self.P, rhs, x, A=self.A)
+        if self.defl is None:
+            iters, resid = self.S(self.P, rhs, x, A=self.A)
+            return x, iters, resid
+        # deflated solve (parity: mpi/subdomain_deflation.hpp:476-479):
+        # x0 = Z E^-1 Z^T rhs;  solve (P A) y = P (rhs - A x0);  x = x0 +
+        # (I - Z E^-1 Z^T A) y
+        defl = self.defl
+        x0 = b.vector(self.A.n_loc)
+        defl.coarse_guess(rhs, x0)
+        r = b.vector(self.A.n_loc)
+        b.residual(rhs, self.A, x0, r)
+        defl.project(r)
+        y = x  # reuse the output buffer for the projected iterate
+        b.clear(y)
+        iters, resid = self.S(self.P, r, y, A=self._proj)
+        Ay = b.vector(self.A.n_loc)
+        b.spmv(1.0, self.A, y, 0.0, Ay)
+        xfinal = defl.post_correct(y, Ay, x0)
+        b.copy(xfinal, x)
         return x, iters, resid
 
     def gather_solution(self, x):

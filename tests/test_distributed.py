@@ -116,3 +116,46 @@ def test_dist_inner_product_single_rank_passthrough():
     strip, b, rb, re_ = am.poisson3d_strip(n, 0, 2)
     assert strip.nrows == n**3 // 2
     assert strip.ncols == n**3
+
+
+def _solve_deflated(rank, world):
+    import numpy as np
+
+    import amgcl_amd as am
+    from amgcl_amd.parallel import make_dist_solver
+
+    n = 20
+    strip, b, row_beg, row_end = am.poisson3d_strip(n, rank, world, rhs="ones")
+    # node coordinates of the local rows (for linear deflation)
+    idx = np.arange(row_beg, row_end)
+    coords = np.stack([idx % n, (idx // n) % n, idx // (n * n)], axis=1)
+    results = {}
+    for defl in (None, {"type": "constant"}, {"type": "linear", "coords_raw": coords}):
+        prm = {"precond": {"class": "amg", "coarse_enough": 200},
+               "solver": {"type": "cg", "tol": 1e-8, "maxiter": 300}}
+        if defl:
+            prm["deflation"] = defl
+        solve = make_dist_solver(strip, prm, backend="cpu")
+        x, iters, resid = solve(b)
+        xg = solve.gather_solution(x)
+        key = defl["type"] if defl else "none"
+        results[key] = (iters, resid, None if xg is None else xg.tolist())
+    return results
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_subdomain_deflation(world):
+    out = spawn(world, _solve_deflated, 29811 + world)
+    import amgcl_amd as am
+
+    n = 20
+    A, b = am.poisson3d(n, rhs="ones")
+    r0 = out[0]
+    for kind in ("none", "constant", "linear"):
+        iters, resid, xg = r0[kind]
+        assert resid < 1e-7, kind
+        x = np.array(xg)
+        assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6, kind
+    # deflation must not be worse than plain block-Jacobi
+    assert r0["constant"][0] <= r0["none"][0] + 2
+    assert r0["linear"][0] <= r0["constant"][0] + 2
