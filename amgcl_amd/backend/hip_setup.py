@@ -168,7 +168,7 @@ def transpose(A):
     return device_csr(A.ncols, A.nrows, tptr, tcol, tval)
 
 
-def spgemm(A, B):
+def spgemm(A, B, sort=True):
     t = _torch()
     dev = A.val.device
     cptr = _new_ptr(A.nrows, dev)
@@ -187,7 +187,8 @@ def spgemm(A, B):
     check(lib().amg_spgemm_fill(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
                                 A.val.data_ptr(), B.ptr.data_ptr(), B.col.data_ptr(),
                                 B.val.data_ptr(), ub.data_ptr(), cptr[1:].data_ptr(),
-                                ccol.data_ptr(), cval.data_ptr(), _stream()),
+                                ccol.data_ptr(), cval.data_ptr(), 1 if sort else 0,
+                                _stream()),
           "spgemm_fill")
     return device_csr(A.nrows, B.ncols, cptr, ccol, cval)
 

@@ -611,7 +611,8 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
                                     const double *__restrict__ bval,
                                     const int *__restrict__ ub,
                                     const int *__restrict__ cptr_scanned,
-                                    int *__restrict__ ccol, double *__restrict__ cval) {
+                                    int *__restrict__ ccol, double *__restrict__ cval,
+                                    int do_sort) {
     __shared__ int keys[32][SSLOTS];
     __shared__ double vals[32][SSLOTS];
     int gid = threadIdx.x / SGRP;
@@ -647,15 +648,37 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
-        // sorted extraction: empties to +inf, bitonic, write the prefix
-        for (int t = lane; t < SSLOTS; t += SGRP)
-            if (tk[t] == -1) tk[t] = 0x7fffffff;
-        lds_bitonic<SGRP, SSLOTS>(tk, tv, lane);
         int base = (row == 0) ? 0 : cptr_scanned[row - 1];
         int len = cptr_scanned[row] - base;
-        for (int t = lane; t < len; t += SGRP) {
-            ccol[base + t] = tk[t];
-            cval[base + t] = tv[t];
+        if (do_sort) {
+            // sorted extraction: empties to +inf, bitonic, write the prefix
+            for (int t = lane; t < SSLOTS; t += SGRP)
+                if (tk[t] == -1) tk[t] = 0x7fffffff;
+            lds_bitonic<SGRP, SSLOTS>(tk, tv, lane);
+            for (int t = lane; t < len; t += SGRP) {
+                ccol[base + t] = tk[t];
+                cval[base + t] = tv[t];
+            }
+        } else {
+            // unsorted compaction (intermediate products: order irrelevant);
+            // per-lane local count -> exclusive shfl prefix -> direct writes
+            int mine = 0;
+            for (int t = lane; t < SSLOTS; t += SGRP)
+                if (tk[t] != -1) ++mine;
+            int off = mine;
+#pragma unroll
+            for (int d = 1; d < SGRP; d <<= 1) {
+                int v = __shfl_up(off, d, SGRP);
+                if (lane >= d) off += v;
+            }
+            off -= mine;  // exclusive prefix within the group
+            int h = base + off;
+            for (int t = lane; t < SSLOTS; t += SGRP)
+                if (tk[t] != -1) {
+                    ccol[h] = tk[t];
+                    cval[h] = tv[t];
+                    ++h;
+                }
         }
         __builtin_amdgcn_wave_barrier();
     }
@@ -742,7 +765,7 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
                               const int *__restrict__ bptr, const int *__restrict__ bcol,
                               const double *__restrict__ bval, const int *__restrict__ ub,
                               const int *__restrict__ cptr_scanned, int *__restrict__ ccol,
-                              double *__restrict__ cval) {
+                              double *__restrict__ cval, int do_sort) {
     __shared__ int keys[4][HSLOTS];
     __shared__ double vals[4][HSLOTS];
     __shared__ int pfx[4][BIGROW + 1];
@@ -794,14 +817,34 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
-        for (int t = lane; t < HSLOTS; t += WAVE)
-            if (tk[t] == -1) tk[t] = 0x7fffffff;
-        lds_bitonic<WAVE, HSLOTS>(tk, tv, lane);
         int base = (row == 0) ? 0 : cptr_scanned[row - 1];
         int len = cptr_scanned[row] - base;
-        for (int t = lane; t < len; t += WAVE) {
-            ccol[base + t] = tk[t];
-            cval[base + t] = tv[t];
+        if (do_sort) {
+            for (int t = lane; t < HSLOTS; t += WAVE)
+                if (tk[t] == -1) tk[t] = 0x7fffffff;
+            lds_bitonic<WAVE, HSLOTS>(tk, tv, lane);
+            for (int t = lane; t < len; t += WAVE) {
+                ccol[base + t] = tk[t];
+                cval[base + t] = tv[t];
+            }
+        } else {
+            int mine = 0;
+            for (int t = lane; t < HSLOTS; t += WAVE)
+                if (tk[t] != -1) ++mine;
+            int off = mine;
+#pragma unroll
+            for (int d = 1; d < WAVE; d <<= 1) {
+                int v = __shfl_up(off, d, WAVE);
+                if (lane >= d) off += v;
+            }
+            off -= mine;
+            int h = base + off;
+            for (int t = lane; t < HSLOTS; t += WAVE)
+                if (tk[t] != -1) {
+                    ccol[h] = tk[t];
+                    cval[h] = tv[t];
+                    ++h;
+                }
         }
         __builtin_amdgcn_wave_barrier();
     }
@@ -861,10 +904,11 @@ extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol, co
 extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol, const double *aval,
                                const int *bptr, const int *bcol, const double *bval,
                                const int *ub, const int *cptr_scanned, int *ccol,
-                               double *cval, hipStream_t s) {
+                               double *cval, int do_sort, hipStream_t s) {
     spgemm_fill_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol,
-                                                        bval, ub, cptr_scanned, ccol, cval);
+                                                        bval, ub, cptr_scanned, ccol, cval,
+                                                        do_sort);
     spgemm_fill_k<<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol, bval,
-                                                  ub, cptr_scanned, ccol, cval);
+                                                  ub, cptr_scanned, ccol, cval, do_sort);
     return (int)hipGetLastError();
 }

@@ -105,7 +105,7 @@ def galerkin(R, A, P):
 
         try:
             with prof.scope("galerkin(dev)"):
-                return hip_setup.spgemm(R, hip_setup.spgemm(A, P))
+                return hip_setup.spgemm(R, hip_setup.spgemm(A, P, sort=False))
         except OverflowError:
             Rh = hip_setup.download(R)
             Ah = hip_setup.download(A)
