@@ -37,6 +37,29 @@ _SIGS = {
                         ctypes.c_void_p],
     "amg_gemv_f64": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
                      ctypes.c_void_p],
+    # --- fp32 variants (mixed precision) + casts ---
+    "amg_spmv_f32": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_void_p, ctypes.c_void_p, ctypes.c_double, ctypes.c_double,
+                     ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+    "amg_residual_f32": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+                         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                         ctypes.c_int, ctypes.c_void_p],
+    "amg_relax_diag_f32": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p,
+                           ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+                           ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p],
+    "amg_axpby_f32": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_double,
+                      ctypes.c_void_p, ctypes.c_void_p],
+    "amg_axpbypcz_f32": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_double,
+                         ctypes.c_void_p, ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p],
+    "amg_vmul_f32": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p,
+                     ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p],
+    "amg_fill_f32": [ctypes.c_int64, ctypes.c_double, ctypes.c_void_p, ctypes.c_void_p],
+    "amg_dot_f32": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
+    "amg_gather_f32": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
+    "amg_scatter_f32": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
+    "amg_gemv_f32": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
+    "amg_cast_d2s": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
+    "amg_cast_s2d": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     # --- device-side setup engine (setup.hip) ---
     "amg_setup_diag": [ctypes.c_int64] + [ctypes.c_void_p] * 5,
     "amg_setup_strong": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,

@@ -75,8 +75,8 @@ class HipBackend:
             return csr
         return DeviceCSR(csr, self.device)
 
-    def vector(self, n):
-        return self.torch.zeros(n, dtype=self.torch.float64, device=self.device)
+    def vector(self, n, dtype=None):
+        return self.torch.zeros(n, dtype=dtype or self.torch.float64, device=self.device)
 
     def from_host(self, a):
         t = self.torch.from_numpy(np.ascontiguousarray(a, dtype=np.float64))
@@ -86,45 +86,66 @@ class HipBackend:
         return v.cpu().numpy()
 
     # --- primitives --------------------------------------------------------
+    @staticmethod
+    def _fn(name, t):
+        import torch
+
+        sfx = "_f32" if t.dtype == torch.float32 else "_f64"
+        return getattr(lib(), "amg_" + name + sfx)
+
     def spmv(self, alpha, A, x, beta, y):
-        check(lib().amg_spmv_f64(A.nrows, A.nnz, A.ptr.data_ptr(), A.col.data_ptr(),
-                                 A.val.data_ptr(), x.data_ptr(), alpha, beta,
-                                 y.data_ptr(), A.subw, _stream()), "spmv")
+        check(self._fn("spmv", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(), A.col.data_ptr(),
+                                      A.val.data_ptr(), x.data_ptr(), alpha, beta,
+                                      y.data_ptr(), A.subw, _stream()), "spmv")
 
     def residual(self, b, A, x, r):
-        check(lib().amg_residual_f64(A.nrows, A.nnz, A.ptr.data_ptr(), A.col.data_ptr(),
-                                     A.val.data_ptr(), b.data_ptr(), x.data_ptr(),
-                                     r.data_ptr(), A.subw, _stream()), "residual")
+        check(self._fn("residual", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(),
+                                          A.col.data_ptr(), A.val.data_ptr(),
+                                          b.data_ptr(), x.data_ptr(), r.data_ptr(),
+                                          A.subw, _stream()), "residual")
 
     def relax_diag(self, A, M, rhs, x, t):
         """t = M∘(rhs - A x); x += t (fused single pass over A + axpby)."""
-        check(lib().amg_relax_diag_f64(A.nrows, A.nnz, A.ptr.data_ptr(), A.col.data_ptr(),
-                                       A.val.data_ptr(), M.data_ptr(), rhs.data_ptr(),
-                                       x.data_ptr(), t.data_ptr(), A.subw, _stream()),
+        check(self._fn("relax_diag", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(),
+                                            A.col.data_ptr(), A.val.data_ptr(),
+                                            M.data_ptr(), rhs.data_ptr(), x.data_ptr(),
+                                            t.data_ptr(), A.subw, _stream()),
               "relax_diag")
         self.axpby(1.0, t, 1.0, x)
 
     def clear(self, x):
-        check(lib().amg_fill_f64(x.numel(), 0.0, x.data_ptr(), _stream()), "fill")
+        check(self._fn("fill", x)(x.numel(), 0.0, x.data_ptr(), _stream()), "fill")
 
     def copy(self, x, y):
         y.copy_(x)
 
     def axpby(self, a, x, b, y):
-        check(lib().amg_axpby_f64(x.numel(), a, x.data_ptr(), b, y.data_ptr(),
-                                  _stream()), "axpby")
+        check(self._fn("axpby", x)(x.numel(), a, x.data_ptr(), b, y.data_ptr(),
+                                   _stream()), "axpby")
 
     def axpbypcz(self, a, x, b, y, c, z):
-        check(lib().amg_axpbypcz_f64(x.numel(), a, x.data_ptr(), b, y.data_ptr(), c,
-                                     z.data_ptr(), _stream()), "axpbypcz")
+        check(self._fn("axpbypcz", x)(x.numel(), a, x.data_ptr(), b, y.data_ptr(), c,
+                                      z.data_ptr(), _stream()), "axpbypcz")
 
     def vmul(self, a, m, x, b, z):
-        check(lib().amg_vmul_f64(x.numel(), a, m.data_ptr(), x.data_ptr(), b,
-                                 z.data_ptr(), _stream()), "vmul")
+        check(self._fn("vmul", x)(x.numel(), a, m.data_ptr(), x.data_ptr(), b,
+                                  z.data_ptr(), _stream()), "vmul")
+
+    def cast(self, src, dst):
+        import torch
+
+        if src.dtype == dst.dtype:
+            dst.copy_(src)
+        elif src.dtype == torch.float64:
+            check(lib().amg_cast_d2s(src.numel(), src.data_ptr(), dst.data_ptr(),
+                                     _stream()), "cast")
+        else:
+            check(lib().amg_cast_s2d(src.numel(), src.data_ptr(), dst.data_ptr(),
+                                     _stream()), "cast")
 
     def dot(self, x, y):
-        check(lib().amg_dot_f64(x.numel(), x.data_ptr(), y.data_ptr(),
-                                self._dotbuf.data_ptr(), _stream()), "dot")
+        check(self._fn("dot", x)(x.numel(), x.data_ptr(), y.data_ptr(),
+                                 self._dotbuf.data_ptr(), _stream()), "dot")
         return float(self._dotbuf[0].item())
 
     def dot2(self, x1, y1, x2, y2):
@@ -135,12 +156,12 @@ class HipBackend:
         return float(self._dothost[0]), float(self._dothost[1])
 
     def gather(self, x, idx, buf):
-        check(lib().amg_gather_f64(idx.numel(), x.data_ptr(), idx.data_ptr(),
-                                   buf.data_ptr(), _stream()), "gather")
+        check(self._fn("gather", x)(idx.numel(), x.data_ptr(), idx.data_ptr(),
+                                    buf.data_ptr(), _stream()), "gather")
 
     def scatter(self, buf, idx, x):
-        check(lib().amg_scatter_f64(idx.numel(), buf.data_ptr(), idx.data_ptr(),
-                                    x.data_ptr(), _stream()), "scatter")
+        check(self._fn("scatter", x)(idx.numel(), buf.data_ptr(), idx.data_ptr(),
+                                     x.data_ptr(), _stream()), "scatter")
 
     # --- coarse direct solver ---------------------------------------------
     def coarse_solver(self, csr):
@@ -168,5 +189,8 @@ class DeviceDenseSolver:
         self.inv = backend.from_host(inv.ravel())
 
     def __call__(self, f, u):
-        check(lib().amg_gemv_f64(self.n, self.inv.data_ptr(), f.data_ptr(),
-                                 u.data_ptr(), _stream()), "gemv")
+        import torch
+
+        fn = lib().amg_gemv_f32 if self.inv.dtype == torch.float32 else lib().amg_gemv_f64
+        check(fn(self.n, self.inv.data_ptr(), f.data_ptr(), u.data_ptr(), _stream()),
+              "gemv")

@@ -92,6 +92,8 @@ class NativeDriver:
         levels = amg.levels
         if not all(isinstance(l.A, DeviceCSR) for l in levels):
             raise TypeError("native driver needs device-resident levels")
+        if getattr(amg, "_mixed", False):
+            raise TypeError("native driver is fp64-only (mixed uses the generic path)")
         for l in levels[:-1]:
             if not isinstance(l.relax, DiagonalSmootherBase):
                 raise TypeError("native driver supports diagonal smoothers only")

@@ -32,11 +32,11 @@ extern "C" int amg_hip_last_error() { return (int)hipGetLastError(); }
 // SUBW lanes cooperate on one row (SUBW=1 -> thread per row). The variant is
 // chosen by the host from the mean row length. BETA0 avoids reading y.
 // ---------------------------------------------------------------------------
-template <int SUBW, bool BETA0>
+template <typename T, int SUBW, bool BETA0>
 __global__ void spmv_k(int nrows, const int *__restrict__ ptr,
-                       const int *__restrict__ col, const double *__restrict__ val,
-                       const double *__restrict__ x, double alpha, double beta,
-                       double *__restrict__ y) {
+                       const int *__restrict__ col, const T *__restrict__ val,
+                       const T *__restrict__ x, double alpha, double beta,
+                       T *__restrict__ y) {
     int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
@@ -44,19 +44,19 @@ __global__ void spmv_k(int nrows, const int *__restrict__ ptr,
     for (; row < nrows; row += stride) {
         double s = 0.0;
         int b = ptr[row], e = ptr[row + 1];
-        for (int j = b + lane; j < e; j += SUBW) s += val[j] * x[col[j]];
+        for (int j = b + lane; j < e; j += SUBW) s += (double)val[j] * (double)x[col[j]];
 #pragma unroll
         for (int off = SUBW / 2; off > 0; off >>= 1) s += __shfl_down(s, off, SUBW);
-        if (lane == 0) y[row] = BETA0 ? alpha * s : alpha * s + beta * y[row];
+        if (lane == 0) y[row] = (T)(BETA0 ? alpha * s : alpha * s + beta * (double)y[row]);
     }
 }
 
 // r = b - A x (fused residual; reference does copy+spmv, hip.hpp:486-491)
-template <int SUBW>
+template <typename T, int SUBW>
 __global__ void residual_k(int nrows, const int *__restrict__ ptr,
-                           const int *__restrict__ col, const double *__restrict__ val,
-                           const double *__restrict__ rhs, const double *__restrict__ x,
-                           double *__restrict__ r) {
+                           const int *__restrict__ col, const T *__restrict__ val,
+                           const T *__restrict__ rhs, const T *__restrict__ x,
+                           T *__restrict__ r) {
     int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
@@ -64,20 +64,20 @@ __global__ void residual_k(int nrows, const int *__restrict__ ptr,
     for (; row < nrows; row += stride) {
         double s = 0.0;
         int b = ptr[row], e = ptr[row + 1];
-        for (int j = b + lane; j < e; j += SUBW) s += val[j] * x[col[j]];
+        for (int j = b + lane; j < e; j += SUBW) s += (double)val[j] * (double)x[col[j]];
 #pragma unroll
         for (int off = SUBW / 2; off > 0; off >>= 1) s += __shfl_down(s, off, SUBW);
-        if (lane == 0) r[row] = rhs[row] - s;
+        if (lane == 0) r[row] = (T)((double)rhs[row] - s);
     }
 }
 
 // t = M ∘ (rhs - A x): the fused SPAI0/Jacobi smoothing step
 // (reference: residual + vmul as separate passes, relaxation/spai0.hpp:85-92)
-template <int SUBW>
+template <typename T, int SUBW>
 __global__ void relax_diag_k(int nrows, const int *__restrict__ ptr,
-                             const int *__restrict__ col, const double *__restrict__ val,
-                             const double *__restrict__ M, const double *__restrict__ rhs,
-                             const double *__restrict__ x, double *__restrict__ t) {
+                             const int *__restrict__ col, const T *__restrict__ val,
+                             const T *__restrict__ M, const T *__restrict__ rhs,
+                             const T *__restrict__ x, T *__restrict__ t) {
     int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
@@ -85,10 +85,10 @@ __global__ void relax_diag_k(int nrows, const int *__restrict__ ptr,
     for (; row < nrows; row += stride) {
         double s = 0.0;
         int b = ptr[row], e = ptr[row + 1];
-        for (int j = b + lane; j < e; j += SUBW) s += val[j] * x[col[j]];
+        for (int j = b + lane; j < e; j += SUBW) s += (double)val[j] * (double)x[col[j]];
 #pragma unroll
         for (int off = SUBW / 2; off > 0; off >>= 1) s += __shfl_down(s, off, SUBW);
-        if (lane == 0) t[row] = M[row] * (rhs[row] - s);
+        if (lane == 0) t[row] = (T)((double)M[row] * ((double)rhs[row] - s));
     }
 }
 
@@ -110,9 +110,9 @@ extern "C" int amg_spmv_f64(int64_t nrows, int64_t nnz, const int *ptr, const in
 #define CASE(SW)                                                                    \
     case SW:                                                                        \
         if (beta == 0.0)                                                            \
-            spmv_k<SW, true><<<grid, B, 0, stream>>>(nrows, ptr, col, val, x, alpha, beta, y);  \
+            spmv_k<double, SW, true><<<grid, B, 0, stream>>>(nrows, ptr, col, val, x, alpha, beta, y);  \
         else                                                                        \
-            spmv_k<SW, false><<<grid, B, 0, stream>>>(nrows, ptr, col, val, x, alpha, beta, y); \
+            spmv_k<double, SW, false><<<grid, B, 0, stream>>>(nrows, ptr, col, val, x, alpha, beta, y); \
         break;
     switch (subw) {
         CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
@@ -130,7 +130,7 @@ extern "C" int amg_residual_f64(int64_t nrows, int64_t nnz, const int *ptr, cons
     const int B = 256;
 #define CASE(SW)                                                                    \
     case SW:                                                                        \
-        residual_k<SW><<<grid, B, 0, stream>>>(nrows, ptr, col, val, rhs, x, r);    \
+        residual_k<double, SW><<<grid, B, 0, stream>>>(nrows, ptr, col, val, rhs, x, r);    \
         break;
     switch (subw) {
         CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
@@ -149,7 +149,7 @@ extern "C" int amg_relax_diag_f64(int64_t nrows, int64_t nnz, const int *ptr,
     const int B = 256;
 #define CASE(SW)                                                                       \
     case SW:                                                                           \
-        relax_diag_k<SW><<<grid, B, 0, stream>>>(nrows, ptr, col, val, M, rhs, x, t);  \
+        relax_diag_k<double, SW><<<grid, B, 0, stream>>>(nrows, ptr, col, val, M, rhs, x, t);  \
         break;
     switch (subw) {
         CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
@@ -162,63 +162,67 @@ extern "C" int amg_relax_diag_f64(int64_t nrows, int64_t nnz, const int *ptr,
 // ---------------------------------------------------------------------------
 // Vector primitives (memory-bound, grid-stride; guide App. B elementwise)
 // ---------------------------------------------------------------------------
-__global__ void axpby_k(int64_t n, double a, const double *__restrict__ x, double b,
-                        double *__restrict__ y) {
+template <typename T>
+__global__ void axpby_k(int64_t n, double a, const T *__restrict__ x, double b,
+                        T *__restrict__ y) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     if (b == 0.0)
-        for (; i < n; i += stride) y[i] = a * x[i];
+        for (; i < n; i += stride) y[i] = (T)(a * (double)x[i]);
     else
-        for (; i < n; i += stride) y[i] = a * x[i] + b * y[i];
+        for (; i < n; i += stride) y[i] = (T)(a * (double)x[i] + b * (double)y[i]);
 }
 
-__global__ void axpbypcz_k(int64_t n, double a, const double *__restrict__ x, double b,
-                           const double *__restrict__ y, double c, double *__restrict__ z) {
+template <typename T>
+__global__ void axpbypcz_k(int64_t n, double a, const T *__restrict__ x, double b,
+                           const T *__restrict__ y, double c, T *__restrict__ z) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     if (c == 0.0)
-        for (; i < n; i += stride) z[i] = a * x[i] + b * y[i];
+        for (; i < n; i += stride) z[i] = (T)(a * (double)x[i] + b * (double)y[i]);
     else
-        for (; i < n; i += stride) z[i] = a * x[i] + b * y[i] + c * z[i];
+        for (; i < n; i += stride) z[i] = (T)(a * (double)x[i] + b * (double)y[i] + c * (double)z[i]);
 }
 
 // z = a*(m∘x) + b*z
-__global__ void vmul_k(int64_t n, double a, const double *__restrict__ m,
-                       const double *__restrict__ x, double b, double *__restrict__ z) {
+template <typename T>
+__global__ void vmul_k(int64_t n, double a, const T *__restrict__ m,
+                       const T *__restrict__ x, double b, T *__restrict__ z) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     if (b == 0.0)
-        for (; i < n; i += stride) z[i] = a * m[i] * x[i];
+        for (; i < n; i += stride) z[i] = (T)(a * (double)m[i] * (double)x[i]);
     else
-        for (; i < n; i += stride) z[i] = a * m[i] * x[i] + b * z[i];
+        for (; i < n; i += stride) z[i] = (T)(a * (double)m[i] * (double)x[i] + b * (double)z[i]);
 }
 
-__global__ void fill_k(int64_t n, double v, double *__restrict__ x) {
+template <typename T>
+__global__ void fill_k(int64_t n, double v, T *__restrict__ x) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) x[i] = v;
+    for (; i < n; i += stride) x[i] = (T)v;
 }
 
 extern "C" int amg_axpby_f64(int64_t n, double a, const double *x, double b, double *y,
                              hipStream_t stream) {
-    axpby_k<<<nblocks(n), 256, 0, stream>>>(n, a, x, b, y);
+    axpby_k<double><<<nblocks(n), 256, 0, stream>>>(n, a, x, b, y);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_axpbypcz_f64(int64_t n, double a, const double *x, double b,
                                 const double *y, double c, double *z, hipStream_t stream) {
-    axpbypcz_k<<<nblocks(n), 256, 0, stream>>>(n, a, x, b, y, c, z);
+    axpbypcz_k<double><<<nblocks(n), 256, 0, stream>>>(n, a, x, b, y, c, z);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_vmul_f64(int64_t n, double a, const double *m, const double *x,
                             double b, double *z, hipStream_t stream) {
-    vmul_k<<<nblocks(n), 256, 0, stream>>>(n, a, m, x, b, z);
+    vmul_k<double><<<nblocks(n), 256, 0, stream>>>(n, a, m, x, b, z);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_fill_f64(int64_t n, double v, double *x, hipStream_t stream) {
-    fill_k<<<nblocks(n), 256, 0, stream>>>(n, v, x);
+    fill_k<double><<<nblocks(n), 256, 0, stream>>>(n, v, x);
     return (int)hipGetLastError();
 }
 
@@ -229,13 +233,14 @@ extern "C" int amg_fill_f64(int64_t n, double v, double *x, hipStream_t stream) 
 // amg_dot2 fuses two inner products into one pass (batch the
 // CG dots into one reduction).
 // ---------------------------------------------------------------------------
-__global__ void dot_k(int64_t n, const double *__restrict__ x, const double *__restrict__ y,
+template <typename T>
+__global__ void dot_k(int64_t n, const T *__restrict__ x, const T *__restrict__ y,
                       double *__restrict__ out) {
     __shared__ double lds[4];  // 256 threads = 4 waves
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     double s = 0.0;
-    for (; i < n; i += stride) s += x[i] * y[i];
+    for (; i < n; i += stride) s += (double)x[i] * (double)y[i];
 #pragma unroll
     for (int off = WAVE / 2; off > 0; off >>= 1) s += __shfl_down(s, off, WAVE);
     int wid = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
@@ -245,16 +250,17 @@ __global__ void dot_k(int64_t n, const double *__restrict__ x, const double *__r
         atomicAdd(out, lds[0] + lds[1] + lds[2] + lds[3]);
 }
 
-__global__ void dot2_k(int64_t n, const double *__restrict__ x1, const double *__restrict__ y1,
-                       const double *__restrict__ x2, const double *__restrict__ y2,
+template <typename T>
+__global__ void dot2_k(int64_t n, const T *__restrict__ x1, const T *__restrict__ y1,
+                       const T *__restrict__ x2, const T *__restrict__ y2,
                        double *__restrict__ out) {
     __shared__ double lds[8];
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     double s1 = 0.0, s2 = 0.0;
     for (; i < n; i += stride) {
-        s1 += x1[i] * y1[i];
-        s2 += x2[i] * y2[i];
+        s1 += (double)x1[i] * (double)y1[i];
+        s2 += (double)x2[i] * (double)y2[i];
     }
 #pragma unroll
     for (int off = WAVE / 2; off > 0; off >>= 1) {
@@ -275,16 +281,16 @@ __global__ void dot2_k(int64_t n, const double *__restrict__ x1, const double *_
 
 extern "C" int amg_dot_f64(int64_t n, const double *x, const double *y, double *out,
                            hipStream_t stream) {
-    fill_k<<<1, 64, 0, stream>>>(1, 0.0, out);
-    dot_k<<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x, y, out);
+    fill_k<double><<<1, 64, 0, stream>>>(1, 0.0, out);
+    dot_k<double><<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x, y, out);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_dot2_f64(int64_t n, const double *x1, const double *y1,
                             const double *x2, const double *y2, double *out,
                             hipStream_t stream) {
-    fill_k<<<1, 64, 0, stream>>>(2, 0.0, out);
-    dot2_k<<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x1, y1, x2, y2, out);
+    fill_k<double><<<1, 64, 0, stream>>>(2, 0.0, out);
+    dot2_k<double><<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x1, y1, x2, y2, out);
     return (int)hipGetLastError();
 }
 
@@ -292,15 +298,17 @@ extern "C" int amg_dot2_f64(int64_t n, const double *x1, const double *y1,
 // gather/scatter by index list — halo pack/unpack for the distributed layer
 // (reference: thrust::gather/scatter, backend/hip.hpp:418-446)
 // ---------------------------------------------------------------------------
-__global__ void gather_k(int64_t n, const double *__restrict__ x,
-                         const int *__restrict__ idx, double *__restrict__ buf) {
+template <typename T>
+__global__ void gather_k(int64_t n, const T *__restrict__ x,
+                         const int *__restrict__ idx, T *__restrict__ buf) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) buf[i] = x[idx[i]];
 }
 
-__global__ void scatter_k(int64_t n, const double *__restrict__ buf,
-                          const int *__restrict__ idx, double *__restrict__ x) {
+template <typename T>
+__global__ void scatter_k(int64_t n, const T *__restrict__ buf,
+                          const int *__restrict__ idx, T *__restrict__ x) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) x[idx[i]] = buf[i];
@@ -308,13 +316,13 @@ __global__ void scatter_k(int64_t n, const double *__restrict__ buf,
 
 extern "C" int amg_gather_f64(int64_t n, const double *x, const int *idx, double *buf,
                               hipStream_t stream) {
-    gather_k<<<nblocks(n), 256, 0, stream>>>(n, x, idx, buf);
+    gather_k<double><<<nblocks(n), 256, 0, stream>>>(n, x, idx, buf);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_scatter_f64(int64_t n, const double *buf, const int *idx, double *x,
                                hipStream_t stream) {
-    scatter_k<<<nblocks(n), 256, 0, stream>>>(n, buf, idx, x);
+    scatter_k<double><<<nblocks(n), 256, 0, stream>>>(n, buf, idx, x);
     return (int)hipGetLastError();
 }
 
@@ -323,23 +331,149 @@ extern "C" int amg_scatter_f64(int64_t n, const double *buf, const int *idx, dou
 // Ainv is ncoarse x ncoarse row-major fp64 (<= ~3000). One wave per row,
 // coalesced row reads, shfl reduce. Memory-bound on Ainv (~72 MB at 3000).
 // ---------------------------------------------------------------------------
-__global__ void gemv_k(int n, const double *__restrict__ a, const double *__restrict__ f,
-                       double *__restrict__ u) {
+template <typename T>
+__global__ void gemv_k(int n, const T *__restrict__ a, const T *__restrict__ f,
+                       T *__restrict__ u) {
     int64_t wid = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int64_t nwaves = ((int64_t)gridDim.x * blockDim.x) / WAVE;
     for (int64_t row = wid; row < n; row += nwaves) {
-        const double *arow = a + row * n;
+        const T *arow = a + row * n;
         double s = 0.0;
-        for (int j = lane; j < n; j += WAVE) s += arow[j] * f[j];
+        for (int j = lane; j < n; j += WAVE) s += (double)arow[j] * (double)f[j];
 #pragma unroll
         for (int off = WAVE / 2; off > 0; off >>= 1) s += __shfl_down(s, off, WAVE);
-        if (lane == 0) u[row] = s;
+        if (lane == 0) u[row] = (T)s;
     }
 }
 
 extern "C" int amg_gemv_f64(int64_t n, const double *a, const double *f, double *u,
                             hipStream_t stream) {
-    gemv_k<<<nblocks(n * WAVE), 256, 0, stream>>>((int)n, a, f, u);
+    gemv_k<double><<<nblocks(n * WAVE), 256, 0, stream>>>((int)n, a, f, u);
+    return (int)hipGetLastError();
+}
+
+
+// ---------------------------------------------------------------------------
+// fp32 entry points (mixed-precision AMG: fp32 hierarchy under a fp64 Krylov
+// loop — SURVEY §5.9, reference examples/mixed_precision.cpp). Reductions
+// accumulate in fp64 regardless of storage type.
+// ---------------------------------------------------------------------------
+extern "C" int amg_spmv_f32(int64_t nrows, int64_t nnz, const int *ptr, const int *col,
+                            const float *val, const float *x, double alpha, double beta,
+                            float *y, int subw, hipStream_t stream) {
+    if (subw <= 0) subw = pick_subw(nrows, nnz);
+    int grid = nblocks(nrows * subw);
+#define CASE(SW)                                                                        \
+    case SW:                                                                            \
+        if (beta == 0.0)                                                                \
+            spmv_k<float, SW, true><<<grid, 256, 0, stream>>>(nrows, ptr, col, val, x,  \
+                                                              alpha, beta, y);          \
+        else                                                                            \
+            spmv_k<float, SW, false><<<grid, 256, 0, stream>>>(nrows, ptr, col, val, x, \
+                                                               alpha, beta, y);         \
+        break;
+    switch (subw) {
+        CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
+        default: return (int)hipErrorInvalidValue;
+    }
+#undef CASE
+    return (int)hipGetLastError();
+}
+
+extern "C" int amg_residual_f32(int64_t nrows, int64_t nnz, const int *ptr, const int *col,
+                                const float *val, const float *rhs, const float *x,
+                                float *r, int subw, hipStream_t stream) {
+    if (subw <= 0) subw = pick_subw(nrows, nnz);
+    int grid = nblocks(nrows * subw);
+#define CASE(SW)                                                                      \
+    case SW:                                                                          \
+        residual_k<float, SW><<<grid, 256, 0, stream>>>(nrows, ptr, col, val, rhs, x, r); \
+        break;
+    switch (subw) {
+        CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
+        default: return (int)hipErrorInvalidValue;
+    }
+#undef CASE
+    return (int)hipGetLastError();
+}
+
+extern "C" int amg_relax_diag_f32(int64_t nrows, int64_t nnz, const int *ptr,
+                                  const int *col, const float *val, const float *M,
+                                  const float *rhs, const float *x, float *t, int subw,
+                                  hipStream_t stream) {
+    if (subw <= 0) subw = pick_subw(nrows, nnz);
+    int grid = nblocks(nrows * subw);
+#define CASE(SW)                                                                          \
+    case SW:                                                                              \
+        relax_diag_k<float, SW><<<grid, 256, 0, stream>>>(nrows, ptr, col, val, M, rhs,   \
+                                                          x, t);                         \
+        break;
+    switch (subw) {
+        CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
+        default: return (int)hipErrorInvalidValue;
+    }
+#undef CASE
+    return (int)hipGetLastError();
+}
+
+extern "C" int amg_axpby_f32(int64_t n, double a, const float *x, double b, float *y,
+                             hipStream_t stream) {
+    axpby_k<float><<<nblocks(n), 256, 0, stream>>>(n, a, x, b, y);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_axpbypcz_f32(int64_t n, double a, const float *x, double b,
+                                const float *y, double c, float *z, hipStream_t stream) {
+    axpbypcz_k<float><<<nblocks(n), 256, 0, stream>>>(n, a, x, b, y, c, z);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_vmul_f32(int64_t n, double a, const float *m, const float *x, double b,
+                            float *z, hipStream_t stream) {
+    vmul_k<float><<<nblocks(n), 256, 0, stream>>>(n, a, m, x, b, z);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_fill_f32(int64_t n, double v, float *x, hipStream_t stream) {
+    fill_k<float><<<nblocks(n), 256, 0, stream>>>(n, v, x);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_dot_f32(int64_t n, const float *x, const float *y, double *out,
+                           hipStream_t stream) {
+    fill_k<double><<<1, 64, 0, stream>>>(1, 0.0, out);
+    dot_k<float><<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x, y, out);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_gather_f32(int64_t n, const float *x, const int *idx, float *buf,
+                              hipStream_t stream) {
+    gather_k<float><<<nblocks(n), 256, 0, stream>>>(n, x, idx, buf);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_scatter_f32(int64_t n, const float *buf, const int *idx, float *x,
+                               hipStream_t stream) {
+    scatter_k<float><<<nblocks(n), 256, 0, stream>>>(n, buf, idx, x);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_gemv_f32(int64_t n, const float *a, const float *f, float *u,
+                            hipStream_t stream) {
+    gemv_k<float><<<nblocks(n * WAVE), 256, 0, stream>>>((int)n, a, f, u);
+    return (int)hipGetLastError();
+}
+
+// precision casts (mixed-precision boundary)
+__global__ void cast_d2s_k(int64_t n, const double *__restrict__ src, float *__restrict__ dst) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) dst[i] = (float)src[i];
+}
+__global__ void cast_s2d_k(int64_t n, const float *__restrict__ src, double *__restrict__ dst) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) dst[i] = (double)src[i];
+}
+extern "C" int amg_cast_d2s(int64_t n, const double *src, float *dst, hipStream_t s) {
+    cast_d2s_k<<<nblocks(n), 256, 0, s>>>(n, src, dst);
+    return (int)hipGetLastError();
+}
+extern "C" int amg_cast_s2d(int64_t n, const float *src, double *dst, hipStream_t s) {
+    cast_s2d_k<<<nblocks(n), 256, 0, s>>>(n, src, dst);
     return (int)hipGetLastError();
 }

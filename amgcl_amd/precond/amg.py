@@ -46,6 +46,7 @@ class AMG:
             "pre_cycles": 1,
             "direct_coarse": True,
             "keep_host_matrices": False,
+            "precision": "fp64",  # "mixed" = fp32 hierarchy under fp64 Krylov
         }
 
     def __init__(self, A, prm=None, backend=None):
@@ -58,7 +59,10 @@ class AMG:
         self.levels = []
         self.coarse_solve = None
         self._coarsening = None
+        self._mixed = False
         self._build(A)
+        if self.prm["precision"] == "mixed":
+            self._to_mixed()
 
     def rebuild(self, A_new):
         """Reuse the transfer operators for a matrix with changed coefficients
@@ -158,6 +162,44 @@ class AMG:
 
                 A_host = hip_setup.download(A_host)
 
+    def _to_mixed(self):
+        """Convert the hierarchy to fp32 storage (mixed precision: fp32
+        preconditioner under a fp64 Krylov loop; parity: SURVEY §5.9,
+        reference examples/mixed_precision.cpp, backend/detail/mixing.hpp)."""
+        if self.backend.name != "hip":
+            raise ValueError("precision='mixed' requires the hip backend")
+        import torch
+
+        from ..backend.hip import DeviceCSR
+        from ..relaxation.spai0 import DiagonalSmootherBase
+
+        f32 = torch.float32
+
+        def conv(m):
+            if m is None:
+                return None
+            return DeviceCSR.from_tensors(m.nrows, m.ncols, m.ptr, m.col,
+                                          m.val.to(f32), m.subw)
+
+        for lvl in self.levels:
+            lvl.A = conv(lvl.A)
+            lvl.P = conv(lvl.P)
+            lvl.R = conv(lvl.R)
+            for name in ("f", "u", "t"):
+                v = getattr(lvl, name)
+                if v is not None:
+                    setattr(lvl, name, v.to(f32))
+            if lvl.relax is not None:
+                if not isinstance(lvl.relax, DiagonalSmootherBase):
+                    raise ValueError("mixed precision supports diagonal smoothers")
+                lvl.relax.M = lvl.relax.M.to(f32)
+        if self.coarse_solve is not None:
+            self.coarse_solve.inv = self.coarse_solve.inv.to(f32)
+        n0 = self.levels[0].rows
+        self._r32 = self.backend.vector(n0, f32)
+        self._x32 = self.backend.vector(n0, f32)
+        self._mixed = True
+
     # --- solve-phase -------------------------------------------------------
     def system_matrix(self):
         return self.levels[0].A
@@ -195,6 +237,13 @@ class AMG:
         """x = M^-1 rhs via pre_cycles cycles from a zero initial guess
         (parity: amgcl/amg.hpp:289-297)."""
         b = self.backend
+        if self._mixed:
+            b.cast(rhs, self._r32)
+            b.clear(self._x32)
+            for _ in range(int(self.prm["pre_cycles"])):
+                self.cycle(0, self._r32, self._x32)
+            b.cast(self._x32, x)
+            return
         b.clear(x)
         for _ in range(int(self.prm["pre_cycles"])):
             self.cycle(0, rhs, x)

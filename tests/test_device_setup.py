@@ -145,3 +145,25 @@ def test_device_setup_iteration_parity_with_host(hip):
     x2, it_dev, _ = s_dev(b)
     # host greedy vs device MIS aggregation may differ slightly
     assert abs(it_dev - it_host) <= 3
+
+
+def test_mixed_precision_hierarchy(hip):
+    """fp32 hierarchy under fp64 CG: converges to fp64-level tolerance
+    (parity: reference mixed_precision.cpp / Serena tutorial)."""
+    import math
+
+    import torch
+
+    A, b = am.poisson3d(48, rhs="random")
+    prm64 = {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}}
+    s64 = am.make_solver(A, prm64, backend=hip)
+    x64, it64, _ = s64(b)
+    prm32 = {"precond": {"class": "amg", "precision": "mixed"},
+             "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}}
+    s32 = am.make_solver(A, prm32, backend=hip)
+    assert s32.P._mixed and s32.P.levels[0].A.val.dtype == torch.float32
+    x32, it32, res32 = s32(b)
+    assert res32 < 1e-8
+    r = b - A @ hip.to_host(x32)
+    assert np.linalg.norm(r) / np.linalg.norm(b) < 1e-7
+    assert it32 <= it64 + 4  # fp32 hierarchy costs at most a few iterations
