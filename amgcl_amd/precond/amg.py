@@ -181,6 +181,10 @@ class AMG:
             return DeviceCSR.from_tensors(m.nrows, m.ncols, m.ptr, m.col,
                                           m.val.to(f32), m.subw)
 
+        # the outer Krylov loop keeps iterating with the fp64 fine operator
+        # (backend/detail/mixing.hpp semantics: solver backend fp64, precond
+        # backend fp32); only the preconditioner's internals go fp32
+        self._A64 = self.levels[0].A
         for lvl in self.levels:
             lvl.A = conv(lvl.A)
             lvl.P = conv(lvl.P)
@@ -202,6 +206,8 @@ class AMG:
 
     # --- solve-phase -------------------------------------------------------
     def system_matrix(self):
+        if self._mixed:
+            return self._A64
         return self.levels[0].A
 
     def cycle(self, i, f, u):
