@@ -54,6 +54,33 @@ class DeviceCSR:
         return (self.ptr.numel() + self.col.numel()) * 4 + self.val.numel() * 8
 
 
+class DeviceBSR:
+    """Block CSR (row-major blocks): the solve-phase container for
+    block-valued matrices (reference: builtin_hybrid / vexcl_static_matrix).
+    BSR SpMV is bandwidth-bound at fp64 (0.25 flop/byte), so the kernels are
+    unrolled block loads, not MFMA — see csrc/hip/block.hip header note."""
+
+    __slots__ = ("nbrows", "bsize", "ptr", "col", "val", "nrows", "ncols")
+
+    def __init__(self, csr: CSR, bsize, device):
+        import torch
+
+        from .. import _core
+
+        bp, bc, bv = _core.csr_to_bsr(csr.nrows, csr.ptr, csr.col, csr.val, int(bsize))
+        self.nbrows = csr.nrows // int(bsize)
+        self.bsize = int(bsize)
+        self.nrows = csr.nrows
+        self.ncols = csr.ncols
+        self.ptr = torch.from_numpy(np.asarray(bp)).to(device)
+        self.col = torch.from_numpy(np.asarray(bc)).to(device)
+        self.val = torch.from_numpy(np.asarray(bv)).to(device)
+
+    @property
+    def nnz(self):
+        return self.val.numel()
+
+
 @register("hip")
 class HipBackend:
     name = "hip"
@@ -94,11 +121,23 @@ class HipBackend:
         return getattr(lib(), "amg_" + name + sfx)
 
     def spmv(self, alpha, A, x, beta, y):
+        if isinstance(A, DeviceBSR):
+            check(lib().amg_bsr_spmv_f64(A.nbrows, A.bsize, A.ptr.data_ptr(),
+                                         A.col.data_ptr(), A.val.data_ptr(),
+                                         x.data_ptr(), alpha, beta, y.data_ptr(),
+                                         _stream()), "bsr_spmv")
+            return
         check(self._fn("spmv", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(), A.col.data_ptr(),
                                       A.val.data_ptr(), x.data_ptr(), alpha, beta,
                                       y.data_ptr(), A.subw, _stream()), "spmv")
 
     def residual(self, b, A, x, r):
+        if isinstance(A, DeviceBSR):
+            check(lib().amg_bsr_residual_f64(A.nbrows, A.bsize, A.ptr.data_ptr(),
+                                             A.col.data_ptr(), A.val.data_ptr(),
+                                             b.data_ptr(), x.data_ptr(), r.data_ptr(),
+                                             _stream()), "bsr_residual")
+            return
         check(self._fn("residual", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(),
                                           A.col.data_ptr(), A.val.data_ptr(),
                                           b.data_ptr(), x.data_ptr(), r.data_ptr(),
@@ -106,6 +145,13 @@ class HipBackend:
 
     def relax_diag(self, A, M, rhs, x, t):
         """t = M∘(rhs - A x); x += t (fused single pass over A + axpby)."""
+        if isinstance(A, DeviceBSR):
+            check(lib().amg_bsr_relax_f64(A.nbrows, A.bsize, A.ptr.data_ptr(),
+                                          A.col.data_ptr(), A.val.data_ptr(),
+                                          M.data_ptr(), rhs.data_ptr(), x.data_ptr(),
+                                          t.data_ptr(), _stream()), "bsr_relax")
+            self.axpby(1.0, t, 1.0, x)
+            return
         check(self._fn("relax_diag", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(),
                                             A.col.data_ptr(), A.val.data_ptr(),
                                             M.data_ptr(), rhs.data_ptr(), x.data_ptr(),

@@ -893,6 +893,78 @@ static py::tuple pointwise_matrix(i64 n, arr<i32> ptr_a, arr<i32> col_a,
     return py::make_tuple(pptr, pcol, pval);
 }
 
+// Scalar CSR -> BSR (block CSR) with zero fill
+// (parity: amgcl/adapter/block_matrix.hpp:44 view + builtin_hybrid storage).
+// Blocks row-major, block columns sorted.
+static py::tuple csr_to_bsr(i64 n, arr<i32> ptr_a, arr<i32> col_a, arr<double> val_a,
+                            i64 b) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    if (n % b) throw std::runtime_error("csr_to_bsr: size not divisible by block");
+    const i64 nb = n / b;
+    arr<i32> bptr(nb + 1);
+    i32 *BP = bptr.mutable_data();
+    BP[0] = 0;
+#pragma omp parallel
+    {
+        RowHash hash(256);
+#pragma omp for schedule(static)
+        for (i64 p = 0; p < nb; ++p) {
+            i64 ub = 0;
+            for (i64 r = p * b; r < (p + 1) * b; ++r) ub += A.ptr[r + 1] - A.ptr[r];
+            hash.ensure(ub);
+            i32 cnt = 0;
+            for (i64 r = p * b; r < (p + 1) * b; ++r)
+                for (i32 j = A.ptr[r]; j < A.ptr[r + 1]; ++j)
+                    cnt += hash.count_add(A.col[j] / (i32)b);
+            hash.reset();
+            BP[p + 1] = cnt;
+        }
+    }
+    scan_i32(BP + 1, nb);
+    const i64 nnzb = BP[nb];
+    arr<i32> bcol = big_arr<i32>(nnzb);
+    arr<double> bval = big_arr<double>(nnzb * b * b);
+    i32 *BC = bcol.mutable_data();
+    double *BV = bval.mutable_data();
+    pretouch(BV, nnzb * b * b * 8);
+#pragma omp parallel
+    {
+        RowHash hash(256);
+#pragma omp for schedule(static)
+        for (i64 p = 0; p < nb; ++p) {
+            // collect sorted distinct block columns
+            i64 ub = 0;
+            for (i64 r = p * b; r < (p + 1) * b; ++r) ub += A.ptr[r + 1] - A.ptr[r];
+            hash.ensure(ub);
+            i32 head = BP[p];
+            i32 cnt = 0;
+            for (i64 r = p * b; r < (p + 1) * b; ++r)
+                for (i32 j = A.ptr[r]; j < A.ptr[r + 1]; ++j)
+                    if (hash.count_add(A.col[j] / (i32)b)) BC[head + cnt++] = A.col[j] / (i32)b;
+            hash.reset();
+            std::sort(BC + head, BC + head + cnt);
+            for (i64 k = (i64)head * b * b; k < (i64)(head + cnt) * b * b; ++k) BV[k] = 0.0;
+            // scatter values: binary search the block column
+            for (i64 r = p * b; r < (p + 1) * b; ++r) {
+                int lr = (int)(r - p * b);
+                for (i32 j = A.ptr[r]; j < A.ptr[r + 1]; ++j) {
+                    i32 q = A.col[j] / (i32)b;
+                    int lc = (int)(A.col[j] - (i64)q * b);
+                    i32 lo = head, hi = head + cnt - 1, pos = -1;
+                    while (lo <= hi) {
+                        i32 mid = (lo + hi) / 2;
+                        if (BC[mid] == q) { pos = mid; break; }
+                        if (BC[mid] < q) lo = mid + 1;
+                        else hi = mid - 1;
+                    }
+                    BV[(i64)pos * b * b + lr * b + lc] = A.val[j];
+                }
+            }
+        }
+    }
+    return py::make_tuple(bptr, bcol, bval);
+}
+
 // Expand a point-level strong mask to the scalar entries: scalar entry (i,j)
 // is strong iff the point pair (i/b, col/b) is strong in the point matrix.
 static arr<uint8_t> expand_strong(i64 n, arr<i32> ptr_a, arr<i32> col_a, i64 b,
@@ -1421,6 +1493,7 @@ PYBIND11_MODULE(_core, m) {
     m.def("ruge_stuben", &ruge_stuben);
     m.def("tentative_nullspace", &tentative_nullspace);
     m.def("pointwise_matrix", &pointwise_matrix);
+    m.def("csr_to_bsr", &csr_to_bsr);
     m.def("expand_strong", &expand_strong);
     m.def("filtered_smoother_matrix", &filtered_smoother_matrix);
     m.def("spmv", &spmv);

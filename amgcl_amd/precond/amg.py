@@ -47,6 +47,7 @@ class AMG:
             "direct_coarse": True,
             "keep_host_matrices": False,
             "precision": "fp64",  # "mixed" = fp32 hierarchy under fp64 Krylov
+            "block_value": 0,      # >1: store level operators as BSR blocks
         }
 
     def __init__(self, A, prm=None, backend=None):
@@ -63,6 +64,8 @@ class AMG:
         self._build(A)
         if self.prm["precision"] == "mixed":
             self._to_mixed()
+        if int(self.prm["block_value"]) > 1:
+            self._to_block(int(self.prm["block_value"]))
 
     def rebuild(self, A_new):
         """Reuse the transfer operators for a matrix with changed coefficients
@@ -203,6 +206,29 @@ class AMG:
         self._r32 = self.backend.vector(n0, f32)
         self._x32 = self.backend.vector(n0, f32)
         self._mixed = True
+
+    def _to_block(self, bsize):
+        """Store level operators as BSR for block-valued solve kernels
+        (parity: amgcl/backend/builtin_hybrid.hpp:43 — scalar setup quality,
+        block solve speed). Only levels whose size is divisible by the block
+        size are converted; transfers stay scalar CSR."""
+        if self.backend.name != "hip":
+            raise ValueError("block_value storage requires the hip backend")
+        if self._mixed:
+            raise ValueError("block_value is fp64-only for now")
+        from ..backend import hip_setup
+        from ..backend.hip import DeviceBSR, DeviceCSR
+
+        for lvl in self.levels:
+            A = lvl.A
+            if A.nrows % bsize:
+                continue
+            host = lvl.A_host
+            if host is None or not isinstance(host, CSR):
+                host = hip_setup.download(A) if isinstance(A, DeviceCSR) else host
+            if host is None:
+                continue
+            lvl.A = DeviceBSR(host, bsize, self.backend.device)
 
     # --- solve-phase -------------------------------------------------------
     def system_matrix(self):
