@@ -1460,6 +1460,299 @@ static py::tuple ilu0_factor(i64 nrows, arr<i32> ptr, arr<i32> col, arr<double> 
     return py::make_tuple(lu, dia_a);
 }
 
+// ILU(k): level-of-fill factorization (parity: amgcl/relaxation/iluk.hpp:49).
+// Returns (ptr, col, lu, dia) of the combined LU factor (unit L implied,
+// U diagonal stored inverted) with rows sorted — compatible with ilu0_solve.
+static py::tuple iluk_factor(i64 n, arr<i32> ptr_a, arr<i32> col_a, arr<double> val_a,
+                             i64 kfill) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    std::vector<std::vector<i32>> cols(n);
+    std::vector<std::vector<double>> vals(n);
+    std::vector<std::vector<i32>> levs(n);
+    std::vector<i32> dia_pos(n, -1);
+
+    std::vector<double> w(n, 0.0);
+    std::vector<i32> wl(n, 0);
+    std::vector<uint8_t> inrow(n, 0);
+    std::vector<i32> touched;
+
+    for (i64 i = 0; i < n; ++i) {
+        touched.clear();
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i32 c = A.col[j];
+            w[c] = A.val[j];
+            wl[c] = 0;
+            inrow[c] = 1;
+            touched.push_back(c);
+        }
+        // IKJ elimination over k < i present in the row
+        std::sort(touched.begin(), touched.end());
+        for (size_t t = 0; t < touched.size(); ++t) {
+            i32 k = touched[t];
+            if (k >= (i64)i) break;
+            if (!inrow[k]) continue;
+            i32 dk = dia_pos[k];
+            double lik = w[k] * vals[k][dk];  // dia stored inverted
+            w[k] = lik;
+            i32 lev_ik = wl[k];
+            for (size_t jj = dk + 1; jj < cols[k].size(); ++jj) {
+                i32 c = cols[k][jj];
+                i32 lev_new = lev_ik + levs[k][jj] + 1;
+                if (inrow[c]) {
+                    w[c] -= lik * vals[k][jj];
+                    if (lev_new < wl[c]) wl[c] = lev_new;
+                } else if (lev_new <= kfill) {
+                    w[c] = -lik * vals[k][jj];
+                    wl[c] = lev_new;
+                    inrow[c] = 1;
+                    touched.insert(std::upper_bound(touched.begin() + t + 1,
+                                                    touched.end(), c), c);
+                }
+            }
+        }
+        // store the row (sorted)
+        for (i32 c : touched) {
+            if (!inrow[c]) continue;
+            if (c == (i64)i) dia_pos[i] = (i32)cols[i].size();
+            cols[i].push_back(c);
+            vals[i].push_back(w[c]);
+            levs[i].push_back(wl[c]);
+            inrow[c] = 0;
+        }
+        if (dia_pos[i] < 0 || vals[i][dia_pos[i]] == 0.0)
+            throw std::runtime_error("iluk: zero pivot");
+        vals[i][dia_pos[i]] = 1.0 / vals[i][dia_pos[i]];
+    }
+    i64 nnz = 0;
+    for (i64 i = 0; i < n; ++i) nnz += cols[i].size();
+    arr<i32> optr(n + 1), ocol(nnz), odia(n);
+    arr<double> oval(nnz);
+    i32 *OP = optr.mutable_data();
+    i32 *OC = ocol.mutable_data();
+    double *OV = oval.mutable_data();
+    i32 *OD = odia.mutable_data();
+    OP[0] = 0;
+    i64 h = 0;
+    for (i64 i = 0; i < n; ++i) {
+        OD[i] = (i32)(h + dia_pos[i]);
+        for (size_t j = 0; j < cols[i].size(); ++j, ++h) {
+            OC[h] = cols[i][j];
+            OV[h] = vals[i][j];
+        }
+        OP[i + 1] = (i32)h;
+    }
+    return py::make_tuple(optr, ocol, oval, odia);
+}
+
+// ILUT(p, tau): threshold ILU (parity: amgcl/relaxation/ilut.hpp:56).
+// Keeps the p*row_nnz largest entries per L/U part above tau*row_norm.
+static py::tuple ilut_factor(i64 n, arr<i32> ptr_a, arr<i32> col_a, arr<double> val_a,
+                             double pfactor, double tau) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    std::vector<std::vector<i32>> cols(n);
+    std::vector<std::vector<double>> vals(n);
+    std::vector<i32> dia_pos(n, -1);
+    std::vector<double> w(n, 0.0);
+    std::vector<uint8_t> inrow(n, 0);
+    std::vector<i32> touched;
+
+    for (i64 i = 0; i < n; ++i) {
+        touched.clear();
+        double nrm = 0.0;
+        i32 lcount = 0, ucount = 0;
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i32 c = A.col[j];
+            w[c] = A.val[j];
+            inrow[c] = 1;
+            touched.push_back(c);
+            nrm += A.val[j] * A.val[j];
+            if (c < (i64)i) ++lcount;
+            else if (c > (i64)i) ++ucount;
+        }
+        nrm = std::sqrt(nrm);
+        const double drop = tau * nrm;
+        const i32 lmax = (i32)(pfactor * lcount) + 1;
+        const i32 umax = (i32)(pfactor * ucount) + 1;
+        std::sort(touched.begin(), touched.end());
+        for (size_t t = 0; t < touched.size(); ++t) {
+            i32 k = touched[t];
+            if (k >= (i64)i) break;
+            if (!inrow[k]) continue;
+            i32 dk = dia_pos[k];
+            double lik = w[k] * vals[k][dk];
+            if (std::abs(lik) < drop) {
+                w[k] = 0.0;
+                inrow[k] = 0;
+                continue;
+            }
+            w[k] = lik;
+            for (size_t jj = dk + 1; jj < cols[k].size(); ++jj) {
+                i32 c = cols[k][jj];
+                double upd = lik * vals[k][jj];
+                if (inrow[c]) {
+                    w[c] -= upd;
+                } else if (std::abs(upd) >= drop) {
+                    w[c] = -upd;
+                    inrow[c] = 1;
+                    touched.insert(std::upper_bound(touched.begin() + t + 1,
+                                                    touched.end(), c), c);
+                }
+            }
+        }
+        // select entries: diagonal always; p largest in each part above drop
+        std::vector<std::pair<double, i32>> lpart, upart;
+        double dval = 0.0;
+        for (i32 c : touched) {
+            if (!inrow[c]) continue;
+            inrow[c] = 0;
+            if (c == (i64)i) {
+                dval = w[c];
+                continue;
+            }
+            if (std::abs(w[c]) < drop) continue;
+            (c < (i64)i ? lpart : upart).push_back({std::abs(w[c]), c});
+        }
+        auto keep = [&](std::vector<std::pair<double, i32>> &part, i32 cap) {
+            if ((i32)part.size() > cap) {
+                std::nth_element(part.begin(), part.begin() + cap, part.end(),
+                                 [](auto &a, auto &b) { return a.first > b.first; });
+                part.resize(cap);
+            }
+            std::sort(part.begin(), part.end(),
+                      [](auto &a, auto &b) { return a.second < b.second; });
+        };
+        keep(lpart, lmax);
+        keep(upart, umax);
+        if (dval == 0.0) throw std::runtime_error("ilut: zero pivot");
+        for (auto &pr : lpart) {
+            cols[i].push_back(pr.second);
+            vals[i].push_back(w[pr.second]);
+        }
+        dia_pos[i] = (i32)cols[i].size();
+        cols[i].push_back((i32)i);
+        vals[i].push_back(1.0 / dval);
+        for (auto &pr : upart) {
+            cols[i].push_back(pr.second);
+            vals[i].push_back(w[pr.second]);
+        }
+    }
+    i64 nnz = 0;
+    for (i64 i = 0; i < n; ++i) nnz += cols[i].size();
+    arr<i32> optr(n + 1), ocol(nnz), odia(n);
+    arr<double> oval(nnz);
+    i32 *OP = optr.mutable_data();
+    i32 *OC = ocol.mutable_data();
+    double *OV = oval.mutable_data();
+    i32 *OD = odia.mutable_data();
+    OP[0] = 0;
+    i64 h = 0;
+    for (i64 i = 0; i < n; ++i) {
+        OD[i] = (i32)(h + dia_pos[i]);
+        for (size_t j = 0; j < cols[i].size(); ++j, ++h) {
+            OC[h] = cols[i][j];
+            OV[h] = vals[i][j];
+        }
+        OP[i + 1] = (i32)h;
+    }
+    return py::make_tuple(optr, ocol, oval, odia);
+}
+
+// SPAI-1: sparse approximate inverse with A's sparsity pattern
+// (parity: amgcl/relaxation/spai1.hpp:54). Per row: least squares
+// min || A(:,J) m - e_i || over J = pattern(i), solved via normal equations
+// (small dense SPD system per row).
+static py::tuple spai1(i64 n, arr<i32> ptr_a, arr<i32> col_a, arr<double> val_a) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    arr<i32> mptr(n + 1);
+    std::memcpy(mptr.mutable_data(), A.ptr, (n + 1) * sizeof(i32));
+    const i64 nnz = (i64)col_a.size();
+    arr<i32> mcol(nnz);
+    std::memcpy(mcol.mutable_data(), A.col, nnz * sizeof(i32));
+    arr<double> mval(nnz);
+    double *MV = mval.mutable_data();
+
+    // A^T pattern access for gathering columns: build transpose once
+    std::vector<i32> tp(n + 1, 0), tc(nnz), tj(nnz);
+    std::vector<double> tv(nnz);
+    for (i64 j = 0; j < nnz; ++j) ++tp[A.col[j] + 1];
+    for (i64 c = 0; c < n; ++c) tp[c + 1] += tp[c];
+    {
+        std::vector<i32> cur(tp.begin(), tp.end() - 1);
+        for (i64 i = 0; i < n; ++i)
+            for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+                i32 h = cur[A.col[j]]++;
+                tc[h] = (i32)i;
+                tv[h] = A.val[j];
+            }
+    }
+
+#pragma omp parallel
+    {
+        std::vector<double> G, rhs, sol;
+#pragma omp for schedule(dynamic, 512)
+        for (i64 i = 0; i < n; ++i) {
+            i32 jb = A.ptr[i], je = A.ptr[i + 1];
+            int k = je - jb;
+            if (k <= 0) continue;
+            // G = (A_J)^T A_J  via sparse column dot products, rhs = (A_J)^T e_i
+            G.assign((size_t)k * k, 0.0);
+            rhs.assign(k, 0.0);
+            for (int a = 0; a < k; ++a) {
+                i32 ca = A.col[jb + a];
+                for (i32 t = tp[ca]; t < tp[ca + 1]; ++t) {
+                    i32 row = tc[t];
+                    double va = tv[t];
+                    if (row == (i32)i) rhs[a] += va;
+                    // dot with other columns: binary search row in each col?
+                    // accumulate via the row's pattern instead:
+                    for (i32 jj = A.ptr[row]; jj < A.ptr[row + 1]; ++jj) {
+                        i32 cb = A.col[jj];
+                        // find cb's position in J (row i's sorted pattern)
+                        i32 lo = jb, hi = je - 1, pos = -1;
+                        while (lo <= hi) {
+                            i32 mid = (lo + hi) / 2;
+                            if (A.col[mid] == cb) { pos = mid; break; }
+                            if (A.col[mid] < cb) lo = mid + 1;
+                            else hi = mid - 1;
+                        }
+                        if (pos >= 0) G[(size_t)a * k + (pos - jb)] += va * A.val[jj];
+                    }
+                }
+            }
+            // solve G m = rhs (Cholesky-free: Gauss with partial pivot)
+            sol = rhs;
+            for (int c = 0; c < k; ++c) {
+                int piv = c;
+                for (int r2 = c + 1; r2 < k; ++r2)
+                    if (std::abs(G[(size_t)r2 * k + c]) > std::abs(G[(size_t)piv * k + c]))
+                        piv = r2;
+                if (piv != c) {
+                    for (int cc = 0; cc < k; ++cc)
+                        std::swap(G[(size_t)c * k + cc], G[(size_t)piv * k + cc]);
+                    std::swap(sol[c], sol[piv]);
+                }
+                double d = G[(size_t)c * k + c];
+                if (d == 0.0) d = 1e-300;
+                for (int r2 = c + 1; r2 < k; ++r2) {
+                    double f = G[(size_t)r2 * k + c] / d;
+                    if (f == 0.0) continue;
+                    for (int cc = c; cc < k; ++cc)
+                        G[(size_t)r2 * k + cc] -= f * G[(size_t)c * k + cc];
+                    sol[r2] -= f * sol[c];
+                }
+            }
+            for (int c = k - 1; c >= 0; --c) {
+                double s = sol[c];
+                for (int cc = c + 1; cc < k; ++cc)
+                    s -= G[(size_t)c * k + cc] * sol[cc];
+                sol[c] = s / (G[(size_t)c * k + c] == 0.0 ? 1e-300 : G[(size_t)c * k + c]);
+            }
+            for (int a = 0; a < k; ++a) MV[jb + a] = sol[a];
+        }
+    }
+    return py::make_tuple(mptr, mcol, mval);
+}
+
 // serial L/U sweeps: solve (LU) z = r with unit L, inverted-diagonal U.
 static void ilu0_solve(i64 nrows, arr<i32> ptr, arr<i32> col, arr<double> lu,
                        arr<i32> dia_a, arr<double> z) {
@@ -1500,6 +1793,9 @@ PYBIND11_MODULE(_core, m) {
     m.def("residual", &residual);
     m.def("gauss_seidel", &gauss_seidel);
     m.def("ilu0_factor", &ilu0_factor);
+    m.def("iluk_factor", &iluk_factor);
+    m.def("ilut_factor", &ilut_factor);
+    m.def("spai1", &spai1);
     m.def("ilu0_solve", &ilu0_solve);
     m.def("omp_threads", []() { return omp_get_max_threads(); });
 }

@@ -6,18 +6,21 @@ apply_post(A, rhs, x, tmp) smooth in place through backend primitives;
 apply(A, rhs, x, tmp) is the single-shot form used when the smoother acts as
 a standalone preconditioner.
 """
-from .spai0 import Spai0
+from .spai0 import Spai0, Spai1
 from .damped_jacobi import DampedJacobi
 from .chebyshev import Chebyshev
 from .gauss_seidel import GaussSeidel
-from .ilu0 import ILU0
+from .ilu0 import ILU0, ILUK, ILUT
 
 REGISTRY = {
     "spai0": Spai0,
+    "spai1": Spai1,
     "damped_jacobi": DampedJacobi,
     "chebyshev": Chebyshev,
     "gauss_seidel": GaussSeidel,
     "ilu0": ILU0,
+    "iluk": ILUK,
+    "ilut": ILUT,
 }
 
 

@@ -21,16 +21,26 @@ class ILU0:
     def defaults():
         return {"damping": 1.0, "solve_iters": 2, "solve_damping": 0.72}
 
+    def _factor(self, A, p):
+        lu, dia = _core.ilu0_factor(A.nrows, A.ptr, A.col, A.val)
+        return A.ptr, A.col, lu, dia
+
     def __init__(self, A, prm, backend):
+        from ..matrix import CSR
+
+        if not isinstance(A, CSR):
+            from ..backend import hip_setup
+
+            A = hip_setup.download(A)  # ILU factorizations are host-side
         p = merge_params(self.defaults(), prm)
         self.damping = float(p["damping"])
         self.backend = backend
-        lu, dia = _core.ilu0_factor(A.nrows, A.ptr, A.col, A.val)
+        fptr, fcol, lu, dia = self._factor(A, p)
         self.n = A.nrows
 
         if backend.name == "cpu":
             self._serial = True
-            self.ptr, self.col = A.ptr, A.col
+            self.ptr, self.col = np.asarray(fptr), np.asarray(fcol)
             self.lu, self.dia = np.asarray(lu), np.asarray(dia)
         else:
             self._serial = False
@@ -40,8 +50,8 @@ class ILU0:
             # strictly-upper U' with inverted diagonal kept separately.
             lu = np.asarray(lu)
             dia = np.asarray(dia).astype(np.int64)
-            ptr = np.asarray(A.ptr).astype(np.int64)
-            col = np.asarray(A.col)
+            ptr = np.asarray(fptr).astype(np.int64)
+            col = np.asarray(fcol)
             row_of = np.repeat(np.arange(self.n, dtype=np.int64), np.diff(ptr))
             idx = np.arange(col.size, dtype=np.int64)
             lower = idx < dia[row_of]
@@ -104,3 +114,34 @@ class ILU0:
             self._solve_serial(x)
         else:
             self._solve_jacobi(x)
+
+
+class ILUK(ILU0):
+    """ILU(k) level-of-fill smoother (parity: amgcl/relaxation/iluk.hpp:49)."""
+
+    @staticmethod
+    def defaults():
+        d = ILU0.defaults()
+        d["k"] = 1
+        return d
+
+    def _factor(self, A, p):
+        fptr, fcol, lu, dia = _core.iluk_factor(A.nrows, A.ptr, A.col, A.val,
+                                                int(p["k"]))
+        return fptr, fcol, lu, dia
+
+
+class ILUT(ILU0):
+    """ILUT(p, tau) threshold smoother (parity: amgcl/relaxation/ilut.hpp:56)."""
+
+    @staticmethod
+    def defaults():
+        d = ILU0.defaults()
+        d["p"] = 2.0
+        d["tau"] = 1e-2
+        return d
+
+    def _factor(self, A, p):
+        fptr, fcol, lu, dia = _core.ilut_factor(A.nrows, A.ptr, A.col, A.val,
+                                                float(p["p"]), float(p["tau"]))
+        return fptr, fcol, lu, dia

@@ -59,3 +59,42 @@ class Spai0(DiagonalSmootherBase):
 
             m = hip_setup.spai0(A)
         self._setup_m(m, backend)
+
+
+class Spai1:
+    """SPAI-1 smoother: approximate inverse with A's sparsity
+    (parity: amgcl/relaxation/spai1.hpp:54). Apply: x += M (rhs - A x)."""
+
+    gpu_supported = True
+
+    @staticmethod
+    def defaults():
+        return {}
+
+    def __init__(self, A, prm, backend):
+        from ..matrix import CSR
+
+        merge_params(self.defaults(), prm)
+        if not isinstance(A, CSR):
+            from ..backend import hip_setup
+
+            A = hip_setup.download(A)
+        mp, mc, mv = _core.spai1(A.nrows, A.ptr, A.col, A.val)
+        self.backend = backend
+        self.M = backend.matrix(CSR(A.nrows, A.ncols, mp, mc, mv))
+        self.n = A.nrows
+
+    def _step(self, A, rhs, x, tmp):
+        b = self.backend
+        b.residual(rhs, A, x, tmp)
+        t2 = getattr(self, "_t2", None)
+        if t2 is None:
+            t2 = self._t2 = b.vector(self.n)
+        b.spmv(1.0, self.M, tmp, 0.0, t2)
+        b.axpby(1.0, t2, 1.0, x)
+
+    apply_pre = _step
+    apply_post = _step
+
+    def apply(self, A, rhs, x, tmp=None):
+        self.backend.spmv(1.0, self.M, rhs, 0.0, x)

@@ -12,6 +12,7 @@ import amgcl_amd as am
 
 COARSENING = ["smoothed_aggregation", "aggregation", "ruge_stuben"]
 RELAX = ["spai0", "damped_jacobi", "chebyshev", "gauss_seidel", "ilu0"]
+RELAX_EXTRA = ["spai1", "iluk", "ilut"]
 SOLVERS = ["cg", "bicgstab", "gmres", "richardson"]
 SOLVERS_EXTRA = ["bicgstabl", "fgmres", "lgmres", "idrs"]
 
@@ -140,3 +141,16 @@ def test_ruge_stuben_iteration_class(problem):
     )
     x, iters, resid = s(b)
     assert resid < 1e-8 and iters <= 20
+
+
+@pytest.mark.parametrize("relax", RELAX_EXTRA)
+def test_extra_smoothers(problem, relax):
+    A, b = problem
+    s = am.make_solver(
+        A,
+        {"precond": {"class": "amg", "relax": {"type": relax}, "coarse_enough": 500},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+    )
+    x, iters, resid = s(b)
+    assert resid < 1e-7
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6
