@@ -21,6 +21,10 @@ def _lazy(kind):
         from .schur import SchurPressureCorrection
 
         REGISTRY[kind] = SchurPressureCorrection
+    if kind == "deflation" and kind not in REGISTRY:
+        from .deflated import DeflatedPrecond
+
+        REGISTRY[kind] = DeflatedPrecond
     if kind in ("cpr", "cpr_drs") and kind not in REGISTRY:
         from .cpr import CPR
 

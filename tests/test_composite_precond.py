@@ -82,3 +82,18 @@ def test_cpr(kind):
     x, iters, resid = s(b)
     assert resid < 1e-7
     assert np.linalg.norm(b - K @ x) / np.linalg.norm(b) < 1e-6
+
+
+def test_shared_memory_deflation():
+    A, b = am.poisson3d(14, rhs="random")
+    rng = np.random.default_rng(2)
+    Z = np.stack([np.ones(A.nrows), rng.random(A.nrows)], axis=1)
+    s = am.make_solver(
+        A,
+        {"precond": {"class": "deflation", "Z_raw": Z,
+                     "inner": {"class": "amg", "coarse_enough": 400}},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+    )
+    x, it, res = s(b)
+    assert res < 1e-7
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6
