@@ -92,3 +92,58 @@ class ScaledProblem:
 
     def unscale_x(self, y):
         return np.asarray(y) * self.dsqrt_inv
+
+
+def complex_to_real(A, rhs=None):
+    """View a complex system as its 2x2-real expansion
+    (parity: amgcl/adapter/complex.hpp:45): each complex entry a+bi becomes
+    the block [[a, -b], [b, a]]; complex unknown z = u+vi becomes (u, v).
+    Returns (A_real CSR, rhs_real) — solve, then recombine with
+    real_to_complex()."""
+    import scipy.sparse as sp
+
+    if isinstance(A, CSR):
+        m = A.to_scipy().astype(np.complex128)
+    else:
+        m = sp.csr_matrix(A, dtype=np.complex128)
+    re, im = m.real.tocoo(), m.imag.tocoo()
+    rows = np.concatenate([2 * re.row, 2 * re.row, 2 * re.row + 1, 2 * re.row + 1])
+    cols = np.concatenate([2 * re.col, 2 * re.col + 1, 2 * re.col, 2 * re.col + 1])
+    vals = np.concatenate([re.data, np.zeros_like(re.data), np.zeros_like(re.data),
+                           re.data])
+    rows = np.concatenate([rows, 2 * im.row, 2 * im.row + 1])
+    cols = np.concatenate([cols, 2 * im.col + 1, 2 * im.col])
+    vals = np.concatenate([vals, -im.data, im.data])
+    n2 = 2 * m.shape[0]
+    mr = sp.coo_matrix((vals, (rows, cols)), shape=(n2, 2 * m.shape[1])).tocsr()
+    mr.sum_duplicates()
+    mr.sort_indices()
+    A_real = CSR.from_scipy(mr)
+    if rhs is None:
+        return A_real, None
+    rhs = np.asarray(rhs, dtype=np.complex128)
+    b = np.empty(n2)
+    b[0::2], b[1::2] = rhs.real, rhs.imag
+    return A_real, b
+
+
+def real_to_complex(x):
+    """Recombine the 2x2-real-expanded solution into complex."""
+    x = np.asarray(x)
+    return x[0::2] + 1j * x[1::2]
+
+
+class CrsBuilder:
+    """Matrix-free row-generator assembly
+    (parity: amgcl/adapter/crs_builder.hpp:105): build a CSR from a callable
+    row(i) -> (cols, vals)."""
+
+    def __init__(self, n, row_func):
+        ptr = np.zeros(n + 1, dtype=np.int64)
+        cols, vals = [], []
+        for i in range(n):
+            c, v = row_func(i)
+            cols.append(np.asarray(c, dtype=np.int32))
+            vals.append(np.asarray(v, dtype=np.float64))
+            ptr[i + 1] = ptr[i] + len(c)
+        self.matrix = CSR(n, n, ptr, np.concatenate(cols), np.concatenate(vals))

@@ -72,6 +72,16 @@ def spai0(A):
     return m
 
 
+def gershgorin(A, scale=False):
+    """Gershgorin spectral-radius bound on the device."""
+    t = _torch()
+    out = t.zeros(1, dtype=t.float64, device=A.val.device)
+    check(lib().amg_gershgorin(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
+                               A.val.data_ptr(), 1 if scale else 0, out.data_ptr(),
+                               _stream()), "gershgorin")
+    return float(out.item())
+
+
 def aggregates(A, eps_strong):
     """Device twin of _core.aggregates_parallel: returns (naggr, id, strong)."""
     t = _torch()

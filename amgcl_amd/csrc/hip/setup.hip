@@ -125,6 +125,52 @@ extern "C" int amg_setup_spai0(int64_t n, const int *ptr, const int *col, const 
     return (int)hipGetLastError();
 }
 
+// Gershgorin spectral-radius bound: max_i sum_j |a_ij| (optionally scaled by
+// 1/|a_ii|), for the Chebyshev smoother (amgcl/backend/builtin.hpp:781).
+__global__ void gersh_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
+                        const double *__restrict__ val, int scale, double *__restrict__ out) {
+    __shared__ double lds[4];
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    double m = 0.0;
+    for (; i < n; i += stride) {
+        double s = 0.0, dia = 1.0;
+        for (int j = ptr[i]; j < ptr[i + 1]; ++j) {
+            s += fabs(val[j]);
+            if (col[j] == (int)i) dia = fabs(val[j]);
+        }
+        if (scale && dia > 0.0) s /= dia;
+        if (s > m) m = s;
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        double o = __shfl_down(m, off, 64);
+        if (o > m) m = o;
+    }
+    int wid = threadIdx.x / 64, lane = threadIdx.x & 63;
+    if (lane == 0) lds[wid] = m;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        double bm = fmax(fmax(lds[0], lds[1]), fmax(lds[2], lds[3]));
+        // device-scope atomic max on double via CAS
+        unsigned long long *o = (unsigned long long *)out;
+        unsigned long long cur = *o;
+        while (__longlong_as_double((long long)cur) < bm) {
+            unsigned long long next = (unsigned long long)__double_as_longlong(bm);
+            unsigned long long old = atomicCAS(o, cur, next);
+            if (old == cur) break;
+            cur = old;
+        }
+    }
+}
+
+extern "C" int amg_gershgorin(int64_t n, const int *ptr, const int *col, const double *val,
+                              int scale, double *out, hipStream_t s) {
+    hipMemsetAsync(out, 0, sizeof(double), s);
+    gersh_k<<<nblk(n, 256, 1024), 256, 0, s>>>(n, ptr, col, val, scale, out);
+    return (int)hipGetLastError();
+}
+
 // ---------------------------------------------------------------------------
 // inclusive scan (i32), hierarchical
 // ---------------------------------------------------------------------------

@@ -18,18 +18,30 @@ class Chebyshev:
         return {"degree": 5, "lower": 1.0 / 30.0, "power_iters": 0, "scale": False}
 
     def __init__(self, A, prm, backend):
+        from ..matrix import CSR
+
         p = merge_params(self.defaults(), prm)
         self.degree = int(p["degree"])
         self.backend = backend
         self.scale = bool(p["scale"])
 
-        d = np.asarray(A.diagonal())
-        if self.scale:
-            self.Dinv = backend.from_host(1.0 / d)
-        else:
-            self.Dinv = None
+        if not isinstance(A, CSR):
+            # device-resident level: Gershgorin + diagonal on the GPU
+            from ..backend import hip_setup
 
-        rho = self._spectral_radius(A, d, int(p["power_iters"]))
+            if self.scale:
+                d32 = hip_setup.diagonal(A)
+                self.Dinv = 1.0 / d32
+            else:
+                self.Dinv = None
+            rho = hip_setup.gershgorin(A, self.scale)
+        else:
+            d = np.asarray(A.diagonal())
+            if self.scale:
+                self.Dinv = backend.from_host(1.0 / d)
+            else:
+                self.Dinv = None
+            rho = self._spectral_radius(A, d, int(p["power_iters"]))
         hi = rho
         lo = rho * float(p["lower"])
         self.theta = 0.5 * (hi + lo)

@@ -127,3 +127,49 @@ def test_cli_mm_with_reorder_scale(tmp_path):
     rc = main(["-A", str(tmp_path / "a.mtx"), "--reorder", "--scale",
                "-p", "solver.type=bicgstab"])
     assert rc == 0
+
+
+def test_complex_adapter():
+    import scipy.sparse as sp
+
+    from amgcl_amd.adapter import complex_to_real, real_to_complex
+
+    rng = np.random.default_rng(4)
+    n = 60
+    m = sp.random(n, n, density=0.1, random_state=rng, format="csr")
+    m = m + m.T + sp.identity(n) * 8.0
+    m = m.astype(np.complex128)
+    m.data = m.data + 1j * 0.2 * rng.standard_normal(m.nnz)
+    m = (m + m.conj().T).tocsr()  # hermitian
+    z = rng.standard_normal(n) + 1j * rng.standard_normal(n)
+    b = m @ z
+    Ar, br = complex_to_real(m, b)
+    s = am.make_solver(Ar, {"solver": {"type": "bicgstab", "tol": 1e-10,
+                                       "maxiter": 500}})
+    xr, it, res = s(br)
+    x = real_to_complex(xr)
+    assert np.linalg.norm(m @ x - b) / np.linalg.norm(b) < 1e-7
+
+
+def test_crs_builder():
+    from amgcl_amd.adapter import CrsBuilder
+
+    n = 50
+
+    def row(i):
+        cols, vals = [i], [2.0]
+        if i > 0:
+            cols.append(i - 1)
+            vals.append(-1.0)
+        if i + 1 < n:
+            cols.append(i + 1)
+            vals.append(-1.0)
+        order = np.argsort(cols)
+        return np.asarray(cols)[order], np.asarray(vals)[order]
+
+    A = CrsBuilder(n, row).matrix
+    b = np.ones(n)
+    s = am.make_solver(A, {"precond": {"class": "relaxation", "type": "ilu0"},
+                           "solver": {"type": "cg", "tol": 1e-10, "maxiter": 200}})
+    x, it, res = s(b)
+    assert np.linalg.norm(b - A @ x) < 1e-8
