@@ -593,19 +593,17 @@ static py::tuple aggregates_parallel(i64 nrows, arr<i32> ptr, arr<i32> col,
             }
         }
 
-        // pass 4: remaining UNDEF nodes adjacent to a FIRM member become
-        // provisional members of that aggregate (greedy 2nd-ring claim);
-        // deterministic choice: the strong neighbor with max root key wins.
-        next_active.clear();
-#pragma omp parallel
-        {
-            std::vector<i32> local_next;
-#pragma omp for schedule(static) nowait
-            for (i64 t = 0; t < na; ++t) {
-                i32 u = active[t];
-                if (id[u] != UNDEF) continue;
+        // pass 4 (two-phase): remaining UNDEF nodes adjacent to a FIRM member
+        // become provisional members (greedy 2nd-ring claim). The mark phase
+        // reads the stable post-claim state; the commit phase applies it —
+        // a fused pass would race with neighbors turning provisional.
+        std::vector<i32> choice(na);
+#pragma omp parallel for schedule(static)
+        for (i64 t = 0; t < na; ++t) {
+            i32 u = active[t];
+            i32 root = -1;
+            if (id[u] == UNDEF) {
                 uint64_t best = 0;
-                i32 root = -1;
                 for (i32 j = A.ptr[u]; j < A.ptr[u + 1]; ++j) {
                     i32 c = A.col[j];
                     if (!S[j] || c == u) continue;
@@ -614,8 +612,19 @@ static py::tuple aggregates_parallel(i64 nrows, arr<i32> ptr, arr<i32> col,
                         if (k > best) { best = k; root = id[c]; }
                     }
                 }
-                if (root >= 0) {
-                    id[u] = root;
+            }
+            choice[t] = root;
+        }
+        next_active.clear();
+#pragma omp parallel
+        {
+            std::vector<i32> local_next;
+#pragma omp for schedule(static) nowait
+            for (i64 t = 0; t < na; ++t) {
+                i32 u = active[t];
+                if (id[u] != UNDEF) continue;
+                if (choice[t] >= 0) {
+                    id[u] = choice[t];
                     prov[u] = 1;
                 } else {
                     local_next.push_back(u);

@@ -319,12 +319,16 @@ __global__ void agg_claim_k(int64_t n, const int *__restrict__ ptr, const int *_
     }
 }
 
-__global__ void agg_adopt_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
-                            const uint8_t *__restrict__ S, int *__restrict__ id,
-                            uint8_t *__restrict__ prov, int *__restrict__ remaining) {
+// Two-phase adoption: the mark pass reads the stable post-claim state and
+// records each node's choice; the commit pass applies it. A single fused
+// pass would race (a neighbor turning provisional concurrently could be
+// mistaken for a firm member, making the result timing-dependent).
+__global__ void agg_adopt_mark_k(int64_t n, const int *__restrict__ ptr,
+                                 const int *__restrict__ col, const uint8_t *__restrict__ S,
+                                 const int *__restrict__ id, const uint8_t *__restrict__ prov,
+                                 int *__restrict__ choice) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    int my_remaining = 0;
     for (; i < n; i += stride) {
         if (id[i] != -1) continue;
         uint64_t best = 0;
@@ -337,6 +341,20 @@ __global__ void agg_adopt_k(int64_t n, const int *__restrict__ ptr, const int *_
                 if (k > best) { best = k; root = id[c]; }
             }
         }
+        choice[i] = root;
+    }
+}
+
+__global__ void agg_adopt_commit_k(int64_t n, int *__restrict__ id,
+                                   uint8_t *__restrict__ prov,
+                                   const int *__restrict__ choice,
+                                   int *__restrict__ remaining) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int my_remaining = 0;
+    for (; i < n; i += stride) {
+        if (id[i] != -1) continue;
+        int root = choice[i];
         if (root >= 0) {
             id[i] = root;
             prov[i] = 1;
@@ -344,7 +362,6 @@ __global__ void agg_adopt_k(int64_t n, const int *__restrict__ ptr, const int *_
             ++my_remaining;
         }
     }
-    // one atomic per wave, not per undecided node (contention)
 #pragma unroll
     for (int off = WAVE / 2; off > 0; off >>= 1)
         my_remaining += __shfl_down(my_remaining, off, WAVE);
@@ -377,7 +394,9 @@ extern "C" int amg_agg_round(int64_t n, const int *ptr, const int *col, const ui
     agg_m1_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, near, m1);
     agg_roots_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, m1, newroot);
     agg_claim_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, prov, newroot);
-    agg_adopt_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, prov, remaining);
+    // reuse the m1 buffer as the choice array (i32 fits in the u64 slots)
+    agg_adopt_mark_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, prov, (int *)m1);
+    agg_adopt_commit_k<<<nblk(n), 256, 0, s>>>(n, id, prov, (int *)m1, remaining);
     return (int)hipGetLastError();
 }
 extern "C" int amg_agg_renumber(int64_t n, int *id, int *mark, hipStream_t s) {
