@@ -158,7 +158,9 @@ def main():
         rank = int(os.environ["RANK"])
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
         if _has_gpu():
-            torch.cuda.set_device(local_rank)
+            # modulo allows oversubscribed single-GPU testing; in production
+            # one rank per GPU (local_rank < device_count)
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
             dist.init_process_group("nccl")
         else:
             dist.init_process_group("gloo")
