@@ -109,11 +109,7 @@ class DenseCoarseSolver:
     """
 
     def __init__(self, csr: CSR, backend):
-        a = np.zeros(csr.shape, dtype=np.float64)
-        for i in range(csr.nrows):
-            lo, hi = csr.ptr[i], csr.ptr[i + 1]
-            a[i, csr.col[lo:hi]] = csr.val[lo:hi]
-        self.inv = np.linalg.inv(a)
+        self.inv = np.linalg.inv(csr.to_scipy().toarray())
 
     def __call__(self, f, u):
         np.matmul(self.inv, f, out=u)

@@ -226,13 +226,10 @@ class DeviceDenseSolver:
     inverse (design note in backend/cpu.py:DenseCoarseSolver)."""
 
     def __init__(self, csr: CSR, backend):
-        a = np.zeros(csr.shape, dtype=np.float64)
-        for i in range(csr.nrows):
-            lo, hi = csr.ptr[i], csr.ptr[i + 1]
-            a[i, csr.col[lo:hi]] = csr.val[lo:hi]
+        a = csr.to_scipy().toarray()
         inv = np.linalg.inv(a)
         self.n = csr.nrows
-        self.inv = backend.from_host(inv.ravel())
+        self.inv = backend.from_host(np.ascontiguousarray(inv).ravel())
 
     def __call__(self, f, u):
         import torch
