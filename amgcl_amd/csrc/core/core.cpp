@@ -157,6 +157,74 @@ static py::tuple poisson3d_strip(i64 n, i64 row_beg, i64 row_end) {
     return py::make_tuple(ptr, col, val);
 }
 
+// Split a distributed row strip (global columns) into the square local part
+// (columns renumbered to local) and the remote part over deduplicated ghost
+// columns (parity: amgcl/mpi/distributed_matrix.hpp:370-430).
+// Returns (lptr, lcol, lval, rptr, rcol, rval, ghost_global[i64]).
+static py::tuple split_strip(i64 nloc, i64 row_beg, i64 row_end, arr<i32> ptr_a,
+                             arr<i32> col_a, arr<double> val_a) {
+    const i32 *ptr = ptr_a.data();
+    const i32 *col = col_a.data();
+    const double *val = val_a.data();
+    (void)row_end;
+
+    arr<i32> lptr(nloc + 1), rptr(nloc + 1);
+    i32 *LP = lptr.mutable_data();
+    i32 *RP = rptr.mutable_data();
+    LP[0] = RP[0] = 0;
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < nloc; ++i) {
+        i32 lc = 0, rc = 0;
+        for (i32 j = ptr[i]; j < ptr[i + 1]; ++j) {
+            i64 c = col[j];
+            if (c >= row_beg && c < row_beg + nloc) ++lc;
+            else ++rc;
+        }
+        LP[i + 1] = lc;
+        RP[i + 1] = rc;
+    }
+    scan_i32(LP + 1, nloc);
+    scan_i32(RP + 1, nloc);
+    const i64 lnnz = LP[nloc], rnnz = RP[nloc];
+    arr<i32> lcol = big_arr<i32>(lnnz), rcol = big_arr<i32>(rnnz);
+    arr<double> lval = big_arr<double>(lnnz), rval = big_arr<double>(rnnz);
+    i32 *LC = lcol.mutable_data();
+    i32 *RC = rcol.mutable_data();
+    double *LV = lval.mutable_data();
+    double *RV = rval.mutable_data();
+    // remote columns keep GLOBAL ids for now (renumbered after dedup)
+#pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < nloc; ++i) {
+        i32 lh = LP[i], rh = RP[i];
+        for (i32 j = ptr[i]; j < ptr[i + 1]; ++j) {
+            i64 c = col[j];
+            if (c >= row_beg && c < row_beg + nloc) {
+                LC[lh] = (i32)(c - row_beg);
+                LV[lh] = val[j];
+                ++lh;
+            } else {
+                RC[rh] = col[j];
+                RV[rh] = val[j];
+                ++rh;
+            }
+        }
+    }
+    // dedup + sort remote global columns
+    std::vector<i32> ghosts(RC, RC + rnnz);
+    std::sort(ghosts.begin(), ghosts.end());
+    ghosts.erase(std::unique(ghosts.begin(), ghosts.end()), ghosts.end());
+    arr<int64_t> ghost_out(ghosts.size());
+    int64_t *G = ghost_out.mutable_data();
+    for (size_t g = 0; g < ghosts.size(); ++g) G[g] = ghosts[g];
+    // renumber remote columns to ghost indices
+#pragma omp parallel for schedule(static)
+    for (i64 j = 0; j < rnnz; ++j) {
+        RC[j] = (i32)(std::lower_bound(ghosts.begin(), ghosts.end(), RC[j]) -
+                      ghosts.begin());
+    }
+    return py::make_tuple(lptr, lcol, lval, rptr, rcol, rval, ghost_out);
+}
+
 // ---------------------------------------------------------------------------
 // diagonal
 // ---------------------------------------------------------------------------
@@ -1784,6 +1852,7 @@ PYBIND11_MODULE(_core, m) {
     m.doc() = "amgcl_amd host setup engine (OpenMP)";
     m.def("poisson3d", &poisson3d, py::arg("n"), py::arg("anisotropy") = 1.0);
     m.def("poisson3d_strip", &poisson3d_strip);
+    m.def("split_strip", &split_strip);
     m.def("diagonal", &diagonal);
     m.def("transpose", &transpose);
     m.def("spgemm", &spgemm);

@@ -34,26 +34,20 @@ class DistMatrix:
         self.row_begs = np.concatenate([[0], np.cumsum(sizes)]).astype(np.int64)
         self.n_loc = n_loc
 
-        # split into local (square) + remote (ghost) parts
-        col = strip.col.astype(np.int64)
-        loc_mask = (col >= self.row_beg) & (col < self.row_end)
-        rem_mask = ~loc_mask
-        row_of = np.repeat(np.arange(n_loc, dtype=np.int64), np.diff(strip.ptr))
+        # split into local (square) + remote (ghost) parts (C++ engine;
+        # parity: distributed_matrix.hpp:370-430)
+        from .. import _core
 
-        def build(mask, cols):
-            ptr = np.zeros(n_loc + 1, dtype=np.int64)
-            np.cumsum(np.bincount(row_of[mask], minlength=n_loc), out=ptr[1:])
-            return ptr, cols, strip.val[mask]
-
-        lp, lc, lv = build(loc_mask, (col[loc_mask] - self.row_beg).astype(np.int32))
-        ghost_global = np.unique(col[rem_mask])
+        lp, lc, lv, rp, rc, rv, ghost_global = _core.split_strip(
+            n_loc, self.row_beg, self.row_end, strip.ptr, strip.col, strip.val
+        )
+        ghost_global = np.asarray(ghost_global)
         self.n_ghost = len(ghost_global)
-        g_idx = np.searchsorted(ghost_global, col[rem_mask]).astype(np.int32)
-        rp, rc, rv = build(rem_mask, g_idx)
 
-        self.A_loc = base.matrix(CSR(n_loc, n_loc, lp, lc, lv))
-        self.A_rem = base.matrix(CSR(n_loc, self.n_ghost, rp, rc, rv)) if self.n_ghost else None
         self.A_loc_host = CSR(n_loc, n_loc, lp, lc, lv)
+        self.A_loc = base.matrix(self.A_loc_host)
+        self.A_rem = (base.matrix(CSR(n_loc, self.n_ghost, rp, rc, rv))
+                      if self.n_ghost else None)
 
         # --- comm pattern (who owns each ghost column; what must we send) ---
         owner = np.searchsorted(self.row_begs, ghost_global, side="right") - 1
