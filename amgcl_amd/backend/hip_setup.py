@@ -29,15 +29,6 @@ def _torch():
     return torch
 
 
-class _Dev:
-    """Lazily-allocated workspace shared across levels."""
-
-    def __init__(self, device):
-        self.device = device
-        t = _torch()
-        self.flag = t.zeros(2, dtype=t.int32, device=device)  # overflow/remaining
-
-
 def device_csr(nrows, ncols, ptr, col, val, subw=0):
     from .hip import DeviceCSR
 
