@@ -64,10 +64,13 @@ def main():
     Ah, bh, coords = elasticity3d(n)
     B = rigid_body_modes(coords)
     bd = hip.from_host(bh)
-    run(f"elasticity {n}^3 nodes CG+SA+RBM+BSR(3)", Ah, bd,
+    run(f"elasticity {n}^3 nodes CG+SA(esr)+RBM+Chebyshev+BSR(3)", Ah, bd,
         {"precond": {"class": "amg", "block_value": 3, "keep_host_matrices": True,
+                     "relax": {"type": "chebyshev"},
                      "coarsening": {"type": "smoothed_aggregation",
-                                    "nullspace_raw": B, "block_size": 3}},
+                                    "nullspace_raw": B, "block_size": 3,
+                                    "estimate_spectral_radius": True,
+                                    "power_iters": 10}},
          "solver": {"type": "cg", "tol": 1e-6, "maxiter": 500}}, hip)
 
     # config #5-class: Schur pressure correction on a stabilized saddle system

@@ -8,14 +8,14 @@ from amgcl_amd.matrix import CSR
 
 
 def stokes_like(n=10):
-    """Stabilized saddle-point-ish system: velocity Poisson (2 comps) coupled
-    with a stabilized pressure block."""
-    rng = np.random.default_rng(5)
+    """Stabilized saddle-point-ish system: velocity Poisson coupled with a
+    stabilized pressure block through a local discrete-gradient-like operator
+    (local coupling keeps the Schur complement sparse, like real Stokes)."""
     Ap, _ = am.poisson3d(n)
     a = Ap.to_scipy()
     nv = a.shape[0]
-    B = sp.random(nv, nv, density=0.002, random_state=rng, format="csr")
-    B.data = 0.1 * rng.standard_normal(B.nnz)
+    # discrete gradient-ish stencil: B = 0.1 (I - shift_x)
+    B = 0.1 * (sp.identity(nv) - sp.diags(np.ones(nv - 1), 1)).tocsr()
     K = sp.bmat([[a, B], [B.T, a + sp.identity(nv)]], format="csr")
     K.sort_indices()
     pmask = np.zeros(2 * nv, dtype=bool)

@@ -199,8 +199,11 @@ def test_block_value_amg_solve(hip):
     B = rigid_body_modes(coords)
     prm = {"precond": {"class": "amg", "coarse_enough": 500, "block_value": 3,
                        "keep_host_matrices": True,
+                       "relax": {"type": "chebyshev"},
                        "coarsening": {"type": "smoothed_aggregation",
-                                      "nullspace_raw": B, "block_size": 3}},
+                                      "nullspace_raw": B, "block_size": 3,
+                                      "estimate_spectral_radius": True,
+                                      "power_iters": 10}},
            "solver": {"type": "cg", "tol": 1e-8, "maxiter": 300}}
     s = am.make_solver(A, prm, backend=hip)
     from amgcl_amd.backend.hip import DeviceBSR

@@ -15,8 +15,11 @@ def test_rbm_improves_elasticity_convergence():
             "solver": {"type": "cg", "tol": 1e-8, "maxiter": 1000}}
     x0, it0, r0 = am.make_solver(A, prm0)(b)
     prm1 = {"precond": {"class": "amg", "coarse_enough": 500,
+                        "relax": {"type": "chebyshev"},
                         "coarsening": {"type": "smoothed_aggregation",
-                                       "nullspace_raw": B, "block_size": 3}},
+                                       "nullspace_raw": B, "block_size": 3,
+                                       "estimate_spectral_radius": True,
+                                       "power_iters": 10}},
             "solver": {"type": "cg", "tol": 1e-8, "maxiter": 1000}}
     x1, it1, r1 = am.make_solver(A, prm1)(b)
     assert r1 < 1e-7
