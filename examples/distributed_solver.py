@@ -10,6 +10,11 @@ Uses RCCL on GPUs (one rank per GPU), gloo on CPU-only hosts.
 import os
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import os
+import sys
+
 import numpy as np
 import torch
 import torch.distributed as dist

@@ -1,6 +1,11 @@
 #!/usr/bin/env python3
 """Elasticity with rigid-body-mode nullspace and block-valued storage
 (reference analogue: tutorial/Nullspace, CoupCons3D block values)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import sys
 
 import amgcl_amd as am

@@ -1,6 +1,11 @@
 #!/usr/bin/env python3
 """Minimal flagship example: AMG-preconditioned CG on 3D Poisson
 (reference analogue: examples/solver.cpp, tutorial/poisson3Db)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import sys
 
 import amgcl_amd as am

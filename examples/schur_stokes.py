@@ -1,6 +1,11 @@
 #!/usr/bin/env python3
 """Schur pressure-correction field split on a stabilized saddle-point system
 (reference analogue: tutorial/Stokes, examples/schur_pressure_correction)."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import sys
 
 import numpy as np
