@@ -60,6 +60,7 @@ _SIGS = {
     "amg_gemv_f32": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
     "amg_cast_d2s": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     "amg_cast_s2d": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
+    "amg_gs_color_f64": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
     # --- block (BSR) solve kernels ---
     "amg_bsr_spmv_f64": [ctypes.c_int64, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
                          ctypes.c_void_p, ctypes.c_void_p, ctypes.c_double,

@@ -213,3 +213,20 @@ def test_block_value_amg_solve(hip):
     assert resid < 1e-8
     r = b - A @ hip.to_host(x)
     assert np.linalg.norm(r) / np.linalg.norm(b) < 1e-6
+
+
+def test_multicolor_gauss_seidel_gpu(hip):
+    """GPU multicolor GS (exceeds the reference, which is CPU-only for GS):
+    must match the convergence class of the serial CPU sweeps."""
+    A, b = am.poisson3d(24, rhs="random")
+    prm = {"precond": {"class": "amg", "relax": {"type": "gauss_seidel"},
+                       "coarse_enough": 500},
+           "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}}
+    s_gpu = am.make_solver(A, prm, backend=hip)
+    x, it_gpu, res = s_gpu(b)
+    assert res < 1e-8
+    r = b - A @ hip.to_host(x)
+    assert np.linalg.norm(r) / np.linalg.norm(b) < 1e-7
+    s_cpu = am.make_solver(A, prm)
+    _, it_cpu, _ = s_cpu(b)
+    assert it_gpu <= it_cpu + 5
