@@ -31,6 +31,8 @@ extern "C" int amg_dot_f64(int64_t, const double *, const double *, double *, hi
 extern "C" int amg_dot2_f64(int64_t, const double *, const double *, const double *,
                             const double *, double *, hipStream_t);
 extern "C" int amg_gemv_f64(int64_t, const double *, const double *, double *, hipStream_t);
+extern "C" int amg_cg_tail_f64(int64_t, double, const double *, const double *, double *,
+                               double *, double *, hipStream_t);
 extern "C" int amg_vmul_f64(int64_t, double, const double *, const double *, double,
                             double *, hipStream_t);
 
@@ -278,9 +280,8 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
         CHK(amg_dot_f64(n, q, p, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
         double alpha = rho1 / dots[0];
-        CHK(amg_axpby_f64(n, alpha, p, 1.0, x, st));
-        CHK(amg_axpby_f64(n, -alpha, q, 1.0, r, st));
-        CHK(amg_dot_f64(n, r, r, D->dotbuf_d, st));
+        // fused x/r update + residual norm (single pass over x,r,p,q)
+        CHK(amg_cg_tail_f64(n, alpha, p, q, x, r, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
         res = sqrt(dots[0]);
         ++iter;

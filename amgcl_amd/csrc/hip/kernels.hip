@@ -509,3 +509,36 @@ extern "C" int amg_gs_color_f64(int64_t nlist, const int *rows, const int *ptr,
     gs_color_k<<<nblocks(nlist), 256, 0, stream>>>(nlist, rows, ptr, col, val, b, x);
     return (int)hipGetLastError();
 }
+
+
+// ---------------------------------------------------------------------------
+// Fused CG tail: x += alpha p ; r -= alpha q ; out[0] += ||r_new||^2
+// (saves one full pass over r and a separate reduction per iteration)
+// ---------------------------------------------------------------------------
+__global__ void cg_tail_k(int64_t n, double alpha, const double *__restrict__ p,
+                          const double *__restrict__ q, double *__restrict__ x,
+                          double *__restrict__ r, double *__restrict__ out) {
+    __shared__ double lds[4];
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    double s = 0.0;
+    for (; i < n; i += stride) {
+        x[i] += alpha * p[i];
+        double rn = r[i] - alpha * q[i];
+        r[i] = rn;
+        s += rn * rn;
+    }
+#pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1) s += __shfl_down(s, off, WAVE);
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
+    if (lane == 0) lds[wid] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) atomicAdd(out, lds[0] + lds[1] + lds[2] + lds[3]);
+}
+
+extern "C" int amg_cg_tail_f64(int64_t n, double alpha, const double *p, const double *q,
+                               double *x, double *r, double *out, hipStream_t stream) {
+    fill_k<double><<<1, 64, 0, stream>>>(1, 0.0, out);
+    cg_tail_k<<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, alpha, p, q, x, r, out);
+    return (int)hipGetLastError();
+}
