@@ -7,11 +7,13 @@ an object with .defaults(), .transfer_operators(A) -> (P, R) and
 from .aggregation import Aggregation
 from .ruge_stuben import RugeStuben
 from .smoothed_aggregation import SmoothedAggregation
+from .smoothed_aggr_emin import SmoothedAggrEMin
 
 REGISTRY = {
     "aggregation": Aggregation,
     "ruge_stuben": RugeStuben,
     "smoothed_aggregation": SmoothedAggregation,
+    "smoothed_aggr_emin": SmoothedAggrEMin,
 }
 
 

@@ -154,3 +154,38 @@ def test_extra_smoothers(problem, relax):
     x, iters, resid = s(b)
     assert resid < 1e-7
     assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6
+
+
+def test_anisotropic_poisson():
+    """Anisotropy exercises the strong-connection filtering
+    (reference fixture supports anisotropy, tests/sample_problem.hpp)."""
+    A, b = am.poisson3d(16, anisotropy=0.25, rhs="random")
+    s = am.make_solver(
+        A, {"precond": {"class": "amg", "coarse_enough": 500},
+            "solver": {"type": "cg", "tol": 1e-8, "maxiter": 200}})
+    x, iters, resid = s(b)
+    assert resid < 1e-8
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-7
+    assert iters < 100
+
+
+def test_smoothed_aggr_emin_nonsymmetric():
+    """Energy-minimizing SA with separately smoothed R on a nonsymmetric
+    convection-diffusion system (parity: coarsening/smoothed_aggr_emin.hpp)."""
+    import scipy.sparse as sp
+
+    A, b = am.poisson3d(14, rhs="random")
+    m = A.to_scipy()
+    n = m.shape[0]
+    conv = 0.4 * (sp.diags(np.ones(n - 1), 1) - sp.diags(np.ones(n - 1), -1))
+    K = (m + conv).tocsr()
+    K.sort_indices()
+    Ak = am.CSR.from_scipy(K)
+    s = am.make_solver(
+        Ak, {"precond": {"class": "amg", "coarsening": {"type": "smoothed_aggr_emin"},
+                         "coarse_enough": 400},
+             "solver": {"type": "bicgstab", "tol": 1e-8, "maxiter": 200}})
+    x, iters, resid = s(b)
+    assert resid < 1e-7
+    assert np.linalg.norm(b - K @ x) / np.linalg.norm(b) < 1e-6
+    assert iters < 40
