@@ -10,7 +10,7 @@ from .spai0 import Spai0, Spai1
 from .damped_jacobi import DampedJacobi
 from .chebyshev import Chebyshev
 from .gauss_seidel import GaussSeidel
-from .ilu0 import ILU0, ILUK, ILUT
+from .ilu0 import ILU0, ILU0ChowPatel, ILUK, ILUT
 
 REGISTRY = {
     "spai0": Spai0,
@@ -21,6 +21,7 @@ REGISTRY = {
     "ilu0": ILU0,
     "iluk": ILUK,
     "ilut": ILUT,
+    "ilu0_chow_patel": ILU0ChowPatel,
 }
 
 

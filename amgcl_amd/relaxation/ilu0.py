@@ -145,3 +145,19 @@ class ILUT(ILU0):
         fptr, fcol, lu, dia = _core.ilut_factor(A.nrows, A.ptr, A.col, A.val,
                                                 float(p["p"]), float(p["tau"]))
         return fptr, fcol, lu, dia
+
+
+class ILU0ChowPatel(ILU0):
+    """Fine-grained (Chow-Patel) parallel ILU(0) factorization
+    (parity: amgcl/relaxation/ilu0_chow_patel.hpp:87)."""
+
+    @staticmethod
+    def defaults():
+        d = ILU0.defaults()
+        d["sweeps"] = 4
+        return d
+
+    def _factor(self, A, p):
+        fptr, fcol, lu, dia = _core.ilu0_chow_patel(A.nrows, A.ptr, A.col, A.val,
+                                                    int(p["sweeps"]))
+        return fptr, fcol, lu, dia

@@ -12,7 +12,7 @@ import amgcl_amd as am
 
 COARSENING = ["smoothed_aggregation", "aggregation", "ruge_stuben"]
 RELAX = ["spai0", "damped_jacobi", "chebyshev", "gauss_seidel", "ilu0"]
-RELAX_EXTRA = ["spai1", "iluk", "ilut"]
+RELAX_EXTRA = ["spai1", "iluk", "ilut", "ilu0_chow_patel"]
 SOLVERS = ["cg", "bicgstab", "gmres", "richardson"]
 SOLVERS_EXTRA = ["bicgstabl", "fgmres", "lgmres", "idrs"]
 
