@@ -134,14 +134,23 @@ static int cycle(Driver *D, int li, const double *f, double **u_io, double **scr
     LevelDesc &L = D->lv[li];
     const bool coarsest = (li + 1 == (int)D->lv.size());
 
+    // relax_zero writes into the swap buffer and swaps, so zero-guess and
+    // general smooths have identical pointer parity (no copy-back needed)
+    auto relax_zero = [&](LevelDesc &LL, const double *ff, double **u2,
+                          double **sc) {
+        relax_zero_k<<<nblocks_d(LL.nrows), 256, 0, D->stream>>>(LL.nrows, LL.M, ff, *sc);
+        double *tmp = *u2;
+        *u2 = *sc;
+        *sc = tmp;
+    };
+
     if (coarsest) {
         if (D->coarse_inv) {
             CHK(amg_gemv_f64(L.nrows, D->coarse_inv, f, *u_io, D->stream));
         } else {
             for (int i = 0; i < D->npre + D->npost; ++i) {
                 if (u_is_zero && i == 0) {
-                    relax_zero_k<<<nblocks_d(L.nrows), 256, 0, D->stream>>>(L.nrows, L.M,
-                                                                            f, *u_io);
+                    relax_zero(L, f, u_io, scratch);
                     continue;
                 }
                 CHK(relax_swap(D, L, f, u_io, scratch));
@@ -153,7 +162,7 @@ static int cycle(Driver *D, int li, const double *f, double **u_io, double **scr
     LevelDesc &N = D->lv[li + 1];
     for (int i = 0; i < D->npre; ++i) {
         if (u_is_zero && i == 0) {
-            relax_zero_k<<<nblocks_d(L.nrows), 256, 0, D->stream>>>(L.nrows, L.M, f, *u_io);
+            relax_zero(L, f, u_io, scratch);
             continue;
         }
         CHK(relax_swap(D, L, f, u_io, scratch));
