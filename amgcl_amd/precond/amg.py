@@ -74,6 +74,12 @@ class AMG:
         relax_factory = make_relaxation_factory(self.prm["relax"])
         A = A_new
         for lvl in self.levels[:-1]:
+            # mirror the device->host handoff of the original build: if this
+            # level's transfers were built on the host, bring A back down
+            if isinstance(lvl.P_build, CSR) and not isinstance(A, CSR):
+                from ..backend import hip_setup
+
+                A = hip_setup.download(A)
             lvl.rows, lvl.nnz = A.nrows, A.nnz
             if lvl.A_host is not None:
                 lvl.A_host = A
@@ -81,6 +87,10 @@ class AMG:
             lvl.relax = relax_factory(A, backend)
             A = self._coarsening.coarse_operator(A, lvl.P_build, lvl.R_build)
         last = self.levels[-1]
+        if not isinstance(A, CSR) and A.nrows <= 20000:
+            from ..backend import hip_setup
+
+            A = hip_setup.download(A)
         last.rows, last.nnz = A.nrows, A.nnz
         if last.A_host is not None:
             last.A_host = A
