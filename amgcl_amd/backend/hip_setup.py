@@ -123,10 +123,10 @@ def smoothed_prolongation(A, S, ids, naggr, omega):
     check(lib().amg_psmooth_count(n, A.ptr.data_ptr(), A.col.data_ptr(), S.data_ptr(),
                                   ids.data_ptr(), pptr[1:].data_ptr(),
                                   overflow.data_ptr(), _stream()), "psmooth_count")
-    if int(overflow.item()) != 0:
-        raise OverflowError("P row exceeds device buffer; use host setup")
     _scan_ptr(pptr)
-    nnz = int(pptr[-1].item())
+    ov, nnz = (int(v) for v in t.cat([overflow, pptr[-1:]]).cpu())
+    if ov != 0:
+        raise OverflowError("P row exceeds device buffer; use host setup")
     pcol = t.empty(nnz, dtype=t.int32, device=dev)
     pval = t.empty(nnz, dtype=t.float64, device=dev)
     check(lib().amg_psmooth_fill(n, A.ptr.data_ptr(), A.col.data_ptr(), A.val.data_ptr(),
@@ -182,10 +182,10 @@ def spgemm(A, B, sort=True):
                                  B.ptr.data_ptr(), B.col.data_ptr(), ub.data_ptr(),
                                  cptr[1:].data_ptr(), overflow.data_ptr(), _stream()),
           "spgemm_count")
-    if int(overflow.item()) != 0:
-        raise OverflowError("spgemm row exceeds LDS hash; use host setup")
     _scan_ptr(cptr)
-    nnz = int(cptr[-1].item())
+    ov, nnz = (int(v) for v in t.cat([overflow, cptr[-1:]]).cpu())
+    if ov != 0:
+        raise OverflowError("spgemm row exceeds LDS hash; use host setup")
     ccol = t.empty(nnz, dtype=t.int32, device=dev)
     cval = t.empty(nnz, dtype=t.float64, device=dev)
     check(lib().amg_spgemm_fill(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
