@@ -94,14 +94,27 @@ class SmoothedAggregation:
         return P, R
 
     def _transfer_operators_device(self, A):
-        """Device twin (backend/hip_setup.py): same algorithm, same keys."""
+        """Device twin (backend/hip_setup.py): same algorithm, same keys.
+        Features the device engine does not cover yet (pointwise/block
+        aggregation, nullspace) raise OverflowError, which AMG handles by
+        downloading the level and continuing with the host engine."""
         from ..backend import hip_setup
         from ..profiler import prof
+
+        if int(self.prm["block_size"]) > 1 or self.B is not None:
+            raise OverflowError("block/nullspace coarsening runs on the host")
 
         with prof.scope("aggregates(dev)"):
             naggr, ids, strong = hip_setup.aggregates(A, self.eps_strong)
         self.eps_strong *= 0.5
-        omega = float(self.prm["relax"]) * (2.0 / 3.0)
+        omega = float(self.prm["relax"])
+        if self.prm["estimate_spectral_radius"]:
+            # power iteration on D^-1 A entirely on the device
+            omega *= (4.0 / 3.0) / _spectral_radius_device(
+                A, int(self.prm["power_iters"])
+            )
+        else:
+            omega *= 2.0 / 3.0
         with prof.scope("smooth_P(dev)"):
             P = hip_setup.smoothed_prolongation(A, strong, ids, naggr, omega)
         with prof.scope("transpose_R(dev)"):
@@ -129,4 +142,31 @@ def spectral_radius_dinv_a(A: CSR, iters):
         if nrm == 0:
             break
         b0, b1 = b1 / nrm, b0
+    return abs(rho)
+
+
+def _spectral_radius_device(A, iters):
+    """Power-iteration estimate of rho(D^-1 A) on a device matrix."""
+    import torch
+
+    from ..backend import hip_setup
+    from ..backend.hip import HipBackend
+
+    hip = HipBackend.__new__(HipBackend)  # op-only use; no state needed
+    d = hip_setup.diagonal(A)
+    dinv = 1.0 / d
+    g = torch.Generator(device=d.device).manual_seed(12345)
+    b0 = torch.rand(A.nrows, dtype=torch.float64, device=d.device, generator=g)
+    b0 /= b0.norm()
+    b1 = torch.empty_like(b0)
+    rho = 2.0
+    for _ in range(max(1, iters)):
+        hip.spmv(1.0, A, b0, 0.0, b1)
+        b1 *= dinv
+        rho = float(b0.dot(b1))
+        nrm = float(b1.norm())
+        if nrm == 0:
+            break
+        b0 = b1 / nrm
+        b1 = torch.empty_like(b0)
     return abs(rho)
