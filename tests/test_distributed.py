@@ -225,3 +225,34 @@ def test_distributed_schur_field_split(world):
     b = rng.standard_normal(2 * nv)
     x = np.array(xg)
     assert np.linalg.norm(b - K @ x) / np.linalg.norm(b) < 1e-6
+
+
+def _solve_bicgstab(rank, world):
+    import amgcl_amd as am
+    from amgcl_amd.parallel import make_dist_solver
+
+    n = 16
+    strip, b, rb, re_ = am.poisson3d_strip(n, rank, world, rhs="ones")
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "amg", "coarse_enough": 300},
+         "solver": {"type": "bicgstab", "tol": 1e-8, "maxiter": 200}},
+        backend="cpu",
+    )
+    x, iters, resid = solve(b)
+    xg = solve.gather_solution(x)
+    return iters, resid, None if xg is None else xg.tolist()
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_bicgstab_fused_dots(world):
+    """Distributed BiCGStab exercises the fused 2-dot allreduce path."""
+    out = spawn(world, _solve_bicgstab, 29961 + world)
+    import amgcl_amd as am
+
+    iters, resid, xg = out[0]
+    assert resid < 1e-7
+    n = 16
+    A, b = am.poisson3d(n, rhs="ones")
+    x = np.array(xg)
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6

@@ -161,3 +161,40 @@ def test_native_library_is_loaded(hip):
     """The .so with hand-written kernels must actually be mapped in-process."""
     maps = open("/proc/self/maps").read()
     assert "libamghip.so" in maps
+
+
+def test_f32_kernels_match_f64(hip):
+    """fp32 kernel variants (mixed-precision hierarchy) vs fp64 reference."""
+    import torch
+
+    from amgcl_amd.backend.hip import DeviceCSR
+
+    rng = np.random.default_rng(23)
+    A, _ = am.poisson3d(10)
+    Ad = DeviceCSR(A, hip.device)
+    A32 = DeviceCSR.from_tensors(Ad.nrows, Ad.ncols, Ad.ptr, Ad.col,
+                                 Ad.val.to(torch.float32))
+    x = rng.standard_normal(A.nrows)
+    x64 = hip.from_host(x)
+    x32 = x64.to(torch.float32)
+    y64 = hip.vector(A.nrows)
+    y32 = hip.vector(A.nrows, torch.float32)
+    hip.spmv(1.0, Ad, x64, 0.0, y64)
+    hip.spmv(1.0, A32, x32, 0.0, y32)
+    assert (y32.double() - y64).abs().max().item() < 1e-4
+    b64 = hip.from_host(rng.standard_normal(A.nrows))
+    b32 = b64.to(torch.float32)
+    r32 = hip.vector(A.nrows, torch.float32)
+    hip.residual(b32, A32, x32, r32)
+    r64 = hip.vector(A.nrows)
+    hip.residual(b64, Ad, x64, r64)
+    assert (r32.double() - r64).abs().max().item() < 1e-4
+    # axpby/vmul/dot f32
+    hip.axpby(1.5, x32, -0.5, r32)
+    assert torch.isfinite(r32).all()
+    d = hip.dot(x32, x32)
+    assert abs(d - float(x64.dot(x64))) < 1e-3
+    # casts
+    z64 = hip.vector(A.nrows)
+    hip.cast(x32, z64)
+    assert (z64 - x32.double()).abs().max().item() == 0.0
