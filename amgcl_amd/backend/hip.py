@@ -201,6 +201,12 @@ class HipBackend:
         self._dothost.copy_(self._dotbuf, non_blocking=False)
         return float(self._dothost[0]), float(self._dothost[1])
 
+    def blkdiag_vmul(self, bsize, M, x, y):
+        """y_i = M_i x_i with M a flat (nblocks*B*B) row-major block-diagonal."""
+        nb = x.numel() // bsize
+        check(lib().amg_blkdiag_vmul_f64(nb, bsize, M.data_ptr(), x.data_ptr(),
+                                         y.data_ptr(), _stream()), "blkdiag_vmul")
+
     def gather(self, x, idx, buf):
         check(self._fn("gather", x)(idx.numel(), x.data_ptr(), idx.data_ptr(),
                                     buf.data_ptr(), _stream()), "gather")

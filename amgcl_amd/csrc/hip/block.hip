@@ -92,6 +92,26 @@ __global__ void bsr_relax_k(int64_t nbrows, const int *__restrict__ ptr,
     }
 }
 
+// block-diagonal matvec y_i = M_i x_i (M: nbrows row-major BxB blocks).
+// Used by the as_block smoother wrapper (reference: relaxation/as_block.hpp,
+// where the block-typed base smoother's M is a block per point).
+template <int B>
+__global__ void blkdiag_vmul_k(int64_t nbrows, const double *__restrict__ M,
+                               const double *__restrict__ x, double *__restrict__ y) {
+    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int lr = (int)(tid % B);
+    int64_t brow = tid / B;
+    int64_t stride = ((int64_t)gridDim.x * blockDim.x) / B;
+    for (; brow < nbrows; brow += stride) {
+        const double *m = M + brow * B * B + lr * B;
+        const double *xb = x + brow * B;
+        double s = 0.0;
+#pragma unroll
+        for (int c = 0; c < B; ++c) s += m[c] * xb[c];
+        y[brow * B + lr] = s;
+    }
+}
+
 #define BSR_LAUNCH(kname, ...)                                              \
     switch (bsize) {                                                        \
         case 2: kname<2><<<grid, 256, 0, s>>>(__VA_ARGS__); break;          \
@@ -135,5 +155,12 @@ extern "C" int amg_bsr_relax_f64(int64_t nbrows, int bsize, const int *ptr, cons
                                  const double *x, double *t, hipStream_t s) {
     int grid = nblk_b(nbrows * bsize);
     BSR_LAUNCH(bsr_relax_k, nbrows, ptr, col, val, M, rhs, x, t)
+    return (int)hipGetLastError();
+}
+
+extern "C" int amg_blkdiag_vmul_f64(int64_t nbrows, int bsize, const double *M,
+                                    const double *x, double *y, hipStream_t s) {
+    int grid = nblk_b(nbrows * bsize);
+    BSR_LAUNCH(blkdiag_vmul_k, nbrows, M, x, y)
     return (int)hipGetLastError();
 }

@@ -10,7 +10,8 @@ from .spai0 import Spai0, Spai1
 from .damped_jacobi import DampedJacobi
 from .chebyshev import Chebyshev
 from .gauss_seidel import GaussSeidel
-from .ilu0 import ILU0, ILU0ChowPatel, ILUK, ILUT
+from .ilu0 import ILU0, ILU0ChowPatel, ILUK, ILUP, ILUT
+from .as_block import AsBlock
 
 REGISTRY = {
     "spai0": Spai0,
@@ -20,8 +21,10 @@ REGISTRY = {
     "gauss_seidel": GaussSeidel,
     "ilu0": ILU0,
     "iluk": ILUK,
+    "ilup": ILUP,
     "ilut": ILUT,
     "ilu0_chow_patel": ILU0ChowPatel,
+    "as_block": AsBlock,
 }
 
 

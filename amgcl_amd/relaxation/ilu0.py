@@ -147,6 +147,48 @@ class ILUT(ILU0):
         return fptr, fcol, lu, dia
 
 
+class ILUP(ILU0):
+    """ILU(p) with fill pattern from the symbolic matrix power A^(k+1)
+    (parity: amgcl/relaxation/ilup.hpp:120-186 — as opposed to iluk, the
+    fill-in positions are the pattern of A multiplied symbolically by itself
+    k times; A's values are scattered onto that pattern and an ILU(0)
+    factorization runs over it)."""
+
+    @staticmethod
+    def defaults():
+        d = ILU0.defaults()
+        d["k"] = 1
+        return d
+
+    def _factor(self, A, p):
+        k = int(p["k"])
+        if k == 0:
+            return super()._factor(A, p)
+        import scipy.sparse as sp
+
+        S = sp.csr_matrix(
+            (np.ones(A.nnz), np.asarray(A.col), np.asarray(A.ptr)), shape=A.shape
+        )
+        P = S
+        for _ in range(k):
+            P = (P @ S).tocsr()
+        P.sort_indices()
+        # scatter A's values onto the power pattern (zero fill elsewhere);
+        # entries not covered by the pattern are dropped, as in the reference
+        ptr = P.indptr.astype(np.int32)
+        col = P.indices.astype(np.int32)
+        val = np.zeros(P.nnz)
+        n = np.int64(A.nrows)
+        key_p = np.repeat(np.arange(n, dtype=np.int64), np.diff(ptr)) * n + col
+        key_a = (np.repeat(np.arange(n, dtype=np.int64), np.diff(A.ptr)) * n
+                 + np.asarray(A.col))
+        pos = np.searchsorted(key_p, key_a)
+        hit = key_p[np.minimum(pos, key_p.size - 1)] == key_a
+        val[pos[hit]] = np.asarray(A.val)[hit]
+        lu, dia = _core.ilu0_factor(A.nrows, ptr, col, val)
+        return ptr, col, lu, dia
+
+
 class ILU0ChowPatel(ILU0):
     """Fine-grained (Chow-Patel) parallel ILU(0) factorization
     (parity: amgcl/relaxation/ilu0_chow_patel.hpp:87)."""

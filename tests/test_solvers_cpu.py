@@ -12,7 +12,7 @@ import amgcl_amd as am
 
 COARSENING = ["smoothed_aggregation", "aggregation", "ruge_stuben"]
 RELAX = ["spai0", "damped_jacobi", "chebyshev", "gauss_seidel", "ilu0"]
-RELAX_EXTRA = ["spai1", "iluk", "ilut", "ilu0_chow_patel"]
+RELAX_EXTRA = ["spai1", "iluk", "ilup", "ilut", "ilu0_chow_patel"]
 SOLVERS = ["cg", "bicgstab", "gmres", "richardson"]
 SOLVERS_EXTRA = ["bicgstabl", "fgmres", "lgmres", "idrs"]
 
@@ -189,3 +189,68 @@ def test_smoothed_aggr_emin_nonsymmetric():
     assert resid < 1e-7
     assert np.linalg.norm(b - K @ x) / np.linalg.norm(b) < 1e-6
     assert iters < 40
+
+
+def test_ilup_fill_pattern(problem):
+    """ILU(p)'s factor pattern is the symbolic power A^(k+1)
+    (reference: relaxation/ilup.hpp symb_product loop)."""
+    from amgcl_amd.backend import make_backend
+    from amgcl_amd.relaxation import ILU0, ILUP
+
+    A, _ = problem
+    cpu = make_backend("cpu")
+    s0 = ILU0(A, {}, cpu)
+    s1 = ILUP(A, {"k": 1}, cpu)
+    assert s1.col.size > s0.col.size  # strictly more fill than ILU(0)
+    # k=0 degenerates to ILU(0)
+    s2 = ILUP(A, {"k": 0}, cpu)
+    assert s2.col.size == s0.col.size
+    np.testing.assert_allclose(s2.lu, s0.lu)
+
+
+def _block_problem(n=10, bsize=2, eps=0.1):
+    """Poisson x identity with intra-block coupling: block-structured SPD."""
+    import scipy.sparse as sp
+
+    from amgcl_amd.matrix import CSR
+
+    A, _ = am.poisson3d(n)
+    C = np.full((bsize, bsize), eps) + (1.0 - eps) * np.eye(bsize)
+    Ab = sp.kron(A.to_scipy(), sp.csr_matrix(C), format="csr")
+    rng = np.random.default_rng(5)
+    return CSR.from_scipy(Ab), rng.standard_normal(Ab.shape[0])
+
+
+@pytest.mark.parametrize("base", ["spai0", "damped_jacobi"])
+def test_as_block_smoother(base):
+    """as_block wrapper: block-valued base smoother over the scalar matrix
+    (reference: relaxation/as_block.hpp)."""
+    A, b = _block_problem()
+    s = am.make_solver(
+        A,
+        {"precond": {"class": "amg", "coarse_enough": 500,
+                     "relax": {"type": "as_block",
+                               "block_size": 2, "base": {"type": base}}},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+    )
+    x, iters, resid = s(b)
+    assert resid < 1e-7
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6
+
+
+def test_as_block_beats_scalar_on_coupled_blocks():
+    """With strong intra-block coupling the block-diagonal inverse must not
+    be slower than the scalar diagonal smoother."""
+    A, b = _block_problem(n=10, bsize=3, eps=0.4)
+    prm = {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 200}}
+    s_blk = am.make_solver(
+        A, {**prm, "precond": {"class": "amg", "coarse_enough": 500,
+                               "relax": {"type": "as_block", "block_size": 3,
+                                         "base": {"type": "damped_jacobi"}}}})
+    _, it_blk, r_blk = s_blk(b)
+    assert r_blk < 1e-8
+    s_sc = am.make_solver(
+        A, {**prm, "precond": {"class": "amg", "coarse_enough": 500,
+                               "relax": {"type": "damped_jacobi"}}})
+    _, it_sc, r_sc = s_sc(b)
+    assert it_blk <= it_sc + 2
