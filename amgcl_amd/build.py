@@ -69,9 +69,29 @@ def build_hip_lib(verbose=True):
     return out
 
 
+def build_capi_lib(verbose=True):
+    """C API shared library (plain C++/OpenMP, no Python dependency) —
+    parity with the reference's C-callable lib/ + Fortran entry points."""
+    src = os.path.join(PKG_DIR, "csrc", "capi", "amgcl_amd_c.cpp")
+    out_dir = os.path.join(PKG_DIR, "_capi")
+    os.makedirs(out_dir, exist_ok=True)
+    out = os.path.join(out_dir, "libamgclamd_c.so")
+    if _newer(out, [src, __file__]):
+        return out
+    cmd = [
+        "g++", "-O3", "-std=c++17", "-shared", "-fPIC", "-fopenmp",
+        "-march=x86-64-v3", src, "-o", out,
+    ]
+    if verbose:
+        print("[amgcl_amd.build] " + " ".join(cmd), flush=True)
+    subprocess.check_call(cmd)
+    return out
+
+
 def build_all(verbose=True):
     build_core_ext(verbose)
     build_hip_lib(verbose)
+    build_capi_lib(verbose)
 
 
 if __name__ == "__main__":
