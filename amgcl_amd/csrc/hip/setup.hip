@@ -399,6 +399,40 @@ extern "C" int amg_agg_round(int64_t n, const int *ptr, const int *col, const ui
     agg_adopt_commit_k<<<nblk(n), 256, 0, s>>>(n, id, prov, (int *)m1, remaining);
     return (int)hipGetLastError();
 }
+// whole MIS round loop driven from C: one ctypes call per level instead of
+// one per round (the Python round loop cost ~11 ms/round of host gaps at
+// 134M rows — launch latency + torch op dispatch + .item() round-trips).
+// Convergence is polled with a pinned-memory readback every `sync_stride`
+// rounds. Returns hipError, or 9999 if max_rounds was exhausted.
+extern "C" int amg_agg_run(int64_t n, const int *ptr, const int *col, const uint8_t *S,
+                           int *id, uint8_t *prov, uint64_t *m1, uint8_t *newroot,
+                           uint8_t *near, int *remaining, int sync_stride,
+                           int max_rounds, int *rounds_out, hipStream_t s) {
+    int *h_rem = nullptr;
+    hipError_t e = hipHostMalloc((void **)&h_rem, sizeof(int), hipHostMallocDefault);
+    if (e != hipSuccess) return (int)e;
+    int rc = 0;
+    bool done = false;
+    int round = 0;
+    for (; round < max_rounds; ++round) {
+        e = hipMemsetAsync(remaining, 0, sizeof(int), s);
+        if (e != hipSuccess) { rc = (int)e; break; }
+        rc = amg_agg_round(n, ptr, col, S, id, prov, m1, newroot, near, remaining, s);
+        if (rc) break;
+        if (round % sync_stride == sync_stride - 1 || round > 8) {
+            e = hipMemcpyAsync(h_rem, remaining, sizeof(int), hipMemcpyDeviceToHost, s);
+            if (e == hipSuccess) e = hipStreamSynchronize(s);
+            if (e != hipSuccess) { rc = (int)e; break; }
+            if (*h_rem == 0) { done = true; break; }
+        }
+    }
+    (void)hipHostFree(h_rem);
+    if (rc) return rc;
+    if (!done) return 9999;
+    if (rounds_out) *rounds_out = round + 1;
+    return 0;
+}
+
 extern "C" int amg_agg_renumber(int64_t n, int *id, int *mark, hipStream_t s) {
     agg_mark_roots_k<<<nblk(n), 256, 0, s>>>(n, id, mark);
     int rc = scan_i32_device(mark, n, s);
