@@ -46,8 +46,10 @@ class DistMatrix:
 
         self.A_loc_host = CSR(n_loc, n_loc, lp, lc, lv)
         self.A_loc = base.matrix(self.A_loc_host)
-        self.A_rem = (base.matrix(CSR(n_loc, self.n_ghost, rp, rc, rv))
-                      if self.n_ghost else None)
+        self.A_rem_host = (CSR(n_loc, self.n_ghost, rp, rc, rv)
+                           if self.n_ghost else None)
+        self.A_rem = base.matrix(self.A_rem_host) if self.n_ghost else None
+        self.ghost_global = ghost_global  # sorted global ids of ghost columns
 
         # --- comm pattern (who owns each ghost column; what must we send) ---
         owner = np.searchsorted(self.row_begs, ghost_global, side="right") - 1

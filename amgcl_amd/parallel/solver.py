@@ -19,7 +19,15 @@ class DistSolver:
             if key not in ("precond", "solver", "deflation"):
                 raise ValueError(f"unknown parameter '{key}'")
         self.A = self.backend.matrix(strip)
-        self.P = LocalBlockPrecond(self.A, prm.get("precond"), self.backend)
+        pprm = dict(prm.get("precond") or {})
+        if pprm.get("class") == "dist_amg":
+            from .dist_amg import DistAMG
+
+            pprm.pop("class")
+            self.P = DistAMG(self.A, pprm, self.backend)
+        else:
+            self.P = LocalBlockPrecond(self.A, pprm or prm.get("precond"),
+                                       self.backend)
         self.inner = DistInnerProduct(backend, group)
         self.S = make_solver_component(
             self.A.n_loc, prm.get("solver"), self.backend, self.inner
