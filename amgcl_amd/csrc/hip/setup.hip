@@ -67,26 +67,38 @@ __device__ __forceinline__ uint64_t agg_key_d(int i) {
 // ---------------------------------------------------------------------------
 // diagonal + strong mask + smoother weights
 // ---------------------------------------------------------------------------
+// One-pass-over-A setup kernels use 4 lanes per row (coalesced col/val
+// loads, like the solve-phase spmv/residual kernels): the thread-per-row
+// version measured only ~450 GB/s on 512^3 where subwarp rows reach multi-TB/s.
+#define SETUP_SW 4
+
 __global__ void diag_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
                        const double *__restrict__ val, double *__restrict__ d) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int lane = (int)(tid & (SETUP_SW - 1));
+    int64_t i = tid / SETUP_SW;
+    int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SETUP_SW;
     for (; i < n; i += stride) {
         double v = 0.0;
-        for (int j = ptr[i]; j < ptr[i + 1]; ++j)
-            if (col[j] == (int)i) { v = val[j]; break; }
-        d[i] = v;
+        for (int j = ptr[i] + lane; j < ptr[i + 1]; j += SETUP_SW)
+            if (col[j] == (int)i) v = val[j];
+#pragma unroll
+        for (int off = SETUP_SW / 2; off > 0; off >>= 1)
+            v += __shfl_down(v, off, SETUP_SW);
+        if (lane == 0) d[i] = v;
     }
 }
 
 __global__ void strong_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
                          const double *__restrict__ val, const double *__restrict__ d,
                          double eps2, uint8_t *__restrict__ S) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int lane = (int)(tid & (SETUP_SW - 1));
+    int64_t i = tid / SETUP_SW;
+    int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SETUP_SW;
     for (; i < n; i += stride) {
         double edi = eps2 * d[i];
-        for (int j = ptr[i]; j < ptr[i + 1]; ++j) {
+        for (int j = ptr[i] + lane; j < ptr[i + 1]; j += SETUP_SW) {
             int c = col[j];
             double v = val[j];
             S[j] = (c != (int)i) && (edi * d[c] < v * v);
@@ -96,32 +108,39 @@ __global__ void strong_k(int64_t n, const int *__restrict__ ptr, const int *__re
 
 __global__ void spai0_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
                         const double *__restrict__ val, double *__restrict__ m) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int lane = (int)(tid & (SETUP_SW - 1));
+    int64_t i = tid / SETUP_SW;
+    int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SETUP_SW;
     for (; i < n; i += stride) {
         double num = 0.0, den = 0.0;
-        for (int j = ptr[i]; j < ptr[i + 1]; ++j) {
+        for (int j = ptr[i] + lane; j < ptr[i + 1]; j += SETUP_SW) {
             double v = val[j];
             den += v * v;
             if (col[j] == (int)i) num = v;
         }
-        m[i] = den > 0.0 ? num / den : 0.0;
+#pragma unroll
+        for (int off = SETUP_SW / 2; off > 0; off >>= 1) {
+            den += __shfl_down(den, off, SETUP_SW);
+            num += __shfl_down(num, off, SETUP_SW);
+        }
+        if (lane == 0) m[i] = den > 0.0 ? num / den : 0.0;
     }
 }
 
 extern "C" int amg_setup_diag(int64_t n, const int *ptr, const int *col, const double *val,
                               double *d, hipStream_t s) {
-    diag_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, val, d);
+    diag_k<<<nblk(n * SETUP_SW), 256, 0, s>>>(n, ptr, col, val, d);
     return (int)hipGetLastError();
 }
 extern "C" int amg_setup_strong(int64_t n, const int *ptr, const int *col, const double *val,
                                 const double *d, double eps2, uint8_t *S, hipStream_t s) {
-    strong_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, val, d, eps2, S);
+    strong_k<<<nblk(n * SETUP_SW), 256, 0, s>>>(n, ptr, col, val, d, eps2, S);
     return (int)hipGetLastError();
 }
 extern "C" int amg_setup_spai0(int64_t n, const int *ptr, const int *col, const double *val,
                                double *m, hipStream_t s) {
-    spai0_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, val, m);
+    spai0_k<<<nblk(n * SETUP_SW), 256, 0, s>>>(n, ptr, col, val, m);
     return (int)hipGetLastError();
 }
 
@@ -702,6 +721,10 @@ __global__ void spgemm_count_small_k(int64_t an, const int *__restrict__ aptr,
     }
 }
 
+// Small-bin fill, templated on the LDS table size and tiered by the exact
+// output length (a 64-slot table suffices for len <= 48 rows and halves the
+// init/extraction work of the dominant fine-level A*P product).
+template <int SLOTS>
 __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
                                     const int *__restrict__ acol,
                                     const double *__restrict__ aval,
@@ -711,9 +734,10 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
                                     const int *__restrict__ ub,
                                     const int *__restrict__ cptr_scanned,
                                     int *__restrict__ ccol, double *__restrict__ cval,
-                                    int do_sort) {
-    __shared__ int keys[32][SSLOTS];
-    __shared__ double vals[32][SSLOTS];
+                                    int do_sort, int len_lo, int len_hi) {
+    __shared__ int keys[32][SLOTS];
+    __shared__ double vals[32][SLOTS];
+    constexpr int SMASK_S = SLOTS - 1;
     int gid = threadIdx.x / SGRP;
     int lane = threadIdx.x & (SGRP - 1);
     int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / SGRP;
@@ -722,7 +746,10 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
     double *tv = vals[gid];
     for (; row < an; row += stride) {
         if (ub[row] > SGSMALL) continue;
-        for (int t = lane; t < SSLOTS; t += SGRP) {
+        int base = (row == 0) ? 0 : cptr_scanned[row - 1];
+        int len = cptr_scanned[row] - base;
+        if (len <= len_lo || len > len_hi) continue;
+        for (int t = lane; t < SLOTS; t += SGRP) {
             tk[t] = -1;
             tv[t] = 0.0;
         }
@@ -734,26 +761,24 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
             for (int jb = bptr[ca] + lane; jb < bptr[ca + 1]; jb += SGRP) {
                 int key = bcol[jb];
                 double v = va * bval[jb];
-                uint32_t h = ((uint32_t)key * 2654435761u) & SSMASK;
+                uint32_t h = ((uint32_t)key * 2654435761u) & SMASK_S;
                 while (true) {
                     int old = atomicCAS(&tk[h], -1, key);
                     if (old == -1 || old == key) {
                         atomicAdd(&tv[h], v);
                         break;
                     }
-                    h = (h + 1) & SSMASK;
+                    h = (h + 1) & SMASK_S;
                 }
             }
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
-        int base = (row == 0) ? 0 : cptr_scanned[row - 1];
-        int len = cptr_scanned[row] - base;
         if (do_sort) {
             // sorted extraction: empties to +inf, bitonic, write the prefix
-            for (int t = lane; t < SSLOTS; t += SGRP)
+            for (int t = lane; t < SLOTS; t += SGRP)
                 if (tk[t] == -1) tk[t] = 0x7fffffff;
-            lds_bitonic<SGRP, SSLOTS>(tk, tv, lane);
+            lds_bitonic<SGRP, SLOTS>(tk, tv, lane);
             for (int t = lane; t < len; t += SGRP) {
                 ccol[base + t] = tk[t];
                 cval[base + t] = tv[t];
@@ -762,7 +787,7 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
             // unsorted compaction (intermediate products: order irrelevant);
             // per-lane local count -> exclusive shfl prefix -> direct writes
             int mine = 0;
-            for (int t = lane; t < SSLOTS; t += SGRP)
+            for (int t = lane; t < SLOTS; t += SGRP)
                 if (tk[t] != -1) ++mine;
             int off = mine;
 #pragma unroll
@@ -772,7 +797,7 @@ __global__ void spgemm_fill_small_k(int64_t an, const int *__restrict__ aptr,
             }
             off -= mine;  // exclusive prefix within the group
             int h = base + off;
-            for (int t = lane; t < SSLOTS; t += SGRP)
+            for (int t = lane; t < SLOTS; t += SGRP)
                 if (tk[t] != -1) {
                     ccol[h] = tk[t];
                     cval[h] = tv[t];
@@ -800,13 +825,20 @@ __device__ __forceinline__ int pfx_find(const int *pfx, int len, int t) {
     return lo;
 }
 
+// Count pass, wave per row, templated on the LDS table size. The exact
+// output length is unknown before counting, so rows tier by the product
+// upper bound: distinct keys <= ub, so any row with ub < SLOTS is safe in
+// the smaller table. (ub_lo, ub_hi] selects this launch's tier.
+template <int SLOTS>
 __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
                                const int *__restrict__ acol, const int *__restrict__ bptr,
                                const int *__restrict__ bcol, const int *__restrict__ ub,
-                               int *__restrict__ cnt, int *__restrict__ overflow) {
-    __shared__ int keys[4][HSLOTS];
+                               int *__restrict__ cnt, int *__restrict__ overflow,
+                               int ub_lo, int ub_hi) {
+    __shared__ int keys[4][SLOTS];
     __shared__ int pfx[4][BIGROW + 1];
     __shared__ int bbeg[4][BIGROW];
+    constexpr int SMASK = SLOTS - 1;
     int wid = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
     int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -815,13 +847,13 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
 
     for (; row < an; row += stride) {
         int total = ub[row];
-        if (total <= SGSMALL) continue;
+        if (total <= ub_lo || total > ub_hi) continue;
         int ab = aptr[row], alen = aptr[row + 1] - ab;
         if (alen > BIGROW) {  // true table overflow is caught by the probes
             if (lane == 0) atomicAdd(overflow, 1);
             continue;
         }
-        for (int t = lane; t < HSLOTS; t += WAVE) tk[t] = -1;
+        for (int t = lane; t < SLOTS; t += WAVE) tk[t] = -1;
         // build the product-space prefix (serial scan by lane 0: alen <= 128)
         if (lane == 0) {
             int acc = 0;
@@ -840,14 +872,14 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
         for (int t = lane; t < total && !ovf; t += WAVE) {
             int ja = pfx_find(pfx[wid], alen, t);
             int key = bcol[bbeg[wid][ja] + (t - pfx[wid][ja])];
-            uint32_t h = ((uint32_t)key * 2654435761u) & HMASK;
+            uint32_t h = ((uint32_t)key * 2654435761u) & SMASK;
             int probes = 0;
             while (true) {
                 int old = atomicCAS(&tk[h], -1, key);
                 if (old == -1) { ++inserted; break; }
                 if (old == key) break;
-                h = (h + 1) & HMASK;
-                if (++probes >= HSLOTS) { ovf = true; break; }
+                h = (h + 1) & SMASK;
+                if (++probes >= SLOTS) { ovf = true; break; }
             }
         }
         if (ovf && lane == 0) atomicAdd(overflow, 1);
@@ -1007,17 +1039,23 @@ extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol, co
                                 hipStream_t s) {
     spgemm_ub_k<<<nblk(an), 256, 0, s>>>(an, aptr, acol, bptr, ub);
     spgemm_count_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub, cnt);
-    spgemm_count_k<<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub, cnt,
-                                                   overflow);
+    // single count tier: measured on 512^3, splitting the count by ub loses
+    // ~10 ms/setup (second full-row sweep) — count is not LDS-limited
+    spgemm_count_k<HSLOTS><<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub,
+                                                           cnt, overflow, SGSMALL, 1 << 30);
     return (int)hipGetLastError();
 }
 extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol, const double *aval,
                                const int *bptr, const int *bcol, const double *bval,
                                const int *ub, const int *cptr_scanned, int *ccol,
                                double *cval, int do_sort, hipStream_t s) {
-    spgemm_fill_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol,
-                                                        bval, ub, cptr_scanned, ccol, cval,
-                                                        do_sort);
+    spgemm_fill_small_k<64><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr,
+                                                            bcol, bval, ub, cptr_scanned,
+                                                            ccol, cval, do_sort, 0, 48);
+    spgemm_fill_small_k<SSLOTS><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr,
+                                                                bcol, bval, ub, cptr_scanned,
+                                                                ccol, cval, do_sort, 48,
+                                                                1 << 30);
     // wave-per-row tiers by exact output length (known after the count pass)
     spgemm_fill_k<128><<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol,
                                                        bval, ub, cptr_scanned, ccol, cval,
