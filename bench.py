@@ -112,11 +112,16 @@ def run_distributed(args, dist_ctx):
         [idx % args.size, (idx // args.size) % args.size, idx // (args.size**2)],
         axis=1,
     ).astype(np.float64)
+    pprm = json.loads(args.precond) if args.precond else {"class": "amg"}
     prm = {
-        "precond": json.loads(args.precond) if args.precond else {"class": "amg"},
+        "precond": pprm,
         "solver": {"type": args.solver, "tol": args.tol, "maxiter": args.maxiter},
         "deflation": {"type": "linear", "coords_raw": coords},
     }
+    if pprm.get("class") == "dist_amg":
+        # one hierarchy over the whole operator already couples the ranks;
+        # subdomain deflation is the block-Jacobi path's accelerator
+        prm.pop("deflation")
     dist.barrier()
     t0 = time.perf_counter()
     solve = make_dist_solver(A_strip, prm, backend=backend_name)
