@@ -218,9 +218,7 @@ class HipBackend:
     # --- coarse direct solver ---------------------------------------------
     def coarse_solver(self, csr):
         if isinstance(csr, DeviceCSR):
-            from . import hip_setup
-
-            csr = hip_setup.download(csr)
+            return DeviceDenseSolver.from_device(csr, self)
         return DeviceDenseSolver(csr, self)
 
     def synchronize(self):
@@ -236,6 +234,24 @@ class DeviceDenseSolver:
         inv = np.linalg.inv(a)
         self.n = csr.nrows
         self.inv = backend.from_host(np.ascontiguousarray(inv).ravel())
+
+    @classmethod
+    def from_device(cls, dcsr, backend):
+        """Densify + invert entirely on the GPU (no download round-trip; the
+        host np.linalg.inv also runs under the GPU box's 16-core CPU quota)."""
+        import torch
+
+        self = cls.__new__(cls)
+        n = dcsr.nrows
+        dense = torch.zeros(n, n, dtype=dcsr.val.dtype, device=dcsr.val.device)
+        lens = dcsr.ptr[1:] - dcsr.ptr[:-1]
+        rows = torch.repeat_interleave(
+            torch.arange(n, device=dcsr.val.device, dtype=torch.int64),
+            lens.to(torch.int64))
+        dense[rows, dcsr.col.to(torch.int64)] = dcsr.val
+        self.n = n
+        self.inv = torch.linalg.inv(dense).contiguous().ravel()
+        return self
 
     def __call__(self, f, u):
         import torch

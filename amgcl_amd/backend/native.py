@@ -45,7 +45,9 @@ def _bind():
     if not _driver_bound:
         L.amg_driver_create.argtypes = [
             ctypes.POINTER(LevelDescC), ctypes.c_int, ctypes.c_void_p, ctypes.c_int64,
-            ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+            ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+            ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_int, ctypes.c_void_p,
         ]
         L.amg_driver_create.restype = ctypes.c_void_p
         L.amg_driver_destroy.argtypes = [ctypes.c_void_p]
@@ -92,8 +94,7 @@ class NativeDriver:
         levels = amg.levels
         if not all(isinstance(l.A, DeviceCSR) for l in levels):
             raise TypeError("native driver needs device-resident levels")
-        if getattr(amg, "_mixed", False):
-            raise TypeError("native driver is fp64-only (mixed uses the generic path)")
+        self._mixed = bool(getattr(amg, "_mixed", False))
         for l in levels[:-1]:
             if not isinstance(l.relax, DiagonalSmootherBase):
                 raise TypeError("native driver supports diagonal smoothers only")
@@ -140,11 +141,21 @@ class NativeDriver:
         prm = amg.prm
         L = _bind()
         stream = torch.cuda.current_stream().cuda_stream
+        if self._mixed:
+            # the Krylov loop keeps iterating with the fp64 fine operator;
+            # the LevelDescs above point at the fp32 hierarchy for the cycle
+            A64 = amg._A64
+            self._keep.extend([A64.ptr, A64.col, A64.val])
+            a64 = (int(A64.nnz), _ptr(A64.ptr), _ptr(A64.col), _ptr(A64.val),
+                   A64.subw or _auto_subw(A64))
+        else:
+            a64 = (0, None, None, None, 0)
         self.handle = L.amg_driver_create(
             descs, len(levels), _ptr(inv),
             amg.coarse_solve.n if amg.coarse_solve is not None else 0,
             int(prm["npre"]), int(prm["npost"]), int(prm["ncycle"]),
-            int(prm["pre_cycles"]), ctypes.c_void_p(stream),
+            int(prm["pre_cycles"]), int(self._mixed), a64[0], a64[1], a64[2],
+            a64[3], a64[4], ctypes.c_void_p(stream),
         )
         if not self.handle:
             raise RuntimeError("amg_driver_create failed")

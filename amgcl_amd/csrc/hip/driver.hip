@@ -35,19 +35,57 @@ extern "C" int amg_cg_tail_f64(int64_t, double, const double *, const double *, 
                                double *, double *, hipStream_t);
 extern "C" int amg_vmul_f64(int64_t, double, const double *, const double *, double,
                             double *, hipStream_t);
+extern "C" int amg_spmv_f32(int64_t, int64_t, const int *, const int *, const float *,
+                            const float *, double, double, float *, int, hipStream_t);
+extern "C" int amg_residual_f32(int64_t, int64_t, const int *, const int *, const float *,
+                                const float *, const float *, float *, int, hipStream_t);
+extern "C" int amg_gemv_f32(int64_t, const float *, const float *, float *, hipStream_t);
+extern "C" int amg_cast_d2s(int64_t, const double *, float *, hipStream_t);
+extern "C" int amg_cast_s2d(int64_t, const float *, double *, hipStream_t);
+
+// value-type dispatch for the cycle (the fp32 hierarchy of mixed precision)
+template <typename T> struct ops;
+template <> struct ops<double> {
+    static int spmv(int64_t n, int64_t nnz, const int *p, const int *c, const double *v,
+                    const double *x, double a, double b, double *y, int sw, hipStream_t s) {
+        return amg_spmv_f64(n, nnz, p, c, v, x, a, b, y, sw, s);
+    }
+    static int residual(int64_t n, int64_t nnz, const int *p, const int *c, const double *v,
+                        const double *rhs, const double *x, double *r, int sw,
+                        hipStream_t s) {
+        return amg_residual_f64(n, nnz, p, c, v, rhs, x, r, sw, s);
+    }
+    static int gemv(int64_t n, const double *inv, const double *f, double *u,
+                    hipStream_t s) {
+        return amg_gemv_f64(n, inv, f, u, s);
+    }
+};
+template <> struct ops<float> {
+    static int spmv(int64_t n, int64_t nnz, const int *p, const int *c, const float *v,
+                    const float *x, double a, double b, float *y, int sw, hipStream_t s) {
+        return amg_spmv_f32(n, nnz, p, c, v, x, a, b, y, sw, s);
+    }
+    static int residual(int64_t n, int64_t nnz, const int *p, const int *c, const float *v,
+                        const float *rhs, const float *x, float *r, int sw, hipStream_t s) {
+        return amg_residual_f32(n, nnz, p, c, v, rhs, x, r, sw, s);
+    }
+    static int gemv(int64_t n, const float *inv, const float *f, float *u, hipStream_t s) {
+        return amg_gemv_f32(n, inv, f, u, s);
+    }
+};
 
 // x_new = x + M ∘ (rhs - A x), written to a separate buffer (pointer swap)
-template <int SUBW>
+template <int SUBW, typename T>
 __global__ void relax_swap_k(int64_t nrows, const int *__restrict__ ptr,
-                             const int *__restrict__ col, const double *__restrict__ val,
-                             const double *__restrict__ M, const double *__restrict__ rhs,
-                             const double *__restrict__ x, double *__restrict__ xn) {
+                             const int *__restrict__ col, const T *__restrict__ val,
+                             const T *__restrict__ M, const T *__restrict__ rhs,
+                             const T *__restrict__ x, T *__restrict__ xn) {
     int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SUBW;
     for (; row < nrows; row += stride) {
-        double s = 0.0;
+        T s = (T)0;
         int b = ptr[row], e = ptr[row + 1];
         for (int j = b + lane; j < e; j += SUBW) s += val[j] * x[col[j]];
 #pragma unroll
@@ -57,8 +95,9 @@ __global__ void relax_swap_k(int64_t nrows, const int *__restrict__ ptr,
 }
 
 // first pre-smooth of a cycle starts from u = 0: x_new = M ∘ rhs (no A pass)
-__global__ void relax_zero_k(int64_t n, const double *__restrict__ M,
-                             const double *__restrict__ rhs, double *__restrict__ xn) {
+template <typename T>
+__global__ void relax_zero_k(int64_t n, const T *__restrict__ M,
+                             const T *__restrict__ rhs, T *__restrict__ xn) {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) xn[i] = M[i] * rhs[i];
@@ -93,12 +132,21 @@ struct LevelDesc {
 
 struct Driver {
     std::vector<LevelDesc> lv;
-    const double *coarse_inv;  // dense n x n (may be null -> smooth coarsest)
+    const void *coarse_inv;  // dense n x n (may be null -> smooth coarsest)
     int64_t ncoarse;
     int npre, npost, ncycle, pre_cycles;
     hipStream_t stream;
     double *dotbuf_d;  // 2 doubles
     double *dotbuf_h;  // pinned host, 2 doubles
+    // mixed precision: the cycle runs fp32 on the (fp32) LevelDescs while the
+    // Krylov loop keeps the fp64 fine operator below + f32 cast buffers
+    int f32;
+    int64_t a64_nnz;
+    const int *a64_ptr;
+    const int *a64_col;
+    const double *a64_val;
+    int a64_subw;
+    float *cb_r, *cb_x, *cb_s;  // level-0-sized cast buffers (f32 mode)
 };
 
 #define CHK(x)                          \
@@ -107,21 +155,23 @@ struct Driver {
         if (_rc) return _rc;            \
     } while (0)
 
-static int relax_swap(Driver *D, const LevelDesc &L, const double *rhs, double **x,
-                      double **xn) {
+template <typename T>
+static int relax_swap(Driver *D, const LevelDesc &L, const T *rhs, T **x, T **xn) {
     int subw = L.subw;
     int grid = nblocks_d(L.nrows * subw);
-#define RCASE(SW)                                                                     \
-    case SW:                                                                          \
-        relax_swap_k<SW><<<grid, 256, 0, D->stream>>>(L.nrows, L.ptr, L.col, L.val,   \
-                                                      L.M, rhs, *x, *xn);             \
+    const T *val = (const T *)L.val;
+    const T *M = (const T *)L.M;
+#define RCASE(SW)                                                                       \
+    case SW:                                                                            \
+        relax_swap_k<SW, T><<<grid, 256, 0, D->stream>>>(L.nrows, L.ptr, L.col, val,    \
+                                                         M, rhs, *x, *xn);              \
         break;
     switch (subw) {
         RCASE(1) RCASE(2) RCASE(4) RCASE(8) RCASE(16) RCASE(32) RCASE(64)
         default: return hipErrorInvalidValue;
     }
 #undef RCASE
-    double *tmp = *x;
+    T *tmp = *x;
     *x = *xn;
     *xn = tmp;
     return (int)hipGetLastError();
@@ -129,31 +179,31 @@ static int relax_swap(Driver *D, const LevelDesc &L, const double *rhs, double *
 
 // one multigrid cycle; *u_io holds the iterate buffer (may be swapped),
 // uses L.t as the swap partner / residual scratch
-static int cycle(Driver *D, int li, const double *f, double **u_io, double **scratch,
-                 bool u_is_zero) {
+template <typename T>
+static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is_zero) {
     LevelDesc &L = D->lv[li];
     const bool coarsest = (li + 1 == (int)D->lv.size());
 
     // relax_zero writes into the swap buffer and swaps, so zero-guess and
     // general smooths have identical pointer parity (no copy-back needed)
-    auto relax_zero = [&](LevelDesc &LL, const double *ff, double **u2,
-                          double **sc) {
-        relax_zero_k<<<nblocks_d(LL.nrows), 256, 0, D->stream>>>(LL.nrows, LL.M, ff, *sc);
-        double *tmp = *u2;
+    auto relax_zero = [&](LevelDesc &LL, const T *ff, T **u2, T **sc) {
+        relax_zero_k<T><<<nblocks_d(LL.nrows), 256, 0, D->stream>>>(
+            LL.nrows, (const T *)LL.M, ff, *sc);
+        T *tmp = *u2;
         *u2 = *sc;
         *sc = tmp;
     };
 
     if (coarsest) {
         if (D->coarse_inv) {
-            CHK(amg_gemv_f64(L.nrows, D->coarse_inv, f, *u_io, D->stream));
+            CHK(ops<T>::gemv(L.nrows, (const T *)D->coarse_inv, f, *u_io, D->stream));
         } else {
             for (int i = 0; i < D->npre + D->npost; ++i) {
                 if (u_is_zero && i == 0) {
                     relax_zero(L, f, u_io, scratch);
                     continue;
                 }
-                CHK(relax_swap(D, L, f, u_io, scratch));
+                CHK(relax_swap<T>(D, L, f, u_io, scratch));
             }
         }
         return (int)hipGetLastError();
@@ -165,37 +215,46 @@ static int cycle(Driver *D, int li, const double *f, double **u_io, double **scr
             relax_zero(L, f, u_io, scratch);
             continue;
         }
-        CHK(relax_swap(D, L, f, u_io, scratch));
+        CHK(relax_swap<T>(D, L, f, u_io, scratch));
     }
     // t = f - A u ; f_next = R t
-    CHK(amg_residual_f64(L.nrows, L.nnz, L.ptr, L.col, L.val, f, *u_io, *scratch, L.subw,
-                         D->stream));
-    CHK(amg_spmv_f64(N.nrows, L.rnnz, L.rptr, L.rcol, L.rval, *scratch, 1.0, 0.0, N.f,
-                     L.rsubw, D->stream));
-    double *nu = N.u;
-    double *nscratch = N.t;
+    CHK(ops<T>::residual(L.nrows, L.nnz, L.ptr, L.col, (const T *)L.val, f, *u_io,
+                         *scratch, L.subw, D->stream));
+    CHK(ops<T>::spmv(N.nrows, L.rnnz, L.rptr, L.rcol, (const T *)L.rval, *scratch, 1.0,
+                     0.0, (T *)N.f, L.rsubw, D->stream));
+    T *nu = (T *)N.u;
+    T *nscratch = (T *)N.t;
     for (int c = 0; c < D->ncycle; ++c) {
-        CHK(cycle(D, li + 1, N.f, &nu, &nscratch, c == 0));
+        CHK(cycle<T>(D, li + 1, (const T *)N.f, &nu, &nscratch, c == 0));
         // after the first sub-cycle the iterate is nonzero
     }
     // u += P u_next
-    CHK(amg_spmv_f64(L.nrows, L.pnnz, L.pptr, L.pcol, L.pval, nu, 1.0, 1.0, *u_io,
-                     L.psubw, D->stream));
+    CHK(ops<T>::spmv(L.nrows, L.pnnz, L.pptr, L.pcol, (const T *)L.pval, nu, 1.0, 1.0,
+                     *u_io, L.psubw, D->stream));
     for (int i = 0; i < D->npost; ++i)
-        CHK(relax_swap(D, L, f, u_io, scratch));
+        CHK(relax_swap<T>(D, L, f, u_io, scratch));
     return (int)hipGetLastError();
 }
 
+// fp64 in/out preconditioner application; in f32 mode the V-cycle runs on the
+// fp32 hierarchy between two cast kernels (backend/detail/mixing.hpp shape)
 static int precond_apply(Driver *D, const double *rhs, double *x, double *x_swap) {
-    // x = M^-1 rhs from zero initial guess (pre_cycles cycles);
-    // result must END in `x` (copy back if the swap parity left it elsewhere)
+    const int64_t n = D->lv[0].nrows;
+    if (D->f32) {
+        CHK(amg_cast_d2s(n, rhs, D->cb_r, D->stream));
+        float *u = D->cb_x;
+        float *scratch = D->cb_s;
+        for (int c = 0; c < D->pre_cycles; ++c)
+            CHK(cycle<float>(D, 0, D->cb_r, &u, &scratch, c == 0));
+        CHK(amg_cast_s2d(n, u, x, D->stream));
+        return 0;
+    }
     double *u = x;
     double *scratch = x_swap;
     for (int c = 0; c < D->pre_cycles; ++c)
-        CHK(cycle(D, 0, rhs, &u, &scratch, c == 0));
+        CHK(cycle<double>(D, 0, rhs, &u, &scratch, c == 0));
     if (u != x)
-        CHK(hipMemcpyAsync(x, u, D->lv[0].nrows * sizeof(double),
-                           hipMemcpyDeviceToDevice, D->stream));
+        CHK(hipMemcpyAsync(x, u, n * sizeof(double), hipMemcpyDeviceToDevice, D->stream));
     return 0;
 }
 
@@ -208,9 +267,11 @@ static int read_dots(Driver *D, int n, double *out) {
 }
 
 extern "C" void *amg_driver_create(const LevelDesc *levels, int nlevels,
-                                   const double *coarse_inv, int64_t ncoarse, int npre,
-                                   int npost, int ncycle, int pre_cycles,
-                                   hipStream_t stream) {
+                                   const void *coarse_inv, int64_t ncoarse, int npre,
+                                   int npost, int ncycle, int pre_cycles, int f32,
+                                   int64_t a64_nnz, const int *a64_ptr,
+                                   const int *a64_col, const double *a64_val,
+                                   int a64_subw, hipStream_t stream) {
     Driver *D = new Driver();
     D->lv.assign(levels, levels + nlevels);
     D->coarse_inv = coarse_inv;
@@ -220,8 +281,22 @@ extern "C" void *amg_driver_create(const LevelDesc *levels, int nlevels,
     D->ncycle = ncycle;
     D->pre_cycles = pre_cycles;
     D->stream = stream;
-    if (hipMalloc((void **)&D->dotbuf_d, 2 * sizeof(double)) != hipSuccess ||
-        hipHostMalloc((void **)&D->dotbuf_h, 2 * sizeof(double)) != hipSuccess) {
+    D->f32 = f32;
+    D->a64_nnz = a64_nnz;
+    D->a64_ptr = a64_ptr;
+    D->a64_col = a64_col;
+    D->a64_val = a64_val;
+    D->a64_subw = a64_subw;
+    D->cb_r = D->cb_x = D->cb_s = nullptr;
+    bool ok = hipMalloc((void **)&D->dotbuf_d, 2 * sizeof(double)) == hipSuccess &&
+              hipHostMalloc((void **)&D->dotbuf_h, 2 * sizeof(double)) == hipSuccess;
+    if (ok && f32) {
+        int64_t n = D->lv[0].nrows;
+        ok = hipMalloc((void **)&D->cb_r, n * sizeof(float)) == hipSuccess &&
+             hipMalloc((void **)&D->cb_x, n * sizeof(float)) == hipSuccess &&
+             hipMalloc((void **)&D->cb_s, n * sizeof(float)) == hipSuccess;
+    }
+    if (!ok) {
         delete D;
         return nullptr;
     }
@@ -231,8 +306,11 @@ extern "C" void *amg_driver_create(const LevelDesc *levels, int nlevels,
 extern "C" void amg_driver_destroy(void *h) {
     Driver *D = (Driver *)h;
     if (!D) return;
-    hipFree(D->dotbuf_d);
-    hipHostFree(D->dotbuf_h);
+    (void)hipFree(D->dotbuf_d);
+    (void)hipHostFree(D->dotbuf_h);
+    if (D->cb_r) (void)hipFree(D->cb_r);
+    if (D->cb_x) (void)hipFree(D->cb_x);
+    if (D->cb_s) (void)hipFree(D->cb_s);
     delete D;
 }
 
@@ -253,6 +331,13 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     Driver *D = (Driver *)h;
     const LevelDesc &L0 = D->lv[0];
     const int64_t n = L0.nrows;
+    // the Krylov loop always iterates with the fp64 fine operator (in mixed
+    // mode L0's LevelDesc holds the fp32 copy used inside the cycle)
+    const int64_t knnz = D->f32 ? D->a64_nnz : L0.nnz;
+    const int *kptr = D->f32 ? D->a64_ptr : L0.ptr;
+    const int *kcol = D->f32 ? D->a64_col : L0.col;
+    const double *kval = D->f32 ? D->a64_val : (const double *)L0.val;
+    const int ksubw = D->f32 ? D->a64_subw : L0.subw;
     hipStream_t st = D->stream;
     double dots[2];
 
@@ -267,7 +352,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
 
-    CHK(amg_residual_f64(n, L0.nnz, L0.ptr, L0.col, L0.val, rhs, x, r, L0.subw, st));
+    CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
     CHK(amg_dot_f64(n, r, r, D->dotbuf_d, st));
     CHK(read_dots(D, 1, dots));
     double res = sqrt(dots[0]);
@@ -285,7 +370,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
         } else {
             CHK(amg_axpby_f64(n, 1.0, s, rho1 / rho2, p, st));
         }
-        CHK(amg_spmv_f64(n, L0.nnz, L0.ptr, L0.col, L0.val, p, 1.0, 0.0, q, L0.subw, st));
+        CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, p, 1.0, 0.0, q, ksubw, st));
         CHK(amg_dot_f64(n, q, p, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
         double alpha = rho1 / dots[0];
@@ -311,6 +396,13 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
     Driver *D = (Driver *)h;
     const LevelDesc &L0 = D->lv[0];
     const int64_t n = L0.nrows;
+    // the Krylov loop always iterates with the fp64 fine operator (in mixed
+    // mode L0's LevelDesc holds the fp32 copy used inside the cycle)
+    const int64_t knnz = D->f32 ? D->a64_nnz : L0.nnz;
+    const int *kptr = D->f32 ? D->a64_ptr : L0.ptr;
+    const int *kcol = D->f32 ? D->a64_col : L0.col;
+    const double *kval = D->f32 ? D->a64_val : (const double *)L0.val;
+    const int ksubw = D->f32 ? D->a64_subw : L0.subw;
     hipStream_t st = D->stream;
     double dots[2];
 
@@ -325,7 +417,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
 
-    CHK(amg_residual_f64(n, L0.nnz, L0.ptr, L0.col, L0.val, rhs, x, r, L0.subw, st));
+    CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
     CHK(hipMemcpyAsync(rh, r, n * sizeof(double), hipMemcpyDeviceToDevice, st));
     CHK(amg_dot_f64(n, r, r, D->dotbuf_d, st));
     CHK(read_dots(D, 1, dots));
@@ -349,7 +441,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         }
         // v = A (M^-1 p);  T = M^-1 p
         CHK(precond_apply(D, p, T, T_swap));
-        CHK(amg_spmv_f64(n, L0.nnz, L0.ptr, L0.col, L0.val, T, 1.0, 0.0, v, L0.subw, st));
+        CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, v, ksubw, st));
         CHK(amg_dot_f64(n, rh, v, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
         alpha = rho1 / dots[0];
@@ -360,8 +452,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         res = sqrt(dots[0]);
         if (res > eps) {
             CHK(precond_apply(D, s2, T, T_swap));
-            CHK(amg_spmv_f64(n, L0.nnz, L0.ptr, L0.col, L0.val, T, 1.0, 0.0, t2, L0.subw,
-                             st));
+            CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, t2, ksubw, st));
             CHK(amg_dot2_f64(n, t2, s2, t2, t2, D->dotbuf_d, st));
             CHK(read_dots(D, 2, dots));
             omega = dots[0] / dots[1];

@@ -233,3 +233,19 @@ def test_blkdiag_vmul_and_as_block(hip):
     assert resid < 1e-8
     assert abs(it_hip - it_cpu) <= 2
     assert np.linalg.norm(b - A @ hip.to_host(x)) / np.linalg.norm(b) < 1e-7
+
+
+def test_device_dense_coarse_inverse(hip):
+    """GPU-resident densify + torch.linalg.inv path (no host round-trip)."""
+    from amgcl_amd.backend.hip import DeviceCSR, DeviceDenseSolver
+
+    rng = np.random.default_rng(31)
+    n = 173
+    a = rng.standard_normal((n, n)) + n * np.eye(n)
+    A = CSR.from_dense(a)
+    solver = DeviceDenseSolver.from_device(DeviceCSR(A, hip.device), hip)
+    f = rng.standard_normal(n)
+    fd, ud = hip.from_host(f), hip.vector(n)
+    solver(fd, ud)
+    np.testing.assert_allclose(hip.to_host(ud), np.linalg.solve(a, f),
+                               rtol=1e-9, atol=1e-9)

@@ -64,3 +64,21 @@ def test_native_rejects_unsupported(hip):
     assert s._native is None  # chebyshev -> generic path
     x, iters, resid = s(b)
     assert resid < 1e-8
+
+
+@pytest.mark.parametrize("solver", ["cg", "bicgstab"])
+def test_native_mixed_precision(hip, solver):
+    """fp32 hierarchy through the native driver (fp64 Krylov + cast-bracketed
+    fp32 V-cycle) matches the generic mixed path exactly."""
+    A, b = am.poisson3d(40, rhs="random")
+    prm = {"precond": {"class": "amg", "precision": "mixed"},
+           "solver": {"type": solver, "tol": 1e-8, "maxiter": 100}}
+    s = am.make_solver(A, prm, backend=hip)
+    assert s._native is not None and s._native._mixed
+    x1, it1, res1 = s(b)
+    s._native = None  # generic mixed path on the same hierarchy
+    x2, it2, res2 = s(b)
+    assert res1 < 1e-8 and res2 < 1e-8
+    assert it1 == it2
+    r = b - A @ hip.to_host(x1)
+    assert np.linalg.norm(r) / np.linalg.norm(b) < 1e-7
