@@ -14,7 +14,12 @@ from ..matrix import CSR
 
 
 class DistMatrix:
-    def __init__(self, strip: CSR, backend, group=None):
+    def __init__(self, strip: CSR, backend, group=None, col_sizes=None):
+        """strip: my rows with GLOBAL columns. By default the column space is
+        partitioned like the rows (square operator). `col_sizes` (one entry
+        per rank) declares a different column partition — the rectangular
+        field blocks of the distributed Schur complement (reference
+        mpi/schur_pressure_correction.hpp works on such blocks)."""
         import torch
         import torch.distributed as dist
 
@@ -33,18 +38,28 @@ class DistMatrix:
         self.n_global = int(np.sum(sizes))
         self.row_begs = np.concatenate([[0], np.cumsum(sizes)]).astype(np.int64)
         self.n_loc = n_loc
+        if col_sizes is None:
+            self.col_begs = self.row_begs
+            col_beg, col_end = self.row_beg, self.row_end
+        else:
+            self.col_begs = np.concatenate([[0], np.cumsum(col_sizes)]).astype(np.int64)
+            col_beg = int(self.col_begs[self.rank])
+            col_end = int(self.col_begs[self.rank + 1])
+        self.col_beg, self.col_end = col_beg, col_end
+        self.n_loc_cols = col_end - col_beg
+        self.n_global_cols = int(self.col_begs[-1])
 
-        # split into local (square) + remote (ghost) parts (C++ engine;
+        # split into local + remote (ghost) parts (C++ engine;
         # parity: distributed_matrix.hpp:370-430)
         from .. import _core
 
         lp, lc, lv, rp, rc, rv, ghost_global = _core.split_strip(
-            n_loc, self.row_beg, self.row_end, strip.ptr, strip.col, strip.val
+            n_loc, col_beg, col_end, strip.ptr, strip.col, strip.val
         )
         ghost_global = np.asarray(ghost_global)
         self.n_ghost = len(ghost_global)
 
-        self.A_loc_host = CSR(n_loc, n_loc, lp, lc, lv)
+        self.A_loc_host = CSR(n_loc, self.n_loc_cols, lp, lc, lv)
         self.A_loc = base.matrix(self.A_loc_host)
         self.A_rem_host = (CSR(n_loc, self.n_ghost, rp, rc, rv)
                            if self.n_ghost else None)
@@ -52,7 +67,7 @@ class DistMatrix:
         self.ghost_global = ghost_global  # sorted global ids of ghost columns
 
         # --- comm pattern (who owns each ghost column; what must we send) ---
-        owner = np.searchsorted(self.row_begs, ghost_global, side="right") - 1
+        owner = np.searchsorted(self.col_begs, ghost_global, side="right") - 1
         self.recv_ranks = []
         self.recv_counts = []
         need_from = [np.empty(0, dtype=np.int64)] * self.world
@@ -74,7 +89,7 @@ class DistMatrix:
             req = gathered[r][self.rank]
             if len(req):
                 self.send_ranks.append(r)
-                send_idx.append((req - self.row_beg).astype(np.int32))
+                send_idx.append((req - self.col_beg).astype(np.int32))
 
         dev = getattr(base, "device", "cpu")
         self._torch = torch
@@ -102,7 +117,7 @@ class DistMatrix:
 
     @property
     def ncols(self):
-        return self.n_loc
+        return self.n_loc_cols
 
     @property
     def nnz(self):

@@ -18,14 +18,21 @@ class DistSolver:
         for key in prm:
             if key not in ("precond", "solver", "deflation"):
                 raise ValueError(f"unknown parameter '{key}'")
-        self.A = self.backend.matrix(strip)
         pprm = dict(prm.get("precond") or {})
         if pprm.get("class") == "dist_amg":
             from .dist_amg import DistAMG
 
             pprm.pop("class")
+            self.A = self.backend.matrix(strip)
             self.P = DistAMG(self.A, pprm, self.backend)
+        elif pprm.get("class") == "schur_pressure_correction":
+            from .schur import DistSchurPressureCorrection
+
+            pprm.pop("class")
+            self.P = DistSchurPressureCorrection(strip, pprm, self.backend, group)
+            self.A = self.P.A  # the full distributed operator
         else:
+            self.A = self.backend.matrix(strip)
             self.P = LocalBlockPrecond(self.A, pprm or prm.get("precond"),
                                        self.backend)
         self.inner = DistInnerProduct(backend, group)
