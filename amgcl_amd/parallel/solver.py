@@ -31,6 +31,12 @@ class DistSolver:
             pprm.pop("class")
             self.P = DistSchurPressureCorrection(strip, pprm, self.backend, group)
             self.A = self.P.A  # the full distributed operator
+        elif pprm.get("class") == "cpr":
+            from .cpr import DistCPR
+
+            pprm.pop("class")
+            self.P = DistCPR(strip, pprm, self.backend, group)
+            self.A = self.P.A
         else:
             self.A = self.backend.matrix(strip)
             self.P = LocalBlockPrecond(self.A, pprm or prm.get("precond"),
