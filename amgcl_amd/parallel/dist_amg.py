@@ -17,10 +17,12 @@ mpi/coarsening/smoothed_aggregation.hpp semantics. Differences by design:
   owners over the matrix's established comm pattern (the same remote-rows
   exchange as mpi/coarsening/detail galerkin, reference pmis.hpp:149-231).
   Each coarse operator is a genuine DistMatrix with its own halo exchange.
-- When the global coarse problem is small it is replicated: every rank
-  gathers the strips and solves redundantly with the dense inverse
-  (replicated direct solve instead of mpi/partition/merge + distributed
-  skyline LU — one all_gather per cycle, no subcommunicator bookkeeping).
+- Coarse levels repartition by REPLICATION: once the global problem fits
+  `repart_threshold`, every rank gathers the strips and continues the SAME
+  hierarchy serially (a replicated AMG tail, applied redundantly — the
+  merge-to-fewer-ranks semantics of mpi/partition/merge.hpp without
+  subcommunicator bookkeeping; one all_gather of the level rhs per cycle).
+  At `coarse_enough` the replicated tail bottoms out in the dense inverse.
 
 Smoothing runs on the distributed operator (halo exchange + fused local
 kernels): SPAI0 / damped Jacobi weights are computed from the FULL row
@@ -48,6 +50,7 @@ class DistAMG:
             "npost": 1,
             "ncycle": 1,
             "coarse_enough": 3000,
+            "repart_threshold": 20000,  # replicate the hierarchy below this
             "max_levels": 20,
         }
 
@@ -133,11 +136,12 @@ class DistAMG:
             sizes = [None] * A.world
             self.dist.all_gather_object(sizes, A.n_loc, group=self.group)
             n_glob = int(np.sum(sizes))
-            small = n_glob <= coarse_enough or len(self.levels) + 1 >= max_levels
-            small = small or min(sizes) < 50  # starved strip: stop coarsening
+            small = (n_glob <= int(self.prm["repart_threshold"])
+                     or len(self.levels) + 1 >= max_levels
+                     or min(sizes) < 50)  # starved strip: stop coarsening
             if small:
                 self.levels.append(L)
-                self._build_coarse_solver(A)
+                self._build_coarse_solver(A, n_glob > coarse_enough)
                 break
 
             # rank-local aggregation + smoothed prolongation (host engine)
@@ -179,8 +183,10 @@ class DistAMG:
             strip = CSR.from_scipy(Ac)
             A = DistMatrix(strip, self.base, self.group)
 
-    def _build_coarse_solver(self, A: DistMatrix):
-        """Replicate the global coarse matrix and precompute a dense solve."""
+    def _build_coarse_solver(self, A: DistMatrix, as_amg_tail=False):
+        """Replicate the global coarse problem on every rank: either a dense
+        inverse (global size <= coarse_enough) or a full serial AMG tail that
+        continues the hierarchy redundantly (merge-repartition semantics)."""
         import scipy.sparse as sp
 
         loc, rem = self._strip_scipy(A)
@@ -203,7 +209,27 @@ class DistAMG:
                 for p, c, v in gathered]
         G = sp.vstack(rows, format="csr")
         G.sort_indices()
-        self.coarse = self.base.coarse_solver(CSR.from_scipy(G))
+        G_csr = CSR.from_scipy(G)
+        if as_amg_tail:
+            # replicated serial AMG tail: every rank continues the SAME
+            # hierarchy below this level and applies it redundantly
+            from ..precond import make_preconditioner
+
+            self.coarse = None
+            self.tail = make_preconditioner(
+                G_csr,
+                {"class": "amg",
+                 "coarse_enough": int(self.prm["coarse_enough"]),
+                 "npre": int(self.prm["npre"]), "npost": int(self.prm["npost"]),
+                 "ncycle": int(self.prm["ncycle"]),
+                 "relax": ({"type": "damped_jacobi",
+                            "damping": float(self.prm["relax"]["damping"])}
+                           if self.prm["relax"]["type"] == "damped_jacobi"
+                           else {"type": self.prm["relax"]["type"]})},
+                self.base)
+        else:
+            self.coarse = self.base.coarse_solver(G_csr)
+            self.tail = None
         self._coarse_n = A.n_global
         self._coarse_sizes = [len(p) - 1 for p, _, _ in gathered]
         self._coarse_beg = int(np.sum(self._coarse_sizes[: A.rank]))
@@ -233,7 +259,10 @@ class DistAMG:
         for r, sz in enumerate(self._coarse_sizes):
             gf[off : off + sz] = self._gbuf[r * pad : r * pad + sz]
             off += sz
-        self.coarse(self._gf, self._gu)
+        if self.coarse is not None:
+            self.coarse(self._gf, self._gu)
+        else:
+            self.tail.apply(self._gf, self._gu)  # replicated hierarchy tail
         self._as_tensor(u)[:] = self._as_tensor(self._gu)[
             self._coarse_beg : self._coarse_beg + t.numel()]
 

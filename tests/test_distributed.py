@@ -266,7 +266,8 @@ def _solve_dist_amg(rank, world):
     strip, b, row_beg, row_end = am.poisson3d_strip(n, rank, world, rhs="ones")
     solve = make_dist_solver(
         strip,
-        {"precond": {"class": "dist_amg", "coarse_enough": 300},
+        {"precond": {"class": "dist_amg", "coarse_enough": 300,
+                     "repart_threshold": 500},
          "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
         backend="cpu",
     )
@@ -314,7 +315,7 @@ def _galerkin_consistency(rank, world):
     strip, _, row_beg, row_end = am.poisson3d_strip(n, rank, world)
     backend = DistBackend(make_backend("cpu"))
     A = backend.matrix(strip)
-    amg = DistAMG(A, {"coarse_enough": 100}, backend)
+    amg = DistAMG(A, {"coarse_enough": 100, "repart_threshold": 150}, backend)
     L1 = amg.levels[1].A  # first coarse DistMatrix
 
     # assemble the global coarse matrix from the per-rank strips
@@ -492,3 +493,44 @@ def test_dist_cpr(world):
          "solver": {"type": "bicgstab", "tol": 1e-8, "maxiter": 200}})
     _, it_serial, _ = s(b)
     assert iters <= 2 * it_serial + 5
+
+
+def _solve_dist_amg_repart(rank, world):
+    import amgcl_amd as am
+    from amgcl_amd.parallel import make_dist_solver
+
+    n = 20
+    strip, b, row_beg, row_end = am.poisson3d_strip(n, rank, world, rhs="ones")
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "dist_amg", "coarse_enough": 200,
+                     "repart_threshold": 2000},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+        backend="cpu",
+    )
+    has_tail = solve.P.tail is not None
+    x, iters, resid = solve(b)
+    xg = solve.gather_solution(x)
+    return iters, resid, has_tail, None if xg is None else xg.tolist()
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_amg_replicated_tail(world):
+    """Merge-style coarse repartition: below repart_threshold the hierarchy
+    is replicated and continued serially per rank (mpi/partition/merge.hpp
+    semantics, redundant-apply variant)."""
+    out = spawn(world, _solve_dist_amg_repart, 30211 + world)
+    iters, resid, has_tail, xg = out[0]
+    assert has_tail
+    assert resid < 1e-8
+    import amgcl_amd as am
+
+    A, b = am.poisson3d(20, rhs="ones")
+    x = np.asarray(xg)
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-7
+    # the replicated tail continues the same hierarchy: iterations stay in
+    # the serial class
+    s = am.make_solver(A, {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 100},
+                           "precond": {"class": "amg", "coarse_enough": 200}})
+    _, it_serial, _ = s(b)
+    assert iters <= it_serial + 6
