@@ -26,10 +26,26 @@ def main(argv=None):
     ap.add_argument("--reorder", action="store_true", help="Cuthill-McKee reorder")
     ap.add_argument("--scale", action="store_true", help="symmetric diagonal scaling")
     ap.add_argument("-o", "--out", help="write solution (binary)")
+    ap.add_argument("--convert", nargs=2, metavar=("SRC", "DST"),
+                    help="convert a matrix between .mtx and .bin and exit")
     args = ap.parse_args(argv)
 
     import amgcl_amd as am
     from amgcl_amd.params import set_kv
+
+    if args.convert:
+        # mm2bin / bin2mm converters (reference examples/{mm2bin,bin2mm}.cpp)
+        from amgcl_amd import io
+
+        src_p, dst = args.convert
+        m = io.read_crs(src_p) if src_p.endswith(".bin") else io.mm_read(src_p)
+        if dst.endswith(".bin"):
+            io.write_crs(dst, m)
+        else:
+            io.mm_write(dst, m)
+        print(f"converted {src_p} -> {dst} ({m.nrows}x{m.ncols}, {m.nnz} nnz)",
+              file=sys.stderr)
+        return 0
 
     if args.poisson:
         A, b = am.poisson3d(args.poisson, rhs="random")
