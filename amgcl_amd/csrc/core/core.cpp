@@ -163,10 +163,12 @@ static py::tuple poisson3d_strip(i64 n, i64 row_beg, i64 row_end) {
 // Returns (lptr, lcol, lval, rptr, rcol, rval, ghost_global[i64]).
 static py::tuple split_strip(i64 nloc, i64 row_beg, i64 row_end, arr<i32> ptr_a,
                              arr<i32> col_a, arr<double> val_a) {
+    // [row_beg, row_end) is the LOCAL COLUMN window: equal to the row range
+    // for square operators, but wider/narrower for rectangular field blocks
+    // (distributed Schur) and transfer strips (cross-rank pmis).
     const i32 *ptr = ptr_a.data();
     const i32 *col = col_a.data();
     const double *val = val_a.data();
-    (void)row_end;
 
     arr<i32> lptr(nloc + 1), rptr(nloc + 1);
     i32 *LP = lptr.mutable_data();
@@ -177,7 +179,7 @@ static py::tuple split_strip(i64 nloc, i64 row_beg, i64 row_end, arr<i32> ptr_a,
         i32 lc = 0, rc = 0;
         for (i32 j = ptr[i]; j < ptr[i + 1]; ++j) {
             i64 c = col[j];
-            if (c >= row_beg && c < row_beg + nloc) ++lc;
+            if (c >= row_beg && c < row_end) ++lc;
             else ++rc;
         }
         LP[i + 1] = lc;
@@ -198,7 +200,7 @@ static py::tuple split_strip(i64 nloc, i64 row_beg, i64 row_end, arr<i32> ptr_a,
         i32 lh = LP[i], rh = RP[i];
         for (i32 j = ptr[i]; j < ptr[i + 1]; ++j) {
             i64 c = col[j];
-            if (c >= row_beg && c < row_beg + nloc) {
+            if (c >= row_beg && c < row_end) {
                 LC[lh] = (i32)(c - row_beg);
                 LV[lh] = val[j];
                 ++lh;

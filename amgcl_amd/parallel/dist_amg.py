@@ -44,7 +44,8 @@ class DistAMG:
     @staticmethod
     def defaults():
         return {
-            "coarsening": {"type": "smoothed_aggregation", "eps_strong": 0.08},
+            "coarsening": {"type": "smoothed_aggregation", "eps_strong": 0.08,
+                           "cross_rank": False},  # True: pmis over the halo
             "relax": {"type": "spai0", "damping": 0.72},
             "npre": 1,
             "npost": 1,
@@ -144,6 +145,12 @@ class DistAMG:
                 self._build_coarse_solver(A, n_glob > coarse_enough)
                 break
 
+            if self.prm["coarsening"]["cross_rank"]:
+                A = self._step_down_pmis(A, L, eps, sizes)
+                eps *= 0.5
+                self.levels.append(L)
+                continue
+
             # rank-local aggregation + smoothed prolongation (host engine)
             Ah = A.A_loc_host
             agg = (_core.aggregates_parallel if Ah.nrows > 100_000
@@ -182,6 +189,27 @@ class DistAMG:
             Ac.sort_indices()
             strip = CSR.from_scipy(Ac)
             A = DistMatrix(strip, self.base, self.group)
+
+    def _step_down_pmis(self, A, L, eps, sizes):
+        """Cross-rank pmis aggregation + distributed transfer operators
+        (parity: mpi/coarsening/pmis.hpp — aggregates cross rank boundaries;
+        P/R are rectangular DistMatrix over the coarse partition and the
+        V-cycle transfers halo-exchange)."""
+        from . import pmis
+
+        ids, S_loc, S_rem = pmis.pmis_aggregates(A, eps, self.dist, self.group)
+        coarse_of, naggr_sizes, coarse_begs = pmis.renumber(
+            A, ids, self.dist, self.group)
+        P = pmis.smoothed_p_strip(A, coarse_of, S_loc, S_rem,
+                                  int(coarse_begs[-1]), 2.0 / 3.0,
+                                  self.dist, self.group)
+        R = pmis.transpose_exchange(A, P, coarse_begs, self.dist, self.group)
+        Ac = pmis.galerkin_strip(A, P, coarse_begs, self.dist, self.group)
+        L.P = DistMatrix(CSR.from_scipy(P), self.base, self.group,
+                         col_sizes=naggr_sizes)
+        L.R = DistMatrix(CSR.from_scipy(R), self.base, self.group,
+                         col_sizes=sizes)
+        return DistMatrix(CSR.from_scipy(Ac), self.base, self.group)
 
     def _build_coarse_solver(self, A: DistMatrix, as_amg_tail=False):
         """Replicate the global coarse problem on every rank: either a dense
@@ -288,9 +316,13 @@ class DistAMG:
             for _ in range(int(self.prm["npre"])):
                 self._relax(L, L.f, L.u)
             b.residual(L.f, L.A, L.u, L.t)
-            self.base.spmv(1.0, L.R, L.t, 0.0, C.f)  # block-diagonal R: local
+            # decoupled transfers are rank-local; cross-rank (pmis) transfers
+            # are rectangular DistMatrix and halo-exchange
+            tb = b if isinstance(L.R, DistMatrix) else self.base
+            tb.spmv(1.0, L.R, L.t, 0.0, C.f)
             self._cycle(k + 1)
-            self.base.spmv(1.0, L.P, C.u, 1.0, L.u)  # block-diagonal P: local
+            tb = b if isinstance(L.P, DistMatrix) else self.base
+            tb.spmv(1.0, L.P, C.u, 1.0, L.u)
             for _ in range(int(self.prm["npost"])):
                 self._relax(L, L.f, L.u)
 
