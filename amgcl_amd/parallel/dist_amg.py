@@ -4,14 +4,13 @@ Parity: amgcl/mpi/amg.hpp (one AMG hierarchy over the whole distributed
 matrix, coarse operators are themselves distributed matrices) with
 mpi/coarsening/smoothed_aggregation.hpp semantics. Differences by design:
 
-- Aggregation is rank-local (decoupled): aggregates never cross rank
-  boundaries, so P and R are block-diagonal over ranks and the grid
-  transfers need no communication. The reference's pmis negotiates
-  cross-boundary aggregates (mpi/coarsening/pmis.hpp:417); the decoupled
-  variant trades slightly higher boundary iteration counts for a
-  communication-free setup of the transfer operators — the standard
-  large-scale compromise, and the xGMI-friendly one (transfers are the
-  inner, latency-sensitive ops of the V-cycle).
+- Aggregation is rank-local (decoupled) by default: aggregates never cross
+  rank boundaries, so P and R are block-diagonal over ranks and the grid
+  transfers need no communication — the xGMI-friendly choice (transfers are
+  the inner, latency-sensitive ops of the V-cycle). With
+  coarsening.cross_rank=True the aggregation instead runs the distributed
+  pmis of `parallel/pmis.py` (reference mpi/coarsening/pmis.hpp semantics,
+  reproduces the single-process MIS exactly); transfers then halo-exchange.
 - The Galerkin product IS fully coupled: Ac = P^T (A_loc P_loc +
   A_rem P_ghost) where the P rows of ghost columns are fetched from their
   owners over the matrix's established comm pattern (the same remote-rows
@@ -67,6 +66,7 @@ class DistAMG:
         self.group = dist_A.group
         self.levels = []
         self.coarse = None
+        self._world = dist_A.world
         self._build(dist_A)
 
     # --- setup ---------------------------------------------------------------
@@ -276,6 +276,12 @@ class DistAMG:
         """All-gather the coarse rhs strips, solve replicated, slice ours."""
         import torch
 
+        if self._world == 1:  # single rank: the strip IS the global problem
+            if self.coarse is not None:
+                self.coarse(f, u)
+            else:
+                self.tail.apply(f, u)
+            return
         t = self._as_tensor(f)
         pad = self._gpad
         send = torch.zeros(pad, dtype=torch.float64, device=self._gbuf.device)

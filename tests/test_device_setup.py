@@ -256,3 +256,44 @@ def test_rebuild_on_device_hierarchy(hip):
     import math
 
     assert math.sqrt(hip.dot(r, r)) / math.sqrt(hip.dot(b, b)) < 1e-7
+
+
+def test_dist_amg_on_device(hip):
+    """DistAMG hierarchy with device-resident levels (single-rank gloo:
+    exchange paths no-op, kernels and replicated tail run on the GPU)."""
+    import os
+
+    import torch
+    import torch.distributed as dist
+
+    import amgcl_amd as am
+    from amgcl_amd.parallel.dist_amg import DistAMG
+    from amgcl_amd.parallel.dist_backend import DistBackend
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29551")
+    created = False
+    if not dist.is_initialized():
+        dist.init_process_group("gloo", rank=0, world_size=1)
+        created = True
+    try:
+        A, b = am.poisson3d(24, rhs="random")
+        backend = DistBackend(hip)
+        Ad = backend.matrix(A)
+        for cross in (False, True):
+            amg = DistAMG(Ad, {"coarse_enough": 300, "repart_threshold": 800,
+                               "coarsening": {"cross_rank": cross}}, backend)
+            bd = hip.from_host(b)
+            x = hip.vector(A.nrows)
+            r = hip.vector(A.nrows)
+            xa = hip.vector(A.nrows)
+            res = []
+            for _ in range(6):
+                backend.residual(bd, Ad, xa, r)
+                res.append(float(r.norm()))
+                amg.apply(r, x)
+                xa += x
+            assert res[-1] < 0.05 * res[0], (cross, res)
+    finally:
+        if created:
+            dist.destroy_process_group()
