@@ -891,13 +891,10 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
         int total = ub[row];
         if (total <= ub_lo || total > ub_hi) continue;
         int ab = aptr[row], alen = aptr[row + 1] - ab;
-        if (alen > BIGROW) {  // true table overflow is caught by the probes
-            if (lane == 0) atomicAdd(overflow, 1);
-            continue;
-        }
         for (int t = lane; t < SLOTS; t += WAVE) tk[t] = -1;
-        // build the product-space prefix (serial scan by lane 0: alen <= 128)
-        if (lane == 0) {
+        const bool huge = alen > BIGROW;
+        if (!huge && lane == 0) {
+            // product-space prefix (serial scan by lane 0)
             int acc = 0;
             for (int j = 0; j < alen; ++j) {
                 pfx[wid][j] = acc;
@@ -911,6 +908,26 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
         __builtin_amdgcn_wave_barrier();
         int inserted = 0;
         bool ovf = false;
+        if (huge) {
+            // rare very-long A rows (big coarse aggregates): iterate A
+            // entries serially, the wave splits each B row — no LDS staging
+            for (int j = 0; j < alen && !ovf; ++j) {
+                int ca = acol[ab + j];
+                for (int jb = bptr[ca] + lane; jb < bptr[ca + 1] && !ovf;
+                     jb += WAVE) {
+                    int key = bcol[jb];
+                    uint32_t h = ((uint32_t)key * 2654435761u) & SMASK;
+                    int probes = 0;
+                    while (true) {
+                        int old = atomicCAS(&tk[h], -1, key);
+                        if (old == -1) { ++inserted; break; }
+                        if (old == key) break;
+                        h = (h + 1) & SMASK;
+                        if (++probes >= SLOTS) { ovf = true; break; }
+                    }
+                }
+            }
+        } else {
         for (int t = lane; t < total && !ovf; t += WAVE) {
             int ja = pfx_find(pfx[wid], alen, t);
             int key = bcol[bbeg[wid][ja] + (t - pfx[wid][ja])];
@@ -923,6 +940,7 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
                 h = (h + 1) & SMASK;
                 if (++probes >= SLOTS) { ovf = true; break; }
             }
+        }
         }
         if (ovf && lane == 0) atomicAdd(overflow, 1);
 #pragma unroll
@@ -965,27 +983,48 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
         int total = ub[row];
         if (total <= SGSMALL) continue;
         int ab = aptr[row], alen = aptr[row + 1] - ab;
-        if (alen > BIGROW) continue;  // host fallback
         int base = (row == 0) ? 0 : cptr_scanned[row - 1];
         int len = cptr_scanned[row] - base;
         if (len <= len_lo || len > len_hi) continue;
+        const bool huge = alen > BIGROW;
         for (int t = lane; t < SLOTS; t += WAVE) {
             tk[t] = -1;
             tv[t] = 0.0;
         }
-        for (int j = lane; j < alen; j += WAVE) av[wid][j] = aval[ab + j];
-        if (lane == 0) {
-            int acc = 0;
-            for (int j = 0; j < alen; ++j) {
-                pfx[wid][j] = acc;
-                int ca = acol[ab + j];
-                bbeg[wid][j] = bptr[ca];
-                acc += bptr[ca + 1] - bptr[ca];
+        if (!huge) {
+            for (int j = lane; j < alen; j += WAVE) av[wid][j] = aval[ab + j];
+            if (lane == 0) {
+                int acc = 0;
+                for (int j = 0; j < alen; ++j) {
+                    pfx[wid][j] = acc;
+                    int ca = acol[ab + j];
+                    bbeg[wid][j] = bptr[ca];
+                    acc += bptr[ca + 1] - bptr[ca];
+                }
+                pfx[wid][alen] = acc;
             }
-            pfx[wid][alen] = acc;
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
+        if (huge) {
+            for (int j = 0; j < alen; ++j) {
+                int ca = acol[ab + j];
+                double va = aval[ab + j];
+                for (int jb = bptr[ca] + lane; jb < bptr[ca + 1]; jb += WAVE) {
+                    int key = bcol[jb];
+                    double v = va * bval[jb];
+                    uint32_t h = ((uint32_t)key * 2654435761u) & SMASK;
+                    while (true) {
+                        int old = atomicCAS(&tk[h], -1, key);
+                        if (old == -1 || old == key) {
+                            atomicAdd(&tv[h], v);
+                            break;
+                        }
+                        h = (h + 1) & SMASK;
+                    }
+                }
+            }
+        } else {
         for (int t = lane; t < total; t += WAVE) {
             int ja = pfx_find(pfx[wid], alen, t);
             int jb = bbeg[wid][ja] + (t - pfx[wid][ja]);
@@ -1000,6 +1039,7 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
                 }
                 h = (h + 1) & SMASK;
             }
+        }
         }
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();

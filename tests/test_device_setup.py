@@ -297,3 +297,33 @@ def test_dist_amg_on_device(hip):
     finally:
         if created:
             dist.destroy_process_group()
+
+
+def test_device_spgemm_huge_rows(hip):
+    """SpGEMM rows whose A row exceeds the LDS staging capacity (BIGROW=256)
+    take the serial-A-entry path — must match the host product exactly."""
+    import scipy.sparse as sp
+
+    from amgcl_amd import _core
+    from amgcl_amd.backend import hip_setup
+    from amgcl_amd.matrix import CSR
+
+    rng = np.random.default_rng(41)
+    n, m, k = 300, 2000, 350
+    A = sp.random(n, m, density=0.02, random_state=rng, format="lil")
+    A[0, :800] = rng.standard_normal(800)   # alen 800 >> BIGROW
+    A[5, :300] = rng.standard_normal(300)   # alen 300 > BIGROW
+    A = A.tocsr()
+    A.sort_indices()
+    B = sp.random(m, k, density=0.02, random_state=rng, format="csr")
+    B.data = rng.standard_normal(B.nnz)
+    B.sort_indices()
+    ref = (A @ B).tocsr()
+    ref.sort_indices()
+
+    Ad = hip.matrix(CSR.from_scipy(A.tocsr()))
+    Bd = hip.matrix(CSR.from_scipy(B))
+    Cd = hip_setup.spgemm(Ad, Bd, sort=True)
+    C = hip_setup.download(Cd).to_scipy()
+    assert C.nnz == ref.nnz
+    np.testing.assert_allclose(C.toarray(), ref.toarray(), rtol=1e-12, atol=1e-13)
