@@ -1,0 +1,142 @@
+"""Property-based tests (hypothesis): invariants that must hold for ANY
+valid input, not just the fixtures — IO roundtrips, adapter involutions,
+host kernel algebra vs scipy, parameter-tree validation."""
+import numpy as np
+import pytest
+import scipy.sparse as sp
+from hypothesis import given, settings, strategies as st
+
+import amgcl_amd as am
+from amgcl_amd import _core
+from amgcl_amd.matrix import CSR
+
+COMMON = dict(deadline=None, max_examples=25)
+
+
+def rand_csr(n, m, density, seed, ensure_diag=False):
+    rng = np.random.default_rng(seed)
+    a = sp.random(n, m, density=density, random_state=rng, format="csr")
+    a.data = rng.standard_normal(a.nnz)
+    if ensure_diag:
+        a = (a + sp.diags(np.full(min(n, m), float(max(n, m))), shape=(n, m))).tocsr()
+    a.sort_indices()
+    return a
+
+
+@settings(**COMMON)
+@given(n=st.integers(2, 40), m=st.integers(2, 40), seed=st.integers(0, 10**6))
+def test_host_spgemm_matches_scipy(n, m, seed):
+    A = rand_csr(n, m, 0.2, seed)
+    B = rand_csr(m, n, 0.2, seed + 1)
+    Ca = CSR.from_scipy(A) @ CSR.from_scipy(B)
+    ref = (A @ B).tocsr()
+    ref.sort_indices()
+    got = Ca.to_scipy()
+    got.sort_indices()
+    assert got.shape == ref.shape
+    assert abs(got - ref).max() < 1e-12 * max(1.0, abs(ref).max())
+
+
+@settings(**COMMON)
+@given(n=st.integers(2, 50), seed=st.integers(0, 10**6))
+def test_transpose_involution(n, seed):
+    A = CSR.from_scipy(rand_csr(n, n, 0.3, seed))
+    T2 = A.transpose().transpose()
+    ref = A.to_scipy()
+    got = T2.to_scipy()
+    got.sort_indices()
+    ref.sort_indices()
+    assert (got != ref).nnz == 0
+
+
+@settings(**COMMON)
+@given(n=st.integers(2, 30), seed=st.integers(0, 10**6))
+def test_io_roundtrips(tmp_path_factory, n, seed):
+    from amgcl_amd import io
+
+    d = tmp_path_factory.mktemp("io")
+    A = CSR.from_scipy(rand_csr(n, n, 0.3, seed, ensure_diag=True))
+    io.mm_write(str(d / "a.mtx"), A)
+    B = io.mm_read(str(d / "a.mtx"))
+    assert abs(B.to_scipy() - A.to_scipy()).max() < 1e-12
+    io.write_crs(str(d / "a.bin"), A)
+    C = io.read_crs(str(d / "a.bin"))
+    assert (C.to_scipy() != A.to_scipy()).nnz == 0
+    # strip read equals the row slice
+    lo, hi = n // 3, max(n // 3 + 1, 2 * n // 3)
+    S = io.read_crs(str(d / "a.bin"), row_beg=lo, row_end=hi)
+    ref = A.to_scipy()[lo:hi]
+    assert abs(S.to_scipy() - ref).max() == 0.0
+
+
+@settings(**COMMON)
+@given(n=st.integers(4, 24), seed=st.integers(0, 10**6))
+def test_amg_cg_solves_random_spd(n, seed):
+    """Any SPD system: AMG-preconditioned CG reaches the tolerance and the
+    returned residual is honest (matches the true residual)."""
+    rng = np.random.default_rng(seed)
+    B = rand_csr(n, n, 0.3, seed)
+    A = (B @ B.T + sp.identity(n)).tocsr()
+    A.sort_indices()
+    b = rng.standard_normal(n)
+    s = am.make_solver(CSR.from_scipy(A),
+                       {"precond": {"class": "amg", "coarse_enough": 10},
+                        "solver": {"type": "cg", "tol": 1e-10, "maxiter": 200}})
+    x, iters, resid = s(b)
+    true = np.linalg.norm(b - A @ x) / max(np.linalg.norm(b), 1e-300)
+    assert resid < 1e-10
+    assert true < 1e-8
+
+
+@settings(**COMMON)
+@given(n=st.integers(3, 30), seed=st.integers(0, 10**6))
+def test_reorder_and_scale_involutions(n, seed):
+    from amgcl_amd.adapter import Reordered, ScaledProblem
+
+    rng = np.random.default_rng(seed)
+    A = CSR.from_scipy(rand_csr(n, n, 0.4, seed, ensure_diag=True))
+    v = rng.standard_normal(n)
+    ro = Reordered(A)
+    np.testing.assert_allclose(ro.inverse(ro.forward(v)), v, atol=1e-14)
+    sc = ScaledProblem(A)
+    # scale then unscale the solution of the scaled system reproduces x
+    x = rng.standard_normal(n)
+    b = A.to_scipy() @ x
+    bs = sc.scale_rhs(b)
+    xs = np.linalg.solve(sc.A.to_scipy().toarray(), bs)
+    np.testing.assert_allclose(sc.unscale_x(xs), x, rtol=1e-6, atol=1e-8)
+
+
+@settings(**COMMON)
+@given(n=st.integers(2, 20), seed=st.integers(0, 10**6))
+def test_complex_adapter_involution(n, seed):
+    from amgcl_amd.adapter import complex_to_real, real_to_complex
+
+    rng = np.random.default_rng(seed)
+    a = rand_csr(n, n, 0.4, seed, ensure_diag=True)
+    Ac = a + 1j * sp.random(n, n, density=0.2, random_state=rng, format="csr")
+    Ar, _ = complex_to_real(Ac.tocsr())
+    z = rng.standard_normal(n) + 1j * rng.standard_normal(n)
+    zr = np.empty(2 * n)
+    zr[0::2], zr[1::2] = z.real, z.imag
+    yr = Ar.to_scipy() @ zr
+    y = Ac @ z
+    np.testing.assert_allclose(yr[0::2] + 1j * yr[1::2], y, rtol=1e-12, atol=1e-12)
+    zz = real_to_complex(zr)
+    np.testing.assert_allclose(zz, z)
+
+
+@settings(**COMMON)
+@given(st.dictionaries(st.sampled_from(["tol", "maxiter"]),
+                       st.floats(0.001, 1.0), max_size=2),
+       st.text(alphabet="abcdef", min_size=1, max_size=8))
+def test_merge_params_unknown_keys_raise(known, bogus):
+    from amgcl_amd.params import UnknownParameter, merge_params
+
+    defaults = {"tol": 1e-8, "maxiter": 100}
+    merged = merge_params(defaults, dict(known))
+    for k, v in known.items():
+        assert merged[k] == v
+    if bogus not in defaults:
+        with pytest.raises(UnknownParameter):
+            merge_params(defaults, {bogus: 1})
