@@ -91,11 +91,28 @@ class CpuBackend:
         x[idx] = buf
 
     # --- coarse direct solver --------------------------------------------
-    def coarse_solver(self, csr: CSR):
+    def coarse_solver(self, csr: CSR, kind="dense"):
+        if kind == "splu":
+            return SpluCoarseSolver(csr, self)
         return DenseCoarseSolver(csr, self)
 
     def synchronize(self):
         pass
+
+
+class SpluCoarseSolver:
+    """Alternative coarsest-level direct solve: scipy SuperLU factorization
+    (parity: solver/eigen.hpp — Eigen SparseLU as alternative coarse solver).
+    Preferable to the dense inverse when coarse_enough is large and the
+    coarse operator is still very sparse."""
+
+    def __init__(self, csr: CSR, backend=None):
+        import scipy.sparse.linalg as spla
+
+        self._lu = spla.splu(csr.to_scipy().tocsc())
+
+    def __call__(self, f, u):
+        u[:] = self._lu.solve(np.asarray(f))
 
 
 class DenseCoarseSolver:

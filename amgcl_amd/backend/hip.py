@@ -216,13 +216,40 @@ class HipBackend:
                                      x.data_ptr(), _stream()), "scatter")
 
     # --- coarse direct solver ---------------------------------------------
-    def coarse_solver(self, csr):
+    def coarse_solver(self, csr, kind="dense"):
+        if kind == "splu":
+            # host SuperLU with a D2H/H2D round-trip per cycle — the
+            # reference's coarse-solve shape (hip.hpp:73-96); only worth it
+            # for very large, very sparse coarse levels
+            return HostSpluSolver(csr, self)
         if isinstance(csr, DeviceCSR):
             return DeviceDenseSolver.from_device(csr, self)
         return DeviceDenseSolver(csr, self)
 
     def synchronize(self):
         self.torch.cuda.synchronize()
+
+
+class HostSpluSolver:
+    """Host SuperLU coarse solve with a device round-trip per application
+    (the reference's coarse-solve shape, backend/hip.hpp:73-96; opt-in via
+    direct_solver='splu')."""
+
+    def __init__(self, csr, backend):
+        import scipy.sparse.linalg as spla
+
+        if isinstance(csr, DeviceCSR):
+            from . import hip_setup
+
+            csr = hip_setup.download(csr)
+        self._lu = spla.splu(csr.to_scipy().tocsc())
+        self.n = csr.nrows
+
+    def __call__(self, f, u):
+        import torch
+
+        sol = self._lu.solve(f.cpu().numpy().astype(np.float64))
+        u.copy_(torch.from_numpy(sol).to(u.device, dtype=u.dtype))
 
 
 class DeviceDenseSolver:

@@ -11,8 +11,16 @@ MI355X-native re-design of the reference's MPI layer (amgcl/mpi/):
   - mpi::block_preconditioner (mpi/block_preconditioner.hpp:49) and
     mpi::subdomain_deflation (mpi/subdomain_deflation.hpp:113)
     -> LocalBlockPrecond / SubdomainDeflation.
+  - mpi::amg + mpi::coarsening::pmis (mpi/amg.hpp:56, mpi/coarsening/
+    pmis.hpp:50) -> DistAMG (one hierarchy over the distributed operator;
+    decoupled or cross-rank pmis aggregation, replicated-tail repartition).
+  - mpi::schur_pressure_correction / mpi::cpr -> DistSchurPressureCorrection
+    / DistCPR (fully-coupled distributed field splits).
 """
 from .dist_matrix import DistMatrix
 from .dist_backend import DistBackend, DistInnerProduct
 from .precond import LocalBlockPrecond
 from .solver import make_dist_solver, DistSolver
+from .dist_amg import DistAMG
+from .schur import DistSchurPressureCorrection
+from .cpr import DistCPR

@@ -45,6 +45,7 @@ class AMG:
             "ncycle": 1,      # 1 = V-cycle, 2 = W-cycle
             "pre_cycles": 1,
             "direct_coarse": True,
+            "direct_solver": "dense",  # or "splu" (scipy SuperLU)
             "keep_host_matrices": False,
             "precision": "fp64",  # "mixed" = fp32 hierarchy under fp64 Krylov
             "block_value": 0,      # >1: store level operators as BSR blocks
@@ -96,7 +97,8 @@ class AMG:
             last.A_host = A
         last.A = backend.matrix(A)
         if self.prm["direct_coarse"]:
-            self.coarse_solve = backend.coarse_solver(A)
+            self.coarse_solve = backend.coarse_solver(
+                A, kind=self.prm["direct_solver"])
         else:
             last.relax = relax_factory(A, backend)
 
@@ -154,7 +156,8 @@ class AMG:
                 lvl.t = backend.vector(A_host.nrows)
                 if prm["direct_coarse"]:
                     with prof.scope("coarse_solver"):
-                        self.coarse_solve = backend.coarse_solver(A_host)
+                        self.coarse_solve = backend.coarse_solver(
+                            A_host, kind=prm["direct_solver"])
                 else:
                     lvl.relax = relax_factory(A_host, backend)
                 self.levels.append(lvl)

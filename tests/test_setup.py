@@ -108,3 +108,18 @@ def test_unknown_param_raises():
         am.make_solver(A, {"precond": {"class": "amg", "coarsening": {"type": "smoothed_aggregation", "bogus": 1}}})
     with pytest.raises(Exception, match="unknown"):
         am.make_solver(A, {"bogus": {}})
+
+
+def test_splu_coarse_solver():
+    """Alternative sparse-LU coarse solve (parity: solver/eigen.hpp class of
+    alternative direct coarse solvers) matches the dense-inverse iterations."""
+    import amgcl_amd as am
+
+    A, b = am.poisson3d(16, rhs="random")
+    prm = {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 100},
+           "precond": {"class": "amg", "coarse_enough": 800}}
+    _, it_dense, r_dense = am.make_solver(A, prm)(b)
+    prm["precond"]["direct_solver"] = "splu"
+    _, it_splu, r_splu = am.make_solver(A, prm)(b)
+    assert r_splu < 1e-8
+    assert it_splu == it_dense
