@@ -1972,6 +1972,114 @@ static void ilu0_solve(i64 nrows, arr<i32> ptr, arr<i32> col, arr<double> lu,
     }
 }
 
+// ---------------------------------------------------------------------------
+// Level schedule of a triangular dependency DAG (parity: relaxation/detail/
+// ilu_solve.hpp:257 sptr_solve level scheduling). For the LOWER solve row i
+// depends on rows c < i among its strictly-lower nonzeros; for the UPPER
+// solve on rows c > i. Rows of one level are independent, so the parallel
+// sweep is BITWISE identical to the serial one (per-row accumulation order
+// unchanged). Returns (lvl_ptr, rows) with rows bucketed by level, ascending
+// row index inside each level.
+// ---------------------------------------------------------------------------
+static py::tuple tri_levels(i64 nrows, arr<i32> ptr_a, arr<i32> col_a,
+                            arr<i32> dia_a, bool lower) {
+    const i32 *P = ptr_a.data(), *C = col_a.data(), *dia = dia_a.data();
+    std::vector<i32> lvl(nrows, 0);
+    i32 nlvl = nrows ? 1 : 0;
+    if (lower) {
+        for (i64 i = 0; i < nrows; ++i) {
+            i32 L = 0;
+            for (i32 j = P[i]; j < dia[i]; ++j)
+                if (lvl[C[j]] + 1 > L) L = lvl[C[j]] + 1;
+            lvl[i] = L;
+            if (L + 1 > nlvl) nlvl = L + 1;
+        }
+    } else {
+        for (i64 i = nrows - 1; i >= 0; --i) {
+            i32 L = 0;
+            for (i32 j = dia[i] + 1; j < P[i + 1]; ++j)
+                if (lvl[C[j]] + 1 > L) L = lvl[C[j]] + 1;
+            lvl[i] = L;
+            if (L + 1 > nlvl) nlvl = L + 1;
+        }
+    }
+    arr<i32> lvl_ptr(nlvl + 1);
+    i32 *LP = lvl_ptr.mutable_data();
+    std::fill(LP, LP + nlvl + 1, 0);
+    for (i64 i = 0; i < nrows; ++i) ++LP[lvl[i] + 1];
+    for (i32 l = 0; l < nlvl; ++l) LP[l + 1] += LP[l];
+    arr<i32> rows(nrows);
+    i32 *R = rows.mutable_data();
+    std::vector<i32> head(LP, LP + nlvl);
+    for (i64 i = 0; i < nrows; ++i) R[head[lvl[i]]++] = (i32)i;
+    return py::make_tuple(lvl_ptr, rows);
+}
+
+// OpenMP level-scheduled ILU(0/k/T) triangular solves — same results as the
+// serial ilu0_solve, levels run in parallel.
+static void ilu0_solve_parallel(i64 nrows, arr<i32> ptr, arr<i32> col,
+                                arr<double> lu, arr<i32> dia_a,
+                                arr<i32> low_ptr_a, arr<i32> low_rows_a,
+                                arr<i32> up_ptr_a, arr<i32> up_rows_a,
+                                arr<double> z) {
+    const i32 *P = ptr.data(), *C = col.data(), *dia = dia_a.data();
+    const double *LU = lu.data();
+    double *Z = z.mutable_data();
+    const i32 *LL = low_ptr_a.data(), *LR = low_rows_a.data();
+    const i32 *UL = up_ptr_a.data(), *UR = up_rows_a.data();
+    const i64 nlow = (i64)low_ptr_a.size() - 1;
+    const i64 nup = (i64)up_ptr_a.size() - 1;
+    for (i64 l = 0; l < nlow; ++l) {
+#pragma omp parallel for schedule(static) if (LL[l + 1] - LL[l] > 512)
+        for (i32 t = LL[l]; t < LL[l + 1]; ++t) {
+            i32 i = LR[t];
+            double s = Z[i];
+            for (i32 j = P[i]; j < dia[i]; ++j) s -= LU[j] * Z[C[j]];
+            Z[i] = s;
+        }
+    }
+    for (i64 l = 0; l < nup; ++l) {
+#pragma omp parallel for schedule(static) if (UL[l + 1] - UL[l] > 512)
+        for (i32 t = UL[l]; t < UL[l + 1]; ++t) {
+            i32 i = UR[t];
+            double s = Z[i];
+            for (i32 j = dia[i] + 1; j < P[i + 1]; ++j) s -= LU[j] * Z[C[j]];
+            Z[i] = s * LU[dia[i]];
+        }
+    }
+}
+
+// Deterministic parallel Gauss-Seidel: multicolor sweep (rows of one color
+// are independent). The reference's level-scheduled parallel GS
+// (gauss_seidel.hpp:185) races on same-level upper reads and is therefore
+// run-to-run nondeterministic; the multicolor ordering (the same one the
+// GPU path uses) is the deterministic CPU-parallel choice.
+static void gauss_seidel_colored(i64 nrows, arr<i32> ptr, arr<i32> col,
+                                 arr<double> val, arr<double> b, arr<double> x,
+                                 arr<i32> order_a, arr<i32> cptr_a, bool forward) {
+    auto A = view(nrows, 0, ptr, col, val);
+    const double *B = b.data();
+    double *X = x.mutable_data();
+    const i32 *order = order_a.data();
+    const i32 *CP = cptr_a.data();
+    const i64 nc = (i64)cptr_a.size() - 1;
+    auto sweep_color = [&](i64 c) {
+#pragma omp parallel for schedule(static) if (CP[c + 1] - CP[c] > 512)
+        for (i32 t = CP[c]; t < CP[c + 1]; ++t) {
+            i32 i = order[t];
+            double s = B[i], d = 1.0;
+            for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+                i32 cc = A.col[j];
+                if (cc == i) d = A.val[j];
+                else s -= A.val[j] * X[cc];
+            }
+            X[i] = s / d;
+        }
+    };
+    if (forward) for (i64 c = 0; c < nc; ++c) sweep_color(c);
+    else         for (i64 c = nc - 1; c >= 0; --c) sweep_color(c);
+}
+
 PYBIND11_MODULE(_core, m) {
     m.doc() = "amgcl_amd host setup engine (OpenMP)";
     m.def("poisson3d", &poisson3d, py::arg("n"), py::arg("anisotropy") = 1.0);
@@ -2001,5 +2109,8 @@ PYBIND11_MODULE(_core, m) {
     m.def("ilu0_chow_patel", &ilu0_chow_patel);
     m.def("spai1", &spai1);
     m.def("ilu0_solve", &ilu0_solve);
+    m.def("tri_levels", &tri_levels);
+    m.def("ilu0_solve_parallel", &ilu0_solve_parallel);
+    m.def("gauss_seidel_colored", &gauss_seidel_colored);
     m.def("omp_threads", []() { return omp_get_max_threads(); });
 }

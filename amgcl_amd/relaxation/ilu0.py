@@ -19,7 +19,11 @@ class ILU0:
 
     @staticmethod
     def defaults():
-        return {"damping": 1.0, "solve_iters": 2, "solve_damping": 0.72}
+        # solve_serial=None: level-scheduled OpenMP triangular sweeps for
+        # large factors (bitwise-identical to serial; parity with
+        # relaxation/detail/ilu_solve.hpp:257), serial below the threshold
+        return {"damping": 1.0, "solve_iters": 2, "solve_damping": 0.72,
+                "solve_serial": None}
 
     def _factor(self, A, p):
         lu, dia = _core.ilu0_factor(A.nrows, A.ptr, A.col, A.val)
@@ -42,6 +46,20 @@ class ILU0:
             self._serial = True
             self.ptr, self.col = np.asarray(fptr), np.asarray(fcol)
             self.lu, self.dia = np.asarray(lu), np.asarray(dia)
+            import os
+
+            ser = p["solve_serial"]
+            if ser is None:
+                ser = os.cpu_count() < 4 or self.n < 20_000
+            self._levels = None
+            if not ser:
+                self._dia32 = np.ascontiguousarray(self.dia, dtype=np.int32)
+                lp, lr = _core.tri_levels(self.n, self.ptr, self.col,
+                                          self._dia32, True)
+                up, ur = _core.tri_levels(self.n, self.ptr, self.col,
+                                          self._dia32, False)
+                self._levels = (np.asarray(lp), np.asarray(lr),
+                                np.asarray(up), np.asarray(ur))
         else:
             self._serial = False
             self.solve_iters = int(p["solve_iters"])
@@ -68,7 +86,12 @@ class ILU0:
             self._t2 = backend.vector(self.n)
 
     def _solve_serial(self, z):
-        _core.ilu0_solve(self.n, self.ptr, self.col, self.lu, self.dia, z)
+        if self._levels is not None:
+            lp, lr, up, ur = self._levels
+            _core.ilu0_solve_parallel(self.n, self.ptr, self.col, self.lu,
+                                      self._dia32, lp, lr, up, ur, z)
+        else:
+            _core.ilu0_solve(self.n, self.ptr, self.col, self.lu, self.dia, z)
 
     def _solve_jacobi(self, z):
         """Damped-Jacobi iterated approximate triangular solves
