@@ -173,3 +173,37 @@ def test_crs_builder():
                            "solver": {"type": "cg", "tol": 1e-10, "maxiter": 200}})
     x, it, res = s(b)
     assert np.linalg.norm(b - A @ x) < 1e-8
+
+
+def test_cli_end_to_end(tmp_path):
+    """The solver CLI (reference examples/solver.cpp): file input, key=value
+    params, reorder + scaling, binary solution output, format conversion."""
+    import json
+    import subprocess
+    import sys
+
+    import amgcl_amd as am
+    from amgcl_amd import io
+
+    A, b = am.poisson3d(10, rhs="random")
+    io.mm_write(str(tmp_path / "A.mtx"), A)
+    io.mm_write(str(tmp_path / "b.mtx"), np.asarray(b).reshape(-1, 1))
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "amgcl_amd.cli",
+         "-A", str(tmp_path / "A.mtx"), "-f", str(tmp_path / "b.mtx"),
+         "-p", "solver.type=bicgstab", "-p", "solver.tol=1e-8",
+         "-p", "precond.coarse_enough=200",
+         "--reorder", "--scale", "-o", str(tmp_path / "x.bin")],
+        capture_output=True, text=True, cwd=repo, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    x = io.read_dense(str(tmp_path / "x.bin")).ravel()
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6
+    # conversion path
+    r = subprocess.run(
+        [sys.executable, "-m", "amgcl_amd.cli", "--convert",
+         str(tmp_path / "A.mtx"), str(tmp_path / "A.bin")],
+        capture_output=True, text=True, cwd=repo, timeout=120)
+    assert r.returncode == 0
+    B = io.read_crs(str(tmp_path / "A.bin"))
+    assert B.nnz == A.nnz
