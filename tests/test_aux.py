@@ -1,0 +1,89 @@
+"""Auxiliary subsystems: tracing/profiler, failure detection,
+verbose solver output, hierarchy report, runtime registries."""
+import numpy as np
+import pytest
+
+import amgcl_amd as am
+from amgcl_amd.matrix import CSR
+
+
+def test_profiler_scopes_and_report():
+    from amgcl_amd.profiler import Profiler
+
+    p = Profiler("t")
+    with p.scope("outer"):
+        with p.scope("inner"):
+            pass
+    rep = p.report()
+    assert "outer" in rep and "inner" in rep and "[t]" in rep
+
+
+def test_zero_pivot_ilu_raises():
+    """Failure detection: singular pivot aborts the factorization loudly
+    (reference throws in ilu0.hpp when the diagonal vanishes)."""
+    import scipy.sparse as sp
+
+    # 2x2 with a zero pivot after elimination: [[1, 1], [1, 1]]
+    A = CSR.from_scipy(sp.csr_matrix(np.array([[1.0, 1.0], [1.0, 1.0]])))
+    from amgcl_amd import _core
+
+    with pytest.raises(RuntimeError, match="pivot"):
+        _core.ilu0_factor(A.nrows, A.ptr, A.col, A.val)
+
+
+def test_empty_aggregation_raises():
+    """A diagonal matrix has no strong connections: the aggregation must
+    refuse to build an empty level instead of looping."""
+    import scipy.sparse as sp
+
+    from amgcl_amd import _core
+
+    A = CSR.from_scipy(sp.identity(50, format="csr"))
+    with pytest.raises(RuntimeError, match="empty level"):
+        _core.aggregates_parallel(A.nrows, A.ptr, A.col, A.val, 0.08)
+
+
+def test_verbose_solver_prints_iterations(capsys):
+    A, b = am.poisson3d(10, rhs="random")
+    s = am.make_solver(A, {"solver": {"type": "cg", "tol": 1e-8,
+                                      "maxiter": 100, "verbose": True},
+                           "precond": {"class": "amg", "coarse_enough": 200}})
+    s(b)
+    out = capsys.readouterr().out
+    assert out.strip()  # per-iteration residual lines
+
+
+def test_hierarchy_report_complexities():
+    A, _ = am.poisson3d(16)
+    s = am.make_solver(A, {"precond": {"class": "amg", "coarse_enough": 300}})
+    rep = str(s)
+    assert "operator complexity" in rep
+    assert "grid complexity" in rep
+    assert "unknowns" in rep
+
+
+def test_runtime_registries_reject_unknown():
+    A, b = am.poisson3d(8)
+    with pytest.raises(ValueError, match="unknown"):
+        am.make_solver(A, {"solver": {"type": "does_not_exist"}})
+    with pytest.raises(ValueError, match="unknown"):
+        am.make_solver(A, {"precond": {"class": "amg",
+                                       "relax": {"type": "nope"}}})
+
+
+def test_missing_hip_lib_fails_loudly(monkeypatch, tmp_path):
+    """GPU ops must never fall back silently: a missing kernel library is a
+    hard error (round-end 'native code not loaded' guard)."""
+    import amgcl_amd.backend._hiplib as hl
+
+    monkeypatch.setattr(hl, "_LIB", None)
+    monkeypatch.setattr(hl.os.path, "exists", lambda p: False)
+
+    def no_build():
+        return None
+
+    import amgcl_amd.build as bld
+
+    monkeypatch.setattr(bld, "build_hip_lib", no_build)
+    with pytest.raises(RuntimeError, match="libamghip.so not found"):
+        hl.lib()
