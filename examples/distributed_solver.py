@@ -25,6 +25,11 @@ from amgcl_amd.parallel import make_dist_solver
 
 def main():
     n = int(sys.argv[1]) if len(sys.argv) > 1 else 32
+    if "RANK" not in os.environ:
+        print("launch under torch.distributed, e.g.:\n"
+              "  torchrun --nnodes=1 --nproc-per-node 2 --master-addr 127.0.0.1 "
+              "examples/distributed_solver.py")
+        return
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if torch.cuda.is_available():

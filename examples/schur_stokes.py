@@ -28,8 +28,10 @@ def main():
     solve = am.make_solver(
         am.CSR.from_scipy(K.tocsr()),
         {"precond": {"class": "schur_pressure_correction", "pmask_raw": pmask,
+                     "usolver": {"precond": {"class": "amg"},
+                                 "solver": {"type": "preonly"}},
                      "psolver": {"precond": {"class": "amg"},
-                                 "solver": {"type": "cg", "tol": 1e-2, "maxiter": 8}}},
+                                 "solver": {"type": "preonly"}}},
          "solver": {"type": "fgmres", "tol": 1e-8, "maxiter": 200}},
         backend=backend,
     )
