@@ -558,7 +558,7 @@ def _pmis_equality(rank, world):
     return np.concatenate([np.asarray(g) for g in got]).tolist()
 
 
-@pytest.mark.parametrize("world", [1, 2, 3])
+@pytest.mark.parametrize("world", [1, 2, 3, 5])
 def test_pmis_matches_serial(world):
     out = spawn(world, _pmis_equality, 30261 + world)
     import amgcl_amd as am

@@ -140,3 +140,54 @@ def test_merge_params_unknown_keys_raise(known, bogus):
     if bogus not in defaults:
         with pytest.raises(UnknownParameter):
             merge_params(defaults, {bogus: 1})
+
+
+@settings(**COMMON)
+@given(n=st.integers(4, 40), seed=st.integers(0, 10**6))
+def test_ilu0_defining_property(n, seed):
+    """ILU(0)'s defining property: (L U) restricted to A's sparsity pattern
+    equals A there (no drop tolerance, no fill)."""
+    A = rand_csr(n, n, 0.25, seed, ensure_diag=True)
+    Ac = CSR.from_scipy(A)
+    lu, dia = _core.ilu0_factor(Ac.nrows, Ac.ptr, Ac.col, Ac.val)
+    lu, dia = np.asarray(lu), np.asarray(dia)
+    row_of = np.repeat(np.arange(n), np.diff(Ac.ptr))
+    idx = np.arange(len(lu))
+    Lm = sp.coo_matrix(
+        (np.where(idx < dia[row_of], lu, 0.0), (row_of, Ac.col)), shape=(n, n)
+    ).tocsr() + sp.identity(n)
+    uvals = np.where(idx > dia[row_of], lu, 0.0)
+    # the U diagonal is stored inverted for the solve
+    uvals[dia] = 1.0 / lu[dia]
+    Um = sp.coo_matrix((uvals, (row_of, Ac.col)), shape=(n, n)).tocsr()
+    prod = (Lm @ Um).tocsr()
+    # compare on A's pattern only
+    mask = sp.csr_matrix((np.ones(Ac.val.size), Ac.col, Ac.ptr), shape=(n, n))
+    diff = (prod.multiply(mask) - A)
+    assert abs(diff).max() < 1e-9 * max(1.0, abs(A).max())
+
+
+@settings(**COMMON)
+@given(n=st.integers(6, 60), k=st.integers(1, 3), seed=st.integers(0, 10**6))
+def test_nullspace_tentative_orthonormal(n, k, seed):
+    """Per-aggregate QR tentative prolongation: P's columns are orthonormal
+    (P^T P = I) and span the nullspace block (P Bnew = B) — the defining
+    properties of tentative_prolongation.hpp's nullspace route."""
+    rng = np.random.default_rng(seed)
+    # synthetic aggregation: contiguous groups of >= k nodes
+    ids = np.sort(rng.integers(0, max(2, n // (2 * k)), size=n)).astype(np.int32)
+    # renumber to consecutive ids
+    _, ids = np.unique(ids, return_inverse=True)
+    ids = ids.astype(np.int32)
+    naggr = int(ids.max()) + 1
+    # ensure every aggregate has at least k members (QR needs full rank)
+    counts = np.bincount(ids, minlength=naggr)
+    if counts.min() < k:
+        return  # hypothesis will try other draws
+    B = rng.standard_normal((n, k))
+    tp, tc, tv, Bnew = _core.tentative_nullspace(n, ids, naggr, B, k)
+    P = CSR(n, naggr * k, tp, tc, tv).to_scipy()
+    G = (P.T @ P).toarray()
+    np.testing.assert_allclose(G, np.eye(naggr * k), atol=1e-10)
+    Bnew = np.asarray(Bnew).reshape(naggr * k, k)
+    np.testing.assert_allclose(P @ Bnew, B, atol=1e-9)
