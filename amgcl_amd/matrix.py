@@ -20,7 +20,11 @@ class CSR:
         self.ncols = int(ncols)
         self.ptr = np.ascontiguousarray(ptr, dtype=np.int32)
         self.col = np.ascontiguousarray(col, dtype=np.int32)
-        self.val = np.ascontiguousarray(val, dtype=np.float64)
+        # complex values are only carried by IO/adapters until
+        # adapter.complex_to_real expands them; casting them to fp64 here
+        # would silently drop the imaginary part
+        dt = np.complex128 if np.iscomplexobj(np.asarray(val)) else np.float64
+        self.val = np.ascontiguousarray(val, dtype=dt)
 
     @property
     def nnz(self):

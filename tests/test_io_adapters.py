@@ -207,3 +207,31 @@ def test_cli_end_to_end(tmp_path):
     assert r.returncode == 0
     B = io.read_crs(str(tmp_path / "A.bin"))
     assert B.nnz == A.nnz
+
+
+def test_mm_pattern_integer_complex(tmp_path):
+    """MatrixMarket field variants (reference io/mm.hpp handles real,
+    integer, pattern, and complex banners)."""
+    from amgcl_amd import io
+
+    p = tmp_path / "pat.mtx"
+    p.write_text(
+        "%%MatrixMarket matrix coordinate pattern general\n"
+        "3 3 4\n1 1\n2 2\n3 3\n1 3\n")
+    A = io.mm_read(str(p))
+    assert A.nnz == 4 and A.val[0] == 1.0
+
+    q = tmp_path / "int.mtx"
+    q.write_text(
+        "%%MatrixMarket matrix coordinate integer general\n"
+        "2 2 3\n1 1 4\n2 2 5\n1 2 -3\n")
+    B = io.mm_read(str(q))
+    assert B.to_scipy()[0, 1] == -3.0 and B.to_scipy()[1, 1] == 5.0
+
+    c = tmp_path / "cplx.mtx"
+    c.write_text(
+        "%%MatrixMarket matrix coordinate complex general\n"
+        "2 2 2\n1 1 1.0 2.0\n2 2 3.0 -1.0\n")
+    C = io.mm_read(str(c))
+    assert np.iscomplexobj(np.asarray(C.val))
+    assert np.asarray(C.val)[0] == 1.0 + 2.0j
