@@ -191,3 +191,81 @@ def test_nullspace_tentative_orthonormal(n, k, seed):
     np.testing.assert_allclose(G, np.eye(naggr * k), atol=1e-10)
     Bnew = np.asarray(Bnew).reshape(naggr * k, k)
     np.testing.assert_allclose(P @ Bnew, B, atol=1e-9)
+
+
+@settings(**COMMON)
+@given(n=st.integers(4, 40), b=st.sampled_from([2, 3, 4]),
+       seed=st.integers(0, 10**6))
+def test_csr_to_bsr_preserves_matrix(n, b, seed):
+    A = rand_csr(n * b, n * b, 0.1, seed, ensure_diag=True)
+    Ac = CSR.from_scipy(A)
+    bp, bc, bv = _core.csr_to_bsr(Ac.nrows, Ac.ptr, Ac.col, Ac.val, b)
+    bp, bc = np.asarray(bp), np.asarray(bc)
+    bv = np.asarray(bv).reshape(-1, b, b)
+    dense = np.zeros((n * b, n * b))
+    for i in range(n):
+        for j in range(bp[i], bp[i + 1]):
+            c = bc[j]
+            dense[i * b:(i + 1) * b, c * b:(c + 1) * b] = bv[j]
+    np.testing.assert_allclose(dense, A.toarray())
+
+
+@settings(**COMMON)
+@given(n=st.integers(4, 60), seed=st.integers(0, 10**6))
+def test_color_graph_is_proper(n, seed):
+    """JP coloring: adjacent rows never share a color (the invariant the
+    multicolor Gauss-Seidel sweeps rely on)."""
+    A = rand_csr(n, n, 0.15, seed, ensure_diag=True)
+    A = (A + A.T).tocsr()  # symmetric adjacency
+    A.sort_indices()
+    Ac = CSR.from_scipy(A)
+    colors, ncolors = _core.color_graph(Ac.nrows, Ac.ptr, Ac.col)
+    colors = np.asarray(colors)
+    assert ncolors >= 1 and colors.min() >= 0 and colors.max() < ncolors
+    row_of = np.repeat(np.arange(n), np.diff(Ac.ptr))
+    off = row_of != Ac.col
+    assert not np.any(colors[row_of[off]] == colors[np.asarray(Ac.col)[off]])
+
+
+@settings(**COMMON)
+@given(n=st.integers(4, 50), seed=st.integers(0, 10**6))
+def test_tri_levels_respect_dependencies(n, seed):
+    A = rand_csr(n, n, 0.25, seed, ensure_diag=True)
+    Ac = CSR.from_scipy(A)
+    lu, dia = _core.ilu0_factor(Ac.nrows, Ac.ptr, Ac.col, Ac.val)
+    dia32 = np.ascontiguousarray(dia, dtype=np.int32)
+    for lower in (True, False):
+        lp, rows = _core.tri_levels(Ac.nrows, Ac.ptr, Ac.col, dia32, lower)
+        lp, rows = np.asarray(lp), np.asarray(rows)
+        assert sorted(rows.tolist()) == list(range(n))
+        level_of = np.empty(n, dtype=int)
+        for l in range(len(lp) - 1):
+            level_of[rows[lp[l]:lp[l + 1]]] = l
+        ptr, col = np.asarray(Ac.ptr), np.asarray(Ac.col)
+        for i in range(n):
+            if lower:
+                deps = col[ptr[i]:dia32[i]]
+            else:
+                deps = col[dia32[i] + 1:ptr[i + 1]]
+            assert np.all(level_of[deps] < level_of[i])
+
+
+@settings(**COMMON)
+@given(n=st.integers(10, 200), seed=st.integers(0, 10**6))
+def test_aggregates_partition_validity(n, seed):
+    """Greedy and MIS aggregation: every id is -2 (isolated) or a valid
+    compact aggregate number; every aggregate is nonempty."""
+    A = rand_csr(n, n, 0.1, seed, ensure_diag=True)
+    A = (A + A.T).tocsr()
+    A.sort_indices()
+    Ac = CSR.from_scipy(A)
+    for agg in (_core.aggregates, _core.aggregates_parallel):
+        try:
+            naggr, ids, strong = agg(Ac.nrows, Ac.ptr, Ac.col, Ac.val, 0.08)
+        except RuntimeError:
+            continue  # empty level: allowed for diagonal-dominant randoms
+        ids = np.asarray(ids)
+        assert naggr >= 1
+        valid = ids[ids >= 0]
+        assert valid.size and valid.max() < naggr
+        assert len(np.unique(valid)) == naggr  # every aggregate nonempty
