@@ -616,3 +616,34 @@ def test_dist_amg_cross_rank_pmis(world):
                            "precond": {"class": "amg", "coarse_enough": 300}})
     _, it_serial, _ = s(b)
     assert iters <= it_serial + 4  # cross-rank hierarchy ~= serial quality
+
+
+def _dist_amg_jacobi_wcycle(rank, world):
+    import amgcl_amd as am
+    from amgcl_amd.parallel import make_dist_solver
+
+    strip, b, rb, re_ = am.poisson3d_strip(16, rank, world, rhs="ones")
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "dist_amg", "coarse_enough": 200,
+                     "repart_threshold": 400, "npre": 2, "npost": 2,
+                     "ncycle": 2, "relax": {"type": "damped_jacobi",
+                                            "damping": 0.72}},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+        backend="cpu")
+    x, iters, resid = solve(b)
+    xg = solve.gather_solution(x)
+    return iters, resid, None if xg is None else xg.tolist()
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_amg_jacobi_wcycle(world):
+    """DistAMG parameter space: damped-Jacobi smoothing, npre/npost=2, W-cycle."""
+    out = spawn(world, _dist_amg_jacobi_wcycle, 30361 + world)
+    iters, resid, xg = out[0]
+    assert resid < 1e-8
+    import amgcl_amd as am
+
+    A, b = am.poisson3d(16, rhs="ones")
+    x = np.asarray(xg)
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-7

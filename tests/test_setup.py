@@ -123,3 +123,36 @@ def test_splu_coarse_solver():
     _, it_splu, r_splu = am.make_solver(A, prm)(b)
     assert r_splu < 1e-8
     assert it_splu == it_dense
+
+
+def test_pre_cycles_and_tuple_input():
+    """pre_cycles > 1 (multiple cycles per apply) and the crs_tuple input
+    adapter (reference adapter/crs_tuple.hpp)."""
+    import amgcl_amd as am
+
+    A, b = am.poisson3d(12, rhs="random")
+    s2 = am.make_solver(
+        (A.ptr, A.col, A.val),  # tuple input
+        {"precond": {"class": "amg", "pre_cycles": 2, "coarse_enough": 200},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}})
+    x, it2, r2 = s2(b)
+    assert r2 < 1e-8
+    s1 = am.make_solver(A, {"precond": {"class": "amg", "coarse_enough": 200},
+                            "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}})
+    _, it1, _ = s1(b)
+    assert it2 <= it1  # the stronger preconditioner cannot need more iterations
+
+
+def test_lagged_preconditioner():
+    """Solve with a matrix different from the setup matrix (reference
+    make_solver.hpp:116 — time stepping with a lagged hierarchy)."""
+    import amgcl_amd as am
+    import numpy as np
+
+    A, b = am.poisson3d(12, rhs="random")
+    s = am.make_solver(A, {"solver": {"type": "cg", "tol": 1e-8, "maxiter": 200},
+                           "precond": {"class": "amg", "coarse_enough": 200}})
+    A2 = am.CSR(A.nrows, A.ncols, A.ptr, A.col, 1.3 * np.asarray(A.val))
+    x, iters, resid = s(b, A=A2)
+    assert resid < 1e-8
+    assert np.linalg.norm(b - A2 @ x) / np.linalg.norm(b) < 1e-7
