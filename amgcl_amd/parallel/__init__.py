@@ -16,6 +16,8 @@ MI355X-native re-design of the reference's MPI layer (amgcl/mpi/):
     decoupled or cross-rank pmis aggregation, replicated-tail repartition).
   - mpi::schur_pressure_correction / mpi::cpr -> DistSchurPressureCorrection
     / DistCPR (fully-coupled distributed field splits).
+  - mpi::partition (ptscotch/parmetis class) -> partition.rcb_partition
+    (geometric recursive coordinate bisection).
 """
 from .dist_matrix import DistMatrix
 from .dist_backend import DistBackend, DistInnerProduct
@@ -24,3 +26,4 @@ from .solver import make_dist_solver, DistSolver
 from .dist_amg import DistAMG
 from .schur import DistSchurPressureCorrection
 from .cpr import DistCPR
+from .partition import rcb_partition, partition_permutation
