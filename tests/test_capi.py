@@ -139,3 +139,28 @@ def test_capi_iteration_parity_with_python(capi):
     capi.amgcl_amd_solver_destroy(s)
     assert res.value < 1e-8
     assert abs(it.value - it_py) <= 5
+
+
+def test_capi_cycle_and_relax_params(capi):
+    """Parameter plumbing: W-cycle, extra sweeps, damped-Jacobi relaxation
+    through the params handle."""
+    A, b = am.poisson3d(14, rhs="random")
+    ptr, col, val = _arrays(A)
+    prm = capi.amgcl_amd_params_create()
+    capi.amgcl_amd_params_seti(prm, b"precond.npre", 2)
+    capi.amgcl_amd_params_seti(prm, b"precond.npost", 2)
+    capi.amgcl_amd_params_seti(prm, b"precond.ncycle", 2)
+    capi.amgcl_amd_params_sets(prm, b"precond.relax.type", b"damped_jacobi")
+    capi.amgcl_amd_params_setf(prm, b"precond.relax.damping", 0.72)
+    capi.amgcl_amd_params_seti(prm, b"precond.coarse_enough", 300)
+    capi.amgcl_amd_params_setf(prm, b"solver.tol", 1e-8)
+    s = capi.amgcl_amd_solver_create(A.nrows, _ptr(ptr), _ptr(col), _ptr(val), prm)
+    capi.amgcl_amd_params_destroy(prm)
+    x = np.zeros(A.nrows)
+    it = ctypes.c_int(0)
+    res = ctypes.c_double(0.0)
+    rc = capi.amgcl_amd_solver_solve(s, _ptr(b), _ptr(x), ctypes.byref(it),
+                                     ctypes.byref(res))
+    capi.amgcl_amd_solver_destroy(s)
+    assert rc == 0 and res.value < 1e-8
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-7
