@@ -157,6 +157,44 @@ static py::tuple poisson3d_strip(i64 n, i64 row_beg, i64 row_end) {
     return py::make_tuple(ptr, col, val);
 }
 
+// Box-grid variant (nx x ny x nz) of the strip generator: the weak-scaling
+// fixture (per-rank cube, domain elongated along z as ranks are added —
+// reference benchmarks.rst weak-scaling setup).
+static py::tuple poisson3d_box_strip(i64 nx, i64 ny, i64 nz, i64 row_beg, i64 row_end) {
+    const i64 nloc = row_end - row_beg;
+    const double dia = 6.0;
+    arr<i32> ptr(nloc + 1);
+    i32 *P = ptr.mutable_data();
+    P[0] = 0;
+#pragma omp parallel for schedule(static)
+    for (i64 r = 0; r < nloc; ++r) {
+        i64 idx = row_beg + r;
+        i64 i = idx % nx, j = (idx / nx) % ny, k = idx / (nx * ny);
+        P[r + 1] = 1 + (k > 0) + (j > 0) + (i > 0) + (i + 1 < nx) + (j + 1 < ny) +
+                   (k + 1 < nz);
+    }
+    scan_i32(P + 1, nloc);
+    const i64 nnz = P[nloc];
+    arr<i32> col = big_arr<i32>(nnz);
+    arr<double> val = big_arr<double>(nnz);
+    i32 *C = col.mutable_data();
+    double *V = val.mutable_data();
+#pragma omp parallel for schedule(static)
+    for (i64 r = 0; r < nloc; ++r) {
+        i64 idx = row_beg + r;
+        i64 i = idx % nx, j = (idx / nx) % ny, k = idx / (nx * ny);
+        i64 h = P[r];
+        if (k > 0)      { C[h] = (i32)(idx - nx * ny); V[h] = -1.0; ++h; }
+        if (j > 0)      { C[h] = (i32)(idx - nx);      V[h] = -1.0; ++h; }
+        if (i > 0)      { C[h] = (i32)(idx - 1);       V[h] = -1.0; ++h; }
+        C[h] = (i32)idx; V[h] = dia; ++h;
+        if (i + 1 < nx) { C[h] = (i32)(idx + 1);       V[h] = -1.0; ++h; }
+        if (j + 1 < ny) { C[h] = (i32)(idx + nx);      V[h] = -1.0; ++h; }
+        if (k + 1 < nz) { C[h] = (i32)(idx + nx * ny); V[h] = -1.0; ++h; }
+    }
+    return py::make_tuple(ptr, col, val);
+}
+
 // Split a distributed row strip (global columns) into the square local part
 // (columns renumbered to local) and the remote part over deduplicated ghost
 // columns (parity: amgcl/mpi/distributed_matrix.hpp:370-430).
@@ -2084,6 +2122,7 @@ PYBIND11_MODULE(_core, m) {
     m.doc() = "amgcl_amd host setup engine (OpenMP)";
     m.def("poisson3d", &poisson3d, py::arg("n"), py::arg("anisotropy") = 1.0);
     m.def("poisson3d_strip", &poisson3d_strip);
+    m.def("poisson3d_box_strip", &poisson3d_box_strip);
     m.def("split_strip", &split_strip);
     m.def("diagonal", &diagonal);
     m.def("transpose", &transpose);

@@ -23,6 +23,24 @@ def poisson3d(n, anisotropy=1.0, rhs="ones"):
     return A, b
 
 
+def poisson3d_box_strip(nx, ny, nz, rank, nranks, rhs="ones"):
+    """Row strip of an nx*ny*nz 7-point Poisson box, cut on z-planes when the
+    grid divides evenly — the weak-scaling fixture (fixed per-rank cube,
+    domain grown along z with the rank count)."""
+    ntot = nx * ny * nz
+    row_beg = (ntot * rank) // nranks
+    row_end = (ntot * (rank + 1)) // nranks
+    ptr, col, val = _core.poisson3d_box_strip(int(nx), int(ny), int(nz),
+                                              int(row_beg), int(row_end))
+    strip = CSR(row_end - row_beg, ntot, ptr, col, val)
+    if rhs == "ones":
+        b = np.ones(row_end - row_beg, dtype=np.float64)
+    else:
+        rng = np.random.default_rng(1234 + rank)
+        b = rng.standard_normal(row_end - row_beg)
+    return strip, b, row_beg, row_end
+
+
 def poisson3d_strip(n, rank, nranks, rhs="ones"):
     """Row strip [row_beg, row_end) of the global n^3 Poisson matrix, for
     distributed tests (parity: examples/mpi/mpi_solver.cpp:47 assembles the

@@ -34,6 +34,9 @@ def parse_args():
     p.add_argument("--maxiter", type=int, default=300)
     p.add_argument("--backend", default=None, choices=[None, "hip", "cpu"])
     p.add_argument("--precond", default=None, help="override precond JSON")
+    p.add_argument("--weak", action="store_true",
+                   help="weak scaling: size^3 unknowns PER RANK (domain grows "
+                        "along z); default is strong scaling at size^3 total")
     return p.parse_args()
 
 
@@ -99,17 +102,30 @@ def run_distributed(args, dist_ctx):
     from amgcl_amd.parallel import make_dist_solver
 
     rank, world = dist_ctx
-    A_strip, _, row_beg, row_end = am.poisson3d_strip(args.size, rank, world, rhs=None)
-    rng = np.random.default_rng(42)
-    b_global = rng.standard_normal(args.size**3)
-    b = b_global[row_beg:row_end]
+    n = args.size
+    if args.weak:
+        # fixed size^3 unknowns per rank, domain elongated along z
+        # (BASELINE.md weak-scaling table shape: ~const work per core)
+        nz = n * world
+        from amgcl_amd.generators import poisson3d_box_strip
+
+        A_strip, _, row_beg, row_end = poisson3d_box_strip(n, n, nz, rank, world,
+                                                           rhs=None)
+        rng = np.random.default_rng(42 + rank)
+        b = rng.standard_normal(row_end - row_beg)
+    else:
+        nz = n
+        A_strip, _, row_beg, row_end = am.poisson3d_strip(n, rank, world, rhs=None)
+        rng = np.random.default_rng(42)
+        b_global = rng.standard_normal(n**3)
+        b = b_global[row_beg:row_end]
 
     backend_name = args.backend or ("hip" if _has_gpu() else "cpu")
     # linear subdomain deflation, the reference's flagship distributed config
     # (MN4 strong scaling uses SDD linear deflation)
     idx = np.arange(row_beg, row_end)
     coords = np.stack(
-        [idx % args.size, (idx // args.size) % args.size, idx // (args.size**2)],
+        [idx % n, (idx // n) % n, idx // (n * n)],
         axis=1,
     ).astype(np.float64)
     pprm = json.loads(args.precond) if args.precond else {"class": "amg"}
@@ -134,12 +150,20 @@ def run_distributed(args, dist_ctx):
     t2 = time.perf_counter()
     # true residual check against the full operator (rank 0)
     xh = solve.gather_solution(x)
+    if args.weak:
+        bs = [None] * world
+        dist.all_gather_object(bs, np.asarray(b))
     true_rel = -1.0
     if xh is not None:
-        A_full, _ = am.poisson3d(args.size)
-        true_rel = float(
-            np.linalg.norm(b_global - A_full @ xh) / np.linalg.norm(b_global)
-        )
+        if args.weak:
+            from amgcl_amd.generators import poisson3d_box_strip
+
+            A_full, _, _, _ = poisson3d_box_strip(n, n, nz, 0, 1)
+            bg = np.concatenate(bs)
+        else:
+            A_full, _ = am.poisson3d(n)
+            bg = b_global
+        true_rel = float(np.linalg.norm(bg - A_full @ xh) / np.linalg.norm(bg))
     return t0, t1, t2, iters, resid, true_rel
 
 
@@ -201,8 +225,11 @@ def main():
 
     if rank == 0:
         value = total / args.steps
+        weak = bool(args.weak and world > 1)
+        metric = (f"setup+solve time (s) & iterations to 1e-6, 3D Poisson "
+                  f"{args.size}^3" + ("/GPU (weak)" if weak else ""))
         out = {
-            "metric": f"setup+solve time (s) & iterations to 1e-6, 3D Poisson {args.size}^3",
+            "metric": metric,
             "value": value,
             "unit": "s",
             "n_gpus": world,
@@ -210,13 +237,13 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": value * 1000.0,
             "higher_is_better": False,
-            "scaling": "strong",
-            "vs_baseline": value / BASELINE_TOTAL_S,
+            "scaling": "weak" if weak else "strong",
+            "vs_baseline": None if weak else value / BASELINE_TOTAL_S,
             "dtype": "fp64",
             "data": "synthetic 7-pt Poisson, random RHS",
             "config": {
                 "model": f"3D Poisson {args.size}^3 (7-point stencil)",
-                "unknowns": args.size**3,
+                "unknowns": args.size**3 * (world if weak else 1),
                 "solver": args.solver,
                 "precond": "smoothed_aggregation AMG + SPAI0",
                 "tol": args.tol,

@@ -35,6 +35,7 @@ def _run_bench(rank, world, port, argv, results):
     (4, []),
     (2, ["--precond", '{"class": "dist_amg", "coarse_enough": 200,'
                       ' "repart_threshold": 400}']),
+    (2, ["--weak"]),
 ])
 def test_bench_distributed_contract(world, extra):
     ctx = mp.get_context("spawn")
@@ -58,15 +59,20 @@ def test_bench_distributed_contract(world, extra):
     for r in range(1, world):
         assert out[r].strip() == ""
     d = json.loads(lines[0])
+    weak = "--weak" in extra
     assert d["n_gpus"] == world
     assert d["steps"] == 2 and d["warmup"] == 1
-    assert d["scaling"] == "strong"
+    assert d["scaling"] == ("weak" if weak else "strong")
     assert d["dtype"] == "fp64"
     assert d["higher_is_better"] is False
     assert d["unit"] == "s"
     assert abs(d["ms_per_step"] - d["value"] * 1000) < 1e-9
-    assert abs(d["vs_baseline"] - d["value"] / 2.03) < 1e-12
-    assert d["config"]["unknowns"] == 24 ** 3
+    if weak:
+        assert d["vs_baseline"] is None
+        assert d["config"]["unknowns"] == 24 ** 3 * world
+    else:
+        assert abs(d["vs_baseline"] - d["value"] / 2.03) < 1e-12
+        assert d["config"]["unknowns"] == 24 ** 3
     assert d["config"]["true_rel_resid"] < 1e-6
     assert "Poisson 24^3" in d["metric"]
 
