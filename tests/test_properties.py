@@ -1,6 +1,8 @@
 """Property-based tests (hypothesis): invariants that must hold for ANY
 valid input, not just the fixtures — IO roundtrips, adapter involutions,
 host kernel algebra vs scipy, parameter-tree validation."""
+import os
+
 import numpy as np
 import pytest
 import scipy.sparse as sp
@@ -10,7 +12,9 @@ import amgcl_amd as am
 from amgcl_amd import _core
 from amgcl_amd.matrix import CSR
 
-COMMON = dict(deadline=None, max_examples=25)
+# AMGCL_HYP_EXAMPLES=500 pytest tests/test_properties.py  -> deeper fuzz
+COMMON = dict(deadline=None,
+              max_examples=int(os.environ.get("AMGCL_HYP_EXAMPLES", "25")))
 
 
 def rand_csr(n, m, density, seed, ensure_diag=False):
