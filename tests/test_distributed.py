@@ -723,3 +723,60 @@ def test_rcb_partitioned_distributed_solve(world):
     x = np.empty(n ** 3)
     x[perm] = np.asarray(xg)  # un-permute
     assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-7
+
+
+def _solve_field_blocked_schur(rank, world):
+    """Extreme split: the global ordering is FIELD-blocked, so one rank owns
+    only velocity rows and another only pressure rows (zero-row field blocks
+    in the rectangular distributed machinery)."""
+    import scipy.sparse as sp
+
+    import amgcl_amd as am
+    from amgcl_amd.matrix import CSR
+    from amgcl_amd.parallel import make_dist_solver
+
+    n = 8
+    Ap, _ = am.poisson3d(n)
+    a = Ap.to_scipy()
+    nv = a.shape[0]
+    B = 0.1 * (sp.identity(nv) - sp.diags(np.ones(nv - 1), 1)).tocsr()
+    K = sp.bmat([[a, B], [B.T, a + sp.identity(nv)]], format="csr")
+    K.sort_indices()
+    n_glob = 2 * nv
+    row_beg = (n_glob * rank) // world
+    row_end = (n_glob * (rank + 1)) // world
+    lo, hi = K.indptr[row_beg], K.indptr[row_end]
+    strip = CSR(row_end - row_beg, n_glob,
+                K.indptr[row_beg:row_end + 1] - lo, K.indices[lo:hi], K.data[lo:hi])
+    pmask_local = np.arange(row_beg, row_end) >= nv
+    b = np.random.default_rng(1).standard_normal(n_glob)
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "schur_pressure_correction", "pmask_raw": pmask_local,
+                     "psolver": {"precond": {"class": "amg", "coarse_enough": 200},
+                                 "solver": {"type": "preonly"}}},
+         "solver": {"type": "fgmres", "tol": 1e-8, "maxiter": 300}},
+        backend="cpu")
+    x, iters, resid = solve(b[row_beg:row_end])
+    xg = solve.gather_solution(x)
+    return iters, resid, None if xg is None else xg.tolist()
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_schur_field_blocked_ordering(world):
+    import scipy.sparse as sp
+
+    import amgcl_amd as am
+
+    out = spawn(world, _solve_field_blocked_schur, 30461 + world)
+    iters, resid, xg = out[0]
+    assert resid < 1e-7
+    n = 8
+    Ap, _ = am.poisson3d(n)
+    a = Ap.to_scipy()
+    nv = a.shape[0]
+    B = 0.1 * (sp.identity(nv) - sp.diags(np.ones(nv - 1), 1)).tocsr()
+    K = sp.bmat([[a, B], [B.T, a + sp.identity(nv)]], format="csr")
+    b = np.random.default_rng(1).standard_normal(2 * nv)
+    x = np.asarray(xg)
+    assert np.linalg.norm(b - K @ x) / np.linalg.norm(b) < 1e-6
