@@ -289,3 +289,35 @@ def test_multicolor_cpu_gauss_seidel():
                "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}}
     _, it_ser, _ = am.make_solver(A, prm_ser)(b)
     assert it1 <= it_ser + 4  # colored ordering stays in the serial class
+
+
+def test_ns_search_finds_nullspace_vector():
+    """ns_search (reference tests/test_solver_ns_builtin.cpp): with a zero
+    right-hand side and a nonzero initial guess, the solver iterates toward a
+    null-space vector of a singular operator instead of returning x = 0."""
+    import scipy.sparse as sp
+
+    from amgcl_amd.matrix import CSR
+
+    # singular pure-Neumann graph Laplacian on a 1D chain: nullspace = const
+    n = 200
+    main = np.full(n, 2.0)
+    main[0] = main[-1] = 1.0
+    L = sp.diags([np.full(n - 1, -1.0), main, np.full(n - 1, -1.0)],
+                 [-1, 0, 1]).tocsr()
+    L.sort_indices()
+    A = CSR.from_scipy(L)
+    rng = np.random.default_rng(3)
+    x0 = rng.standard_normal(n)
+    s = am.make_solver(
+        A, {"precond": {"class": "relaxation", "type": "damped_jacobi"},
+            "solver": {"type": "cg", "tol": 1e-10, "maxiter": 2000,
+                       "ns_search": True}})
+    x, iters, resid = s(np.zeros(n), x=x0.copy())
+    x = np.asarray(x)
+    assert np.linalg.norm(x) > 1e-6          # NOT the trivial solution
+    assert np.linalg.norm(L @ x) < 1e-6 * np.linalg.norm(x)  # in the nullspace
+    # the found vector is (up to scale) the constant vector
+    xn = x / np.linalg.norm(x)
+    const = np.ones(n) / np.sqrt(n)
+    assert min(np.linalg.norm(xn - const), np.linalg.norm(xn + const)) < 1e-4
