@@ -780,3 +780,52 @@ def test_dist_schur_field_blocked_ordering(world):
     b = np.random.default_rng(1).standard_normal(2 * nv)
     x = np.asarray(xg)
     assert np.linalg.norm(b - K @ x) / np.linalg.norm(b) < 1e-6
+
+
+def _solve_dist_elasticity(rank, world):
+    """Distributed elasticity with per-rank rigid-body nullspace (the
+    reference's NullspaceMPI tutorial shape: strips cut on node boundaries,
+    RBMs from the local coordinates)."""
+    import amgcl_amd as am
+    from amgcl_amd.generators import elasticity3d, rigid_body_modes
+    from amgcl_amd.matrix import CSR
+    from amgcl_amd.parallel import make_dist_solver
+
+    A, b, coords = elasticity3d(10)
+    n = A.nrows
+    nnodes = n // 3
+    node_beg = (nnodes * rank) // world
+    node_end = (nnodes * (rank + 1)) // world
+    row_beg, row_end = node_beg * 3, node_end * 3
+    m = A.to_scipy()
+    lo, hi = m.indptr[row_beg], m.indptr[row_end]
+    strip = CSR(row_end - row_beg, n, m.indptr[row_beg:row_end + 1] - lo,
+                m.indices[lo:hi], m.data[lo:hi])
+    B_loc = rigid_body_modes(coords[node_beg:node_end])
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "amg", "coarse_enough": 300,
+                     "coarsening": {"type": "smoothed_aggregation",
+                                    "block_size": 3, "nullspace_raw": B_loc,
+                                    "estimate_spectral_radius": True,
+                                    "power_iters": 10},
+                     "relax": {"type": "chebyshev"}},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 300}},
+        backend="cpu")
+    x, iters, resid = solve(b[row_beg:row_end])
+    xg = solve.gather_solution(x)
+    return iters, resid, None if xg is None else xg.tolist()
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_elasticity_nullspace(world):
+    out = spawn(world, _solve_dist_elasticity, 30511 + world)
+    import amgcl_amd as am
+    from amgcl_amd.generators import elasticity3d
+
+    iters, resid, xg = out[0]
+    assert resid < 1e-8
+    assert iters < 120
+    A, b, _ = elasticity3d(10)
+    x = np.asarray(xg)
+    assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-7
