@@ -115,7 +115,12 @@ Csr spgemm(const Csr &A, const Csr &B) {
     C.val.resize(C.ptr[C.n]);
 #pragma omp parallel
     {
+        // Position marker plus a row-ownership marker: the position test alone
+        // (marker[cb] < beg) is only safe when each thread visits rows in
+        // increasing order, which OpenMP>=5 dynamic schedules do not guarantee
+        // (non-monotonic by default).  marker_row pins validity to this row.
         std::vector<int> marker(B.m, -1);
+        std::vector<int> marker_row(B.m, -1);
 #pragma omp for schedule(dynamic, 256)
         for (int i = 0; i < A.n; ++i) {
             int beg = C.ptr[i], end = beg;
@@ -124,7 +129,8 @@ Csr spgemm(const Csr &A, const Csr &B) {
                 double va = A.val[ja];
                 for (int jb = B.ptr[ca]; jb < B.ptr[ca + 1]; ++jb) {
                     int cb = B.col[jb];
-                    if (marker[cb] < beg) {
+                    if (marker_row[cb] != i) {
+                        marker_row[cb] = i;
                         marker[cb] = end;
                         C.col[end] = cb;
                         C.val[end] = va * B.val[jb];

@@ -686,7 +686,12 @@ static py::tuple aggregates_parallel(i64 nrows, arr<i32> ptr, arr<i32> col,
 
         // pass 3: new roots claim their strong 1-ring FIRM, overwriting
         // provisional members (greedy: "later claimed by other aggregates").
-        // No conflicts: two new roots are never within distance 2.
+        // With a *symmetric* strength mask two new roots are never within
+        // distance 2 and claims cannot conflict; the mask is row-wise, so for
+        // nonsymmetric matrices two roots CAN claim the same node.  Resolve
+        // claims deterministically with an atomic max on the root key
+        // (m1 is free after pass 2): clear -> max -> commit, so the winner is
+        // independent of thread count / schedule.
 #pragma omp parallel for schedule(static)
         for (i64 t = 0; t < na; ++t) {
             i32 u = active[t];
@@ -694,7 +699,30 @@ static py::tuple aggregates_parallel(i64 nrows, arr<i32> ptr, arr<i32> col,
             for (i32 j = A.ptr[u]; j < A.ptr[u + 1]; ++j) {
                 i32 c = A.col[j];
                 if (!S[j]) continue;
-                if (id[c] == UNDEF || prov[c]) {
+                if (id[c] == UNDEF || prov[c]) m1[c].store(0, std::memory_order_relaxed);
+            }
+        }
+#pragma omp parallel for schedule(static)
+        for (i64 t = 0; t < na; ++t) {
+            i32 u = active[t];
+            if (id[u] != u) continue;
+            uint64_t key = agg_key(u);
+            for (i32 j = A.ptr[u]; j < A.ptr[u + 1]; ++j) {
+                i32 c = A.col[j];
+                if (!S[j]) continue;
+                if (id[c] == UNDEF || prov[c]) atomic_max_u64(m1[c], key);
+            }
+        }
+#pragma omp parallel for schedule(static)
+        for (i64 t = 0; t < na; ++t) {
+            i32 u = active[t];
+            if (id[u] != u) continue;
+            uint64_t key = agg_key(u);
+            for (i32 j = A.ptr[u]; j < A.ptr[u + 1]; ++j) {
+                i32 c = A.col[j];
+                if (!S[j]) continue;
+                if ((id[c] == UNDEF || prov[c]) &&
+                    m1[c].load(std::memory_order_relaxed) == key) {
                     id[c] = u;
                     prov[c] = 0;
                 }

@@ -213,7 +213,9 @@ class AMG:
                 if not isinstance(lvl.relax, DiagonalSmootherBase):
                     raise ValueError("mixed precision supports diagonal smoothers")
                 lvl.relax.M = lvl.relax.M.to(f32)
-        if self.coarse_solve is not None:
+        if self.coarse_solve is not None and hasattr(self.coarse_solve, "inv"):
+            # HostSpluSolver (direct_solver="splu") has no dense inverse to
+            # convert; its factorized solve simply stays fp64.
             self.coarse_solve.inv = self.coarse_solve.inv.to(f32)
         n0 = self.levels[0].rows
         self._r32 = self.backend.vector(n0, f32)

@@ -97,3 +97,19 @@ def test_shared_memory_deflation():
     x, it, res = s(b)
     assert res < 1e-7
     assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-6
+
+
+def test_cpr_quasi_impes_weights():
+    """Quasi-IMPES decoupling (amgcl/preconditioner/cpr.hpp): the per-cell
+    weights are the first row of the inverted diagonal block, so w^T D = e_p^T
+    and the non-pressure couplings cancel within each cell."""
+    from amgcl_amd.precond.cpr import CPR
+
+    K = block_reservoir(6)
+    m = K.to_scipy().tocsr()
+    b = 2
+    nc = K.nrows // b
+    w = CPR._quasi_impes_weights(m, nc, b).reshape(nc, b)
+    for i in [0, 3, nc - 1]:
+        D = m[i * b : (i + 1) * b, i * b : (i + 1) * b].toarray()
+        np.testing.assert_allclose(w[i] @ D, np.eye(b)[0], atol=1e-10)

@@ -235,3 +235,22 @@ def test_mm_pattern_integer_complex(tmp_path):
     C = io.mm_read(str(c))
     assert np.iscomplexobj(np.asarray(C.val))
     assert np.asarray(C.val)[0] == 1.0 + 2.0j
+
+
+def test_binary_reference_layout(tmp_path):
+    """Files in the upstream amgcl binary layout (size_t n, int64 ptr/col,
+    float64 val, no magic — amgcl/io/binary.hpp, examples/mm2bin.cpp) load
+    directly, including strip reads."""
+    rng = np.random.default_rng(3)
+    A = rand_csr(rng, 40)
+    p = str(tmp_path / "ref.bin")
+    with open(p, "wb") as f:
+        np.array([A.nrows], dtype=np.int64).tofile(f)
+        A.ptr.astype(np.int64).tofile(f)
+        A.col.astype(np.int64).tofile(f)
+        A.val.astype(np.float64).tofile(f)
+    B = io.read_crs(p)
+    assert abs(B.to_scipy() - A.to_scipy()).max() < 1e-15
+    S = io.read_crs(p, row_beg=5, row_end=25)
+    assert S.nrows == 20
+    assert abs(S.to_scipy() - A.to_scipy()[5:25]).max() < 1e-15
