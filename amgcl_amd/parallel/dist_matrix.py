@@ -134,9 +134,13 @@ class DistMatrix:
         dist = self.dist
         for buf, idx in zip(self.send_bufs, self.send_idx):
             self.backend.gather(x, idx, buf)
-        if hasattr(self.backend, "synchronize") and self.backend.name == "hip":
-            # RCCL launches on its own internal stream; make packed buffers
-            # visible (cheap: the gather kernels are tiny)
+        # No host sync needed before posting: ProcessGroupNCCL records an
+        # event on the current stream and makes the RCCL stream wait on it,
+        # so the gather kernels are ordered before the sends while the host
+        # keeps going (launches the overlapped local SpMV immediately).
+        # AMGCL_HALO_SYNC=1 restores a full sync for debugging.
+        if (self.backend.name == "hip"
+                and __import__("os").environ.get("AMGCL_HALO_SYNC")):
             self._torch.cuda.current_stream().synchronize()
         ops = []
         P2POp = dist.P2POp
