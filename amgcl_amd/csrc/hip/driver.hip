@@ -13,6 +13,8 @@
 
 #include <hip/hip_runtime.h>
 
+#include "amg_common.h"
+
 #include <cmath>
 #include <cstdint>
 #include <cstdlib>
@@ -80,14 +82,15 @@ __global__ void relax_swap_k(int64_t nrows, const int *__restrict__ ptr,
                              const int *__restrict__ col, const T *__restrict__ val,
                              const T *__restrict__ M, const T *__restrict__ rhs,
                              const T *__restrict__ x, T *__restrict__ xn) {
-    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t tid = amg_logical_block() * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SUBW;
     for (; row < nrows; row += stride) {
         T s = (T)0;
         int b = ptr[row], e = ptr[row + 1];
-        for (int j = b + lane; j < e; j += SUBW) s += val[j] * x[col[j]];
+        for (int j = b + lane; j < e; j += SUBW)
+            s += AMG_STREAM_LD(&val[j]) * x[AMG_STREAM_LD(&col[j])];
 #pragma unroll
         for (int off = SUBW / 2; off > 0; off >>= 1) s += __shfl_down(s, off, SUBW);
         if (lane == 0) xn[row] = x[row] + M[row] * (rhs[row] - s);
@@ -104,8 +107,7 @@ __global__ void relax_zero_k(int64_t n, const T *__restrict__ M,
 }
 
 static inline int nblocks_d(int64_t work, int block = 256, int cap = 2048) {
-    int64_t b = (work + block - 1) / block;
-    return (int)(b < 1 ? 1 : (b > cap ? cap : b));
+    return amg_nblocks(work, block, cap);
 }
 
 struct LevelDesc {
@@ -130,6 +132,13 @@ struct LevelDesc {
     double *t;  // workspace (f/u unused at level 0)
 };
 
+struct GraphEntry {
+    const double *rhs;
+    double *x;
+    double *xswap;
+    hipGraphExec_t exec;
+};
+
 struct Driver {
     std::vector<LevelDesc> lv;
     const void *coarse_inv;  // dense n x n (may be null -> smooth coarsest)
@@ -138,6 +147,12 @@ struct Driver {
     hipStream_t stream;
     double *dotbuf_d;  // 2 doubles
     double *dotbuf_h;  // pinned host, 2 doubles
+    // hipGraph cache for the V-cycle: the cycle is a fixed kernel sequence
+    // over fixed pointers, so one capture per (rhs, x, x_swap) triple turns
+    // the whole preconditioner application (dozens of launches, the coarse
+    // tail being pure launch overhead) into a single graph launch.
+    std::vector<GraphEntry> graphs;
+    bool use_graphs;
     // mixed precision: the cycle runs fp32 on the (fp32) LevelDescs while the
     // Krylov loop keeps the fp64 fine operator below + f32 cast buffers
     int f32;
@@ -258,6 +273,38 @@ static int precond_apply(Driver *D, const double *rhs, double *x, double *x_swap
     return 0;
 }
 
+// Graph-cached preconditioner application.  The per-cycle launch sequence is
+// deterministic (the relax pointer-swap count is fixed), so a capture is
+// valid for every later call with the same buffer triple.  Falls back to the
+// direct path if capture fails (e.g. nested capture) or AMGCL_NO_GRAPH=1.
+static int precond_apply_graphed(Driver *D, const double *rhs, double *x,
+                                 double *x_swap) {
+    if (!D->use_graphs) return precond_apply(D, rhs, x, x_swap);
+    for (const GraphEntry &g : D->graphs)
+        if (g.rhs == rhs && g.x == x && g.xswap == x_swap)
+            return (int)hipGraphLaunch(g.exec, D->stream);
+    if (D->graphs.size() >= 8) return precond_apply(D, rhs, x, x_swap);
+    if (hipStreamBeginCapture(D->stream, hipStreamCaptureModeThreadLocal) !=
+        hipSuccess)
+        return precond_apply(D, rhs, x, x_swap);
+    int rc = precond_apply(D, rhs, x, x_swap);
+    hipGraph_t graph;
+    hipError_t ec = hipStreamEndCapture(D->stream, &graph);
+    if (rc) return rc;
+    if (ec != hipSuccess) return (int)ec;
+    hipGraphExec_t exec;
+    ec = hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0);
+    (void)hipGraphDestroy(graph);
+    if (ec != hipSuccess) {
+        // capture succeeded but instantiation failed: run directly
+        (void)hipGetLastError();
+        D->use_graphs = false;
+        return precond_apply(D, rhs, x, x_swap);
+    }
+    D->graphs.push_back({rhs, x, x_swap, exec});
+    return (int)hipGraphLaunch(exec, D->stream);
+}
+
 static int read_dots(Driver *D, int n, double *out) {
     CHK(hipMemcpyAsync(D->dotbuf_h, D->dotbuf_d, n * sizeof(double),
                        hipMemcpyDeviceToHost, D->stream));
@@ -288,6 +335,8 @@ extern "C" void *amg_driver_create(const LevelDesc *levels, int nlevels,
     D->a64_val = a64_val;
     D->a64_subw = a64_subw;
     D->cb_r = D->cb_x = D->cb_s = nullptr;
+    const char *ng = getenv("AMGCL_NO_GRAPH");
+    D->use_graphs = !(ng && ng[0] && ng[0] != '0');
     bool ok = hipMalloc((void **)&D->dotbuf_d, 2 * sizeof(double)) == hipSuccess &&
               hipHostMalloc((void **)&D->dotbuf_h, 2 * sizeof(double)) == hipSuccess;
     if (ok && f32) {
@@ -308,6 +357,7 @@ extern "C" void amg_driver_destroy(void *h) {
     if (!D) return;
     (void)hipFree(D->dotbuf_d);
     (void)hipHostFree(D->dotbuf_h);
+    for (GraphEntry &g : D->graphs) (void)hipGraphExecDestroy(g.exec);
     if (D->cb_r) (void)hipFree(D->cb_r);
     if (D->cb_x) (void)hipFree(D->cb_x);
     if (D->cb_s) (void)hipFree(D->cb_s);
@@ -319,7 +369,7 @@ extern "C" void amg_driver_destroy(void *h) {
 // runs natively)
 extern "C" int amg_driver_precond(void *h, const double *rhs, double *x, double *x_swap) {
     Driver *D = (Driver *)h;
-    return precond_apply(D, rhs, x, x_swap);
+    return precond_apply_graphed(D, rhs, x, x_swap);
 }
 
 // Preconditioned CG (parity: amgcl/solver/cg.hpp:152-204).
@@ -360,7 +410,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     double rho1 = 0.0, rho2 = 0.0;
     int64_t iter = 0;
     while (res > eps && iter < maxiter) {
-        CHK(precond_apply(D, r, s, s_swap));
+        CHK(precond_apply_graphed(D, r, s, s_swap));
         rho2 = rho1;
         CHK(amg_dot_f64(n, r, s, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
@@ -440,7 +490,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
             CHK(amg_axpbypcz_f64(n, 1.0, r, -beta * omega, v, beta, p, st));
         }
         // v = A (M^-1 p);  T = M^-1 p
-        CHK(precond_apply(D, p, T, T_swap));
+        CHK(precond_apply_graphed(D, p, T, T_swap));
         CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, v, ksubw, st));
         CHK(amg_dot_f64(n, rh, v, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
@@ -451,7 +501,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         CHK(read_dots(D, 1, dots));
         res = sqrt(dots[0]);
         if (res > eps) {
-            CHK(precond_apply(D, s2, T, T_swap));
+            CHK(precond_apply_graphed(D, s2, T, T_swap));
             CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, t2, ksubw, st));
             CHK(amg_dot2_f64(n, t2, s2, t2, t2, D->dotbuf_d, st));
             CHK(read_dots(D, 2, dots));

@@ -16,13 +16,10 @@
 #include <cstdint>
 #include <cstdio>
 
-#define WAVE 64
+#include "amg_common.h"
 
-// Launch geometry for memory-bound grid-stride kernels (guide §6 G11):
-// cap at ~8 blocks/CU and grid-stride the rest.
 static inline int nblocks(int64_t work, int block = 256, int cap = 2048) {
-    int64_t b = (work + block - 1) / block;
-    return (int)(b < 1 ? 1 : (b > cap ? cap : b));
+    return amg_nblocks(work, block, cap);
 }
 
 extern "C" int amg_hip_last_error() { return (int)hipGetLastError(); }
@@ -37,14 +34,14 @@ __global__ void spmv_k(int nrows, const int *__restrict__ ptr,
                        const int *__restrict__ col, const T *__restrict__ val,
                        const T *__restrict__ x, double alpha, double beta,
                        T *__restrict__ y) {
-    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t tid = amg_logical_block() * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SUBW;
     for (; row < nrows; row += stride) {
         double s = 0.0;
         int b = ptr[row], e = ptr[row + 1];
-        for (int j = b + lane; j < e; j += SUBW) s += (double)val[j] * (double)x[col[j]];
+        for (int j = b + lane; j < e; j += SUBW) s += (double)AMG_STREAM_LD(&val[j]) * (double)x[AMG_STREAM_LD(&col[j])];
 #pragma unroll
         for (int off = SUBW / 2; off > 0; off >>= 1) s += __shfl_down(s, off, SUBW);
         if (lane == 0) y[row] = (T)(BETA0 ? alpha * s : alpha * s + beta * (double)y[row]);
@@ -57,14 +54,14 @@ __global__ void residual_k(int nrows, const int *__restrict__ ptr,
                            const int *__restrict__ col, const T *__restrict__ val,
                            const T *__restrict__ rhs, const T *__restrict__ x,
                            T *__restrict__ r) {
-    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t tid = amg_logical_block() * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SUBW;
     for (; row < nrows; row += stride) {
         double s = 0.0;
         int b = ptr[row], e = ptr[row + 1];
-        for (int j = b + lane; j < e; j += SUBW) s += (double)val[j] * (double)x[col[j]];
+        for (int j = b + lane; j < e; j += SUBW) s += (double)AMG_STREAM_LD(&val[j]) * (double)x[AMG_STREAM_LD(&col[j])];
 #pragma unroll
         for (int off = SUBW / 2; off > 0; off >>= 1) s += __shfl_down(s, off, SUBW);
         if (lane == 0) r[row] = (T)((double)rhs[row] - s);
@@ -78,14 +75,14 @@ __global__ void relax_diag_k(int nrows, const int *__restrict__ ptr,
                              const int *__restrict__ col, const T *__restrict__ val,
                              const T *__restrict__ M, const T *__restrict__ rhs,
                              const T *__restrict__ x, T *__restrict__ t) {
-    int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t tid = amg_logical_block() * blockDim.x + threadIdx.x;
     int lane = (int)(tid & (SUBW - 1));
     int64_t row = tid / SUBW;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SUBW;
     for (; row < nrows; row += stride) {
         double s = 0.0;
         int b = ptr[row], e = ptr[row + 1];
-        for (int j = b + lane; j < e; j += SUBW) s += (double)val[j] * (double)x[col[j]];
+        for (int j = b + lane; j < e; j += SUBW) s += (double)AMG_STREAM_LD(&val[j]) * (double)x[AMG_STREAM_LD(&col[j])];
 #pragma unroll
         for (int off = SUBW / 2; off > 0; off >>= 1) s += __shfl_down(s, off, SUBW);
         if (lane == 0) t[row] = (T)((double)M[row] * ((double)rhs[row] - s));

@@ -84,6 +84,10 @@ class SubdomainDeflation:
         dist.all_gather_object(blocks, eblock, group=dist_A.group)
         E = np.concatenate(blocks, axis=0)  # m x m
         self.Einv = np.linalg.pinv(E)
+        # device copy so the per-iteration coarse solve (lam = Einv @ z) stays
+        # on the GPU: no host round-trip inside the projected operator
+        self.Einv_d = (torch.from_numpy(self.Einv).to(base.device)
+                       if is_hip else None)
 
         self._lam = torch.zeros(m, dtype=torch.float64,
                                 device=base.device if is_hip else "cpu")
@@ -100,25 +104,21 @@ class SubdomainDeflation:
         full = t.zeros(self.m, dtype=t.float64, device=local.device)
         full[self.col0 : self.col0 + self.ndv] = local
         dist.all_reduce(full, group=self.A.group)
-        lam = self.Einv @ full.cpu().numpy()
-        return lam
+        if self.base.name == "hip":
+            return self.Einv_d @ full  # stays on device, no host sync
+        return self.Einv @ full.numpy()
 
     def project(self, w):
         """w <- w - AZ E^-1 Z^T w  (P w)."""
         lam = self._coarse(w)
-        if self.base.name == "hip":
-            lam_d = self._t.from_numpy(lam).to(self.base.device)
-            w -= self.AZ @ lam_d
-        else:
-            w -= self.AZ @ lam
+        w -= self.AZ @ lam
 
     def coarse_guess(self, rhs, x):
         """x = Z E^-1 Z^T rhs (coarse-grid initial guess)."""
         lam = self._coarse(rhs)
         mine = lam[self.col0 : self.col0 + self.ndv]
         if self.base.name == "hip":
-            mine_d = self._t.from_numpy(mine).to(self.base.device)
-            x.copy_(self.Z @ mine_d)
+            x.copy_(self.Z @ mine)
         else:
             np.copyto(x, self.Z @ mine)
 
@@ -126,16 +126,9 @@ class SubdomainDeflation:
         """x = x0 + y - Z E^-1 Z^T (A y)."""
         lam = self._coarse(Ay)
         mine = lam[self.col0 : self.col0 + self.ndv]
-        if self.base.name == "hip":
-            mine_d = self._t.from_numpy(mine).to(self.base.device)
-            corr = self.Z @ mine_d
-        else:
-            corr = self.Z @ mine
+        corr = self.Z @ mine
         self.base.axpby(1.0, y, 1.0, x0)
-        if self.base.name == "hip":
-            x0 -= corr
-        else:
-            x0 -= corr
+        x0 -= corr
         return x0
 
 
