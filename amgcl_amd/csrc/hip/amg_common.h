@@ -11,8 +11,13 @@
 // of the logical index space: the x-gather bands of neighboring CSR rows
 // then land in a single XCD's L2 instead of being replicated in all eight.
 // Affects speed only, never correctness.
+// MEASURED (kbw384, r02): the swizzle LOST bandwidth on every level
+// (L0 4107->3337 GB/s): the natural round-robin dispatch already places
+// same-XCD blocks ~8 blocks (~512 rows) apart, inside the stencil's +-n
+// x-band, so L2 locality is better WITHOUT the remap.  Off by default,
+// kept for A/B (-DAMGCL_SWIZZLE).
 __device__ static inline int64_t amg_logical_block() {
-#ifndef AMGCL_NO_SWIZZLE
+#ifdef AMGCL_SWIZZLE
     unsigned g = gridDim.x;
     if ((g & 7u) == 0u)
         return (int64_t)(blockIdx.x & 7u) * (int64_t)(g >> 3) + (blockIdx.x >> 3);
@@ -20,10 +25,12 @@ __device__ static inline int64_t amg_logical_block() {
     return blockIdx.x;
 }
 
-// Streaming loads for the matrix arrays (val/col are read exactly once per
-// SpMV): the nontemporal hint keeps them from evicting the x vector, which
-// is the only array with reuse.
-#ifndef AMGCL_NO_NT
+// MEASURED (kbw384, r02): nontemporal val/col loads LOST bandwidth badly on
+// cache-resident coarse levels (L2-level 3653->2088 GB/s: a 15 MB matrix is
+// L2/LLC-resident across V-cycle passes, and the evict-first hint destroys
+// that residency) and did not help the streaming fine level either.  Off by
+// default, kept for A/B (-DAMGCL_NT).
+#ifdef AMGCL_NT
 #define AMG_STREAM_LD(p) __builtin_nontemporal_load(p)
 #else
 #define AMG_STREAM_LD(p) (*(p))

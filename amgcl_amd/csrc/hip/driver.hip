@@ -153,6 +153,13 @@ struct Driver {
     // tail being pure launch overhead) into a single graph launch.
     std::vector<GraphEntry> graphs;
     bool use_graphs;
+    // The solve runs on an internal non-blocking stream, event-chained to the
+    // caller's (torch) stream at entry/exit.  The torch default stream is the
+    // HIP *legacy* stream, which hipStreamBeginCapture cannot capture
+    // (hipErrorStreamCaptureUnsupported, measured r02); an owned stream both
+    // enables graph capture and decouples the solve from torch's stream.
+    hipStream_t ext_stream;  // caller's stream (ordering boundary only)
+    hipEvent_t ev_in, ev_out;
     // mixed precision: the cycle runs fp32 on the (fp32) LevelDescs while the
     // Krylov loop keeps the fp64 fine operator below + f32 cast buffers
     int f32;
@@ -305,6 +312,21 @@ static int precond_apply_graphed(Driver *D, const double *rhs, double *x,
     return (int)hipGraphLaunch(exec, D->stream);
 }
 
+// order the internal stream after the caller's stream (entry) and the
+// caller's stream after the internal one (exit)
+static int enter_driver(Driver *D) {
+    if (D->stream == D->ext_stream) return 0;
+    CHK((int)hipEventRecord(D->ev_in, D->ext_stream));
+    CHK((int)hipStreamWaitEvent(D->stream, D->ev_in, 0));
+    return 0;
+}
+static int leave_driver(Driver *D) {
+    if (D->stream == D->ext_stream) return 0;
+    CHK((int)hipEventRecord(D->ev_out, D->stream));
+    CHK((int)hipStreamWaitEvent(D->ext_stream, D->ev_out, 0));
+    return 0;
+}
+
 static int read_dots(Driver *D, int n, double *out) {
     CHK(hipMemcpyAsync(D->dotbuf_h, D->dotbuf_d, n * sizeof(double),
                        hipMemcpyDeviceToHost, D->stream));
@@ -337,6 +359,17 @@ extern "C" void *amg_driver_create(const LevelDesc *levels, int nlevels,
     D->cb_r = D->cb_x = D->cb_s = nullptr;
     const char *ng = getenv("AMGCL_NO_GRAPH");
     D->use_graphs = !(ng && ng[0] && ng[0] != '0');
+    D->ext_stream = stream;
+    D->ev_in = D->ev_out = nullptr;
+    hipStream_t st_int = nullptr;
+    if (hipStreamCreateWithFlags(&st_int, hipStreamNonBlocking) == hipSuccess &&
+        hipEventCreateWithFlags(&D->ev_in, hipEventDisableTiming) == hipSuccess &&
+        hipEventCreateWithFlags(&D->ev_out, hipEventDisableTiming) == hipSuccess) {
+        D->stream = st_int;  // all driver work goes to the internal stream
+    } else {
+        if (st_int) (void)hipStreamDestroy(st_int);
+        D->use_graphs = false;  // legacy stream cannot be captured
+    }
     bool ok = hipMalloc((void **)&D->dotbuf_d, 2 * sizeof(double)) == hipSuccess &&
               hipHostMalloc((void **)&D->dotbuf_h, 2 * sizeof(double)) == hipSuccess;
     if (ok && f32) {
@@ -358,6 +391,9 @@ extern "C" void amg_driver_destroy(void *h) {
     (void)hipFree(D->dotbuf_d);
     (void)hipHostFree(D->dotbuf_h);
     for (GraphEntry &g : D->graphs) (void)hipGraphExecDestroy(g.exec);
+    if (D->stream != D->ext_stream) (void)hipStreamDestroy(D->stream);
+    if (D->ev_in) (void)hipEventDestroy(D->ev_in);
+    if (D->ev_out) (void)hipEventDestroy(D->ev_out);
     if (D->cb_r) (void)hipFree(D->cb_r);
     if (D->cb_x) (void)hipFree(D->cb_x);
     if (D->cb_s) (void)hipFree(D->cb_s);
@@ -369,7 +405,9 @@ extern "C" void amg_driver_destroy(void *h) {
 // runs natively)
 extern "C" int amg_driver_precond(void *h, const double *rhs, double *x, double *x_swap) {
     Driver *D = (Driver *)h;
-    return precond_apply_graphed(D, rhs, x, x_swap);
+    CHK(enter_driver(D));
+    CHK(precond_apply_graphed(D, rhs, x, x_swap));
+    return leave_driver(D);
 }
 
 // Preconditioned CG (parity: amgcl/solver/cg.hpp:152-204).
@@ -379,6 +417,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
                              double abstol, int maxiter, int64_t *iters_out,
                              double *resid_out) {
     Driver *D = (Driver *)h;
+    CHK(enter_driver(D));
     const LevelDesc &L0 = D->lv[0];
     const int64_t n = L0.nrows;
     // the Krylov loop always iterates with the fp64 fine operator (in mixed
@@ -398,6 +437,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
         CHK(amg_fill_f64(n, 0.0, x, st));
         *iters_out = 0;
         *resid_out = 0.0;
+        CHK(leave_driver(D));
         return hipStreamSynchronize(st);
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
@@ -432,6 +472,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     }
     *iters_out = iter;
     *resid_out = res / norm_rhs;
+    CHK(leave_driver(D));
     return hipStreamSynchronize(st);
 }
 
@@ -444,6 +485,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
                                    double abstol, int maxiter, int64_t *iters_out,
                                    double *resid_out) {
     Driver *D = (Driver *)h;
+    CHK(enter_driver(D));
     const LevelDesc &L0 = D->lv[0];
     const int64_t n = L0.nrows;
     // the Krylov loop always iterates with the fp64 fine operator (in mixed
@@ -463,6 +505,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         CHK(amg_fill_f64(n, 0.0, x, st));
         *iters_out = 0;
         *resid_out = 0.0;
+        CHK(leave_driver(D));
         return hipStreamSynchronize(st);
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
@@ -517,5 +560,6 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
     }
     *iters_out = iter;
     *resid_out = res / norm_rhs;
+    CHK(leave_driver(D));
     return hipStreamSynchronize(st);
 }
