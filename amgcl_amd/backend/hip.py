@@ -25,7 +25,8 @@ def _stream():
 
 
 class DeviceCSR:
-    __slots__ = ("nrows", "ncols", "ptr", "col", "val", "subw")
+    __slots__ = ("nrows", "ncols", "ptr", "col", "val", "subw",
+                 "nslice", "soff", "scol", "sval")
 
     def __init__(self, csr: CSR, device, subw=0):
         import torch
@@ -36,6 +37,8 @@ class DeviceCSR:
         self.col = torch.from_numpy(np.asarray(csr.col)).to(device)
         self.val = torch.from_numpy(np.asarray(csr.val)).to(device)
         self.subw = subw  # 0 = auto by mean row length
+        self.nslice = 0
+        self.soff = self.scol = self.sval = None
 
     @classmethod
     def from_tensors(cls, nrows, ncols, ptr, col, val, subw=0):
@@ -44,6 +47,8 @@ class DeviceCSR:
         self.ncols = int(ncols)
         self.ptr, self.col, self.val = ptr, col, val
         self.subw = subw
+        self.nslice = 0
+        self.soff = self.scol = self.sval = None
         return self
 
     @property
@@ -51,7 +56,40 @@ class DeviceCSR:
         return self.col.numel()
 
     def bytes(self):
-        return (self.ptr.numel() + self.col.numel()) * 4 + self.val.numel() * 8
+        b = (self.ptr.numel() + self.col.numel()) * 4 + self.val.numel() * 8
+        if self.nslice:
+            b += self.soff.numel() * 8 + self.scol.numel() * 4
+            b += self.sval.numel() * self.sval.element_size()
+        return b
+
+    def build_sell(self):
+        """Build the SELL-64 image of this matrix (kernels.hip rationale:
+        wave-native layout; lane = row, slices column-major).  The CSR arrays
+        are kept — transfers and non-SELL paths still use them."""
+        if self.nslice:
+            return self
+        import torch
+
+        n = self.nrows
+        dev = self.val.device
+        nslice = (n + 63) // 64
+        lens = (self.ptr[1:] - self.ptr[:-1]).to(torch.int64)
+        pad = nslice * 64 - n
+        if pad:
+            lens = torch.cat([lens, torch.zeros(pad, dtype=torch.int64, device=dev)])
+        w = lens.view(nslice, 64).max(dim=1).values
+        soff = torch.zeros(nslice + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(w * 64, dim=0, out=soff[1:])
+        total = int(soff[-1].item())
+        scol = torch.zeros(total, dtype=torch.int32, device=dev)
+        sval = torch.zeros(total, dtype=self.val.dtype, device=dev)
+        fn = (lib().amg_sell_fill_f32 if self.val.dtype == torch.float32
+              else lib().amg_sell_fill_f64)
+        check(fn(n, self.ptr.data_ptr(), self.col.data_ptr(), self.val.data_ptr(),
+                 soff.data_ptr(), scol.data_ptr(), sval.data_ptr(), _stream()),
+              "sell_fill")
+        self.nslice, self.soff, self.scol, self.sval = nslice, soff, scol, sval
+        return self
 
 
 class DeviceBSR:
@@ -127,6 +165,12 @@ class HipBackend:
                                          x.data_ptr(), alpha, beta, y.data_ptr(),
                                          _stream()), "bsr_spmv")
             return
+        if getattr(A, "nslice", 0):
+            check(self._fn("sell_spmv", A.sval)(A.nrows, A.nslice, A.soff.data_ptr(),
+                                                A.scol.data_ptr(), A.sval.data_ptr(),
+                                                x.data_ptr(), alpha, beta,
+                                                y.data_ptr(), _stream()), "sell_spmv")
+            return
         check(self._fn("spmv", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(), A.col.data_ptr(),
                                       A.val.data_ptr(), x.data_ptr(), alpha, beta,
                                       y.data_ptr(), A.subw, _stream()), "spmv")
@@ -137,6 +181,13 @@ class HipBackend:
                                              A.col.data_ptr(), A.val.data_ptr(),
                                              b.data_ptr(), x.data_ptr(), r.data_ptr(),
                                              _stream()), "bsr_residual")
+            return
+        if getattr(A, "nslice", 0):
+            check(self._fn("sell_residual", A.sval)(A.nrows, A.nslice,
+                                                    A.soff.data_ptr(), A.scol.data_ptr(),
+                                                    A.sval.data_ptr(), b.data_ptr(),
+                                                    x.data_ptr(), r.data_ptr(),
+                                                    _stream()), "sell_residual")
             return
         check(self._fn("residual", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(),
                                           A.col.data_ptr(), A.val.data_ptr(),
@@ -151,6 +202,15 @@ class HipBackend:
                                           M.data_ptr(), rhs.data_ptr(), x.data_ptr(),
                                           t.data_ptr(), _stream()), "bsr_relax")
             self.axpby(1.0, t, 1.0, x)
+            return
+        if getattr(A, "nslice", 0):
+            # SELL relax writes x_new = x + M(rhs - Ax) into t, then copy back
+            check(self._fn("sell_relax", A.sval)(A.nrows, A.nslice, A.soff.data_ptr(),
+                                                 A.scol.data_ptr(), A.sval.data_ptr(),
+                                                 M.data_ptr(), rhs.data_ptr(),
+                                                 x.data_ptr(), t.data_ptr(),
+                                                 _stream()), "sell_relax")
+            x.copy_(t)
             return
         check(self._fn("relax_diag", A.val)(A.nrows, A.nnz, A.ptr.data_ptr(),
                                             A.col.data_ptr(), A.val.data_ptr(),

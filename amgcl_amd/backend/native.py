@@ -33,6 +33,10 @@ class LevelDescC(ctypes.Structure):
         ("f", ctypes.c_void_p),
         ("u", ctypes.c_void_p),
         ("t", ctypes.c_void_p),
+        ("nslice", ctypes.c_int64),
+        ("soff", ctypes.c_void_p),
+        ("scol", ctypes.c_void_p),
+        ("sval", ctypes.c_void_p),
     ]
 
 
@@ -124,6 +128,10 @@ class NativeDriver:
                 d.rnnz = int(R.nnz)
                 d.rptr, d.rcol, d.rval = _ptr(R.ptr), _ptr(R.col), _ptr(R.val)
                 d.rsubw = R.subw or _auto_subw(R)
+            if A.nslice:
+                d.nslice = A.nslice
+                d.soff, d.scol, d.sval = _ptr(A.soff), _ptr(A.scol), _ptr(A.sval)
+                self._keep.extend([A.soff, A.scol, A.sval])
             relax = l.relax
             d.M = _ptr(relax.M if relax is not None else None)
             d.f = _ptr(l.f)

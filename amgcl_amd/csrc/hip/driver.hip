@@ -44,6 +44,18 @@ extern "C" int amg_residual_f32(int64_t, int64_t, const int *, const int *, cons
 extern "C" int amg_gemv_f32(int64_t, const float *, const float *, float *, hipStream_t);
 extern "C" int amg_cast_d2s(int64_t, const double *, float *, hipStream_t);
 extern "C" int amg_cast_s2d(int64_t, const float *, double *, hipStream_t);
+extern "C" int amg_sell_residual_f64(int64_t, int64_t, const int64_t *, const int *,
+                                     const double *, const double *, const double *,
+                                     double *, hipStream_t);
+extern "C" int amg_sell_relax_f64(int64_t, int64_t, const int64_t *, const int *,
+                                  const double *, const double *, const double *,
+                                  const double *, double *, hipStream_t);
+extern "C" int amg_sell_residual_f32(int64_t, int64_t, const int64_t *, const int *,
+                                     const float *, const float *, const float *,
+                                     float *, hipStream_t);
+extern "C" int amg_sell_relax_f32(int64_t, int64_t, const int64_t *, const int *,
+                                  const float *, const float *, const float *,
+                                  const float *, float *, hipStream_t);
 
 // value-type dispatch for the cycle (the fp32 hierarchy of mixed precision)
 template <typename T> struct ops;
@@ -61,6 +73,16 @@ template <> struct ops<double> {
                     hipStream_t s) {
         return amg_gemv_f64(n, inv, f, u, s);
     }
+    static int sell_residual(int64_t n, int64_t ns, const int64_t *soff, const int *c,
+                             const double *v, const double *rhs, const double *x,
+                             double *r, hipStream_t s) {
+        return amg_sell_residual_f64(n, ns, soff, c, v, rhs, x, r, s);
+    }
+    static int sell_relax(int64_t n, int64_t ns, const int64_t *soff, const int *c,
+                          const double *v, const double *M, const double *rhs,
+                          const double *x, double *xn, hipStream_t s) {
+        return amg_sell_relax_f64(n, ns, soff, c, v, M, rhs, x, xn, s);
+    }
 };
 template <> struct ops<float> {
     static int spmv(int64_t n, int64_t nnz, const int *p, const int *c, const float *v,
@@ -73,6 +95,16 @@ template <> struct ops<float> {
     }
     static int gemv(int64_t n, const float *inv, const float *f, float *u, hipStream_t s) {
         return amg_gemv_f32(n, inv, f, u, s);
+    }
+    static int sell_residual(int64_t n, int64_t ns, const int64_t *soff, const int *c,
+                             const float *v, const float *rhs, const float *x,
+                             float *r, hipStream_t s) {
+        return amg_sell_residual_f32(n, ns, soff, c, v, rhs, x, r, s);
+    }
+    static int sell_relax(int64_t n, int64_t ns, const int64_t *soff, const int *c,
+                          const float *v, const float *M, const float *rhs,
+                          const float *x, float *xn, hipStream_t s) {
+        return amg_sell_relax_f32(n, ns, soff, c, v, M, rhs, x, xn, s);
     }
 };
 
@@ -130,6 +162,11 @@ struct LevelDesc {
     double *f;
     double *u;
     double *t;  // workspace (f/u unused at level 0)
+    // optional SELL-64 image of A (coarse levels; see kernels.hip)
+    int64_t nslice;  // 0 = no SELL
+    const int64_t *soff;
+    const int *scol;
+    const void *sval;
 };
 
 struct GraphEntry {
@@ -179,6 +216,14 @@ struct Driver {
 
 template <typename T>
 static int relax_swap(Driver *D, const LevelDesc &L, const T *rhs, T **x, T **xn) {
+    if (L.nslice) {
+        CHK(ops<T>::sell_relax(L.nrows, L.nslice, L.soff, L.scol, (const T *)L.sval,
+                               (const T *)L.M, rhs, *x, *xn, D->stream));
+        T *tmp = *x;
+        *x = *xn;
+        *xn = tmp;
+        return 0;
+    }
     int subw = L.subw;
     int grid = nblocks_d(L.nrows * subw);
     const T *val = (const T *)L.val;
@@ -240,8 +285,12 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
         CHK(relax_swap<T>(D, L, f, u_io, scratch));
     }
     // t = f - A u ; f_next = R t
-    CHK(ops<T>::residual(L.nrows, L.nnz, L.ptr, L.col, (const T *)L.val, f, *u_io,
-                         *scratch, L.subw, D->stream));
+    if (L.nslice)
+        CHK(ops<T>::sell_residual(L.nrows, L.nslice, L.soff, L.scol, (const T *)L.sval,
+                                  f, *u_io, *scratch, D->stream));
+    else
+        CHK(ops<T>::residual(L.nrows, L.nnz, L.ptr, L.col, (const T *)L.val, f, *u_io,
+                             *scratch, L.subw, D->stream));
     CHK(ops<T>::spmv(N.nrows, L.rnnz, L.rptr, L.rcol, (const T *)L.rval, *scratch, 1.0,
                      0.0, (T *)N.f, L.rsubw, D->stream));
     T *nu = (T *)N.u;

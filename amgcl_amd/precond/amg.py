@@ -49,6 +49,12 @@ class AMG:
             "keep_host_matrices": False,
             "precision": "fp64",  # "mixed" = fp32 hierarchy under fp64 Krylov
             "block_value": 0,      # >1: store level operators as BSR blocks
+            # SELL-64 image of coarse-level operators (wave-native layout;
+            # kernels.hip rationale). "auto" converts HIP levels with
+            # mean row length >= sell_min_mean and >= sell_min_rows rows.
+            "sell": "auto",
+            "sell_min_rows": 4096,
+            "sell_min_mean": 12.0,
         }
 
     def __init__(self, A, prm=None, backend=None):
@@ -67,6 +73,28 @@ class AMG:
             self._to_mixed()
         if int(self.prm["block_value"]) > 1:
             self._to_block(int(self.prm["block_value"]))
+        self._build_sell()
+
+    def _build_sell(self):
+        """Attach SELL-64 images to coarse-level operators (measured: CSR
+        sub-wave SpMV reaches only ~2 TB/s on SA coarse levels because the
+        x-gather scatters across the sub-wave; see csrc/hip/kernels.hip)."""
+        import os
+
+        prm = self.prm
+        if prm["sell"] in (False, "off") or os.environ.get("AMGCL_NO_SELL"):
+            return
+        if self.backend.name != "hip":
+            return
+        from ..backend.hip import DeviceCSR
+
+        for lvl in self.levels:
+            A = lvl.A
+            if not isinstance(A, DeviceCSR) or A.nrows < int(prm["sell_min_rows"]):
+                continue
+            if A.nnz / max(A.nrows, 1) < float(prm["sell_min_mean"]):
+                continue
+            A.build_sell()
 
     def rebuild(self, A_new):
         """Reuse the transfer operators for a matrix with changed coefficients
@@ -101,6 +129,7 @@ class AMG:
                 A, kind=self.prm["direct_solver"])
         else:
             last.relax = relax_factory(A, backend)
+        self._build_sell()
 
     # --- setup (host) ------------------------------------------------------
     def _build(self, A: CSR):

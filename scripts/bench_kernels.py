@@ -65,6 +65,8 @@ def main():
         gb_spmv = (nnz * 12 + (n + 1) * 4 + min(nnz, Ad.ncols) * 8 + n * 8) / 1e9
         row = f"L{li} n={n:>9} nnz={nnz:>10} mean={nnz/n:5.1f} | "
         best = (None, 0)
+        sell_state = (Ad.nslice, Ad.soff, Ad.scol, Ad.sval)
+        Ad.nslice = 0  # force the CSR kernels for the subw sweep
         for subw in (1, 2, 4, 8, 16, 32, 64):
             Ad.subw = subw
             dt = timeit(lambda: hip.spmv(1.0, Ad, x, 0.0, y))
@@ -85,6 +87,24 @@ def main():
         dt = timeit(lambda: hip.residual(b, Ad, x, y))
         print(f"residual: {(gb_spmv + n*8/1e9)/dt:5.0f} GB/s")
         Ad.subw = 0
+        # SELL-64 image (same nominal-bytes denominator as CSR for a fair
+        # apples-to-apples effective number; padding inflates actual bytes)
+        Ad.nslice, Ad.soff, Ad.scol, Ad.sval = sell_state
+        if not Ad.nslice:
+            try:
+                Ad.build_sell()
+            except Exception as e:
+                print(f"   (sell build failed: {e})")
+        if Ad.nslice:
+            pad = Ad.sval.numel() / max(nnz, 1)
+            dt = timeit(lambda: hip.spmv(1.0, Ad, x, 0.0, y))
+            s_spmv = gb_spmv / dt
+            dt = timeit(lambda: hip.relax_diag(Ad, m, b, y, t))
+            s_rel = gb / dt
+            dt = timeit(lambda: hip.residual(b, Ad, x, y))
+            s_res = (gb_spmv + n * 8 / 1e9) / dt
+            print(f"   SELL (pad x{pad:4.2f}): spmv {s_spmv:5.0f}  "
+                  f"relax {s_rel:5.0f}  residual {s_res:5.0f} GB/s")
 
     n = amg.levels[0].rows
     x = torch.rand(n, dtype=torch.float64, device=hip.device)
