@@ -37,6 +37,14 @@ class LevelDescC(ctypes.Structure):
         ("soff", ctypes.c_void_p),
         ("scol", ctypes.c_void_p),
         ("sval", ctypes.c_void_p),
+        ("pnslice", ctypes.c_int64),
+        ("psoff", ctypes.c_void_p),
+        ("pscol", ctypes.c_void_p),
+        ("psval", ctypes.c_void_p),
+        ("rnslice", ctypes.c_int64),
+        ("rsoff", ctypes.c_void_p),
+        ("rscol", ctypes.c_void_p),
+        ("rsval", ctypes.c_void_p),
     ]
 
 
@@ -132,6 +140,16 @@ class NativeDriver:
                 d.nslice = A.nslice
                 d.soff, d.scol, d.sval = _ptr(A.soff), _ptr(A.scol), _ptr(A.sval)
                 self._keep.extend([A.soff, A.scol, A.sval])
+            if l.P is not None and getattr(l.P, "nslice", 0):
+                d.pnslice = l.P.nslice
+                d.psoff, d.pscol, d.psval = (_ptr(l.P.soff), _ptr(l.P.scol),
+                                             _ptr(l.P.sval))
+                self._keep.extend([l.P.soff, l.P.scol, l.P.sval])
+            if l.R is not None and getattr(l.R, "nslice", 0):
+                d.rnslice = l.R.nslice
+                d.rsoff, d.rscol, d.rsval = (_ptr(l.R.soff), _ptr(l.R.scol),
+                                             _ptr(l.R.sval))
+                self._keep.extend([l.R.soff, l.R.scol, l.R.sval])
             relax = l.relax
             d.M = _ptr(relax.M if relax is not None else None)
             d.f = _ptr(l.f)

@@ -44,6 +44,12 @@ extern "C" int amg_residual_f32(int64_t, int64_t, const int *, const int *, cons
 extern "C" int amg_gemv_f32(int64_t, const float *, const float *, float *, hipStream_t);
 extern "C" int amg_cast_d2s(int64_t, const double *, float *, hipStream_t);
 extern "C" int amg_cast_s2d(int64_t, const float *, double *, hipStream_t);
+extern "C" int amg_sell_spmv_f64(int64_t, int64_t, const int64_t *, const int *,
+                                 const double *, const double *, double, double,
+                                 double *, hipStream_t);
+extern "C" int amg_sell_spmv_f32(int64_t, int64_t, const int64_t *, const int *,
+                                 const float *, const float *, double, double,
+                                 float *, hipStream_t);
 extern "C" int amg_sell_residual_f64(int64_t, int64_t, const int64_t *, const int *,
                                      const double *, const double *, const double *,
                                      double *, hipStream_t);
@@ -73,6 +79,11 @@ template <> struct ops<double> {
                     hipStream_t s) {
         return amg_gemv_f64(n, inv, f, u, s);
     }
+    static int sell_spmv(int64_t n, int64_t ns, const int64_t *soff, const int *c,
+                         const double *v, const double *x, double a, double b,
+                         double *y, hipStream_t s) {
+        return amg_sell_spmv_f64(n, ns, soff, c, v, x, a, b, y, s);
+    }
     static int sell_residual(int64_t n, int64_t ns, const int64_t *soff, const int *c,
                              const double *v, const double *rhs, const double *x,
                              double *r, hipStream_t s) {
@@ -95,6 +106,11 @@ template <> struct ops<float> {
     }
     static int gemv(int64_t n, const float *inv, const float *f, float *u, hipStream_t s) {
         return amg_gemv_f32(n, inv, f, u, s);
+    }
+    static int sell_spmv(int64_t n, int64_t ns, const int64_t *soff, const int *c,
+                         const float *v, const float *x, double a, double b,
+                         float *y, hipStream_t s) {
+        return amg_sell_spmv_f32(n, ns, soff, c, v, x, a, b, y, s);
     }
     static int sell_residual(int64_t n, int64_t ns, const int64_t *soff, const int *c,
                              const float *v, const float *rhs, const float *x,
@@ -162,11 +178,19 @@ struct LevelDesc {
     double *f;
     double *u;
     double *t;  // workspace (f/u unused at level 0)
-    // optional SELL-64 image of A (coarse levels; see kernels.hip)
+    // optional SELL-64 images of A / P / R (see kernels.hip)
     int64_t nslice;  // 0 = no SELL
     const int64_t *soff;
     const int *scol;
     const void *sval;
+    int64_t pnslice;
+    const int64_t *psoff;
+    const int *pscol;
+    const void *psval;
+    int64_t rnslice;
+    const int64_t *rsoff;
+    const int *rscol;
+    const void *rsval;
 };
 
 struct GraphEntry {
@@ -291,8 +315,12 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
     else
         CHK(ops<T>::residual(L.nrows, L.nnz, L.ptr, L.col, (const T *)L.val, f, *u_io,
                              *scratch, L.subw, D->stream));
-    CHK(ops<T>::spmv(N.nrows, L.rnnz, L.rptr, L.rcol, (const T *)L.rval, *scratch, 1.0,
-                     0.0, (T *)N.f, L.rsubw, D->stream));
+    if (L.rnslice)
+        CHK(ops<T>::sell_spmv(N.nrows, L.rnslice, L.rsoff, L.rscol, (const T *)L.rsval,
+                              *scratch, 1.0, 0.0, (T *)N.f, D->stream));
+    else
+        CHK(ops<T>::spmv(N.nrows, L.rnnz, L.rptr, L.rcol, (const T *)L.rval, *scratch,
+                         1.0, 0.0, (T *)N.f, L.rsubw, D->stream));
     T *nu = (T *)N.u;
     T *nscratch = (T *)N.t;
     for (int c = 0; c < D->ncycle; ++c) {
@@ -300,8 +328,12 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
         // after the first sub-cycle the iterate is nonzero
     }
     // u += P u_next
-    CHK(ops<T>::spmv(L.nrows, L.pnnz, L.pptr, L.pcol, (const T *)L.pval, nu, 1.0, 1.0,
-                     *u_io, L.psubw, D->stream));
+    if (L.pnslice)
+        CHK(ops<T>::sell_spmv(L.nrows, L.pnslice, L.psoff, L.pscol, (const T *)L.psval,
+                              nu, 1.0, 1.0, *u_io, D->stream));
+    else
+        CHK(ops<T>::spmv(L.nrows, L.pnnz, L.pptr, L.pcol, (const T *)L.pval, nu, 1.0,
+                         1.0, *u_io, L.psubw, D->stream));
     for (int i = 0; i < D->npost; ++i)
         CHK(relax_swap<T>(D, L, f, u_io, scratch));
     return (int)hipGetLastError();
@@ -476,6 +508,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     const int *kcol = D->f32 ? D->a64_col : L0.col;
     const double *kval = D->f32 ? D->a64_val : (const double *)L0.val;
     const int ksubw = D->f32 ? D->a64_subw : L0.subw;
+    const int64_t ksell = D->f32 ? 0 : L0.nslice;  // fp64 SELL fine operator
     hipStream_t st = D->stream;
     double dots[2];
 
@@ -491,7 +524,11 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
 
-    CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
+    if (ksell)
+        CHK(amg_sell_residual_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
+                                  rhs, x, r, st));
+    else
+        CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
     CHK(amg_dot_f64(n, r, r, D->dotbuf_d, st));
     CHK(read_dots(D, 1, dots));
     double res = sqrt(dots[0]);
@@ -509,7 +546,11 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
         } else {
             CHK(amg_axpby_f64(n, 1.0, s, rho1 / rho2, p, st));
         }
-        CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, p, 1.0, 0.0, q, ksubw, st));
+        if (ksell)
+            CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
+                                  p, 1.0, 0.0, q, st));
+        else
+            CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, p, 1.0, 0.0, q, ksubw, st));
         CHK(amg_dot_f64(n, q, p, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
         double alpha = rho1 / dots[0];
@@ -544,6 +585,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
     const int *kcol = D->f32 ? D->a64_col : L0.col;
     const double *kval = D->f32 ? D->a64_val : (const double *)L0.val;
     const int ksubw = D->f32 ? D->a64_subw : L0.subw;
+    const int64_t ksell = D->f32 ? 0 : L0.nslice;  // fp64 SELL fine operator
     hipStream_t st = D->stream;
     double dots[2];
 
@@ -559,7 +601,11 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
 
-    CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
+    if (ksell)
+        CHK(amg_sell_residual_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
+                                  rhs, x, r, st));
+    else
+        CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
     CHK(hipMemcpyAsync(rh, r, n * sizeof(double), hipMemcpyDeviceToDevice, st));
     CHK(amg_dot_f64(n, r, r, D->dotbuf_d, st));
     CHK(read_dots(D, 1, dots));
@@ -583,7 +629,11 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         }
         // v = A (M^-1 p);  T = M^-1 p
         CHK(precond_apply_graphed(D, p, T, T_swap));
-        CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, v, ksubw, st));
+        if (ksell)
+            CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
+                                  T, 1.0, 0.0, v, st));
+        else
+            CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, v, ksubw, st));
         CHK(amg_dot_f64(n, rh, v, D->dotbuf_d, st));
         CHK(read_dots(D, 1, dots));
         alpha = rho1 / dots[0];
@@ -594,7 +644,11 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         res = sqrt(dots[0]);
         if (res > eps) {
             CHK(precond_apply_graphed(D, s2, T, T_swap));
-            CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, t2, ksubw, st));
+            if (ksell)
+                CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol,
+                                      (const double *)L0.sval, T, 1.0, 0.0, t2, st));
+            else
+                CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, t2, ksubw, st));
             CHK(amg_dot2_f64(n, t2, s2, t2, t2, D->dotbuf_d, st));
             CHK(read_dots(D, 2, dots));
             omega = dots[0] / dots[1];

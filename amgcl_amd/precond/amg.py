@@ -49,12 +49,16 @@ class AMG:
             "keep_host_matrices": False,
             "precision": "fp64",  # "mixed" = fp32 hierarchy under fp64 Krylov
             "block_value": 0,      # >1: store level operators as BSR blocks
-            # SELL-64 image of coarse-level operators (wave-native layout;
-            # kernels.hip rationale). "auto" converts HIP levels with
-            # mean row length >= sell_min_mean and >= sell_min_rows rows.
+            # SELL-64 images of level operators (wave-native layout;
+            # kernels.hip rationale).  "auto" converts HIP-resident operators
+            # with >= sell_min_rows rows: measured +25% on the fine level and
+            # +17% on level 1, but a LOSS below ~8k slices (parallelism
+            # starvation), hence the row threshold.  sell_min_mean kept as an
+            # extra gate (0 = off).
             "sell": "auto",
-            "sell_min_rows": 4096,
-            "sell_min_mean": 12.0,
+            "sell_min_rows": 500000,
+            "sell_min_mean": 0.0,
+            "sell_transfers": True,  # also convert P/R transfer operators
         }
 
     def __init__(self, A, prm=None, backend=None):
@@ -88,13 +92,21 @@ class AMG:
             return
         from ..backend.hip import DeviceCSR
 
+        min_rows = int(prm["sell_min_rows"])
+        min_mean = float(prm["sell_min_mean"])
+
+        def eligible(M):
+            if not isinstance(M, DeviceCSR) or M.nrows < min_rows:
+                return False
+            return M.nnz / max(M.nrows, 1) >= min_mean
+
         for lvl in self.levels:
-            A = lvl.A
-            if not isinstance(A, DeviceCSR) or A.nrows < int(prm["sell_min_rows"]):
-                continue
-            if A.nnz / max(A.nrows, 1) < float(prm["sell_min_mean"]):
-                continue
-            A.build_sell()
+            if eligible(lvl.A):
+                lvl.A.build_sell()
+            if prm["sell_transfers"]:
+                for M in (lvl.P, lvl.R):
+                    if M is not None and eligible(M):
+                        M.build_sell()
 
     def rebuild(self, A_new):
         """Reuse the transfer operators for a matrix with changed coefficients
