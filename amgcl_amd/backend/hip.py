@@ -81,8 +81,9 @@ class DeviceCSR:
         soff = torch.zeros(nslice + 1, dtype=torch.int64, device=dev)
         torch.cumsum(w * 64, dim=0, out=soff[1:])
         total = int(soff[-1].item())
-        scol = torch.zeros(total, dtype=torch.int32, device=dev)
-        sval = torch.zeros(total, dtype=self.val.dtype, device=dev)
+        # empty, not zeros: the fill kernel writes the padding itself
+        scol = torch.empty(total, dtype=torch.int32, device=dev)
+        sval = torch.empty(total, dtype=self.val.dtype, device=dev)
         fn = (lib().amg_sell_fill_f32 if self.val.dtype == torch.float32
               else lib().amg_sell_fill_f64)
         check(fn(n, self.ptr.data_ptr(), self.col.data_ptr(), self.val.data_ptr(),

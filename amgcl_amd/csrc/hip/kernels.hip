@@ -253,24 +253,38 @@ extern "C" int amg_sell_relax_f32(int64_t nrows, int64_t nslice, const int64_t *
 }
 #undef SELL_LAUNCH
 
-// CSR -> SELL-64 fill: one thread per row scatters its nonzeros into the
-// column-major slice block (setup-time; write pattern is strided but the
-// kernel runs once per level per setup).
+// CSR -> SELL-64 fill: one wave per slice, iterating the element index i so
+// every store of a wave is fully coalesced (lane-major); the CSR reads are
+// gathers of 64 rows' i-th entries, which land close together for sorted
+// matrices (setup-time cost matters: the fine level alone is ~11 GB).
 template <typename T>
 __global__ void sell_fill_k(int64_t nrows, const int *__restrict__ ptr,
                             const int *__restrict__ col, const T *__restrict__ val,
                             const int64_t *__restrict__ soff, int *__restrict__ scol,
                             T *__restrict__ sval) {
-    int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; row < nrows; row += stride) {
-        int64_t s = row / WAVE;
-        int lane = (int)(row & (WAVE - 1));
-        int64_t base = soff[s] + lane;
-        int b = ptr[row], e = ptr[row + 1];
-        for (int k = 0; k < e - b; ++k) {
-            scol[base + (int64_t)k * WAVE] = col[b + k];
-            sval[base + (int64_t)k * WAVE] = val[b + k];
+    const int wpb = blockDim.x / WAVE;
+    int wid = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
+    int64_t nslice = (nrows + WAVE - 1) / WAVE;
+    int64_t s = (int64_t)blockIdx.x * wpb + wid;
+    int64_t sstride = (int64_t)gridDim.x * wpb;
+    for (; s < nslice; s += sstride) {
+        int64_t row = s * WAVE + lane;
+        int b = 0, len = 0;
+        if (row < nrows) {
+            b = ptr[row];
+            len = ptr[row + 1] - b;
+        }
+        int64_t beg = soff[s];
+        int w = (int)((soff[s + 1] - beg) / WAVE);
+        for (int i = 0; i < w; ++i) {
+            int64_t dst = beg + (int64_t)i * WAVE + lane;
+            if (i < len) {
+                scol[dst] = col[b + i];
+                sval[dst] = val[b + i];
+            } else {  // padding written here so the host can use empty buffers
+                scol[dst] = 0;
+                sval[dst] = (T)0;
+            }
         }
     }
 }
@@ -278,16 +292,20 @@ __global__ void sell_fill_k(int64_t nrows, const int *__restrict__ ptr,
 extern "C" int amg_sell_fill_f64(int64_t nrows, const int *ptr, const int *col,
                                  const double *val, const int64_t *soff, int *scol,
                                  double *sval, hipStream_t stream) {
-    sell_fill_k<double><<<nblocks(nrows), 256, 0, stream>>>(nrows, ptr, col, val, soff,
-                                                            scol, sval);
+    int64_t nslice = (nrows + WAVE - 1) / WAVE;
+    sell_fill_k<double><<<nblocks(nslice * WAVE), 256, 0, stream>>>(nrows, ptr, col,
+                                                                    val, soff, scol,
+                                                                    sval);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_sell_fill_f32(int64_t nrows, const int *ptr, const int *col,
                                  const float *val, const int64_t *soff, int *scol,
                                  float *sval, hipStream_t stream) {
-    sell_fill_k<float><<<nblocks(nrows), 256, 0, stream>>>(nrows, ptr, col, val, soff,
-                                                           scol, sval);
+    int64_t nslice = (nrows + WAVE - 1) / WAVE;
+    sell_fill_k<float><<<nblocks(nslice * WAVE), 256, 0, stream>>>(nrows, ptr, col,
+                                                                   val, soff, scol,
+                                                                   sval);
     return (int)hipGetLastError();
 }
 
