@@ -94,10 +94,38 @@ class CpuBackend:
     def coarse_solver(self, csr: CSR, kind="dense"):
         if kind == "splu":
             return SpluCoarseSolver(csr, self)
+        if kind == "skyline":
+            return SkylineCoarseSolver(csr, self)
         return DenseCoarseSolver(csr, self)
 
     def synchronize(self):
         pass
+
+
+class SkylineCoarseSolver:
+    """Coarsest-level direct solve by skyline (profile) LU after a
+    Cuthill-McKee reorder (parity: amgcl/solver/skyline_lu.hpp:85, the
+    reference's default coarse solver): memory O(profile) instead of the
+    dense inverse's O(n^2), factorization in native C++ (_core)."""
+
+    def __init__(self, csr: CSR, backend=None):
+        from .. import _core
+        from ..adapter import Reordered
+
+        self._re = Reordered(csr)
+        Ap = self._re.A
+        self.n = csr.nrows
+        self._fac = _core.skyline_factor(Ap.nrows, Ap.ptr, Ap.col, Ap.val)
+
+    def __call__(self, f, u):
+        from .. import _core
+
+        y = _core.skyline_solve(*self._fac, self._re.forward(np.asarray(f)))
+        np.copyto(u, self._re.inverse(y))
+
+    def bytes(self):
+        sp = self._fac[0]
+        return int(sp[-1]) * 16 + self.n * 8
 
 
 class SpluCoarseSolver:

@@ -278,6 +278,10 @@ class HipBackend:
 
     # --- coarse direct solver ---------------------------------------------
     def coarse_solver(self, csr, kind="dense"):
+        if kind == "skyline":
+            # host profile-LU with a round-trip per cycle (reference's coarse
+            # shape, hip.hpp:73-96); O(profile) memory vs dense O(n^2)
+            return HostSkylineSolver(csr, self)
         if kind == "splu":
             # host SuperLU with a D2H/H2D round-trip per cycle — the
             # reference's coarse-solve shape (hip.hpp:73-96); only worth it
@@ -289,6 +293,29 @@ class HipBackend:
 
     def synchronize(self):
         self.torch.cuda.synchronize()
+
+
+class HostSkylineSolver:
+    """Host skyline (profile) LU coarse solve with a device round-trip per
+    application (parity: amgcl/solver/skyline_lu.hpp via backend/hip.hpp
+    hip_skyline_lu:73)."""
+
+    def __init__(self, csr, backend):
+        from .cpu import SkylineCoarseSolver
+
+        if isinstance(csr, DeviceCSR):
+            from . import hip_setup
+
+            csr = hip_setup.download(csr)
+        self._s = SkylineCoarseSolver(csr)
+        self.n = csr.nrows
+
+    def __call__(self, f, u):
+        import torch
+
+        out = np.empty(self.n)
+        self._s(f.cpu().numpy().astype(np.float64), out)
+        u.copy_(torch.from_numpy(out).to(u.device, dtype=u.dtype))
 
 
 class HostSpluSolver:

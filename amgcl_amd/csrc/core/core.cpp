@@ -2047,6 +2047,114 @@ static void ilu0_solve(i64 nrows, arr<i32> ptr, arr<i32> col, arr<double> lu,
 // unchanged). Returns (lvl_ptr, rows) with rows bucketed by level, ascending
 // row index inside each level.
 // ---------------------------------------------------------------------------
+
+// ---------------------------------------------------------------------------
+// Skyline (profile) LU coarse solver.
+// Crout factorization inside the symmetric envelope of the (CM-permuted)
+// matrix: L stored by rows (non-unit, diagonal kept inverted in D), U by
+// columns (unit diagonal), sharing one profile pointer array.  Role parity:
+// amgcl/solver/skyline_lu.hpp:85 (same storage scheme, the classic skyline
+// algorithm); memory O(profile) instead of the dense inverse's O(n^2).
+// The caller applies a bandwidth-reducing permutation first.
+static py::tuple skyline_factor(i64 n, arr<i32> ptr_a, arr<i32> col_a,
+                                arr<double> val_a) {
+    auto A = view(n, n, ptr_a, col_a, val_a);
+    // envelope: len[i] = max reach below the diagonal in row i / above in col i
+    std::vector<i64> len(n + 1, 0);
+    for (i64 i = 0; i < n; ++i)
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i64 c = A.col[j];
+            i64 d = i > c ? i - c : c - i;
+            i64 t = i > c ? i : c;
+            if (len[t + 1] < d) len[t + 1] = d;  // profile of index t
+        }
+    arr<i64> sp_a(n + 1);
+    i64 *sp = sp_a.mutable_data();
+    sp[0] = 0;
+    for (i64 i = 0; i < n; ++i) sp[i + 1] = sp[i] + len[i + 1];
+    const i64 total = sp[n];
+    arr<double> L_a(total), U_a(total), D_a(n);
+    double *L = L_a.mutable_data();
+    double *U = U_a.mutable_data();
+    double *D = D_a.mutable_data();
+    std::fill(L, L + total, 0.0);
+    std::fill(U, U + total, 0.0);
+    std::fill(D, D + n, 0.0);
+    // scatter CSR entries into the profile
+    auto lo = [&](i64 i) { return i - (sp[i + 1] - sp[i]); };
+    for (i64 i = 0; i < n; ++i)
+        for (i32 j = A.ptr[i]; j < A.ptr[i + 1]; ++j) {
+            i64 c = A.col[j];
+            double v = A.val[j];
+            if (c == i) D[i] = v;
+            else if (c < i) L[sp[i] + (c - lo(i))] = v;       // row i of L
+            else U[sp[c] + (i - lo(c))] = v;                  // column c of U
+        }
+    // Crout: advance index m; compute U column m, L row m, then D[m]
+    if (D[0] == 0.0) throw std::runtime_error("zero pivot in skyline_lu");
+    D[0] = 1.0 / D[0];
+    for (i64 m = 1; m < n; ++m) {
+        const i64 lom = lo(m);
+        // column m of U: U[i,m] = (A[i,m] - sum_j L[i,j] U[j,m]) * D[i]
+        for (i64 i = lom; i < m; ++i) {
+            const i64 loi = lo(i);
+            i64 jb = lom > loi ? lom : loi;
+            double s = U[sp[m] + (i - lom)];
+            const double *Lr = L + sp[i] - loi;   // Lr[j] = L[i,j]
+            const double *Uc = U + sp[m] - lom;   // Uc[j] = U[j,m]
+            for (i64 j = jb; j < i; ++j) s -= Lr[j] * Uc[j];
+            U[sp[m] + (i - lom)] = s * D[i];
+        }
+        // row m of L: L[m,i] = A[m,i] - sum_j L[m,j] U[j,i]
+        for (i64 i = lom; i < m; ++i) {
+            const i64 loi = lo(i);
+            i64 jb = lom > loi ? lom : loi;
+            double s = L[sp[m] + (i - lom)];
+            const double *Lr = L + sp[m] - lom;   // Lr[j] = L[m,j]
+            const double *Uc = U + sp[i] - loi;   // Uc[j] = U[j,i]
+            for (i64 j = jb; j < i; ++j) s -= Lr[j] * Uc[j];
+            L[sp[m] + (i - lom)] = s;
+        }
+        // pivot
+        double s = D[m];
+        const double *Lr = L + sp[m] - lom;
+        const double *Uc = U + sp[m] - lom;
+        for (i64 j = lom; j < m; ++j) s -= Lr[j] * Uc[j];
+        if (s == 0.0) throw std::runtime_error("zero pivot in skyline_lu");
+        D[m] = 1.0 / s;
+    }
+    return py::make_tuple(sp_a, L_a, U_a, D_a);
+}
+
+static arr<double> skyline_solve(arr<i64> sp_a, arr<double> L_a, arr<double> U_a,
+                                 arr<double> D_a, arr<double> b_a) {
+    const i64 n = (i64)D_a.size();
+    const i64 *sp = sp_a.data();
+    const double *L = L_a.data();
+    const double *U = U_a.data();
+    const double *D = D_a.data();
+    const double *b = b_a.data();
+    arr<double> y_a(n);
+    double *y = y_a.mutable_data();
+    auto lo = [&](i64 i) { return i - (sp[i + 1] - sp[i]); };
+    // forward: y = L^{-1} b (L non-unit, D = inverted diagonal)
+    for (i64 i = 0; i < n; ++i) {
+        double s = b[i];
+        const i64 loi = lo(i);
+        const double *Lr = L + sp[i] - loi;
+        for (i64 j = loi; j < i; ++j) s -= Lr[j] * y[j];
+        y[i] = s * D[i];
+    }
+    // backward: y = U^{-1} y (U unit diagonal, stored by columns)
+    for (i64 c = n - 1; c >= 0; --c) {
+        const i64 loc = lo(c);
+        const double *Uc = U + sp[c] - loc;
+        const double yc = y[c];
+        for (i64 i = loc; i < c; ++i) y[i] -= Uc[i] * yc;
+    }
+    return y_a;
+}
+
 static py::tuple tri_levels(i64 nrows, arr<i32> ptr_a, arr<i32> col_a,
                             arr<i32> dia_a, bool lower) {
     const i32 *P = ptr_a.data(), *C = col_a.data(), *dia = dia_a.data();
@@ -2177,6 +2285,8 @@ PYBIND11_MODULE(_core, m) {
     m.def("spai1", &spai1);
     m.def("ilu0_solve", &ilu0_solve);
     m.def("tri_levels", &tri_levels);
+    m.def("skyline_factor", &skyline_factor);
+    m.def("skyline_solve", &skyline_solve);
     m.def("ilu0_solve_parallel", &ilu0_solve_parallel);
     m.def("gauss_seidel_colored", &gauss_seidel_colored);
     m.def("omp_threads", []() { return omp_get_max_threads(); });
