@@ -61,6 +61,10 @@ _SIGS = {
     "amg_cast_d2s": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     "amg_cast_s2d": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     "amg_gs_color_f64": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
+    # exact level-scheduled triangular solve (cooperative kernel)
+    "amg_sptrsv_f64": [ctypes.c_int64] + [ctypes.c_void_p] * 7
+                      + [ctypes.c_int, ctypes.c_void_p],
+    "amg_coop_supported": [],
     # --- SELL-64 (wave-native sliced-ELL) solve kernels ---
     "amg_sell_spmv_f64": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p,
                           ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,

@@ -310,6 +310,85 @@ extern "C" int amg_sell_fill_f32(int64_t nrows, const int *ptr, const int *col,
 }
 
 // ---------------------------------------------------------------------------
+// Exact sparse triangular solve (level-scheduled), for ILU0 smoothing.
+//
+// The reference's exact GPU ILU0 uses the vendor SpSV machinery
+// (amgcl/relaxation/rocsparse_ilu0.hpp:225-300, hipsparseSpSV).  Here the
+// level schedule (rows grouped by dependency depth, computed on the host by
+// _core.tri_levels) is executed by ONE cooperative kernel that loops over
+// levels with a grid-wide sync between them — no per-level launch storm
+// (a 512^3 ILU0 has ~1500 levels) and no vendor analysis object.
+// In-place on z: lower solve uses the strictly-lower unit-diagonal factor,
+// upper solve the strictly-upper factor with the inverted diagonal.
+// ---------------------------------------------------------------------------
+#include <hip/hip_cooperative_groups.h>
+
+template <typename T, bool LOWER>
+__global__ void sptrsv_levels_k(int64_t nlev, const int *__restrict__ lptr,
+                                const int *__restrict__ rows,
+                                const int64_t *__restrict__ mp,
+                                const int *__restrict__ mc,
+                                const T *__restrict__ mv,
+                                const T *__restrict__ dinv, T *__restrict__ z) {
+    cooperative_groups::grid_group grid = cooperative_groups::this_grid();
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t lev = 0; lev < nlev; ++lev) {
+        for (int64_t r = lptr[lev] + tid; r < lptr[lev + 1]; r += nthreads) {
+            int row = rows[r];
+            double s = (double)z[row];
+            for (int64_t j = mp[row]; j < mp[row + 1]; ++j)
+                s -= (double)mv[j] * (double)z[mc[j]];
+            z[row] = LOWER ? (T)s : (T)((double)dinv[row] * s);
+        }
+        grid.sync();
+    }
+}
+
+template <typename T, bool LOWER>
+static int launch_sptrsv(int64_t nlev, const int *lptr, const int *rows,
+                         const int64_t *mp, const int *mc, const T *mv,
+                         const T *dinv, T *z, hipStream_t stream) {
+    static int grid_blocks = 0;
+    const int block = 256;
+    if (!grid_blocks) {
+        int per_cu = 0, ncu = 0;
+        hipError_t e = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+            &per_cu, (const void *)sptrsv_levels_k<T, LOWER>, block, 0);
+        if (e != hipSuccess) return (int)e;
+        hipDeviceProp_t prop;
+        if (hipGetDeviceProperties(&prop, 0) != hipSuccess) return -1;
+        ncu = prop.multiProcessorCount;
+        grid_blocks = per_cu * ncu;
+        if (grid_blocks < 1) grid_blocks = 1;
+        if (grid_blocks > 2048) grid_blocks = 2048;
+    }
+    void *args[] = {&nlev, (void *)&lptr, (void *)&rows, (void *)&mp,
+                    (void *)&mc, (void *)&mv, (void *)&dinv, (void *)&z};
+    return (int)hipLaunchCooperativeKernel((const void *)sptrsv_levels_k<T, LOWER>,
+                                           dim3(grid_blocks), dim3(block), args, 0,
+                                           stream);
+}
+
+extern "C" int amg_sptrsv_f64(int64_t nlev, const int *lptr, const int *rows,
+                              const int64_t *mp, const int *mc, const double *mv,
+                              const double *dinv, double *z, int lower,
+                              hipStream_t stream) {
+    return lower ? launch_sptrsv<double, true>(nlev, lptr, rows, mp, mc, mv, dinv, z,
+                                               stream)
+                 : launch_sptrsv<double, false>(nlev, lptr, rows, mp, mc, mv, dinv, z,
+                                                stream);
+}
+
+extern "C" int amg_coop_supported() {
+    int v = 0;
+    if (hipDeviceGetAttribute(&v, hipDeviceAttributeCooperativeLaunch, 0) !=
+        hipSuccess)
+        return 0;
+    return v;
+}
+
+// ---------------------------------------------------------------------------
 // Vector primitives (memory-bound, grid-stride; guide App. B elementwise)
 // ---------------------------------------------------------------------------
 template <typename T>

@@ -22,8 +22,12 @@ class ILU0:
         # solve_serial=None: level-scheduled OpenMP triangular sweeps for
         # large factors (bitwise-identical to serial; parity with
         # relaxation/detail/ilu_solve.hpp:257), serial below the threshold
+        # solve="jacobi": the reference's GPU-native iterated approximate
+        # triangular solve (ilu_solve.hpp:44-124). solve="exact": exact
+        # level-scheduled solve by one cooperative GPU kernel (parity with
+        # the vendor-SpSV exact path, amgcl/relaxation/rocsparse_ilu0.hpp).
         return {"damping": 1.0, "solve_iters": 2, "solve_damping": 0.72,
-                "solve_serial": None}
+                "solve_serial": None, "solve": "jacobi"}
 
     def _factor(self, A, p):
         lu, dia = _core.ilu0_factor(A.nrows, A.ptr, A.col, A.val)
@@ -84,6 +88,22 @@ class ILU0:
             self._t0 = backend.vector(self.n)
             self._t1 = backend.vector(self.n)
             self._t2 = backend.vector(self.n)
+            self._exact = str(p["solve"]) == "exact"
+            if self._exact:
+                import torch
+
+                from ..backend._hiplib import lib as _hl
+
+                if not _hl().amg_coop_supported():
+                    raise RuntimeError("exact GPU sptrsv needs cooperative launch")
+                dia32 = np.ascontiguousarray(dia, dtype=np.int32)
+                lp, lr = _core.tri_levels(self.n, fptr, fcol, dia32, True)
+                up, ur = _core.tri_levels(self.n, fptr, fcol, dia32, False)
+                dev = backend.device
+                self._lev = tuple(
+                    torch.from_numpy(np.ascontiguousarray(a)).to(dev)
+                    for a in (lp, lr, up, ur))
+                self._nlev = (len(lp) - 1, len(up) - 1)
 
     def _solve_serial(self, z):
         if self._levels is not None:
@@ -92,6 +112,23 @@ class ILU0:
                                       self._dia32, lp, lr, up, ur, z)
         else:
             _core.ilu0_solve(self.n, self.ptr, self.col, self.lu, self.dia, z)
+
+    def _solve_exact(self, z):
+        """Exact level-scheduled triangular solves on the GPU (one
+        cooperative kernel per factor; bitwise semantics of the serial
+        sweeps — see kernels.hip sptrsv_levels_k)."""
+        from ..backend._hiplib import check, lib
+        from ..backend.hip import _stream
+
+        lp, lr, up, ur = self._lev
+        check(lib().amg_sptrsv_f64(self._nlev[0], lp.data_ptr(), lr.data_ptr(),
+                                   self.L.ptr.data_ptr(), self.L.col.data_ptr(),
+                                   self.L.val.data_ptr(), 0, z.data_ptr(), 1,
+                                   _stream()), "sptrsv_lower")
+        check(lib().amg_sptrsv_f64(self._nlev[1], up.data_ptr(), ur.data_ptr(),
+                                   self.U.ptr.data_ptr(), self.U.col.data_ptr(),
+                                   self.U.val.data_ptr(), self.Dinv.data_ptr(),
+                                   z.data_ptr(), 0, _stream()), "sptrsv_upper")
 
     def _solve_jacobi(self, z):
         """Damped-Jacobi iterated approximate triangular solves
@@ -120,6 +157,8 @@ class ILU0:
         b.residual(rhs, A, x, tmp)
         if self._serial:
             self._solve_serial(tmp)
+        elif getattr(self, "_exact", False):
+            self._solve_exact(tmp)
         else:
             self._solve_jacobi(tmp)
         b.axpby(self.damping, tmp, 1.0, x)
@@ -135,6 +174,8 @@ class ILU0:
         b.copy(rhs, x)
         if self._serial:
             self._solve_serial(x)
+        elif getattr(self, "_exact", False):
+            self._solve_exact(x)
         else:
             self._solve_jacobi(x)
 

@@ -307,3 +307,46 @@ def test_sell_in_amg_solve(hip):
     assert r2 < 1e-8
     xa = hip.to_host(x2) if not isinstance(x2, np.ndarray) else x2
     assert np.linalg.norm(b - A @ xa) / np.linalg.norm(b) < 1e-7
+
+
+def test_ilu0_exact_gpu_sptrsv(hip):
+    """Exact level-scheduled GPU triangular solve (cooperative kernel)
+    reproduces the serial host ILU0 application to 1e-12 — the reference's
+    vendor-SpSV exact path (amgcl/relaxation/rocsparse_ilu0.hpp:225-300)
+    rebuilt without a vendor analysis object."""
+    from amgcl_amd.relaxation.ilu0 import ILU0
+    from amgcl_amd.backend import make_backend
+
+    A, _ = am.poisson3d(16)
+    rng = np.random.default_rng(5)
+    z = rng.standard_normal(A.nrows)
+
+    cpu = make_backend("cpu")
+    r_cpu = ILU0(A, {"solve_serial": True}, cpu)
+    x_cpu = z.copy()
+    r_cpu._solve_serial(x_cpu)
+
+    r_gpu = ILU0(A, {"solve": "exact"}, hip)
+    zd = hip.from_host(z)
+    r_gpu._solve_exact(zd)
+    np.testing.assert_allclose(hip.to_host(zd), x_cpu, rtol=1e-12, atol=1e-12)
+
+
+def test_ilu0_exact_vs_jacobi_iterations(hip):
+    """On an anisotropic problem the exact triangular solve must be at
+    least as strong per iteration as the approximate Jacobi variant."""
+    A, b = am.poisson3d(24, rhs="random", anisotropy=(1.0, 1.0, 100.0))
+
+    def solve(solve_kind):
+        s = am.make_solver(
+            A, {"precond": {"class": "relaxation",
+                            "type": "ilu0", "solve": solve_kind},
+                "solver": {"type": "bicgstab", "tol": 1e-8, "maxiter": 500}},
+            backend=hip)
+        x, iters, resid = s(b)
+        assert resid < 1e-8
+        return iters
+
+    it_exact = solve("exact")
+    it_jac = solve("jacobi")
+    assert it_exact <= it_jac + 1, (it_exact, it_jac)
