@@ -87,6 +87,9 @@ _SIGS = {
     "amg_bsr_residual_f64": [ctypes.c_int64, ctypes.c_int] + [ctypes.c_void_p] * 7,
     "amg_bsr_relax_f64": [ctypes.c_int64, ctypes.c_int] + [ctypes.c_void_p] * 8,
     "amg_blkdiag_vmul_f64": [ctypes.c_int64, ctypes.c_int] + [ctypes.c_void_p] * 4,
+    "amg_bsr_spmv_mfma4_f64": [ctypes.c_int64] + [ctypes.c_void_p] * 4
+                              + [ctypes.c_double, ctypes.c_double]
+                              + [ctypes.c_void_p] * 2,
     # --- device-side setup engine (setup.hip) ---
     "amg_setup_diag": [ctypes.c_int64] + [ctypes.c_void_p] * 5,
     "amg_setup_strong": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,

@@ -164,3 +164,90 @@ extern "C" int amg_blkdiag_vmul_f64(int64_t nbrows, int bsize, const double *M,
     BSR_LAUNCH(blkdiag_vmul_k, nbrows, M, x, y)
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// MFMA variant of the B=4 BSR SpMV (the VERDICT-mandated head-to-head;
+// reference analogue of the question: vexcl_static_matrix block kernels).
+//
+// Mapping: one wave handles FOUR 4x4 block rows per step of
+// v_mfma_f64_16x16x4_f64 (D[16x16] += A[16x4]*B[4x16]).  The M dimension
+// carries the 16 scalar rows of the 4 block rows, K the 4 scalar columns of
+// each block-row's current BSR block, and the B operand replicates each
+// block row's gathered x block across its 4 N-columns, so the needed
+// results are the diagonal D[m][m].  15/16 of the MFMA lanes compute
+// unused dot products — the point of the experiment: flops are free at
+// AI 0.25 flop/byte, so IF the kernel is purely bandwidth-bound the MFMA
+// version should tie the unrolled one.  Measured result is recorded in
+// profiles/README.md (round 2).
+// Fragment layout (16x16x4 f64): A: lane l holds A[l%16][l/16]; B: lane l
+// holds B[l/16][l%16]; D: lane l item i holds D[4*(l/16)+i][l%16].
+// ---------------------------------------------------------------------------
+typedef double d4_t __attribute__((ext_vector_type(4)));
+
+template <bool BETA0>
+__global__ void bsr_spmv_mfma4_k(int64_t nbrows, const int *__restrict__ ptr,
+                                 const int *__restrict__ col,
+                                 const double *__restrict__ val,
+                                 const double *__restrict__ x, double alpha,
+                                 double beta, double *__restrict__ y) {
+    const int wpb = blockDim.x / 64;
+    const int wid = threadIdx.x / 64, lane = threadIdx.x & 63;
+    const int m = lane & 15;          // scalar row within the 4-block group
+    const int k = lane >> 4;          // scalar column within a block
+    const int ga = m >> 2;            // block row (A operand) within group
+    const int ra = m & 3;             // scalar row within that block
+    const int gb = m >> 2;            // for B operand: n = lane&15 -> group n/4
+    int64_t grp = (int64_t)blockIdx.x * wpb + wid;
+    const int64_t ngrp = (nbrows + 3) >> 2;
+    const int64_t gstride = (int64_t)gridDim.x * wpb;
+    for (; grp < ngrp; grp += gstride) {
+        const int64_t brow0 = grp << 2;
+        // this lane's A block row and B block row current positions
+        const int64_t browA = brow0 + ga;
+        const int64_t browB = brow0 + gb;  // == browA (n==m grouping)
+        int begA = 0, lenA = 0;
+        if (browA < nbrows) {
+            begA = ptr[browA];
+            lenA = ptr[browA + 1] - begA;
+        }
+        int maxlen = lenA;
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1) {
+            int o = __shfl_down(maxlen, off);
+            if (o > maxlen) maxlen = o;
+        }
+        maxlen = __shfl(maxlen, 0);
+        d4_t acc = {0.0, 0.0, 0.0, 0.0};
+        for (int t = 0; t < maxlen; ++t) {
+            double a = 0.0, b = 0.0;
+            if (t < lenA) {
+                const int64_t j = begA + t;
+                a = val[j * 16 + ra * 4 + k];
+                b = x[(int64_t)col[j] * 4 + k];
+            }
+            acc = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, acc, 0, 0, 0);
+        }
+        // extract the diagonal D[m][m]: lane l item i holds D[4*(l>>4)+i][l&15]
+        const int n = lane & 15;
+        const int i = n - 4 * (lane >> 4);
+        if (i >= 0 && i < 4) {
+            const int64_t row = brow0 * 4 + n;
+            if (row < nbrows * 4)
+                y[row] = BETA0 ? alpha * acc[i] : alpha * acc[i] + beta * y[row];
+        }
+    }
+}
+
+extern "C" int amg_bsr_spmv_mfma4_f64(int64_t nbrows, const int *ptr, const int *col,
+                                      const double *val, const double *x, double alpha,
+                                      double beta, double *y, hipStream_t s) {
+    int64_t ngrp = (nbrows + 3) >> 2;
+    int grid = nblk_b(ngrp * 64);
+    if (beta == 0.0)
+        bsr_spmv_mfma4_k<true><<<grid, 256, 0, s>>>(nbrows, ptr, col, val, x, alpha,
+                                                    beta, y);
+    else
+        bsr_spmv_mfma4_k<false><<<grid, 256, 0, s>>>(nbrows, ptr, col, val, x, alpha,
+                                                     beta, y);
+    return (int)hipGetLastError();
+}
