@@ -179,8 +179,11 @@ extern "C" int amg_blkdiag_vmul_f64(int64_t nbrows, int bsize, const double *M,
 // AI 0.25 flop/byte, so IF the kernel is purely bandwidth-bound the MFMA
 // version should tie the unrolled one.  Measured result is recorded in
 // profiles/README.md (round 2).
-// Fragment layout (16x16x4 f64): A: lane l holds A[l%16][l/16]; B: lane l
-// holds B[l/16][l%16]; D: lane l item i holds D[4*(l/16)+i][l%16].
+// Fragment layout for v_mfma_f64_16x16x4_f64, decoded EMPIRICALLY on
+// gfx950 (scripts/mfma_probe, gpurun_out/mfma_probe.log):
+//   A: lane l holds A[4*(l&3) + ((l>>2)&3)][l>>4]   (4x4-transposed sub-index)
+//   B: lane l holds B[l>>4][l&15]
+//   D: lane l item i holds D[4*(l>>4)+i][l&15]
 // ---------------------------------------------------------------------------
 typedef double d4_t __attribute__((ext_vector_type(4)));
 
@@ -192,23 +195,27 @@ __global__ void bsr_spmv_mfma4_k(int64_t nbrows, const int *__restrict__ ptr,
                                  double beta, double *__restrict__ y) {
     const int wpb = blockDim.x / 64;
     const int wid = threadIdx.x / 64, lane = threadIdx.x & 63;
-    const int m = lane & 15;          // scalar row within the 4-block group
-    const int k = lane >> 4;          // scalar column within a block
-    const int ga = m >> 2;            // block row (A operand) within group
-    const int ra = m & 3;             // scalar row within that block
-    const int gb = m >> 2;            // for B operand: n = lane&15 -> group n/4
+    const int sub = lane & 15;
+    const int k = lane >> 4;               // scalar column within a block
+    const int ma = 4 * (sub & 3) + (sub >> 2);  // A row this lane feeds
+    const int ga = ma >> 2;                // A-side block row within group
+    const int ra = ma & 3;                 // scalar row within that block
+    const int gb = sub >> 2;               // B-side: col n = sub -> group n/4
     int64_t grp = (int64_t)blockIdx.x * wpb + wid;
     const int64_t ngrp = (nbrows + 3) >> 2;
     const int64_t gstride = (int64_t)gridDim.x * wpb;
     for (; grp < ngrp; grp += gstride) {
         const int64_t brow0 = grp << 2;
-        // this lane's A block row and B block row current positions
         const int64_t browA = brow0 + ga;
-        const int64_t browB = brow0 + gb;  // == browA (n==m grouping)
-        int begA = 0, lenA = 0;
+        const int64_t browB = brow0 + gb;
+        int begA = 0, lenA = 0, begB = 0, lenB = 0;
         if (browA < nbrows) {
             begA = ptr[browA];
             lenA = ptr[browA + 1] - begA;
+        }
+        if (browB < nbrows) {
+            begB = ptr[browB];
+            lenB = ptr[browB + 1] - begB;
         }
         int maxlen = lenA;
 #pragma unroll
@@ -220,11 +227,8 @@ __global__ void bsr_spmv_mfma4_k(int64_t nbrows, const int *__restrict__ ptr,
         d4_t acc = {0.0, 0.0, 0.0, 0.0};
         for (int t = 0; t < maxlen; ++t) {
             double a = 0.0, b = 0.0;
-            if (t < lenA) {
-                const int64_t j = begA + t;
-                a = val[j * 16 + ra * 4 + k];
-                b = x[(int64_t)col[j] * 4 + k];
-            }
+            if (t < lenA) a = val[(int64_t)(begA + t) * 16 + ra * 4 + k];
+            if (t < lenB) b = x[(int64_t)col[begB + t] * 4 + k];
             acc = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, acc, 0, 0, 0);
         }
         // extract the diagonal D[m][m]: lane l item i holds D[4*(l>>4)+i][l&15]

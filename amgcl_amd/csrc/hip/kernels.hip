@@ -326,7 +326,7 @@ extern "C" int amg_sell_fill_f32(int64_t nrows, const int *ptr, const int *col,
 template <typename T, bool LOWER>
 __global__ void sptrsv_levels_k(int64_t nlev, const int *__restrict__ lptr,
                                 const int *__restrict__ rows,
-                                const int64_t *__restrict__ mp,
+                                const int *__restrict__ mp,
                                 const int *__restrict__ mc,
                                 const T *__restrict__ mv,
                                 const T *__restrict__ dinv, T *__restrict__ z) {
@@ -337,7 +337,7 @@ __global__ void sptrsv_levels_k(int64_t nlev, const int *__restrict__ lptr,
         for (int64_t r = lptr[lev] + tid; r < lptr[lev + 1]; r += nthreads) {
             int row = rows[r];
             double s = (double)z[row];
-            for (int64_t j = mp[row]; j < mp[row + 1]; ++j)
+            for (int j = mp[row]; j < mp[row + 1]; ++j)
                 s -= (double)mv[j] * (double)z[mc[j]];
             z[row] = LOWER ? (T)s : (T)((double)dinv[row] * s);
         }
@@ -347,7 +347,7 @@ __global__ void sptrsv_levels_k(int64_t nlev, const int *__restrict__ lptr,
 
 template <typename T, bool LOWER>
 static int launch_sptrsv(int64_t nlev, const int *lptr, const int *rows,
-                         const int64_t *mp, const int *mc, const T *mv,
+                         const int *mp, const int *mc, const T *mv,
                          const T *dinv, T *z, hipStream_t stream) {
     static int grid_blocks = 0;
     const int block = 256;
@@ -371,7 +371,7 @@ static int launch_sptrsv(int64_t nlev, const int *lptr, const int *rows,
 }
 
 extern "C" int amg_sptrsv_f64(int64_t nlev, const int *lptr, const int *rows,
-                              const int64_t *mp, const int *mc, const double *mv,
+                              const int *mp, const int *mc, const double *mv,
                               const double *dinv, double *z, int lower,
                               hipStream_t stream) {
     return lower ? launch_sptrsv<double, true>(nlev, lptr, rows, mp, mc, mv, dinv, z,

@@ -335,7 +335,7 @@ def test_ilu0_exact_gpu_sptrsv(hip):
 def test_ilu0_exact_vs_jacobi_iterations(hip):
     """On an anisotropic problem the exact triangular solve must be at
     least as strong per iteration as the approximate Jacobi variant."""
-    A, b = am.poisson3d(24, rhs="random", anisotropy=(1.0, 1.0, 100.0))
+    A, b = am.poisson3d(24, rhs="random", anisotropy=100.0)
 
     def solve(solve_kind):
         s = am.make_solver(
