@@ -7,10 +7,11 @@
 // Design note (MI355X): fp64 BSR SpMV is memory-bandwidth-bound — a BxB
 // block contributes 2*B*B flops for 8*B*B value bytes, i.e. arithmetic
 // intensity 0.25 flop/byte, ~1e-3 of the fp64 compute:bandwidth balance
-// point. Matrix cores (MFMA) therefore cannot speed this kernel up; the
-// speed-of-light design is maximal-bandwidth block loads (val rows
-// contiguous, one scalar row per lane within the block row) plus the x-reuse
-// that BSR gives for free (each x block is loaded once per block row).
+// point.  MEASURED (round 2, profiles/mfma_vs_unrolled_bsr_r02.log): a
+// correct v_mfma_f64_16x16x4_f64 variant (below) reaches 2586 GB/s vs
+// 5680 GB/s for these unrolled kernels on the elasticity config — the
+// unrolled form IS the speed-of-light design (val rows contiguous, one
+// scalar row per lane within the block row, x-block reuse for free).
 //
 // Layout: block values row-major within each block, blocks in CSR order.
 // B lanes cooperate on one block row (lane = scalar row within the block).
