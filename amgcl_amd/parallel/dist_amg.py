@@ -52,6 +52,13 @@ class DistAMG:
             "coarse_enough": 3000,
             "repart_threshold": 20000,  # replicate the hierarchy below this
             "max_levels": 20,
+            # >1: only every coarse_group_size-th rank (the group master)
+            # builds and applies the coarse tail; slaves receive the solution
+            # by an intra-group broadcast.  Parity role:
+            # amgcl/mpi/direct_solver/solver_base.hpp:69-163
+            # (slaves_per_master comm split) — tail memory O(n^2) only on
+            # masters instead of every rank.
+            "coarse_group_size": 1,
         }
 
     def __init__(self, dist_A: DistMatrix, prm=None, backend=None):
@@ -217,6 +224,21 @@ class DistAMG:
         continues the hierarchy redundantly (merge-repartition semantics)."""
         import scipy.sparse as sp
 
+        g = int(self.prm["coarse_group_size"])
+        self._cg_master = None
+        self._cg_group = None
+        self._cg_rank = A.rank
+        is_master = True
+        if g > 1 and A.world > 1:
+            my_master = (A.rank // g) * g
+            for m0 in range(0, A.world, g):
+                ranks = list(range(m0, min(m0 + g, A.world)))
+                grp = self.dist.new_group(ranks=ranks)
+                if m0 == my_master:
+                    self._cg_group = grp
+            self._cg_master = my_master
+            is_master = A.rank == my_master
+
         loc, rem = self._strip_scipy(A)
         # re-express the strip with global columns
         cols_g = np.asarray(A.A_loc_host.col, dtype=np.int64) + A.row_beg
@@ -233,12 +255,20 @@ class DistAMG:
         self.dist.all_gather_object(
             gathered, (strip.indptr, strip.indices, strip.data),
             group=self.group)
-        rows = [sp.csr_matrix((v, c, p), shape=(len(p) - 1, A.n_global))
-                for p, c, v in gathered]
-        G = sp.vstack(rows, format="csr")
-        G.sort_indices()
-        G_csr = CSR.from_scipy(G)
-        if as_amg_tail:
+        if not is_master:
+            # slaves hold no coarse system at all (subcommunicator mode)
+            self.coarse = None
+            self.tail = None
+            G_csr = None
+        else:
+            rows = [sp.csr_matrix((v, c, p), shape=(len(p) - 1, A.n_global))
+                    for p, c, v in gathered]
+            G = sp.vstack(rows, format="csr")
+            G.sort_indices()
+            G_csr = CSR.from_scipy(G)
+        if not is_master:
+            pass
+        elif as_amg_tail:
             # replicated serial AMG tail: every rank continues the SAME
             # hierarchy below this level and applies it redundantly
             from ..precond import make_preconditioner
@@ -258,6 +288,7 @@ class DistAMG:
         else:
             self.coarse = self.base.coarse_solver(G_csr)
             self.tail = None
+        self._is_cg_master = is_master
         self._coarse_n = A.n_global
         self._coarse_sizes = [len(p) - 1 for p, _, _ in gathered]
         self._coarse_beg = int(np.sum(self._coarse_sizes[: A.rank]))
@@ -293,7 +324,17 @@ class DistAMG:
         for r, sz in enumerate(self._coarse_sizes):
             gf[off : off + sz] = self._gbuf[r * pad : r * pad + sz]
             off += sz
-        if self.coarse is not None:
+        if self._cg_master is not None:
+            # subcommunicator mode: only the group master solves; the group
+            # receives the global solution by broadcast (solver_base shape)
+            if self._is_cg_master:
+                if self.coarse is not None:
+                    self.coarse(self._gf, self._gu)
+                else:
+                    self.tail.apply(self._gf, self._gu)
+            self.dist.broadcast(self._as_tensor(self._gu), src=self._cg_master,
+                                group=self._cg_group)
+        elif self.coarse is not None:
             self.coarse(self._gf, self._gu)
         else:
             self.tail.apply(self._gf, self._gu)  # replicated hierarchy tail

@@ -895,3 +895,42 @@ def test_graph_partitioned_distributed_solve(world):
     x = np.empty(n ** 3)
     x[perm] = np.asarray(xg)
     assert np.linalg.norm(b - A @ x) / np.linalg.norm(b) < 1e-7
+
+
+def _solve_dist_amg_group_coarse(rank, world):
+    import amgcl_amd as am
+    from amgcl_amd.parallel import make_dist_solver
+
+    n = 16
+    strip, b, row_beg, row_end = am.poisson3d_strip(n, rank, world, rhs="ones")
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "dist_amg", "coarse_enough": 400,
+                     "coarse_group_size": 2},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+        backend="cpu")
+    x, iters, resid = solve(b)
+    xg = solve.gather_solution(x)
+    P = solve.P
+    has_tail = P.coarse is not None or P.tail is not None
+    return iters, resid, None if xg is None else xg.tolist(), has_tail
+
+
+@pytest.mark.parametrize("world", [4])
+def test_dist_amg_subcommunicator_coarse(world):
+    """Subcommunicator coarse solve (solver_base.hpp shape): only group
+    masters hold the coarse tail; slaves receive the solution by an
+    intra-group broadcast.  Same convergence as the replicated path."""
+    out = spawn(world, _solve_dist_amg_group_coarse, 30511 + world)
+    import amgcl_amd as am
+
+    n = 16
+    A, b = am.poisson3d(n, rhs="ones")
+    iters, resid, xg, _ = out[0]
+    assert resid < 1e-8
+    xg = np.array(xg)
+    assert np.linalg.norm(b - A @ xg) / np.linalg.norm(b) < 1e-7
+    # masters (ranks 0, 2) hold the tail, slaves (1, 3) hold nothing
+    tails = {r: out[r][3] for r in range(world)}
+    assert tails[0] and tails[2]
+    assert not tails[1] and not tails[3]
