@@ -20,6 +20,10 @@ REGISTRY = {
 def make_coarsening(prm=None):
     prm = dict(prm or {})
     kind = prm.pop("type", "smoothed_aggregation")
+    if kind == "as_scalar":
+        from .as_scalar import AsScalar
+
+        return AsScalar(prm)
     if kind not in REGISTRY:
         raise ValueError(f"unknown coarsening '{kind}'")
     return REGISTRY[kind](prm)

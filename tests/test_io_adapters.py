@@ -254,3 +254,32 @@ def test_binary_reference_layout(tmp_path):
     S = io.read_crs(p, row_beg=5, row_end=25)
     assert S.nrows == 20
     assert abs(S.to_scipy() - A.to_scipy()[5:25]).max() < 1e-15
+
+
+def test_unblock_bsr_and_as_scalar():
+    """Block-valued (BSR) input path (parity: amgcl/coarsening/as_scalar.hpp
+    + adapter/block_matrix.hpp direction): unblock a BSR matrix to scalar
+    CSR exactly, and run AMG through the registry's as_scalar wrapper."""
+    import scipy.sparse as sp
+
+    from amgcl_amd.coarsening.as_scalar import unblock_bsr
+
+    rng = np.random.default_rng(0)
+    nb, B = 40, 3
+    a = sp.random(nb, nb, density=0.1, random_state=rng, format="csr")
+    a = (a + sp.identity(nb)).tocsr()
+    a.data[:] = 1.0
+    blocks = rng.standard_normal((a.nnz, B, B))
+    m = sp.bsr_matrix((blocks, a.indices, a.indptr), shape=(nb * B, nb * B)).tocsr()
+    A = unblock_bsr(nb, B, a.indptr, a.indices, blocks)
+    assert abs(A.to_scipy() - m).max() < 1e-15
+
+    Ah, b = am.poisson3d(16, rhs="ones")
+    s = am.make_solver(
+        Ah, {"precond": {"class": "amg", "coarse_enough": 300,
+                         "coarsening": {"type": "as_scalar",
+                                        "base": {"type": "smoothed_aggregation"}}},
+             "solver": {"type": "cg", "tol": 1e-8}})
+    x, it, r = s(b)
+    assert r < 1e-8
+    assert np.linalg.norm(b - Ah @ x) / np.linalg.norm(b) < 1e-7
