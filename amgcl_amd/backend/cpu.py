@@ -18,9 +18,9 @@ class CpuBackend:
     device = "cpu"
 
     def __init__(self, dtype=np.float64):
-        if dtype not in (np.float64,):
-            raise ValueError("cpu backend is fp64 (reference parity)")
-        self.dtype = np.float64
+        if dtype not in (np.float64, np.complex128):
+            raise ValueError("cpu backend supports fp64 and complex128")
+        self.dtype = dtype
 
     # --- containers -------------------------------------------------------
     def matrix(self, csr: CSR):
@@ -37,9 +37,15 @@ class CpuBackend:
 
     # --- primitives (amgcl/backend/interface.hpp:253-443) ----------------
     def spmv(self, alpha, A, x, beta, y):
+        if np.iscomplexobj(A.val) or np.iscomplexobj(x):
+            A.spmv(alpha, x, beta, y)
+            return
         _core.spmv(alpha, A.nrows, A.ptr, A.col, A.val, x, beta, y)
 
     def residual(self, b, A, x, r):
+        if np.iscomplexobj(A.val) or np.iscomplexobj(x):
+            A.residual(b, x, r)
+            return
         _core.residual(A.nrows, A.ptr, A.col, A.val, b, x, r)
 
     def clear(self, x):
@@ -49,12 +55,18 @@ class CpuBackend:
         np.copyto(y, x)
 
     def dot(self, x, y):
+        if np.iscomplexobj(x) or np.iscomplexobj(y):
+            # adjoint inner product (math::inner_product conjugates the
+            # first argument for complex value types)
+            return complex(np.vdot(x, y))
         return float(np.dot(x, y))
 
     def dot2(self, x1, y1, x2, y2):
-        return float(np.dot(x1, y1)), float(np.dot(x2, y2))
+        return self.dot(x1, y1), self.dot(x2, y2)
 
     def norm(self, x):
+        if np.iscomplexobj(x):
+            return float(np.linalg.norm(x))
         return float(np.sqrt(np.dot(x, x)))
 
     def axpby(self, a, x, b, y):

@@ -36,6 +36,8 @@ class SmoothedAggregation:
 
         if not isinstance(A, CSR):
             return self._transfer_operators_device(A)
+        if A.is_complex:
+            return self._transfer_operators_complex(A)
         with prof.scope("aggregates"):
             # large levels: deterministic parallel MIS(2) aggregation
             # (mirrors the reference's own distributed pmis design);
@@ -92,6 +94,49 @@ class SmoothedAggregation:
         with prof.scope("transpose_R"):
             R = P.transpose()
         return P, R
+
+    def _transfer_operators_complex(self, A):
+        """Native complex-valued coarsening (parity:
+        amgcl/value_type/complex.hpp — the reference instantiates the same
+        templates over std::complex).  Strength uses |a_ij|^2 >
+        eps^2 |a_ii||a_jj| (the complex specialization of the criterion);
+        aggregation runs on the magnitude matrix; P is Jacobi-smoothed with
+        complex arithmetic; R = P^H (adjoint)."""
+        import scipy.sparse as sp
+
+        if int(self.prm["block_size"]) > 1 or self.B is not None:
+            raise ValueError("complex path supports scalar SA without nullspace")
+        n = A.nrows
+        mag = np.abs(A.val)
+        agg = _core.aggregates_parallel if n > 100_000 else _core.aggregates
+        naggr, aggr_id, strong = agg(n, A.ptr, A.col, mag, self.eps_strong)
+        self.eps_strong *= 0.5
+        aggr_id = np.asarray(aggr_id)
+        strong = np.asarray(strong).astype(bool)
+        omega = float(self.prm["relax"]) * (2.0 / 3.0)
+
+        m = A.to_scipy().tocsr()
+        row_of = np.repeat(np.arange(n), np.diff(m.indptr))
+        offdiag = m.indices != row_of
+        # filtered matrix: weak off-diagonal entries folded into the diagonal
+        weak = offdiag & ~strong
+        dia = np.asarray(m.diagonal()).copy()
+        np.add.at(dia, row_of[weak], m.data[weak])
+        keep = strong.copy()
+        Af = sp.csr_matrix((m.data[keep], m.indices[keep],
+                            np.concatenate([[0], np.cumsum(
+                                np.bincount(row_of[keep], minlength=n))])),
+                           shape=(n, n))
+        # P_tent: one 1 per row at its aggregate (removed rows empty)
+        ok = aggr_id >= 0
+        Pt = sp.csr_matrix((np.ones(ok.sum()), (np.flatnonzero(ok), aggr_id[ok])),
+                           shape=(n, naggr))
+        dinv = np.where(dia != 0, 1.0 / np.where(dia == 0, 1, dia), 0)
+        S = sp.identity(n, format="csr") - sp.diags(omega * dinv) @ Af
+        P = (S @ Pt).tocsr()
+        P.sort_indices()
+        Pc = CSR.from_scipy(P)
+        return Pc, Pc.transpose()  # transpose() is the adjoint for complex
 
     def _transfer_operators_device(self, A):
         """Device twin (backend/hip_setup.py): same algorithm, same keys.

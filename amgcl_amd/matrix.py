@@ -37,10 +37,23 @@ class CSR:
     def bytes(self):
         return self.ptr.nbytes + self.col.nbytes + self.val.nbytes
 
+    @property
+    def is_complex(self):
+        return np.iscomplexobj(self.val)
+
     def diagonal(self):
+        if self.is_complex:
+            return self.to_scipy().diagonal()
         return _core.diagonal(self.nrows, self.ptr, self.col, self.val)
 
     def transpose(self):
+        """Adjoint transpose (parity: backend::transpose applies
+        math::adjoint — for complex values this is the conjugate
+        transpose, amgcl/backend/builtin.hpp:348)."""
+        if self.is_complex:
+            m = self.to_scipy().conj().T.tocsr()
+            m.sort_indices()
+            return CSR.from_scipy(m)
         tp, tc, tv = _core.transpose(self.nrows, self.ncols, self.ptr, self.col, self.val)
         return CSR(self.ncols, self.nrows, tp, tc, tv)
 
@@ -48,21 +61,37 @@ class CSR:
         if isinstance(other, CSR):
             if self.ncols != other.nrows:
                 raise ValueError("dimension mismatch in CSR product")
+            if self.is_complex or other.is_complex:
+                m = (self.to_scipy() @ other.to_scipy()).tocsr()
+                m.sort_indices()
+                return CSR.from_scipy(m)
             cp, cc, cv = _core.spgemm(
                 self.nrows, self.ncols, other.ncols,
                 self.ptr, self.col, self.val,
                 other.ptr, other.col, other.val,
             )
             return CSR(self.nrows, other.ncols, cp, cc, cv)
+        if self.is_complex or np.iscomplexobj(other):
+            return self.to_scipy() @ np.asarray(other)
         x = np.ascontiguousarray(other, dtype=np.float64)
         y = np.empty(self.nrows, dtype=np.float64)
         _core.spmv(1.0, self.nrows, self.ptr, self.col, self.val, x, 0.0, y)
         return y
 
     def spmv(self, alpha, x, beta, y):
+        if self.is_complex or np.iscomplexobj(x):
+            if beta == 0.0:
+                np.copyto(y, alpha * (self.to_scipy() @ x))
+            else:
+                y *= beta
+                y += alpha * (self.to_scipy() @ x)
+            return
         _core.spmv(alpha, self.nrows, self.ptr, self.col, self.val, x, beta, y)
 
     def residual(self, b, x, r):
+        if self.is_complex or np.iscomplexobj(x):
+            np.copyto(r, b - self.to_scipy() @ x)
+            return
         _core.residual(self.nrows, self.ptr, self.col, self.val, b, x, r)
 
     def to_scipy(self):

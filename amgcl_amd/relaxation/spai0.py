@@ -53,7 +53,17 @@ class Spai0(DiagonalSmootherBase):
 
         merge_params(self.defaults(), prm)
         if isinstance(A, CSR):
-            m = _core.spai0(A.nrows, A.ptr, A.col, A.val)
+            if A.is_complex:
+                # complex SPAI-0: m_i = conj(a_ii) / sum_j |a_ij|^2
+                # (minimizes ||I - M A||_F with diagonal M over C)
+                import numpy as np
+
+                row_of = np.repeat(np.arange(A.nrows), np.diff(A.ptr))
+                s = np.zeros(A.nrows)
+                np.add.at(s, row_of, np.abs(A.val) ** 2)
+                m = np.conj(np.asarray(A.diagonal())) / np.where(s == 0, 1, s)
+            else:
+                m = _core.spai0(A.nrows, A.ptr, A.col, A.val)
         else:  # device-resident level matrix: compute weights on device
             from ..backend import hip_setup
 

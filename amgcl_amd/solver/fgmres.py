@@ -24,10 +24,12 @@ class FGMRES(SolverBase):
         self.w = b.vector(n)
         self.v = [b.vector(n) for _ in range(m + 1)]
         self.z = [b.vector(n) for _ in range(m)]
-        self.H = np.zeros((m + 1, m))
-        self.cs = np.zeros(m + 1)
-        self.sn = np.zeros(m + 1)
-        self.g = np.zeros(m + 1)
+        dt = getattr(backend, "dtype", np.float64)
+        self._cplx = np.issubdtype(np.dtype(dt), np.complexfloating)
+        self.H = np.zeros((m + 1, m), dtype=dt)
+        self.cs = np.zeros(m + 1)  # real also in the complex rotation
+        self.sn = np.zeros(m + 1, dtype=dt)
+        self.g = np.zeros(m + 1, dtype=dt)
 
     def solve(self, A, P, rhs, x):
         b = self.backend
@@ -64,24 +66,40 @@ class FGMRES(SolverBase):
                 self.H[i + 1, i] = hn
                 if hn != 0.0:
                     b.axpby(1.0 / hn, self.w, 0.0, self.v[i + 1])
-                for k in range(i):
-                    h0 = self.cs[k] * self.H[k, i] + self.sn[k] * self.H[k + 1, i]
-                    h1 = -self.sn[k] * self.H[k, i] + self.cs[k] * self.H[k + 1, i]
-                    self.H[k, i], self.H[k + 1, i] = h0, h1
-                d = math.hypot(self.H[i, i], self.H[i + 1, i])
-                self.cs[i] = self.H[i, i] / d if d else 1.0
-                self.sn[i] = self.H[i + 1, i] / d if d else 0.0
-                self.H[i, i] = d
-                self.H[i + 1, i] = 0.0
-                g0 = self.cs[i] * self.g[i]
-                g1 = -self.sn[i] * self.g[i]
+                if self._cplx:
+                    for k in range(i):
+                        h0 = (self.cs[k] * self.H[k, i]
+                              + self.sn[k] * self.H[k + 1, i])
+                        h1 = (-np.conj(self.sn[k]) * self.H[k, i]
+                              + self.cs[k] * self.H[k + 1, i])
+                        self.H[k, i], self.H[k + 1, i] = h0, h1
+                    cs_, sn_, d = _complex_givens(self.H[i, i], self.H[i + 1, i])
+                    if d == 0.0:
+                        cs_, sn_ = 1.0, 0.0
+                    self.cs[i], self.sn[i] = cs_, sn_
+                    self.H[i, i] = d
+                    self.H[i + 1, i] = 0.0
+                    g0 = cs_ * self.g[i]
+                    g1 = -np.conj(sn_) * self.g[i]
+                else:
+                    for k in range(i):
+                        h0 = self.cs[k] * self.H[k, i] + self.sn[k] * self.H[k + 1, i]
+                        h1 = -self.sn[k] * self.H[k, i] + self.cs[k] * self.H[k + 1, i]
+                        self.H[k, i], self.H[k + 1, i] = h0, h1
+                    d = math.hypot(self.H[i, i].real, self.H[i + 1, i].real)
+                    self.cs[i] = self.H[i, i].real / d if d else 1.0
+                    self.sn[i] = self.H[i + 1, i].real / d if d else 0.0
+                    self.H[i, i] = d
+                    self.H[i + 1, i] = 0.0
+                    g0 = self.cs[i] * self.g[i]
+                    g1 = -self.sn[i] * self.g[i]
                 self.g[i], self.g[i + 1] = g0, g1
                 res = abs(g1)
                 iters += 1
                 if res <= eps:
                     break
             # x += sum y_k z_k
-            y = np.zeros(i + 1)
+            y = np.zeros(i + 1, dtype=self.H.dtype)
             for k in range(i, -1, -1):
                 y[k] = (self.g[k] - self.H[k, k + 1 : i + 1] @ y[k + 1 : i + 1]) / self.H[k, k]
             for k in range(i + 1):
@@ -89,3 +107,6 @@ class FGMRES(SolverBase):
             if res <= eps:
                 break
         return iters, res / norm_rhs
+
+
+from .gmres import _complex_givens  # noqa: E402

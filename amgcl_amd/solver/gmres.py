@@ -25,10 +25,12 @@ class GMRES(SolverBase):
         self.w = b.vector(n)
         self.tmp = b.vector(n)
         self.v = [b.vector(n) for _ in range(m + 1)]
-        self.H = np.zeros((m + 1, m))
-        self.cs = np.zeros(m + 1)
-        self.sn = np.zeros(m + 1)
-        self.g = np.zeros(m + 1)
+        dt = getattr(backend, "dtype", np.float64)
+        self._cplx = np.issubdtype(np.dtype(dt), np.complexfloating)
+        self.H = np.zeros((m + 1, m), dtype=dt)
+        self.cs = np.zeros(m + 1)  # real also in the complex rotation
+        self.sn = np.zeros(m + 1, dtype=dt)
+        self.g = np.zeros(m + 1, dtype=dt)
 
     def _restart(self, A, P, rhs, x, left):
         b = self.backend
@@ -63,17 +65,31 @@ class GMRES(SolverBase):
         if hnext != 0.0:
             b.axpby(1.0 / hnext, self.w, 0.0, self.v[i + 1])
         # apply existing Givens rotations to column i
+        if self._cplx:
+            for k in range(i):
+                h0 = self.cs[k] * self.H[k, i] + self.sn[k] * self.H[k + 1, i]
+                h1 = (-np.conj(self.sn[k]) * self.H[k, i]
+                      + self.cs[k] * self.H[k + 1, i])
+                self.H[k, i], self.H[k + 1, i] = h0, h1
+            cs, sn, d = _complex_givens(self.H[i, i], self.H[i + 1, i])
+            self.cs[i], self.sn[i] = cs, sn
+            self.H[i, i] = d
+            self.H[i + 1, i] = 0.0
+            g0 = cs * self.g[i]
+            g1 = -np.conj(sn) * self.g[i]
+            self.g[i], self.g[i + 1] = g0, g1
+            return abs(g1)
         for k in range(i):
             h0 = self.cs[k] * self.H[k, i] + self.sn[k] * self.H[k + 1, i]
             h1 = -self.sn[k] * self.H[k, i] + self.cs[k] * self.H[k + 1, i]
             self.H[k, i], self.H[k + 1, i] = h0, h1
         # new rotation
-        d = math.hypot(self.H[i, i], self.H[i + 1, i])
+        d = math.hypot(self.H[i, i].real, self.H[i + 1, i].real)
         if d == 0.0:
             self.cs[i], self.sn[i] = 1.0, 0.0
         else:
-            self.cs[i] = self.H[i, i] / d
-            self.sn[i] = self.H[i + 1, i] / d
+            self.cs[i] = self.H[i, i].real / d
+            self.sn[i] = self.H[i + 1, i].real / d
         self.H[i, i] = d
         self.H[i + 1, i] = 0.0
         g0 = self.cs[i] * self.g[i]
@@ -83,7 +99,7 @@ class GMRES(SolverBase):
 
     def _update(self, x, P, i, left):
         b = self.backend
-        y = np.zeros(i + 1)
+        y = np.zeros(i + 1, dtype=self.H.dtype)
         for k in range(i, -1, -1):
             s = self.g[k] - self.H[k, k + 1 : i + 1] @ y[k + 1 : i + 1]
             y[k] = s / self.H[k, k]
@@ -126,3 +142,19 @@ class GMRES(SolverBase):
             if res <= eps:
                 break
         return iters, res / norm_rhs
+
+
+def _complex_givens(h0, h1):
+    """Complex Givens rotation: returns (cs real, sn complex, d) with
+    cs*h0 + sn*h1 = d and -conj(sn)*h0 + cs*h1 = 0."""
+    import numpy as _np
+
+    a0, a1 = abs(h0), abs(h1)
+    r = _np.hypot(a0, a1)
+    if r == 0.0:
+        return 1.0, 0.0 + 0.0j, 0.0
+    if a0 == 0.0:
+        return 0.0, _np.conj(h1) / a1, a1
+    cs = a0 / r
+    sn = (h0 / a0) * _np.conj(h1) / r
+    return cs, sn, (h0 / a0) * r
