@@ -60,7 +60,7 @@ def build_hip_lib(verbose=True):
         return out
     cmd = [
         HIPCC, f"--offload-arch={GPU_ARCH}", "-O3", "-std=c++17",
-        "-shared", "-fPIC", "-ffast-math",
+        "-shared", "-fPIC", "-ffast-math", "-fopenmp",
         # A/B experiment knobs, e.g. "-DAMGCL_NO_SWIZZLE -DAMGCL_NO_NT"
         *os.environ.get("AMGCL_HIP_DEFINES", "").split(),
         *srcs, "-o", out,

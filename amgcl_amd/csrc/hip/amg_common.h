@@ -46,3 +46,42 @@ static inline int amg_nblocks(int64_t work, int block = 256, int cap = 2048) {
     if (b > 8) b = (b + 7) & ~(int64_t)7;
     return (int)b;
 }
+
+// Per-level operator descriptor consumed by the native solve driver
+// (driver.hip amg_driver_create); filled either from Python (backend/
+// native.py, torch tensors) or from the torch-free GPU C API
+// (capi_gpu.hip, raw hipMalloc buffers).
+struct LevelDesc {
+    int64_t nrows, nnz;
+    const int *ptr;
+    const int *col;
+    const double *val;
+    int subw;
+    int64_t pnnz;  // P: nrows x next->nrows
+    const int *pptr;
+    const int *pcol;
+    const double *pval;
+    int psubw;
+    int64_t rnnz;  // R: next->nrows x nrows
+    const int *rptr;
+    const int *rcol;
+    const double *rval;
+    int rsubw;
+    const double *M;  // diagonal smoother weights
+    double *f;
+    double *u;
+    double *t;  // workspace (f/u unused at level 0)
+    // optional SELL-64 images of A / P / R (see kernels.hip)
+    int64_t nslice;  // 0 = no SELL
+    const int64_t *soff;
+    const int *scol;
+    const void *sval;
+    int64_t pnslice;
+    const int64_t *psoff;
+    const int *pscol;
+    const void *psval;
+    int64_t rnslice;
+    const int64_t *rsoff;
+    const int *rscol;
+    const void *rsval;
+};

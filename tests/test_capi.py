@@ -189,3 +189,55 @@ def test_fortran_module_compiles_and_solves(tmp_path):
         cwd=str(tmp_path))
     out = subprocess.check_output([exe], text=True)
     assert "FORTRAN_OK" in out
+
+
+@pytest.mark.gpu
+def test_torch_free_gpu_solve():
+    """A full AMG+CG GPU solve with torch ABSENT from the process
+    (VERDICT r01 #3 'Done' criterion): a subprocess using only
+    ctypes+numpy drives the torch-free GPU C API (capi_gpu.hip ->
+    libamghip.so), which builds the hierarchy with the host C++ engine,
+    uploads it with raw hipMalloc and solves through the native driver."""
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    prog = r"""
+import ctypes, sys
+import numpy as np
+assert "torch" not in sys.modules
+lib = ctypes.CDLL(r"%s/amgcl_amd/_hip/libamghip.so")
+lib.amgcl_amd_gpu_solver_create.restype = ctypes.c_void_p
+lib.amgcl_amd_gpu_solver_create.argtypes = [ctypes.c_int] + [ctypes.c_void_p]*3 + [ctypes.c_char_p]
+lib.amgcl_amd_gpu_solver_solve.restype = ctypes.c_int
+lib.amgcl_amd_gpu_solver_solve.argtypes = [ctypes.c_void_p]*3 + [ctypes.POINTER(ctypes.c_int), ctypes.POINTER(ctypes.c_double)]
+m = 32
+n = m**3
+import itertools
+ptr = [0]; col = []; val = []
+for k in range(m):
+    for j in range(m):
+        for i in range(m):
+            r = (k*m + j)*m + i
+            for (dk,dj,di,v) in ((-1,0,0,-1.),(0,-1,0,-1.),(0,0,-1,-1.),(0,0,0,6.),(0,0,1,-1.),(0,1,0,-1.),(1,0,0,-1.)):
+                kk,jj,ii = k+dk, j+dj, i+di
+                if 0 <= kk < m and 0 <= jj < m and 0 <= ii < m:
+                    col.append((kk*m + jj)*m + ii); val.append(v)
+            ptr.append(len(col))
+ptr = np.asarray(ptr, dtype=np.int32); col = np.asarray(col, dtype=np.int32)
+val = np.asarray(val, dtype=np.float64)
+b = np.ones(n); x = np.zeros(n)
+h = lib.amgcl_amd_gpu_solver_create(n, ptr.ctypes.data, col.ctypes.data, val.ctypes.data,
+                                    b"solver.type=cg;solver.tol=1e-8;precond.coarse_enough=1000")
+assert h, "create failed"
+it = ctypes.c_int(0); res = ctypes.c_double(0.0)
+rc = lib.amgcl_amd_gpu_solver_solve(h, b.ctypes.data, x.ctypes.data,
+                                    ctypes.byref(it), ctypes.byref(res))
+assert rc == 0, rc
+assert res.value < 1e-8 and it.value < 40, (it.value, res.value)
+assert "torch" not in sys.modules
+print("TORCHFREE_OK", it.value, res.value)
+""" % root
+    out = subprocess.check_output([sys.executable, "-c", prog], text=True,
+                                  stderr=subprocess.STDOUT, timeout=300)
+    assert "TORCHFREE_OK" in out
