@@ -6,6 +6,7 @@ backend with 1-based (Fortran) entry points; these tests exercise the same
 surface on amgcl_amd's plain-C++/OpenMP shim.
 """
 import ctypes
+import os
 
 import numpy as np
 import pytest
