@@ -81,9 +81,14 @@ class DeviceCSR:
         soff = torch.zeros(nslice + 1, dtype=torch.int64, device=dev)
         torch.cumsum(w * 64, dim=0, out=soff[1:])
         total = int(soff[-1].item())
-        # empty, not zeros: the fill kernel writes the padding itself
-        scol = torch.empty(total, dtype=torch.int32, device=dev)
-        sval = torch.empty(total, dtype=self.val.dtype, device=dev)
+        # ONE allocation for col+val (empty, not zeros: the fill kernel
+        # writes the padding itself).  Measured: separate multi-GB
+        # allocations stalled the GPU ~11 ms each at 512^3 (allocator
+        # growth), see profiles/README.md round 2.
+        esize = self.val.element_size()
+        blob = torch.empty(total * (4 + esize), dtype=torch.uint8, device=dev)
+        scol = blob[: total * 4].view(torch.int32)
+        sval = blob[total * 4 :].view(self.val.dtype)
         fn = (lib().amg_sell_fill_f32 if self.val.dtype == torch.float32
               else lib().amg_sell_fill_f64)
         check(fn(n, self.ptr.data_ptr(), self.col.data_ptr(), self.val.data_ptr(),
