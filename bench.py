@@ -18,10 +18,6 @@ import os
 import sys
 import time
 
-# expandable segments make the setup's multi-GB allocations (SELL images,
-# SpGEMM outputs) grow the arena instead of fresh hipMalloc segments
-os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
-
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 BASELINE_TOTAL_S = 2.03  # MN4 6144 cores: 0.68 setup + 1.35 solve
