@@ -68,8 +68,15 @@ class DeviceCSR:
         are kept — transfers and non-SELL paths still use them."""
         if self.nslice:
             return self
+        import os
+        import time
+
         import torch
 
+        dbg = os.environ.get("AMGCL_DEBUG_SELL")
+        if dbg:
+            torch.cuda.synchronize()
+            _t0 = time.perf_counter()
         n = self.nrows
         dev = self.val.device
         nslice = (n + 63) // 64
@@ -80,7 +87,11 @@ class DeviceCSR:
         w = lens.view(nslice, 64).max(dim=1).values
         soff = torch.zeros(nslice + 1, dtype=torch.int64, device=dev)
         torch.cumsum(w * 64, dim=0, out=soff[1:])
+        if dbg:
+            _t1 = time.perf_counter()  # torch ops enqueued
         total = int(soff[-1].item())
+        if dbg:
+            _t2 = time.perf_counter()  # sync + D2H of the total
         # ONE allocation for col+val (empty, not zeros: the fill kernel
         # writes the padding itself).  Measured: separate multi-GB
         # allocations stalled the GPU ~11 ms each at 512^3 (allocator
@@ -89,12 +100,20 @@ class DeviceCSR:
         blob = torch.empty(total * (4 + esize), dtype=torch.uint8, device=dev)
         scol = blob[: total * 4].view(torch.int32)
         sval = blob[total * 4 :].view(self.val.dtype)
+        if dbg:
+            _t3 = time.perf_counter()  # allocation
         fn = (lib().amg_sell_fill_f32 if self.val.dtype == torch.float32
               else lib().amg_sell_fill_f64)
         check(fn(n, self.ptr.data_ptr(), self.col.data_ptr(), self.val.data_ptr(),
                  soff.data_ptr(), scol.data_ptr(), sval.data_ptr(), _stream()),
               "sell_fill")
         self.nslice, self.soff, self.scol, self.sval = nslice, soff, scol, sval
+        if dbg:
+            torch.cuda.synchronize()
+            _t4 = time.perf_counter()
+            print(f"[sell] n={n:>10} total={total:>11} enq={(_t1-_t0)*1e3:7.1f} "
+                  f"item={(_t2-_t1)*1e3:7.1f} alloc={(_t3-_t2)*1e3:7.1f} "
+                  f"fill={(_t4-_t3)*1e3:7.1f} ms", flush=True)
         return self
 
 

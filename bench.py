@@ -18,6 +18,11 @@ import os
 import sys
 import time
 
+# Multi-GB setup blocks (SELL images, SpGEMM outputs) must be reusable
+# across steps: forbid splitting big cached blocks, otherwise the warm
+# steps pay fresh hipMalloc (~28 ms/GB measured, profiles/README r02).
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "max_split_size_mb:512")
+
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 BASELINE_TOTAL_S = 2.03  # MN4 6144 cores: 0.68 setup + 1.35 solve
