@@ -19,6 +19,15 @@ from ..coarsening import make_coarsening
 from ..relaxation import make_relaxation_factory
 
 
+def _host_handoff():
+    """Row threshold below which coarse levels leave the device engine for
+    the host (tiny kernels are launch-bound; the host engine also supports
+    every smoother/solver).  Env-tunable for A/B measurements."""
+    import os
+
+    return int(os.environ.get("AMGCL_HOST_HANDOFF", "20000"))
+
+
 class Level:
     __slots__ = ("A", "P", "R", "P_build", "R_build", "f", "u", "t", "relax",
                  "rows", "nnz", "A_host")
@@ -128,7 +137,7 @@ class AMG:
             lvl.relax = relax_factory(A, backend)
             A = self._coarsening.coarse_operator(A, lvl.P_build, lvl.R_build)
         last = self.levels[-1]
-        if not isinstance(A, CSR) and A.nrows <= 20000:
+        if not isinstance(A, CSR) and A.nrows <= _host_handoff():
             from ..backend import hip_setup
 
             A = hip_setup.download(A)
@@ -214,7 +223,7 @@ class AMG:
             # device-setup path: hand small coarse levels back to the host
             # engine (tiny kernels are launch-bound; the host path also
             # supports every smoother/coarse solver)
-            if not isinstance(A_host, CSR) and A_host.nrows <= 20000:
+            if not isinstance(A_host, CSR) and A_host.nrows <= _host_handoff():
                 from ..backend import hip_setup
 
                 A_host = hip_setup.download(A_host)
