@@ -376,8 +376,10 @@ class DeviceDenseSolver:
 
     @classmethod
     def from_device(cls, dcsr, backend):
-        """Densify + invert entirely on the GPU (no download round-trip; the
-        host np.linalg.inv also runs under the GPU box's 16-core CPU quota)."""
+        """Densify on the GPU, invert where it is cheapest: hipSOLVER/MAGMA
+        has tens of ms of fixed overhead, so small coarse systems (the
+        common case, n <= 2048) round-trip to numpy instead — measured
+        faster despite the copies."""
         import torch
 
         self = cls.__new__(cls)
@@ -389,7 +391,12 @@ class DeviceDenseSolver:
             lens.to(torch.int64))
         dense[rows, dcsr.col.to(torch.int64)] = dcsr.val
         self.n = n
-        self.inv = torch.linalg.inv(dense).contiguous().ravel()
+        if n <= 2048:
+            inv = np.linalg.inv(dense.cpu().numpy())
+            self.inv = torch.from_numpy(np.ascontiguousarray(inv).ravel()).to(
+                dcsr.val.device, dtype=dcsr.val.dtype)
+        else:
+            self.inv = torch.linalg.inv(dense).contiguous().ravel()
         return self
 
     def __call__(self, f, u):
