@@ -95,5 +95,32 @@ module amgcl_amd
        import c_ptr
        type(c_ptr), value :: amg
      end subroutine
+
+     ! torch-free GPU solver (link libamghip.so; 0-based CSR indices)
+     function amgcl_amd_gpu_solver_create(n, ptr, col, val, config) &
+          bind(c, name="amgcl_amd_gpu_solver_create")
+       import c_ptr, c_int, c_double, c_char
+       type(c_ptr) :: amgcl_amd_gpu_solver_create
+       integer(c_int), value :: n
+       integer(c_int) :: ptr(*), col(*)
+       real(c_double) :: val(*)
+       character(kind=c_char) :: config(*)
+     end function
+
+     function amgcl_amd_gpu_solver_solve(solver, rhs, x, iters, resid) &
+          bind(c, name="amgcl_amd_gpu_solver_solve")
+       import c_ptr, c_int, c_double
+       integer(c_int) :: amgcl_amd_gpu_solver_solve
+       type(c_ptr), value :: solver
+       real(c_double) :: rhs(*), x(*)
+       integer(c_int) :: iters
+       real(c_double) :: resid
+     end function
+
+     subroutine amgcl_amd_gpu_solver_destroy(solver) &
+          bind(c, name="amgcl_amd_gpu_solver_destroy")
+       import c_ptr
+       type(c_ptr), value :: solver
+     end subroutine
   end interface
 end module amgcl_amd

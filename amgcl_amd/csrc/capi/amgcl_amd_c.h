@@ -68,6 +68,16 @@ int amgcl_amd_solver_solve_mtx(amgcl_amd_handle solver, const int *A_ptr,
 int amgcl_amd_solver_report(amgcl_amd_handle solver, char *buf, int len);
 void amgcl_amd_solver_destroy(amgcl_amd_handle solver);
 
+/* Torch-free GPU solver (exported by libamghip.so, NOT libamgclamd_c.so):
+ * host hierarchy assembly + raw hipMalloc upload + the native gfx950 solve
+ * driver.  `config` is "key=value;key=value" with the same keys as the
+ * Params object (solver.type/tol/maxiter, precond.coarse_enough, ...). */
+amgcl_amd_handle amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *col,
+                                             const double *val, const char *config);
+int amgcl_amd_gpu_solver_solve(amgcl_amd_handle solver, const double *rhs,
+                               double *x, int *iters, double *resid);
+void amgcl_amd_gpu_solver_destroy(amgcl_amd_handle solver);
+
 #ifdef __cplusplus
 }
 #endif

@@ -52,10 +52,12 @@ class LGMRES(SolverBase):
                 break
             naug = min(len(self.aug), self.k)
             msize = self.m + naug
-            H = np.zeros((msize + 1, msize))
+            dt = getattr(self.backend, "dtype", np.float64)
+            cplx = np.issubdtype(np.dtype(dt), np.complexfloating)
+            H = np.zeros((msize + 1, msize), dtype=dt)
             cs = np.zeros(msize + 1)
-            sn = np.zeros(msize + 1)
-            g = np.zeros(msize + 1)
+            sn = np.zeros(msize + 1, dtype=dt)
+            g = np.zeros(msize + 1, dtype=dt)
             g[0] = beta
             b.axpby(1.0 / beta, self.r, 0.0, self.v[0])
             dx_dirs = []
@@ -77,22 +79,35 @@ class LGMRES(SolverBase):
                 H[i + 1, i] = hn
                 if hn != 0.0:
                     b.axpby(1.0 / hn, self.w, 0.0, self.v[i + 1])
-                for kk in range(i):
-                    h0 = cs[kk] * H[kk, i] + sn[kk] * H[kk + 1, i]
-                    h1 = -sn[kk] * H[kk, i] + cs[kk] * H[kk + 1, i]
-                    H[kk, i], H[kk + 1, i] = h0, h1
-                d = math.hypot(H[i, i], H[i + 1, i])
-                cs[i] = H[i, i] / d if d else 1.0
-                sn[i] = H[i + 1, i] / d if d else 0.0
-                H[i, i] = d
-                H[i + 1, i] = 0.0
-                g0, g1 = cs[i] * g[i], -sn[i] * g[i]
+                if cplx:
+                    for kk in range(i):
+                        h0 = cs[kk] * H[kk, i] + sn[kk] * H[kk + 1, i]
+                        h1 = -np.conj(sn[kk]) * H[kk, i] + cs[kk] * H[kk + 1, i]
+                        H[kk, i], H[kk + 1, i] = h0, h1
+                    cs_, sn_, d = _complex_givens(H[i, i], H[i + 1, i])
+                    if d == 0.0:
+                        cs_, sn_ = 1.0, 0.0
+                    cs[i], sn[i] = cs_, sn_
+                    H[i, i] = d
+                    H[i + 1, i] = 0.0
+                    g0, g1 = cs_ * g[i], -np.conj(sn_) * g[i]
+                else:
+                    for kk in range(i):
+                        h0 = cs[kk] * H[kk, i] + sn[kk] * H[kk + 1, i]
+                        h1 = -sn[kk] * H[kk, i] + cs[kk] * H[kk + 1, i]
+                        H[kk, i], H[kk + 1, i] = h0, h1
+                    d = math.hypot(H[i, i].real, H[i + 1, i].real)
+                    cs[i] = H[i, i].real / d if d else 1.0
+                    sn[i] = H[i + 1, i].real / d if d else 0.0
+                    H[i, i] = d
+                    H[i + 1, i] = 0.0
+                    g0, g1 = cs[i] * g[i], -sn[i] * g[i]
                 g[i], g[i + 1] = g0, g1
                 res = abs(g1)
                 iters += 1
                 if res <= eps:
                     break
-            y = np.zeros(i + 1)
+            y = np.zeros(i + 1, dtype=H.dtype)
             for kk in range(i, -1, -1):
                 y[kk] = (g[kk] - H[kk, kk + 1 : i + 1] @ y[kk + 1 : i + 1]) / H[kk, kk]
             # outer correction dx = sum y_k dir_k (store as next aug vector)
@@ -108,3 +123,6 @@ class LGMRES(SolverBase):
             if res <= eps:
                 break
         return iters, res / norm_rhs
+
+
+from .gmres import _complex_givens  # noqa: E402

@@ -21,7 +21,7 @@ def helmholtz(n=16, shift=0.4 + 0.35j):
     return CSR.from_scipy(m), m
 
 
-@pytest.mark.parametrize("solver", ["bicgstab", "gmres", "fgmres"])
+@pytest.mark.parametrize("solver", ["bicgstab", "gmres", "fgmres", "lgmres", "richardson"])
 def test_native_complex_solve(solver):
     A, m = helmholtz(14)
     rng = np.random.default_rng(0)
