@@ -99,7 +99,8 @@ _SIGS = {
     "amg_agg_init": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
     "amg_agg_round": [ctypes.c_int64] + [ctypes.c_void_p] * 10,
     "amg_agg_run": [ctypes.c_int64] + [ctypes.c_void_p] * 9 + [
-        ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p],
+        ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p],
     "amg_agg_renumber": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     "amg_psmooth_count": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
     "amg_psmooth_fill": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,

@@ -97,10 +97,12 @@ def aggregates(A, eps_strong):
     import ctypes
 
     rounds = ctypes.c_int(0)
+    # frontier-compaction scratch: two ping-pong active lists + counter
+    lists = t.empty(2 * n + 2, dtype=t.int32, device=dev)
     rc = lib().amg_agg_run(n, A.ptr.data_ptr(), A.col.data_ptr(), S.data_ptr(),
                            ids.data_ptr(), prov.data_ptr(), m1.data_ptr(),
                            newroot.data_ptr(), near.data_ptr(), remaining.data_ptr(),
-                           2, 64, ctypes.byref(rounds), _stream())
+                           2, 64, ctypes.byref(rounds), lists.data_ptr(), _stream())
     if rc == 9999:
         raise RuntimeError("device aggregation did not converge")
     check(rc, "agg_run")

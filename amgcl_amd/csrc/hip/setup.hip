@@ -260,12 +260,28 @@ __global__ void agg_init_k(int64_t n, const int *__restrict__ ptr, const uint8_t
 // m1 needs to be valid for every node within distance 1 of an UNDEF node.
 // `near` marks those nodes (set during the previous round's passes or the
 // init pass); everything else keeps m1 = 0 without re-reading its row.
-__global__ void agg_m1_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
+// Frontier lists: after the first rounds, the active set (UNDEF nodes and
+// their strong 1-ring) shrinks monotonically, so the round kernels iterate
+// a compacted index list instead of re-scanning all n nodes (the host
+// engine's sparse branch, on the device).  list == nullptr -> full range.
+#define AGG_FOREACH(i)                                                              int64_t _t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;                    int64_t _stride = (int64_t)gridDim.x * blockDim.x;                              for (; _t < nwork; _t += _stride)                                                   if (int64_t i = (list ? (int64_t)list[_t] : _t); true)
+
+__global__ void agg_clear_near_k(int64_t nwork, const int *__restrict__ list,
+                                 uint8_t *__restrict__ near) {
+    AGG_FOREACH(i) near[i] = 0;
+}
+
+__global__ void agg_compact_k(int64_t nwork, const int *__restrict__ list,
+                              const uint8_t *__restrict__ near,
+                              int *__restrict__ out, int *__restrict__ cnt) {
+    AGG_FOREACH(i) if (near[i]) out[atomicAdd(cnt, 1)] = (int)i;
+}
+
+__global__ void agg_m1_k(int64_t nwork, const int *__restrict__ list,
+                         const int *__restrict__ ptr, const int *__restrict__ col,
                          const uint8_t *__restrict__ S, const int *__restrict__ id,
                          const uint8_t *__restrict__ near, uint64_t *__restrict__ m1) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) {
+    AGG_FOREACH(i) {
         if (!near[i]) {
             m1[i] = 0;
             continue;
@@ -284,12 +300,11 @@ __global__ void agg_m1_k(int64_t n, const int *__restrict__ ptr, const int *__re
 }
 
 // near[i] = i is UNDEF or has an UNDEF strong neighbor
-__global__ void agg_near_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
+__global__ void agg_near_k(int64_t nwork, const int *__restrict__ list,
+                           const int *__restrict__ ptr, const int *__restrict__ col,
                            const uint8_t *__restrict__ S, const int *__restrict__ id,
                            uint8_t *__restrict__ near) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) {
+    AGG_FOREACH(i) {
         if (id[i] == -1) {
             near[i] = 1;
             for (int j = ptr[i]; j < ptr[i + 1]; ++j)
@@ -298,12 +313,11 @@ __global__ void agg_near_k(int64_t n, const int *__restrict__ ptr, const int *__
     }
 }
 
-__global__ void agg_roots_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
+__global__ void agg_roots_k(int64_t nwork, const int *__restrict__ list,
+                            const int *__restrict__ ptr, const int *__restrict__ col,
                             const uint8_t *__restrict__ S, int *__restrict__ id,
                             const uint64_t *__restrict__ m1, uint8_t *__restrict__ newroot) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) {
+    AGG_FOREACH(i) {
         newroot[i] = 0;
         if (id[i] != -1) continue;
         uint64_t key = agg_key_d((int)i);
@@ -320,12 +334,11 @@ __global__ void agg_roots_k(int64_t n, const int *__restrict__ ptr, const int *_
     }
 }
 
-__global__ void agg_claim_k(int64_t n, const int *__restrict__ ptr, const int *__restrict__ col,
+__global__ void agg_claim_k(int64_t nwork, const int *__restrict__ list,
+                            const int *__restrict__ ptr, const int *__restrict__ col,
                             const uint8_t *__restrict__ S, int *__restrict__ id,
                             uint8_t *__restrict__ prov, const uint8_t *__restrict__ newroot) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) {
+    AGG_FOREACH(i) {
         if (!newroot[i]) continue;
         for (int j = ptr[i]; j < ptr[i + 1]; ++j) {
             int c = col[j];
@@ -342,13 +355,12 @@ __global__ void agg_claim_k(int64_t n, const int *__restrict__ ptr, const int *_
 // records each node's choice; the commit pass applies it. A single fused
 // pass would race (a neighbor turning provisional concurrently could be
 // mistaken for a firm member, making the result timing-dependent).
-__global__ void agg_adopt_mark_k(int64_t n, const int *__restrict__ ptr,
+__global__ void agg_adopt_mark_k(int64_t nwork, const int *__restrict__ list,
+                                 const int *__restrict__ ptr,
                                  const int *__restrict__ col, const uint8_t *__restrict__ S,
                                  const int *__restrict__ id, const uint8_t *__restrict__ prov,
                                  int *__restrict__ choice) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) {
+    AGG_FOREACH(i) {
         if (id[i] != -1) continue;
         uint64_t best = 0;
         int root = -1;
@@ -364,14 +376,13 @@ __global__ void agg_adopt_mark_k(int64_t n, const int *__restrict__ ptr,
     }
 }
 
-__global__ void agg_adopt_commit_k(int64_t n, int *__restrict__ id,
+__global__ void agg_adopt_commit_k(int64_t nwork, const int *__restrict__ list,
+                                   int *__restrict__ id,
                                    uint8_t *__restrict__ prov,
                                    const int *__restrict__ choice,
                                    int *__restrict__ remaining) {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int my_remaining = 0;
-    for (; i < n; i += stride) {
+    AGG_FOREACH(i) {
         if (id[i] != -1) continue;
         int root = choice[i];
         if (root >= 0) {
@@ -405,38 +416,94 @@ extern "C" int amg_agg_init(int64_t n, const int *ptr, const uint8_t *S, int *id
     agg_init_k<<<nblk(n), 256, 0, s>>>(n, ptr, S, id);
     return (int)hipGetLastError();
 }
+static int agg_round_impl(int64_t nwork, const int *list, const int *ptr,
+                          const int *col, const uint8_t *S, int *id, uint8_t *prov,
+                          uint64_t *m1, uint8_t *newroot, uint8_t *near,
+                          int *remaining, hipStream_t s) {
+    if (list)
+        agg_clear_near_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, near);
+    else
+        (void)hipMemsetAsync(near, 0, nwork, s);
+    agg_near_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, ptr, col, S, id, near);
+    agg_m1_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, ptr, col, S, id, near, m1);
+    agg_roots_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, ptr, col, S, id, m1, newroot);
+    agg_claim_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, ptr, col, S, id, prov, newroot);
+    // reuse the m1 buffer as the choice array (i32 fits in the u64 slots)
+    agg_adopt_mark_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, ptr, col, S, id, prov,
+                                                 (int *)m1);
+    agg_adopt_commit_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, id, prov, (int *)m1,
+                                                   remaining);
+    return (int)hipGetLastError();
+}
+
 extern "C" int amg_agg_round(int64_t n, const int *ptr, const int *col, const uint8_t *S,
                              int *id, uint8_t *prov, uint64_t *m1, uint8_t *newroot,
                              uint8_t *near, int *remaining, hipStream_t s) {
-    hipMemsetAsync(near, 0, n, s);
-    agg_near_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, near);
-    agg_m1_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, near, m1);
-    agg_roots_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, m1, newroot);
-    agg_claim_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, prov, newroot);
-    // reuse the m1 buffer as the choice array (i32 fits in the u64 slots)
-    agg_adopt_mark_k<<<nblk(n), 256, 0, s>>>(n, ptr, col, S, id, prov, (int *)m1);
-    agg_adopt_commit_k<<<nblk(n), 256, 0, s>>>(n, id, prov, (int *)m1, remaining);
-    return (int)hipGetLastError();
+    return agg_round_impl(n, nullptr, ptr, col, S, id, prov, m1, newroot, near,
+                          remaining, s);
 }
 // whole MIS round loop driven from C: one ctypes call per level instead of
 // one per round (the Python round loop cost ~11 ms/round of host gaps at
 // 134M rows — launch latency + torch op dispatch + .item() round-trips).
 // Convergence is polled with a pinned-memory readback every `sync_stride`
 // rounds. Returns hipError, or 9999 if max_rounds was exhausted.
+// lists: caller-provided scratch of (2n + 2) ints — two ping-pong frontier
+// lists plus a device counter; nullptr disables frontier compaction.
 extern "C" int amg_agg_run(int64_t n, const int *ptr, const int *col, const uint8_t *S,
                            int *id, uint8_t *prov, uint64_t *m1, uint8_t *newroot,
                            uint8_t *near, int *remaining, int sync_stride,
-                           int max_rounds, int *rounds_out, hipStream_t s) {
+                           int max_rounds, int *rounds_out, int *lists,
+                           hipStream_t s) {
     int *h_rem = nullptr;
-    hipError_t e = hipHostMalloc((void **)&h_rem, sizeof(int), hipHostMallocDefault);
+    hipError_t e = hipHostMalloc((void **)&h_rem, 2 * sizeof(int),
+                                 hipHostMallocDefault);
     if (e != hipSuccess) return (int)e;
+    int *cnt_d = lists ? lists + 2 * n : nullptr;
+    int *L[2] = {lists, lists ? lists + n : nullptr};
+    int cur = -1;          // -1: full range; 0/1: active list index
+    int64_t nwork = n;
     int rc = 0;
     bool done = false;
     int round = 0;
     for (; round < max_rounds; ++round) {
         e = hipMemsetAsync(remaining, 0, sizeof(int), s);
         if (e != hipSuccess) { rc = (int)e; break; }
-        rc = amg_agg_round(n, ptr, col, S, id, prov, m1, newroot, near, remaining, s);
+        const int *list = cur < 0 ? nullptr : L[cur];
+        // refresh near first (it defines the next frontier superset)
+        if (list)
+            agg_clear_near_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, near);
+        else
+            (void)hipMemsetAsync(near, 0, n, s);
+        agg_near_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, ptr, col, S, id, near);
+        // every 4th round (after warm-up) compact the frontier: the active
+        // set (UNDEF + strong 1-ring) shrinks monotonically, so later
+        // rounds stop re-scanning all n nodes
+        if (lists && round >= 2 && (round - 2) % 4 == 0) {
+            int nxt = cur < 0 ? 0 : 1 - cur;
+            e = hipMemsetAsync(cnt_d, 0, sizeof(int), s);
+            if (e != hipSuccess) { rc = (int)e; break; }
+            agg_compact_k<<<nblk(nwork), 256, 0, s>>>(nwork, list, near, L[nxt],
+                                                      cnt_d);
+            e = hipMemcpyAsync(h_rem + 1, cnt_d, sizeof(int),
+                               hipMemcpyDeviceToHost, s);
+            if (e == hipSuccess) e = hipStreamSynchronize(s);
+            if (e != hipSuccess) { rc = (int)e; break; }
+            if (h_rem[1] > 0 && h_rem[1] < nwork) {
+                cur = nxt;
+                nwork = h_rem[1];
+            }
+        }
+        const int *list2 = cur < 0 ? nullptr : L[cur];
+        agg_m1_k<<<nblk(nwork), 256, 0, s>>>(nwork, list2, ptr, col, S, id, near, m1);
+        agg_roots_k<<<nblk(nwork), 256, 0, s>>>(nwork, list2, ptr, col, S, id, m1,
+                                                newroot);
+        agg_claim_k<<<nblk(nwork), 256, 0, s>>>(nwork, list2, ptr, col, S, id, prov,
+                                                newroot);
+        agg_adopt_mark_k<<<nblk(nwork), 256, 0, s>>>(nwork, list2, ptr, col, S, id,
+                                                     prov, (int *)m1);
+        agg_adopt_commit_k<<<nblk(nwork), 256, 0, s>>>(nwork, list2, id, prov,
+                                                       (int *)m1, remaining);
+        rc = (int)hipGetLastError();
         if (rc) break;
         if (round % sync_stride == sync_stride - 1 || round > 8) {
             e = hipMemcpyAsync(h_rem, remaining, sizeof(int), hipMemcpyDeviceToHost, s);
