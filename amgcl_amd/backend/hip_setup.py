@@ -98,7 +98,7 @@ def aggregates(A, eps_strong):
 
     rounds = ctypes.c_int(0)
     # frontier-compaction scratch: two ping-pong active lists + counter
-    lists = t.empty(2 * n + 2, dtype=t.int32, device=dev)
+    lists = t.empty(3 * n, dtype=t.int32, device=dev)
     rc = lib().amg_agg_run(n, A.ptr.data_ptr(), A.col.data_ptr(), S.data_ptr(),
                            ids.data_ptr(), prov.data_ptr(), m1.data_ptr(),
                            newroot.data_ptr(), near.data_ptr(), remaining.data_ptr(),
