@@ -66,20 +66,18 @@ _SIGS = {
                       + [ctypes.c_int, ctypes.c_void_p],
     "amg_coop_supported": [],
     # --- SELL-64 (wave-native sliced-ELL) solve kernels ---
-    "amg_sell_spmv_f64": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p,
-                          ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
-                          ctypes.c_double, ctypes.c_double, ctypes.c_void_p,
-                          ctypes.c_void_p],
-    "amg_sell_residual_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 7,
-    "amg_sell_relax_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
-    "amg_sell_fill_f64": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
-    "amg_sell_spmv_f32": [ctypes.c_int64, ctypes.c_int64, ctypes.c_void_p,
-                          ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
-                          ctypes.c_double, ctypes.c_double, ctypes.c_void_p,
-                          ctypes.c_void_p],
-    "amg_sell_residual_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 7,
-    "amg_sell_relax_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
-    "amg_sell_fill_f32": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
+    "amg_sell_spmv_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+                         + [ctypes.c_double, ctypes.c_double]
+                         + [ctypes.c_void_p] * 2,
+    "amg_sell_residual_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
+    "amg_sell_relax_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 9,
+    "amg_sell_fill_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
+    "amg_sell_spmv_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+                         + [ctypes.c_double, ctypes.c_double]
+                         + [ctypes.c_void_p] * 2,
+    "amg_sell_residual_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
+    "amg_sell_relax_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 9,
+    "amg_sell_fill_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
     # --- block (BSR) solve kernels ---
     "amg_bsr_spmv_f64": [ctypes.c_int64, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
                          ctypes.c_void_p, ctypes.c_void_p, ctypes.c_double,

@@ -26,7 +26,7 @@ def _stream():
 
 class DeviceCSR:
     __slots__ = ("nrows", "ncols", "ptr", "col", "val", "subw",
-                 "nslice", "soff", "scol", "sval")
+                 "nslice", "soff", "scol", "sval", "srows")
 
     def __init__(self, csr: CSR, device, subw=0):
         import torch
@@ -38,7 +38,7 @@ class DeviceCSR:
         self.val = torch.from_numpy(np.asarray(csr.val)).to(device)
         self.subw = subw  # 0 = auto by mean row length
         self.nslice = 0
-        self.soff = self.scol = self.sval = None
+        self.soff = self.scol = self.sval = self.srows = None
 
     @classmethod
     def from_tensors(cls, nrows, ncols, ptr, col, val, subw=0):
@@ -48,7 +48,7 @@ class DeviceCSR:
         self.ptr, self.col, self.val = ptr, col, val
         self.subw = subw
         self.nslice = 0
-        self.soff = self.scol = self.sval = None
+        self.soff = self.scol = self.sval = self.srows = None
         return self
 
     @property
@@ -90,6 +90,32 @@ class DeviceCSR:
         if dbg:
             _t1 = time.perf_counter()  # torch ops enqueued
         total = int(soff[-1].item())
+        srows = None
+        nnz = self.col.numel()
+        if total > 1.08 * nnz and n > (1 << 16):
+            # sigma-sorted slices: order rows by length inside sigma-sized
+            # windows to cut the slice padding of ragged (coarse SA) levels;
+            # the permutation stays window-local, so the scattered x/rhs/y
+            # accesses remain L2-local
+            sigma = 8192
+            npad = ((n + sigma - 1) // sigma) * sigma
+            lens2 = (self.ptr[1:] - self.ptr[:-1]).to(torch.int64)
+            if npad > n:
+                lens2 = torch.cat([lens2, torch.zeros(npad - n, dtype=torch.int64,
+                                                      device=dev)])
+            lw = lens2.view(-1, sigma)
+            slens, order = torch.sort(lw, dim=1, descending=True, stable=True)
+            base = (torch.arange(lw.shape[0], device=dev, dtype=torch.int64)
+                    * sigma).unsqueeze(1)
+            rowids = (order + base).reshape(-1)
+            srows = torch.where(rowids < n, rowids,
+                                torch.full_like(rowids, -1)).to(torch.int32)
+            srows = srows.contiguous()
+            nslice = npad // 64
+            w = slens.reshape(-1).view(nslice, 64).max(dim=1).values
+            soff = torch.zeros(nslice + 1, dtype=torch.int64, device=dev)
+            torch.cumsum(w * 64, dim=0, out=soff[1:])
+            total = int(soff[-1].item())
         if dbg:
             _t2 = time.perf_counter()  # sync + D2H of the total
         # ONE allocation for col+val (empty, not zeros: the fill kernel
@@ -104,10 +130,12 @@ class DeviceCSR:
             _t3 = time.perf_counter()  # allocation
         fn = (lib().amg_sell_fill_f32 if self.val.dtype == torch.float32
               else lib().amg_sell_fill_f64)
-        check(fn(n, self.ptr.data_ptr(), self.col.data_ptr(), self.val.data_ptr(),
-                 soff.data_ptr(), scol.data_ptr(), sval.data_ptr(), _stream()),
-              "sell_fill")
+        check(fn(n, nslice, self.ptr.data_ptr(), self.col.data_ptr(),
+                 self.val.data_ptr(), soff.data_ptr(),
+                 srows.data_ptr() if srows is not None else 0,
+                 scol.data_ptr(), sval.data_ptr(), _stream()), "sell_fill")
         self.nslice, self.soff, self.scol, self.sval = nslice, soff, scol, sval
+        self.srows = srows
         if dbg:
             torch.cuda.synchronize()
             _t4 = time.perf_counter()
@@ -193,6 +221,7 @@ class HipBackend:
         if getattr(A, "nslice", 0):
             check(self._fn("sell_spmv", A.sval)(A.nrows, A.nslice, A.soff.data_ptr(),
                                                 A.scol.data_ptr(), A.sval.data_ptr(),
+                                                A.srows.data_ptr() if A.srows is not None else 0,
                                                 x.data_ptr(), alpha, beta,
                                                 y.data_ptr(), _stream()), "sell_spmv")
             return
@@ -210,7 +239,9 @@ class HipBackend:
         if getattr(A, "nslice", 0):
             check(self._fn("sell_residual", A.sval)(A.nrows, A.nslice,
                                                     A.soff.data_ptr(), A.scol.data_ptr(),
-                                                    A.sval.data_ptr(), b.data_ptr(),
+                                                    A.sval.data_ptr(),
+                                                    A.srows.data_ptr() if A.srows is not None else 0,
+                                                    b.data_ptr(),
                                                     x.data_ptr(), r.data_ptr(),
                                                     _stream()), "sell_residual")
             return
@@ -232,6 +263,7 @@ class HipBackend:
             # SELL relax writes x_new = x + M(rhs - Ax) into t, then copy back
             check(self._fn("sell_relax", A.sval)(A.nrows, A.nslice, A.soff.data_ptr(),
                                                  A.scol.data_ptr(), A.sval.data_ptr(),
+                                                 A.srows.data_ptr() if A.srows is not None else 0,
                                                  M.data_ptr(), rhs.data_ptr(),
                                                  x.data_ptr(), t.data_ptr(),
                                                  _stream()), "sell_relax")

@@ -45,6 +45,9 @@ class LevelDescC(ctypes.Structure):
         ("rsoff", ctypes.c_void_p),
         ("rscol", ctypes.c_void_p),
         ("rsval", ctypes.c_void_p),
+        ("srows", ctypes.c_void_p),
+        ("psrows", ctypes.c_void_p),
+        ("rsrows", ctypes.c_void_p),
     ]
 
 
@@ -139,17 +142,20 @@ class NativeDriver:
             if A.nslice:
                 d.nslice = A.nslice
                 d.soff, d.scol, d.sval = _ptr(A.soff), _ptr(A.scol), _ptr(A.sval)
-                self._keep.extend([A.soff, A.scol, A.sval])
+                d.srows = _ptr(A.srows)
+                self._keep.extend([A.soff, A.scol, A.sval, A.srows])
             if l.P is not None and getattr(l.P, "nslice", 0):
                 d.pnslice = l.P.nslice
                 d.psoff, d.pscol, d.psval = (_ptr(l.P.soff), _ptr(l.P.scol),
                                              _ptr(l.P.sval))
-                self._keep.extend([l.P.soff, l.P.scol, l.P.sval])
+                d.psrows = _ptr(l.P.srows)
+                self._keep.extend([l.P.soff, l.P.scol, l.P.sval, l.P.srows])
             if l.R is not None and getattr(l.R, "nslice", 0):
                 d.rnslice = l.R.nslice
                 d.rsoff, d.rscol, d.rsval = (_ptr(l.R.soff), _ptr(l.R.scol),
                                              _ptr(l.R.sval))
-                self._keep.extend([l.R.soff, l.R.scol, l.R.sval])
+                d.rsrows = _ptr(l.R.srows)
+                self._keep.extend([l.R.soff, l.R.scol, l.R.sval, l.R.srows])
             relax = l.relax
             d.M = _ptr(relax.M if relax is not None else None)
             d.f = _ptr(l.f)

@@ -84,4 +84,8 @@ struct LevelDesc {
     const int64_t *rsoff;
     const int *rscol;
     const void *rsval;
+    // optional sigma-sort permutations (slot -> row id, < 0 = padding)
+    const int *srows;
+    const int *psrows;
+    const int *rsrows;
 };

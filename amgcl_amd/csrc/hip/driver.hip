@@ -45,23 +45,25 @@ extern "C" int amg_gemv_f32(int64_t, const float *, const float *, float *, hipS
 extern "C" int amg_cast_d2s(int64_t, const double *, float *, hipStream_t);
 extern "C" int amg_cast_s2d(int64_t, const float *, double *, hipStream_t);
 extern "C" int amg_sell_spmv_f64(int64_t, int64_t, const int64_t *, const int *,
-                                 const double *, const double *, double, double,
-                                 double *, hipStream_t);
+                                 const double *, const int *, const double *, double,
+                                 double, double *, hipStream_t);
 extern "C" int amg_sell_spmv_f32(int64_t, int64_t, const int64_t *, const int *,
-                                 const float *, const float *, double, double,
-                                 float *, hipStream_t);
+                                 const float *, const int *, const float *, double,
+                                 double, float *, hipStream_t);
 extern "C" int amg_sell_residual_f64(int64_t, int64_t, const int64_t *, const int *,
-                                     const double *, const double *, const double *,
-                                     double *, hipStream_t);
+                                     const double *, const int *, const double *,
+                                     const double *, double *, hipStream_t);
 extern "C" int amg_sell_relax_f64(int64_t, int64_t, const int64_t *, const int *,
-                                  const double *, const double *, const double *,
-                                  const double *, double *, hipStream_t);
+                                  const double *, const int *, const double *,
+                                  const double *, const double *, double *,
+                                  hipStream_t);
 extern "C" int amg_sell_residual_f32(int64_t, int64_t, const int64_t *, const int *,
-                                     const float *, const float *, const float *,
-                                     float *, hipStream_t);
+                                     const float *, const int *, const float *,
+                                     const float *, float *, hipStream_t);
 extern "C" int amg_sell_relax_f32(int64_t, int64_t, const int64_t *, const int *,
-                                  const float *, const float *, const float *,
-                                  const float *, float *, hipStream_t);
+                                  const float *, const int *, const float *,
+                                  const float *, const float *, float *,
+                                  hipStream_t);
 
 // value-type dispatch for the cycle (the fp32 hierarchy of mixed precision)
 template <typename T> struct ops;
@@ -80,19 +82,20 @@ template <> struct ops<double> {
         return amg_gemv_f64(n, inv, f, u, s);
     }
     static int sell_spmv(int64_t n, int64_t ns, const int64_t *soff, const int *c,
-                         const double *v, const double *x, double a, double b,
-                         double *y, hipStream_t s) {
-        return amg_sell_spmv_f64(n, ns, soff, c, v, x, a, b, y, s);
+                         const double *v, const int *sr, const double *x, double a,
+                         double b, double *y, hipStream_t s) {
+        return amg_sell_spmv_f64(n, ns, soff, c, v, sr, x, a, b, y, s);
     }
     static int sell_residual(int64_t n, int64_t ns, const int64_t *soff, const int *c,
-                             const double *v, const double *rhs, const double *x,
-                             double *r, hipStream_t s) {
-        return amg_sell_residual_f64(n, ns, soff, c, v, rhs, x, r, s);
+                             const double *v, const int *sr, const double *rhs,
+                             const double *x, double *r, hipStream_t s) {
+        return amg_sell_residual_f64(n, ns, soff, c, v, sr, rhs, x, r, s);
     }
     static int sell_relax(int64_t n, int64_t ns, const int64_t *soff, const int *c,
-                          const double *v, const double *M, const double *rhs,
-                          const double *x, double *xn, hipStream_t s) {
-        return amg_sell_relax_f64(n, ns, soff, c, v, M, rhs, x, xn, s);
+                          const double *v, const int *sr, const double *M,
+                          const double *rhs, const double *x, double *xn,
+                          hipStream_t s) {
+        return amg_sell_relax_f64(n, ns, soff, c, v, sr, M, rhs, x, xn, s);
     }
 };
 template <> struct ops<float> {
@@ -108,19 +111,20 @@ template <> struct ops<float> {
         return amg_gemv_f32(n, inv, f, u, s);
     }
     static int sell_spmv(int64_t n, int64_t ns, const int64_t *soff, const int *c,
-                         const float *v, const float *x, double a, double b,
-                         float *y, hipStream_t s) {
-        return amg_sell_spmv_f32(n, ns, soff, c, v, x, a, b, y, s);
+                         const float *v, const int *sr, const float *x, double a,
+                         double b, float *y, hipStream_t s) {
+        return amg_sell_spmv_f32(n, ns, soff, c, v, sr, x, a, b, y, s);
     }
     static int sell_residual(int64_t n, int64_t ns, const int64_t *soff, const int *c,
-                             const float *v, const float *rhs, const float *x,
-                             float *r, hipStream_t s) {
-        return amg_sell_residual_f32(n, ns, soff, c, v, rhs, x, r, s);
+                             const float *v, const int *sr, const float *rhs,
+                             const float *x, float *r, hipStream_t s) {
+        return amg_sell_residual_f32(n, ns, soff, c, v, sr, rhs, x, r, s);
     }
     static int sell_relax(int64_t n, int64_t ns, const int64_t *soff, const int *c,
-                          const float *v, const float *M, const float *rhs,
-                          const float *x, float *xn, hipStream_t s) {
-        return amg_sell_relax_f32(n, ns, soff, c, v, M, rhs, x, xn, s);
+                          const float *v, const int *sr, const float *M,
+                          const float *rhs, const float *x, float *xn,
+                          hipStream_t s) {
+        return amg_sell_relax_f32(n, ns, soff, c, v, sr, M, rhs, x, xn, s);
     }
 };
 
@@ -209,7 +213,7 @@ template <typename T>
 static int relax_swap(Driver *D, const LevelDesc &L, const T *rhs, T **x, T **xn) {
     if (L.nslice) {
         CHK(ops<T>::sell_relax(L.nrows, L.nslice, L.soff, L.scol, (const T *)L.sval,
-                               (const T *)L.M, rhs, *x, *xn, D->stream));
+                               L.srows, (const T *)L.M, rhs, *x, *xn, D->stream));
         T *tmp = *x;
         *x = *xn;
         *xn = tmp;
@@ -278,13 +282,13 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
     // t = f - A u ; f_next = R t
     if (L.nslice)
         CHK(ops<T>::sell_residual(L.nrows, L.nslice, L.soff, L.scol, (const T *)L.sval,
-                                  f, *u_io, *scratch, D->stream));
+                                  L.srows, f, *u_io, *scratch, D->stream));
     else
         CHK(ops<T>::residual(L.nrows, L.nnz, L.ptr, L.col, (const T *)L.val, f, *u_io,
                              *scratch, L.subw, D->stream));
     if (L.rnslice)
         CHK(ops<T>::sell_spmv(N.nrows, L.rnslice, L.rsoff, L.rscol, (const T *)L.rsval,
-                              *scratch, 1.0, 0.0, (T *)N.f, D->stream));
+                              L.rsrows, *scratch, 1.0, 0.0, (T *)N.f, D->stream));
     else
         CHK(ops<T>::spmv(N.nrows, L.rnnz, L.rptr, L.rcol, (const T *)L.rval, *scratch,
                          1.0, 0.0, (T *)N.f, L.rsubw, D->stream));
@@ -297,7 +301,7 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
     // u += P u_next
     if (L.pnslice)
         CHK(ops<T>::sell_spmv(L.nrows, L.pnslice, L.psoff, L.pscol, (const T *)L.psval,
-                              nu, 1.0, 1.0, *u_io, D->stream));
+                              L.psrows, nu, 1.0, 1.0, *u_io, D->stream));
     else
         CHK(ops<T>::spmv(L.nrows, L.pnnz, L.pptr, L.pcol, (const T *)L.pval, nu, 1.0,
                          1.0, *u_io, L.psubw, D->stream));
@@ -493,7 +497,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
 
     if (ksell)
         CHK(amg_sell_residual_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
-                                  rhs, x, r, st));
+                                  L0.srows, rhs, x, r, st));
     else
         CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
     CHK(amg_dot_f64(n, r, r, D->dotbuf_d, st));
@@ -515,7 +519,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
         }
         if (ksell)
             CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
-                                  p, 1.0, 0.0, q, st));
+                                  L0.srows, p, 1.0, 0.0, q, st));
         else
             CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, p, 1.0, 0.0, q, ksubw, st));
         CHK(amg_dot_f64(n, q, p, D->dotbuf_d, st));
@@ -570,7 +574,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
 
     if (ksell)
         CHK(amg_sell_residual_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
-                                  rhs, x, r, st));
+                                  L0.srows, rhs, x, r, st));
     else
         CHK(amg_residual_f64(n, knnz, kptr, kcol, kval, rhs, x, r, ksubw, st));
     CHK(hipMemcpyAsync(rh, r, n * sizeof(double), hipMemcpyDeviceToDevice, st));
@@ -598,7 +602,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         CHK(precond_apply_graphed(D, p, T, T_swap));
         if (ksell)
             CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
-                                  T, 1.0, 0.0, v, st));
+                                  L0.srows, T, 1.0, 0.0, v, st));
         else
             CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, v, ksubw, st));
         CHK(amg_dot_f64(n, rh, v, D->dotbuf_d, st));
@@ -613,7 +617,8 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
             CHK(precond_apply_graphed(D, s2, T, T_swap));
             if (ksell)
                 CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol,
-                                      (const double *)L0.sval, T, 1.0, 0.0, t2, st));
+                                      (const double *)L0.sval, L0.srows, T, 1.0, 0.0,
+                                      t2, st));
             else
                 CHK(amg_spmv_f64(n, knnz, kptr, kcol, kval, T, 1.0, 0.0, t2, ksubw, st));
             CHK(amg_dot2_f64(n, t2, s2, t2, t2, D->dotbuf_d, st));

@@ -171,10 +171,15 @@ extern "C" int amg_relax_diag_f64(int64_t nrows, int64_t nnz, const int *ptr,
 // reference's hybrid-ELL GPU format (amgcl/backend/vexcl_static_matrix.hpp
 // csr2ell_kernel :450), redesigned for 64-wide CDNA4 wavefronts.
 // ---------------------------------------------------------------------------
+// srows: optional sigma-sort row permutation (slot -> row id; < 0 = pad
+// slot).  Sorting rows by length inside sigma-sized windows cuts the slice
+// padding of ragged coarse levels; the permutation stays within a window,
+// so the scattered x/rhs/y accesses remain L2-local.
 template <typename T, int MODE>  // 0: y=aAx+by  1: r=rhs-Ax  2: xn=x+M(rhs-Ax)
 __global__ void sell_k(int64_t nrows, int64_t nslice,
                        const int64_t *__restrict__ soff, const int *__restrict__ col,
-                       const T *__restrict__ val, const T *__restrict__ x,
+                       const T *__restrict__ val, const int *__restrict__ srows,
+                       const T *__restrict__ x,
                        double alpha, double beta, const T *__restrict__ rhs,
                        const T *__restrict__ M, T *__restrict__ y) {
     const int wpb = blockDim.x / WAVE;
@@ -186,8 +191,8 @@ __global__ void sell_k(int64_t nrows, int64_t nslice,
         double acc = 0.0;
         for (int64_t j = beg + lane; j < end; j += WAVE)
             acc += (double)val[j] * (double)x[col[j]];
-        int64_t row = s * WAVE + lane;
-        if (row < nrows) {
+        int64_t row = srows ? (int64_t)srows[s * WAVE + lane] : s * WAVE + lane;
+        if (row >= 0 && row < nrows) {
             if (MODE == 0)
                 y[row] = (T)(beta == 0.0 ? alpha * acc
                                          : alpha * acc + beta * (double)y[row]);
@@ -204,51 +209,56 @@ __global__ void sell_k(int64_t nrows, int64_t nslice,
     sell_k<T, MODE><<<nblocks(nslice * WAVE), 256, 0, stream>>>(__VA_ARGS__)
 
 extern "C" int amg_sell_spmv_f64(int64_t nrows, int64_t nslice, const int64_t *soff,
-                                 const int *col, const double *val, const double *x,
-                                 double alpha, double beta, double *y,
+                                 const int *col, const double *val, const int *srows,
+                                 const double *x, double alpha, double beta, double *y,
                                  hipStream_t stream) {
-    SELL_LAUNCH(double, 0, nrows, nslice, soff, col, val, x, alpha, beta, nullptr,
-                nullptr, y);
+    SELL_LAUNCH(double, 0, nrows, nslice, soff, col, val, srows, x, alpha, beta,
+                nullptr, nullptr, y);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_sell_residual_f64(int64_t nrows, int64_t nslice, const int64_t *soff,
-                                     const int *col, const double *val,
+                                     const int *col, const double *val, const int *srows,
                                      const double *rhs, const double *x, double *r,
                                      hipStream_t stream) {
-    SELL_LAUNCH(double, 1, nrows, nslice, soff, col, val, x, 0.0, 0.0, rhs, nullptr, r);
+    SELL_LAUNCH(double, 1, nrows, nslice, soff, col, val, srows, x, 0.0, 0.0, rhs,
+                nullptr, r);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_sell_relax_f64(int64_t nrows, int64_t nslice, const int64_t *soff,
-                                  const int *col, const double *val, const double *M,
-                                  const double *rhs, const double *x, double *xn,
-                                  hipStream_t stream) {
-    SELL_LAUNCH(double, 2, nrows, nslice, soff, col, val, x, 0.0, 0.0, rhs, M, xn);
+                                  const int *col, const double *val, const int *srows,
+                                  const double *M, const double *rhs, const double *x,
+                                  double *xn, hipStream_t stream) {
+    SELL_LAUNCH(double, 2, nrows, nslice, soff, col, val, srows, x, 0.0, 0.0, rhs, M,
+                xn);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_sell_spmv_f32(int64_t nrows, int64_t nslice, const int64_t *soff,
-                                 const int *col, const float *val, const float *x,
-                                 double alpha, double beta, float *y,
+                                 const int *col, const float *val, const int *srows,
+                                 const float *x, double alpha, double beta, float *y,
                                  hipStream_t stream) {
-    SELL_LAUNCH(float, 0, nrows, nslice, soff, col, val, x, alpha, beta, nullptr,
-                nullptr, y);
+    SELL_LAUNCH(float, 0, nrows, nslice, soff, col, val, srows, x, alpha, beta,
+                nullptr, nullptr, y);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_sell_residual_f32(int64_t nrows, int64_t nslice, const int64_t *soff,
-                                     const int *col, const float *val, const float *rhs,
-                                     const float *x, float *r, hipStream_t stream) {
-    SELL_LAUNCH(float, 1, nrows, nslice, soff, col, val, x, 0.0, 0.0, rhs, nullptr, r);
+                                     const int *col, const float *val, const int *srows,
+                                     const float *rhs, const float *x, float *r,
+                                     hipStream_t stream) {
+    SELL_LAUNCH(float, 1, nrows, nslice, soff, col, val, srows, x, 0.0, 0.0, rhs,
+                nullptr, r);
     return (int)hipGetLastError();
 }
 
 extern "C" int amg_sell_relax_f32(int64_t nrows, int64_t nslice, const int64_t *soff,
-                                  const int *col, const float *val, const float *M,
-                                  const float *rhs, const float *x, float *xn,
-                                  hipStream_t stream) {
-    SELL_LAUNCH(float, 2, nrows, nslice, soff, col, val, x, 0.0, 0.0, rhs, M, xn);
+                                  const int *col, const float *val, const int *srows,
+                                  const float *M, const float *rhs, const float *x,
+                                  float *xn, hipStream_t stream) {
+    SELL_LAUNCH(float, 2, nrows, nslice, soff, col, val, srows, x, 0.0, 0.0, rhs, M,
+                xn);
     return (int)hipGetLastError();
 }
 #undef SELL_LAUNCH
@@ -258,19 +268,19 @@ extern "C" int amg_sell_relax_f32(int64_t nrows, int64_t nslice, const int64_t *
 // gathers of 64 rows' i-th entries, which land close together for sorted
 // matrices (setup-time cost matters: the fine level alone is ~11 GB).
 template <typename T>
-__global__ void sell_fill_k(int64_t nrows, const int *__restrict__ ptr,
+__global__ void sell_fill_k(int64_t nrows, int64_t nslice, const int *__restrict__ ptr,
                             const int *__restrict__ col, const T *__restrict__ val,
-                            const int64_t *__restrict__ soff, int *__restrict__ scol,
+                            const int64_t *__restrict__ soff,
+                            const int *__restrict__ srows, int *__restrict__ scol,
                             T *__restrict__ sval) {
     const int wpb = blockDim.x / WAVE;
     int wid = threadIdx.x / WAVE, lane = threadIdx.x & (WAVE - 1);
-    int64_t nslice = (nrows + WAVE - 1) / WAVE;
     int64_t s = (int64_t)blockIdx.x * wpb + wid;
     int64_t sstride = (int64_t)gridDim.x * wpb;
     for (; s < nslice; s += sstride) {
-        int64_t row = s * WAVE + lane;
+        int64_t row = srows ? (int64_t)srows[s * WAVE + lane] : s * WAVE + lane;
         int b = 0, len = 0;
-        if (row < nrows) {
+        if (row >= 0 && row < nrows) {
             b = ptr[row];
             len = ptr[row + 1] - b;
         }
@@ -289,23 +299,21 @@ __global__ void sell_fill_k(int64_t nrows, const int *__restrict__ ptr,
     }
 }
 
-extern "C" int amg_sell_fill_f64(int64_t nrows, const int *ptr, const int *col,
-                                 const double *val, const int64_t *soff, int *scol,
+extern "C" int amg_sell_fill_f64(int64_t nrows, int64_t nslice, const int *ptr,
+                                 const int *col, const double *val,
+                                 const int64_t *soff, const int *srows, int *scol,
                                  double *sval, hipStream_t stream) {
-    int64_t nslice = (nrows + WAVE - 1) / WAVE;
-    sell_fill_k<double><<<nblocks(nslice * WAVE), 256, 0, stream>>>(nrows, ptr, col,
-                                                                    val, soff, scol,
-                                                                    sval);
+    sell_fill_k<double><<<nblocks(nslice * WAVE), 256, 0, stream>>>(
+        nrows, nslice, ptr, col, val, soff, srows, scol, sval);
     return (int)hipGetLastError();
 }
 
-extern "C" int amg_sell_fill_f32(int64_t nrows, const int *ptr, const int *col,
-                                 const float *val, const int64_t *soff, int *scol,
+extern "C" int amg_sell_fill_f32(int64_t nrows, int64_t nslice, const int *ptr,
+                                 const int *col, const float *val,
+                                 const int64_t *soff, const int *srows, int *scol,
                                  float *sval, hipStream_t stream) {
-    int64_t nslice = (nrows + WAVE - 1) / WAVE;
-    sell_fill_k<float><<<nblocks(nslice * WAVE), 256, 0, stream>>>(nrows, ptr, col,
-                                                                   val, soff, scol,
-                                                                   sval);
+    sell_fill_k<float><<<nblocks(nslice * WAVE), 256, 0, stream>>>(
+        nrows, nslice, ptr, col, val, soff, srows, scol, sval);
     return (int)hipGetLastError();
 }
 
