@@ -66,13 +66,13 @@ _SIGS = {
                       + [ctypes.c_int, ctypes.c_void_p],
     "amg_coop_supported": [],
     # --- SELL-64 (wave-native sliced-ELL) solve kernels ---
-    "amg_sell_spmv_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+    "amg_sell_spmv_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 5
                          + [ctypes.c_double, ctypes.c_double]
                          + [ctypes.c_void_p] * 2,
     "amg_sell_residual_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
     "amg_sell_relax_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 9,
     "amg_sell_fill_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,
-    "amg_sell_spmv_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+    "amg_sell_spmv_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 5
                          + [ctypes.c_double, ctypes.c_double]
                          + [ctypes.c_void_p] * 2,
     "amg_sell_residual_f32": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8,

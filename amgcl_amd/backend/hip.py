@@ -62,10 +62,18 @@ class DeviceCSR:
             b += self.sval.numel() * self.sval.element_size()
         return b
 
-    def build_sell(self):
+    def build_sell(self, sigma=0):
         """Build the SELL-64 image of this matrix (kernels.hip rationale:
         wave-native layout; lane = row, slices column-major).  The CSR arrays
-        are kept — transfers and non-SELL paths still use them."""
+        are kept — transfers and non-SELL paths still use them.
+
+        sigma > 0 sorts rows by length inside sigma-sized windows to cut
+        slice padding.  MEASURED (r02): a NET LOSS on the 512^3 hierarchy
+        (solve 0.43 -> 0.64 s) — the permutation turns the perfectly
+        coalesced per-wave x/rhs/y accesses into window-local gathers,
+        which costs far more than the ~24% padding it saves.  Off by
+        default; kept as an explicit option (and for matrices so ragged
+        that padding dominates)."""
         if self.nslice:
             return self
         import os
@@ -92,12 +100,11 @@ class DeviceCSR:
         total = int(soff[-1].item())
         srows = None
         nnz = self.col.numel()
-        if total > 1.08 * nnz and n > (1 << 16):
+        if sigma and total > 1.08 * nnz and n > (1 << 16):
             # sigma-sorted slices: order rows by length inside sigma-sized
             # windows to cut the slice padding of ragged (coarse SA) levels;
             # the permutation stays window-local, so the scattered x/rhs/y
             # accesses remain L2-local
-            sigma = 8192
             npad = ((n + sigma - 1) // sigma) * sigma
             lens2 = (self.ptr[1:] - self.ptr[:-1]).to(torch.int64)
             if npad > n:

@@ -409,7 +409,7 @@ def test_sell_sigma_sorted_matches(hip):
     val = rng.standard_normal(nnz)
     A = CSR(n, n, ptr, col, val)
     Ad = hip.matrix(A)
-    Ad.build_sell()
+    Ad.build_sell(sigma=8192)
     assert Ad.srows is not None, "sigma sort should engage on this shape"
     x = rng.standard_normal(n)
     b = rng.standard_normal(n)
