@@ -109,8 +109,9 @@ _SIGS = {
     "amg_transpose_count": [ctypes.c_int64] + [ctypes.c_void_p] * 3,
     "amg_transpose_scatter": [ctypes.c_int64] + [ctypes.c_void_p] * 7,
     "amg_sort_rows": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
-    "amg_spgemm_count": [ctypes.c_int64] + [ctypes.c_void_p] * 8,
-    "amg_spgemm_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 10 + [ctypes.c_int, ctypes.c_void_p],
+    "amg_spgemm_count": [ctypes.c_int64] + [ctypes.c_void_p] * 9,
+    "amg_spgemm_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 10 + [ctypes.c_int]
+                       + [ctypes.c_void_p] * 2,
     "amg_gershgorin": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p],
     "amg_poisson_cnt": [ctypes.c_int64] + [ctypes.c_void_p] * 2,
     "amg_poisson_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 4,

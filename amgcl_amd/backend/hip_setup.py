@@ -179,9 +179,12 @@ def spgemm(A, B, sort=True):
     cptr = _new_ptr(A.nrows, dev)
     ub = t.empty(A.nrows, dtype=t.int32, device=dev)
     overflow = t.zeros(1, dtype=t.int32, device=dev)
+    # flags/scan + big-row worklist + device-side count (no host sync)
+    bigscratch = t.empty(2 * A.nrows + 1, dtype=t.int32, device=dev)
     check(lib().amg_spgemm_count(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
                                  B.ptr.data_ptr(), B.col.data_ptr(), ub.data_ptr(),
-                                 cptr[1:].data_ptr(), overflow.data_ptr(), _stream()),
+                                 cptr[1:].data_ptr(), overflow.data_ptr(),
+                                 bigscratch.data_ptr(), _stream()),
           "spgemm_count")
     _scan_ptr(cptr)
     ov, nnz = (int(v) for v in t.cat([overflow, cptr[-1:]]).cpu())
@@ -193,7 +196,7 @@ def spgemm(A, B, sort=True):
                                 A.val.data_ptr(), B.ptr.data_ptr(), B.col.data_ptr(),
                                 B.val.data_ptr(), ub.data_ptr(), cptr[1:].data_ptr(),
                                 ccol.data_ptr(), cval.data_ptr(), 1 if sort else 0,
-                                _stream()),
+                                bigscratch.data_ptr(), _stream()),
           "spgemm_fill")
     return device_csr(A.nrows, B.ncols, cptr, ccol, cval)
 

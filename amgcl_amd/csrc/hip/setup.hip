@@ -955,7 +955,9 @@ __device__ __forceinline__ int pfx_find(const int *pfx, int len, int t) {
 // upper bound: distinct keys <= ub, so any row with ub < SLOTS is safe in
 // the smaller table. (ub_lo, ub_hi] selects this launch's tier.
 template <int SLOTS>
-__global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
+__global__ void spgemm_count_k(int64_t an, const int *__restrict__ biglist,
+                               const int *__restrict__ nbig,
+                               const int *__restrict__ aptr,
                                const int *__restrict__ acol, const int *__restrict__ bptr,
                                const int *__restrict__ bcol, const int *__restrict__ ub,
                                int *__restrict__ cnt, int *__restrict__ overflow,
@@ -966,11 +968,13 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
     constexpr int SMASK = SLOTS - 1;
     int wid = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
-    int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+    const int64_t nwork = *nbig;
     int *tk = keys[wid];
 
-    for (; row < an; row += stride) {
+    for (; idx < nwork; idx += stride) {
+        int64_t row = biglist[idx];
         int total = ub[row];
         if (total <= ub_lo || total > ub_hi) continue;
         int ab = aptr[row], alen = aptr[row + 1] - ab;
@@ -1042,7 +1046,9 @@ __global__ void spgemm_count_k(int64_t an, const int *__restrict__ aptr,
 #define MIDLEN 96
 
 template <int SLOTS>
-__global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
+__global__ void spgemm_fill_k(int64_t an, const int *__restrict__ biglist,
+                              const int *__restrict__ nbig,
+                              const int *__restrict__ aptr,
                               const int *__restrict__ acol, const double *__restrict__ aval,
                               const int *__restrict__ bptr, const int *__restrict__ bcol,
                               const double *__restrict__ bval, const int *__restrict__ ub,
@@ -1057,12 +1063,14 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ aptr,
     constexpr int SMASK = SLOTS - 1;
     int wid = threadIdx.x / WAVE;
     int lane = threadIdx.x & (WAVE - 1);
-    int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+    const int64_t nwork = *nbig;
     int *tk = keys[wid];
     double *tv = vals[wid];
 
-    for (; row < an; row += stride) {
+    for (; idx < nwork; idx += stride) {
+        int64_t row = biglist[idx];
         int total = ub[row];
         if (total <= SGSMALL) continue;
         int ab = aptr[row], alen = aptr[row + 1] - ab;
@@ -1199,21 +1207,32 @@ extern "C" int amg_poisson_fill(int64_t n, const int *ptr_scanned, int *col, dou
     return (int)hipGetLastError();
 }
 
+// bigscratch: caller scratch of (2*an + 1) ints — flags/scan buffer,
+// the big-row list, and its device-side count (kernels read the count from
+// device memory, so no host sync is needed to size the launches).
 extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol, const int *bptr,
                                 const int *bcol, int *ub, int *cnt, int *overflow,
-                                hipStream_t s) {
+                                int *bigscratch, hipStream_t s) {
+    int *flags = bigscratch;
+    int *biglist = bigscratch + an;
+    int *nbig = bigscratch + 2 * an;
     spgemm_ub_k<<<nblk(an), 256, 0, s>>>(an, aptr, acol, bptr, ub);
+    spg_bigflag_k<<<nblk(an), 256, 0, s>>>(an, ub, flags);
+    int rc = scan_i32_device(flags, an, s);
+    if (rc) return rc;
+    spg_bigscatter_k<<<nblk(an), 256, 0, s>>>(an, ub, flags, biglist, nbig);
     spgemm_count_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub, cnt);
-    // single count tier: measured on 512^3, splitting the count by ub loses
-    // ~10 ms/setup (second full-row sweep) — count is not LDS-limited
-    spgemm_count_k<HSLOTS><<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub,
-                                                           cnt, overflow, SGSMALL, 1 << 30);
+    spgemm_count_k<HSLOTS><<<nblk(an * WAVE, 256, 512), 256, 0, s>>>(
+        an, biglist, nbig, aptr, acol, bptr, bcol, ub, cnt, overflow, SGSMALL, 1 << 30);
     return (int)hipGetLastError();
 }
 extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol, const double *aval,
                                const int *bptr, const int *bcol, const double *bval,
                                const int *ub, const int *cptr_scanned, int *ccol,
-                               double *cval, int do_sort, hipStream_t s) {
+                               double *cval, int do_sort, const int *bigscratch,
+                               hipStream_t s) {
+    const int *biglist = bigscratch + an;
+    const int *nbig = bigscratch + 2 * an;
     spgemm_fill_small_k<64><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr,
                                                             bcol, bval, ub, cptr_scanned,
                                                             ccol, cval, do_sort, 0, 48);
@@ -1221,12 +1240,12 @@ extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol, con
                                                                 bcol, bval, ub, cptr_scanned,
                                                                 ccol, cval, do_sort, 48,
                                                                 1 << 30);
-    // wave-per-row tiers by exact output length (known after the count pass)
-    spgemm_fill_k<128><<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol,
-                                                       bval, ub, cptr_scanned, ccol, cval,
-                                                       do_sort, 0, MIDLEN);
-    spgemm_fill_k<HSLOTS><<<nblk(an * WAVE), 256, 0, s>>>(an, aptr, acol, aval, bptr, bcol,
-                                                          bval, ub, cptr_scanned, ccol,
-                                                          cval, do_sort, MIDLEN, 1 << 30);
+    // wave-per-row tiers by exact output length, over the big-row list only
+    spgemm_fill_k<128><<<nblk(an * WAVE, 256, 512), 256, 0, s>>>(
+        an, biglist, nbig, aptr, acol, aval, bptr, bcol, bval, ub, cptr_scanned, ccol,
+        cval, do_sort, 0, MIDLEN);
+    spgemm_fill_k<HSLOTS><<<nblk(an * WAVE, 256, 512), 256, 0, s>>>(
+        an, biglist, nbig, aptr, acol, aval, bptr, bcol, bval, ub, cptr_scanned, ccol,
+        cval, do_sort, MIDLEN, 1 << 30);
     return (int)hipGetLastError();
 }
