@@ -954,6 +954,29 @@ __device__ __forceinline__ int pfx_find(const int *pfx, int len, int t) {
 // output length is unknown before counting, so rows tier by the product
 // upper bound: distinct keys <= ub, so any row with ub < SLOTS is safe in
 // the smaller table. (ub_lo, ub_hi] selects this launch's tier.
+// Big-row worklist: rows with ub > SGSMALL (the wave-per-row tiers).
+// Measured: letting the big-tier kernels scan-and-skip ALL rows wave-per-row
+// costs ~33 ms per fine-level product where only 0.03% of rows qualify.
+// Built by flag -> scan -> scatter (no global-atomic counter); the count
+// lives in device memory so the launches need no host sync.
+__global__ void spg_bigflag_k(int64_t an, const int *__restrict__ ub,
+                              int *__restrict__ flags) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < an; i += stride) flags[i] = ub[i] > SGSMALL ? 1 : 0;
+}
+
+__global__ void spg_bigscatter_k(int64_t an, const int *__restrict__ ub,
+                                 const int *__restrict__ flags_scanned,
+                                 int *__restrict__ list, int *__restrict__ nbig) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < an; i += stride) {
+        if (ub[i] > SGSMALL) list[flags_scanned[i] - 1] = (int)i;
+        if (i == an - 1) *nbig = flags_scanned[i];
+    }
+}
+
 template <int SLOTS>
 __global__ void spgemm_count_k(int64_t an, const int *__restrict__ biglist,
                                const int *__restrict__ nbig,
