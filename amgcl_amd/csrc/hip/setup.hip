@@ -1245,7 +1245,7 @@ extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol, co
     if (rc) return rc;
     spg_bigscatter_k<<<nblk(an), 256, 0, s>>>(an, ub, flags, biglist, nbig);
     spgemm_count_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub, cnt);
-    spgemm_count_k<HSLOTS><<<nblk(an * WAVE, 256, 512), 256, 0, s>>>(
+    spgemm_count_k<HSLOTS><<<nblk(an * WAVE), 256, 0, s>>>(
         an, biglist, nbig, aptr, acol, bptr, bcol, ub, cnt, overflow, SGSMALL, 1 << 30);
     return (int)hipGetLastError();
 }
@@ -1264,10 +1264,10 @@ extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol, con
                                                                 ccol, cval, do_sort, 48,
                                                                 1 << 30);
     // wave-per-row tiers by exact output length, over the big-row list only
-    spgemm_fill_k<128><<<nblk(an * WAVE, 256, 512), 256, 0, s>>>(
+    spgemm_fill_k<128><<<nblk(an * WAVE), 256, 0, s>>>(
         an, biglist, nbig, aptr, acol, aval, bptr, bcol, bval, ub, cptr_scanned, ccol,
         cval, do_sort, 0, MIDLEN);
-    spgemm_fill_k<HSLOTS><<<nblk(an * WAVE, 256, 512), 256, 0, s>>>(
+    spgemm_fill_k<HSLOTS><<<nblk(an * WAVE), 256, 0, s>>>(
         an, biglist, nbig, aptr, acol, aval, bptr, bcol, bval, ub, cptr_scanned, ccol,
         cval, do_sort, MIDLEN, 1 << 30);
     return (int)hipGetLastError();
