@@ -75,6 +75,30 @@ class DistAMG:
         self.coarse = None
         self._world = dist_A.world
         self._build(dist_A)
+        self._build_sell()
+
+    def _build_sell(self):
+        """SELL-64 images for the distributed levels' device-resident parts
+        (same policy/measurements as precond/amg.py: >= 500k local rows).
+        The DistBackend's spmv dispatches on A_loc, so the conversion is
+        transparent to the V-cycle."""
+        import os
+
+        if getattr(self.base, "name", "") != "hip" or os.environ.get("AMGCL_NO_SELL"):
+            return
+        from ..backend.hip import DeviceCSR
+
+        def conv(M):
+            if isinstance(M, DeviceCSR) and M.nrows >= 500_000:
+                M.build_sell()
+            elif isinstance(M, DistMatrix) and isinstance(M.A_loc, DeviceCSR) \
+                    and M.A_loc.nrows >= 500_000:
+                M.A_loc.build_sell()
+
+        for L in self.levels:
+            for M in (L.A, L.P, L.R):
+                if M is not None:
+                    conv(M)
 
     # --- setup ---------------------------------------------------------------
     def _strip_scipy(self, A: DistMatrix):
