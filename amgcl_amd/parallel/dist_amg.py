@@ -96,7 +96,8 @@ class DistAMG:
                 M.A_loc.build_sell()
 
         for L in self.levels:
-            for M in (L.A, L.P, L.R):
+            for name in ("A", "P", "R"):
+                M = getattr(L, name, None)  # __slots__: coarsest has no P/R
                 if M is not None:
                     conv(M)
 
