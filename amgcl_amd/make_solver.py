@@ -61,6 +61,18 @@ class MakeSolver:
     def solve(self, rhs, x=None, A=None):
         return self(rhs, x, A)
 
+    def rebuild(self, A_new):
+        """Rebuild the preconditioner for a matrix with changed coefficients
+        (reuses the transfer operators; parity: amgcl/amg.hpp rebuild) and
+        refresh the native driver, which holds raw pointers into the old
+        level tensors."""
+        self.P.rebuild(A_new)
+        self.A_host = A_new
+        if self._native is not None:
+            from .backend.native import try_native
+
+            self._native = try_native(self)
+
     def system_matrix(self):
         return self.P.system_matrix()
 

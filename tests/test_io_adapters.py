@@ -283,3 +283,17 @@ def test_unblock_bsr_and_as_scalar():
     x, it, r = s(b)
     assert r < 1e-8
     assert np.linalg.norm(b - Ah @ x) / np.linalg.norm(b) < 1e-7
+
+
+def test_make_solver_rebuild_refreshes():
+    """MakeSolver.rebuild: new coefficients, reused transfers, correct
+    solutions afterwards (the preconditioner-level rebuild alone would
+    leave a native driver with stale pointers on the GPU backend)."""
+    A, b = am.poisson3d(12, rhs="ones")
+    s = am.make_solver(A, {"precond": {"class": "amg", "coarse_enough": 200},
+                           "solver": {"type": "cg", "tol": 1e-9, "maxiter": 100}})
+    x1, it1, _ = s(b)
+    A2 = am.matrix.CSR(A.nrows, A.ncols, A.ptr, A.col, 2.0 * A.val)
+    s.rebuild(A2)
+    x2, it2, _ = s(b)
+    np.testing.assert_allclose(x2, x1 / 2.0, rtol=1e-7, atol=1e-10)
