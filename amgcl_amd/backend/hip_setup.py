@@ -35,6 +35,23 @@ def device_csr(nrows, ncols, ptr, col, val, subw=0):
     return DeviceCSR.from_tensors(nrows, ncols, ptr, col, val, subw)
 
 
+_SCRATCH = {}
+
+
+def _scratch_i32(n, dev):
+    """Process-lifetime int32 scratch (grow-only, keyed by device).  The
+    aggregation frontier lists and the SpGEMM big-row worklists are pure
+    scratch, but allocating them per call (up to 1.6 GB at 512^3) can miss
+    the torch allocator cache after interleaved allocations — a cold
+    hipMalloc costs ~28 ms/GB (profiles/README r02)."""
+    t = _torch()
+    key = str(dev)
+    buf = _SCRATCH.get(key)
+    if buf is None or buf.numel() < n:
+        _SCRATCH[key] = buf = t.empty(int(n * 1.25), dtype=t.int32, device=dev)
+    return buf[:n]
+
+
 def _new_ptr(n, device):
     t = _torch()
     return t.zeros(n + 1, dtype=t.int32, device=device)
@@ -98,7 +115,7 @@ def aggregates(A, eps_strong):
 
     rounds = ctypes.c_int(0)
     # frontier-compaction scratch: two ping-pong active lists + counter
-    lists = t.empty(3 * n, dtype=t.int32, device=dev)
+    lists = _scratch_i32(3 * n, dev)
     rc = lib().amg_agg_run(n, A.ptr.data_ptr(), A.col.data_ptr(), S.data_ptr(),
                            ids.data_ptr(), prov.data_ptr(), m1.data_ptr(),
                            newroot.data_ptr(), near.data_ptr(), remaining.data_ptr(),
@@ -180,7 +197,7 @@ def spgemm(A, B, sort=True):
     ub = t.empty(A.nrows, dtype=t.int32, device=dev)
     overflow = t.zeros(1, dtype=t.int32, device=dev)
     # flags/scan + big-row worklist + device-side count (no host sync)
-    bigscratch = t.empty(2 * A.nrows + 1, dtype=t.int32, device=dev)
+    bigscratch = _scratch_i32(2 * A.nrows + 1, dev)
     check(lib().amg_spgemm_count(A.nrows, A.ptr.data_ptr(), A.col.data_ptr(),
                                  B.ptr.data_ptr(), B.col.data_ptr(), ub.data_ptr(),
                                  cptr[1:].data_ptr(), overflow.data_ptr(),
