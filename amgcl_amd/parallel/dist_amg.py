@@ -59,6 +59,9 @@ class DistAMG:
             # (slaves_per_master comm split) — tail memory O(n^2) only on
             # masters instead of every rank.
             "coarse_group_size": 1,
+            # coarse direct solver on the master ranks: "dense" (device
+            # GEMV inverse), "skyline" (profile LU) or "splu"
+            "direct_solver": "dense",
         }
 
     def __init__(self, dist_A: DistMatrix, prm=None, backend=None):
@@ -311,7 +314,8 @@ class DistAMG:
                            else {"type": self.prm["relax"]["type"]})},
                 self.base)
         else:
-            self.coarse = self.base.coarse_solver(G_csr)
+            self.coarse = self.base.coarse_solver(
+                G_csr, kind=str(self.prm["direct_solver"]))
             self.tail = None
         self._is_cg_master = is_master
         self._coarse_n = A.n_global

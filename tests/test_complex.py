@@ -83,3 +83,23 @@ def test_complex_adjoint_transpose():
     A = CSR.from_scipy(a)
     T = A.transpose()
     assert abs(T.to_scipy() - a.conj().T).max() < 1e-15
+
+
+def test_cli_complex_mm(tmp_path, capsys):
+    """CLI end-to-end on a complex MatrixMarket system: the front door
+    auto-selects the complex128 CPU backend (reference: examples/solver.cpp
+    compiled for std::complex)."""
+    from amgcl_amd import cli, io
+
+    A, m = helmholtz(8)
+    p = str(tmp_path / "helm.mtx")
+    io.mm_write(p, A)
+    rc = cli.main(["-A", p, "-p", "solver.type=bicgstab",
+                   "-p", "solver.tol=1e-8",
+                   "-p", "precond.coarse_enough=200"])
+    assert rc in (0, None)
+    out = capsys.readouterr().out
+    assert "iters:" in out
+    resid = float([l for l in out.splitlines() if l.startswith("error:")][0]
+                  .split()[1])
+    assert resid < 1e-8

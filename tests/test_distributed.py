@@ -934,3 +934,35 @@ def test_dist_amg_subcommunicator_coarse(world):
     tails = {r: out[r][3] for r in range(world)}
     assert tails[0] and tails[2]
     assert not tails[1] and not tails[3]
+
+
+def _solve_dist_amg_skyline(rank, world):
+    import amgcl_amd as am
+    from amgcl_amd.parallel import make_dist_solver
+
+    n = 14
+    strip, b, row_beg, row_end = am.poisson3d_strip(n, rank, world, rhs="ones")
+    solve = make_dist_solver(
+        strip,
+        {"precond": {"class": "dist_amg", "coarse_enough": 400,
+                     "direct_solver": "skyline"},
+         "solver": {"type": "cg", "tol": 1e-8, "maxiter": 100}},
+        backend="cpu")
+    x, iters, resid = solve(b)
+    xg = solve.gather_solution(x)
+    return iters, resid, None if xg is None else xg.tolist()
+
+
+@pytest.mark.parametrize("world", [2])
+def test_dist_amg_skyline_coarse(world):
+    """Distributed coarse solve through the skyline (profile) LU instead of
+    the dense inverse (parity: mpi/direct_solver/skyline_lu.hpp)."""
+    out = spawn(world, _solve_dist_amg_skyline, 30561 + world)
+    import amgcl_amd as am
+
+    n = 14
+    A, b = am.poisson3d(n, rhs="ones")
+    iters, resid, xg = out[0]
+    assert resid < 1e-8
+    xg = np.array(xg)
+    assert np.linalg.norm(b - A @ xg) / np.linalg.norm(b) < 1e-7
