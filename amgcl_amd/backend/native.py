@@ -48,6 +48,16 @@ class LevelDescC(ctypes.Structure):
         ("srows", ctypes.c_void_p),
         ("psrows", ctypes.c_void_p),
         ("rsrows", ctypes.c_void_p),
+        ("cheb_degree", ctypes.c_int),
+        ("cheb_theta", ctypes.c_double),
+        ("cheb_delta", ctypes.c_double),
+        ("cheb_sigma1", ctypes.c_double),
+        ("cheb_d", ctypes.c_void_p),
+        ("bsize", ctypes.c_int),
+        ("nbrows", ctypes.c_int64),
+        ("bptr", ctypes.c_void_p),
+        ("bcol", ctypes.c_void_p),
+        ("bval", ctypes.c_void_p),
     ]
 
 
@@ -96,8 +106,9 @@ class NativeDriver:
     def __init__(self, amg, backend, solver_kind="cg", solver_prm=None):
         import torch
 
+        from ..relaxation.chebyshev import Chebyshev
         from ..relaxation.spai0 import DiagonalSmootherBase
-        from .hip import DeviceCSR, DeviceDenseSolver
+        from .hip import DeviceBSR, DeviceCSR, DeviceDenseSolver
 
         self.backend = backend
         self.solver_kind = solver_kind
@@ -107,20 +118,21 @@ class NativeDriver:
         self.maxiter = int(solver_prm["maxiter"])
 
         levels = amg.levels
-        if not all(isinstance(l.A, DeviceCSR) for l in levels):
+        if not all(isinstance(l.A, (DeviceCSR, DeviceBSR)) for l in levels):
             raise TypeError("native driver needs device-resident levels")
         self._mixed = bool(getattr(amg, "_mixed", False))
         for l in levels[:-1]:
-            if not isinstance(l.relax, DiagonalSmootherBase):
-                raise TypeError("native driver supports diagonal smoothers only")
+            if not isinstance(l.relax, (DiagonalSmootherBase, Chebyshev)):
+                raise TypeError(
+                    "native driver supports diagonal/Chebyshev smoothers")
         if amg.coarse_solve is not None and not isinstance(
             amg.coarse_solve, DeviceDenseSolver
         ):
             raise TypeError("native driver needs the dense coarse solver")
         if amg.coarse_solve is None and not isinstance(
-            levels[-1].relax, DiagonalSmootherBase
+            levels[-1].relax, (DiagonalSmootherBase, Chebyshev)
         ):
-            raise TypeError("native coarsest smoother must be diagonal")
+            raise TypeError("native coarsest smoother must be diagonal/Chebyshev")
 
         self._keep = []  # tensor refs
         descs = (LevelDescC * len(levels))()
@@ -128,9 +140,14 @@ class NativeDriver:
         for i, l in enumerate(levels):
             d = descs[i]
             A = l.A
-            d.nrows, d.nnz = A.nrows, int(A.nnz)
-            d.ptr, d.col, d.val = _ptr(A.ptr), _ptr(A.col), _ptr(A.val)
-            d.subw = A.subw or _auto_subw(A)
+            if isinstance(A, DeviceBSR):
+                d.nrows, d.nnz = A.nrows, int(A.nnz)
+                d.bsize, d.nbrows = A.bsize, A.nbrows
+                d.bptr, d.bcol, d.bval = _ptr(A.ptr), _ptr(A.col), _ptr(A.val)
+            else:
+                d.nrows, d.nnz = A.nrows, int(A.nnz)
+                d.ptr, d.col, d.val = _ptr(A.ptr), _ptr(A.col), _ptr(A.val)
+                d.subw = A.subw or _auto_subw(A)
             if l.P is not None:
                 P, R = l.P, l.R
                 d.pnnz = int(P.nnz)
@@ -139,7 +156,7 @@ class NativeDriver:
                 d.rnnz = int(R.nnz)
                 d.rptr, d.rcol, d.rval = _ptr(R.ptr), _ptr(R.col), _ptr(R.val)
                 d.rsubw = R.subw or _auto_subw(R)
-            if A.nslice:
+            if getattr(A, "nslice", 0):
                 d.nslice = A.nslice
                 d.soff, d.scol, d.sval = _ptr(A.soff), _ptr(A.scol), _ptr(A.sval)
                 d.srows = _ptr(A.srows)
@@ -157,14 +174,27 @@ class NativeDriver:
                 d.rsrows = _ptr(l.R.srows)
                 self._keep.extend([l.R.soff, l.R.scol, l.R.sval, l.R.srows])
             relax = l.relax
-            d.M = _ptr(relax.M if relax is not None else None)
+            if isinstance(relax, Chebyshev):
+                d.cheb_degree = relax.degree
+                d.cheb_theta = relax.theta
+                d.cheb_delta = relax.delta
+                d.cheb_sigma1 = relax.sigma1
+                cheb_d = torch.empty(A.nrows, dtype=torch.float64,
+                                     device=backend.device)
+                d.cheb_d = _ptr(cheb_d)
+                self._keep.append(cheb_d)
+                d.M = _ptr(relax.Dinv)  # optional D^-1 scaling (may be null)
+                if relax.Dinv is not None:
+                    self._keep.append(relax.Dinv)
+            else:
+                d.M = _ptr(relax.M if relax is not None else None)
             d.f = _ptr(l.f)
             d.u = _ptr(l.u)
             d.t = _ptr(l.t)
             self._keep.extend([A.ptr, A.col, A.val, l.f, l.u, l.t])
             if l.P is not None:
                 self._keep.extend([l.P.ptr, l.P.col, l.P.val, l.R.ptr, l.R.col, l.R.val])
-            if relax is not None:
+            if relax is not None and not isinstance(relax, Chebyshev):
                 self._keep.append(relax.M)
 
         inv = amg.coarse_solve.inv if amg.coarse_solve is not None else None

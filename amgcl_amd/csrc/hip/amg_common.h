@@ -88,4 +88,17 @@ struct LevelDesc {
     const int *srows;
     const int *psrows;
     const int *rsrows;
+    // Chebyshev smoothing (cheb_degree > 0 replaces the diagonal smoother;
+    // M then holds the optional D^-1 scaling or null)
+    int cheb_degree;
+    double cheb_theta;
+    double cheb_delta;
+    double cheb_sigma1;
+    void *cheb_d;  // extra per-level work vector for the recurrence
+    // BSR storage of A (bsize > 0; ptr/col/val above are then unused for A)
+    int bsize;
+    int64_t nbrows;
+    const int *bptr;
+    const int *bcol;
+    const double *bval;
 };

@@ -50,6 +50,20 @@ extern "C" int amg_sell_spmv_f64(int64_t, int64_t, const int64_t *, const int *,
 extern "C" int amg_sell_spmv_f32(int64_t, int64_t, const int64_t *, const int *,
                                  const float *, const int *, const float *, double,
                                  double, float *, hipStream_t);
+extern "C" int amg_bsr_spmv_f64(int64_t, int, const int *, const int *,
+                                const double *, const double *, double, double,
+                                double *, hipStream_t);
+extern "C" int amg_bsr_residual_f64(int64_t, int, const int *, const int *,
+                                    const double *, const double *, const double *,
+                                    double *, hipStream_t);
+extern "C" int amg_bsr_relax_f64(int64_t, int, const int *, const int *,
+                                 const double *, const double *, const double *,
+                                 const double *, double *, hipStream_t);
+extern "C" int amg_axpby_f32(int64_t, double, const float *, double, float *,
+                             hipStream_t);
+extern "C" int amg_vmul_f32(int64_t, double, const float *, const float *, double,
+                            float *, hipStream_t);
+extern "C" int amg_fill_f32(int64_t, double, float *, hipStream_t);
 extern "C" int amg_sell_residual_f64(int64_t, int64_t, const int64_t *, const int *,
                                      const double *, const int *, const double *,
                                      const double *, double *, hipStream_t);
@@ -97,6 +111,32 @@ template <> struct ops<double> {
                           hipStream_t s) {
         return amg_sell_relax_f64(n, ns, soff, c, v, sr, M, rhs, x, xn, s);
     }
+    static int axpby(int64_t n, double a, const double *x, double b, double *y,
+                     hipStream_t s) {
+        return amg_axpby_f64(n, a, x, b, y, s);
+    }
+    static int vmul(int64_t n, const double *m, const double *x, double *z,
+                    hipStream_t s) {
+        return amg_vmul_f64(n, 1.0, m, x, 0.0, z, s);
+    }
+    static int fill(int64_t n, double *x, hipStream_t s) {
+        return amg_fill_f64(n, 0.0, x, s);
+    }
+    static int bsr_spmv(int64_t nb, int bs, const int *p, const int *c,
+                        const double *v, const double *x, double a, double b,
+                        double *y, hipStream_t s) {
+        return amg_bsr_spmv_f64(nb, bs, p, c, v, x, a, b, y, s);
+    }
+    static int bsr_residual(int64_t nb, int bs, const int *p, const int *c,
+                            const double *v, const double *rhs, const double *x,
+                            double *r, hipStream_t s) {
+        return amg_bsr_residual_f64(nb, bs, p, c, v, rhs, x, r, s);
+    }
+    static int bsr_relax(int64_t nb, int bs, const int *p, const int *c,
+                         const double *v, const double *M, const double *rhs,
+                         const double *x, double *t, hipStream_t s) {
+        return amg_bsr_relax_f64(nb, bs, p, c, v, M, rhs, x, t, s);
+    }
 };
 template <> struct ops<float> {
     static int spmv(int64_t n, int64_t nnz, const int *p, const int *c, const float *v,
@@ -125,6 +165,31 @@ template <> struct ops<float> {
                           const float *rhs, const float *x, float *xn,
                           hipStream_t s) {
         return amg_sell_relax_f32(n, ns, soff, c, v, sr, M, rhs, x, xn, s);
+    }
+    static int axpby(int64_t n, double a, const float *x, double b, float *y,
+                     hipStream_t s) {
+        return amg_axpby_f32(n, a, x, b, y, s);
+    }
+    static int vmul(int64_t n, const float *m, const float *x, float *z,
+                    hipStream_t s) {
+        return amg_vmul_f32(n, 1.0, m, x, 0.0, z, s);
+    }
+    static int fill(int64_t n, float *x, hipStream_t s) {
+        return amg_fill_f32(n, 0.0, x, s);
+    }
+    // BSR is fp64-only (block_value + mixed precision is rejected upstream)
+    static int bsr_spmv(int64_t, int, const int *, const int *, const float *,
+                        const float *, double, double, float *, hipStream_t) {
+        return (int)hipErrorInvalidValue;
+    }
+    static int bsr_residual(int64_t, int, const int *, const int *, const float *,
+                            const float *, const float *, float *, hipStream_t) {
+        return (int)hipErrorInvalidValue;
+    }
+    static int bsr_relax(int64_t, int, const int *, const int *, const float *,
+                         const float *, const float *, const float *, float *,
+                         hipStream_t) {
+        return (int)hipErrorInvalidValue;
     }
 };
 
@@ -209,8 +274,59 @@ struct Driver {
         if (_rc) return _rc;            \
     } while (0)
 
+// residual through whichever storage this level's A carries (BSR > SELL > CSR)
+template <typename T>
+static int level_residual(Driver *D, const LevelDesc &L, const T *rhs, const T *x,
+                          T *r) {
+    if (L.bsize)
+        return ops<T>::bsr_residual(L.nbrows, L.bsize, L.bptr, L.bcol,
+                                    (const T *)L.bval, rhs, x, r, D->stream);
+    if (L.nslice)
+        return ops<T>::sell_residual(L.nrows, L.nslice, L.soff, L.scol,
+                                     (const T *)L.sval, L.srows, rhs, x, r,
+                                     D->stream);
+    return ops<T>::residual(L.nrows, L.nnz, L.ptr, L.col, (const T *)L.val, rhs, x, r,
+                            L.subw, D->stream);
+}
+
+// Chebyshev polynomial smoothing, in place on x (driver twin of
+// relaxation/chebyshev.py _polynomial; scratch r provided by the caller,
+// d lives in the level)
+template <typename T>
+static int cheb_apply(Driver *D, const LevelDesc &L, const T *rhs, T *x, T *r) {
+    T *d = (T *)L.cheb_d;
+    CHK(level_residual<T>(D, L, rhs, x, r));
+    if (L.M) CHK(ops<T>::vmul(L.nrows, (const T *)L.M, r, r, D->stream));
+    double rho = 1.0 / L.cheb_sigma1;
+    CHK(ops<T>::axpby(L.nrows, 1.0 / L.cheb_theta, r, 0.0, d, D->stream));
+    for (int k = 0; k < L.cheb_degree; ++k) {
+        CHK(ops<T>::axpby(L.nrows, 1.0, d, 1.0, x, D->stream));
+        CHK(level_residual<T>(D, L, rhs, x, r));
+        if (L.M) CHK(ops<T>::vmul(L.nrows, (const T *)L.M, r, r, D->stream));
+        double rho_next = 1.0 / (2.0 * L.cheb_sigma1 - rho);
+        CHK(ops<T>::axpby(L.nrows, 2.0 * rho_next / L.cheb_delta, r, rho_next * rho,
+                          d, D->stream));
+        rho = rho_next;
+    }
+    return 0;
+}
+
 template <typename T>
 static int relax_swap(Driver *D, const LevelDesc &L, const T *rhs, T **x, T **xn) {
+    if (L.cheb_degree) {  // in place, no pointer swap
+        CHK(cheb_apply<T>(D, L, rhs, *x, *xn));
+        return 0;
+    }
+    if (L.bsize) {
+        // xn = M o (rhs - A x); xn += x; swap
+        CHK(ops<T>::bsr_relax(L.nbrows, L.bsize, L.bptr, L.bcol, (const T *)L.bval,
+                              (const T *)L.M, rhs, *x, *xn, D->stream));
+        CHK(ops<T>::axpby(L.nrows, 1.0, *x, 1.0, *xn, D->stream));
+        T *tmp = *x;
+        *x = *xn;
+        *xn = tmp;
+        return 0;
+    }
     if (L.nslice) {
         CHK(ops<T>::sell_relax(L.nrows, L.nslice, L.soff, L.scol, (const T *)L.sval,
                                L.srows, (const T *)L.M, rhs, *x, *xn, D->stream));
@@ -248,12 +364,18 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
 
     // relax_zero writes into the swap buffer and swaps, so zero-guess and
     // general smooths have identical pointer parity (no copy-back needed)
-    auto relax_zero = [&](LevelDesc &LL, const T *ff, T **u2, T **sc) {
+    auto relax_zero = [&](LevelDesc &LL, const T *ff, T **u2, T **sc) -> int {
+        if (LL.cheb_degree || LL.bsize) {
+            // no fused zero-guess form for these smoothers: clear + general
+            CHK(ops<T>::fill(LL.nrows, *u2, D->stream));
+            return relax_swap<T>(D, LL, ff, u2, sc);
+        }
         relax_zero_k<T><<<nblocks_d(LL.nrows), 256, 0, D->stream>>>(
             LL.nrows, (const T *)LL.M, ff, *sc);
         T *tmp = *u2;
         *u2 = *sc;
         *sc = tmp;
+        return 0;
     };
 
     if (coarsest) {
@@ -262,7 +384,7 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
         } else {
             for (int i = 0; i < D->npre + D->npost; ++i) {
                 if (u_is_zero && i == 0) {
-                    relax_zero(L, f, u_io, scratch);
+                    CHK(relax_zero(L, f, u_io, scratch));
                     continue;
                 }
                 CHK(relax_swap<T>(D, L, f, u_io, scratch));
@@ -274,18 +396,13 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
     LevelDesc &N = D->lv[li + 1];
     for (int i = 0; i < D->npre; ++i) {
         if (u_is_zero && i == 0) {
-            relax_zero(L, f, u_io, scratch);
+            CHK(relax_zero(L, f, u_io, scratch));
             continue;
         }
         CHK(relax_swap<T>(D, L, f, u_io, scratch));
     }
     // t = f - A u ; f_next = R t
-    if (L.nslice)
-        CHK(ops<T>::sell_residual(L.nrows, L.nslice, L.soff, L.scol, (const T *)L.sval,
-                                  L.srows, f, *u_io, *scratch, D->stream));
-    else
-        CHK(ops<T>::residual(L.nrows, L.nnz, L.ptr, L.col, (const T *)L.val, f, *u_io,
-                             *scratch, L.subw, D->stream));
+    CHK(level_residual<T>(D, L, f, *u_io, *scratch));
     if (L.rnslice)
         CHK(ops<T>::sell_spmv(N.nrows, L.rnslice, L.rsoff, L.rscol, (const T *)L.rsval,
                               L.rsrows, *scratch, 1.0, 0.0, (T *)N.f, D->stream));
@@ -480,6 +597,7 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     const double *kval = D->f32 ? D->a64_val : (const double *)L0.val;
     const int ksubw = D->f32 ? D->a64_subw : L0.subw;
     const int64_t ksell = D->f32 ? 0 : L0.nslice;  // fp64 SELL fine operator
+    const int kbsr = D->f32 ? 0 : L0.bsize;        // fp64 BSR fine operator
     hipStream_t st = D->stream;
     double dots[2];
 
@@ -495,7 +613,10 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
 
-    if (ksell)
+    if (kbsr)
+        CHK(amg_bsr_residual_f64(L0.nbrows, kbsr, L0.bptr, L0.bcol,
+                                 (const double *)L0.bval, rhs, x, r, st));
+    else if (ksell)
         CHK(amg_sell_residual_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
                                   L0.srows, rhs, x, r, st));
     else
@@ -517,7 +638,10 @@ extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r, d
         } else {
             CHK(amg_axpby_f64(n, 1.0, s, rho1 / rho2, p, st));
         }
-        if (ksell)
+        if (kbsr)
+            CHK(amg_bsr_spmv_f64(L0.nbrows, kbsr, L0.bptr, L0.bcol,
+                                 (const double *)L0.bval, p, 1.0, 0.0, q, st));
+        else if (ksell)
             CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
                                   L0.srows, p, 1.0, 0.0, q, st));
         else
@@ -557,6 +681,7 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
     const double *kval = D->f32 ? D->a64_val : (const double *)L0.val;
     const int ksubw = D->f32 ? D->a64_subw : L0.subw;
     const int64_t ksell = D->f32 ? 0 : L0.nslice;  // fp64 SELL fine operator
+    const int kbsr = D->f32 ? 0 : L0.bsize;        // fp64 BSR fine operator
     hipStream_t st = D->stream;
     double dots[2];
 
@@ -572,7 +697,10 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
     }
     double eps = tol * norm_rhs > abstol ? tol * norm_rhs : abstol;
 
-    if (ksell)
+    if (kbsr)
+        CHK(amg_bsr_residual_f64(L0.nbrows, kbsr, L0.bptr, L0.bcol,
+                                 (const double *)L0.bval, rhs, x, r, st));
+    else if (ksell)
         CHK(amg_sell_residual_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
                                   L0.srows, rhs, x, r, st));
     else
@@ -600,7 +728,10 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         }
         // v = A (M^-1 p);  T = M^-1 p
         CHK(precond_apply_graphed(D, p, T, T_swap));
-        if (ksell)
+        if (kbsr)
+            CHK(amg_bsr_spmv_f64(L0.nbrows, kbsr, L0.bptr, L0.bcol,
+                                 (const double *)L0.bval, T, 1.0, 0.0, v, st));
+        else if (ksell)
             CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol, (const double *)L0.sval,
                                   L0.srows, T, 1.0, 0.0, v, st));
         else
@@ -615,7 +746,10 @@ extern "C" int amg_driver_bicgstab(void *h, const double *rhs, double *x, double
         res = sqrt(dots[0]);
         if (res > eps) {
             CHK(precond_apply_graphed(D, s2, T, T_swap));
-            if (ksell)
+            if (kbsr)
+                CHK(amg_bsr_spmv_f64(L0.nbrows, kbsr, L0.bptr, L0.bcol,
+                                     (const double *)L0.bval, T, 1.0, 0.0, t2, st));
+            else if (ksell)
                 CHK(amg_sell_spmv_f64(n, ksell, L0.soff, L0.scol,
                                       (const double *)L0.sval, L0.srows, T, 1.0, 0.0,
                                       t2, st));
