@@ -401,7 +401,7 @@ def test_sell_sigma_sorted_matches(hip):
     sort actually engages (pad > 1.08, n > 65536)."""
     rng = np.random.default_rng(17)
     n = 140_000
-    lens = rng.integers(1, 64, size=n)
+    lens = rng.integers(0, 64, size=n)  # includes EMPTY rows
     ptr = np.zeros(n + 1, dtype=np.int64)
     np.cumsum(lens, out=ptr[1:])
     nnz = int(ptr[-1])
