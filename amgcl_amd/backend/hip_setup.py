@@ -231,6 +231,12 @@ def download(A):
 
 
 def poisson3d_device(n, device="cuda"):
+    if 7 * n**3 >= 2**31:
+        raise ValueError(
+            f"n={n}: {7*n**3/1e9:.2f}G nonzeros exceed the framework's int32 "
+            "index design (2^31); the single-GPU ceiling is ~673^3 (305M "
+            "unknowns, 2.1G nnz) — shard larger problems across ranks "
+            "(bench.py --gpus N)")
     """7-point Poisson fixture generated directly on the GPU."""
     t = _torch()
     n3 = n**3
