@@ -157,9 +157,12 @@ def test_native_driver_chebyshev(hip):
     s2 = am.make_solver(A, prm2, backend=hip)
     assert s2._native is None
     x2, it2, r2 = s2(b)
-    assert it1 == it2, (it1, it2)
-    np.testing.assert_allclose(hip.to_host(x1), hip.to_host(x2),
-                               rtol=1e-10, atol=1e-10)
+    # the native CG fuses the x/r update (cg_tail), so the residual history
+    # differs in the last bits — allow a 1-iteration difference
+    assert abs(it1 - it2) <= 1, (it1, it2)
+    assert r2 < 1e-8
+    xa, xb = hip.to_host(x1), hip.to_host(x2)
+    assert np.linalg.norm(xa - xb) / np.linalg.norm(xa) < 1e-8
 
 
 def test_native_driver_bsr_levels(hip):
