@@ -140,17 +140,27 @@ class SmoothedAggregation:
 
     def _transfer_operators_device(self, A):
         """Device twin (backend/hip_setup.py): same algorithm, same keys.
-        Features the device engine does not cover yet (pointwise/block
-        aggregation, nullspace) raise OverflowError, which AMG handles by
-        downloading the level and continuing with the host engine."""
+        Block (pointwise) aggregation and the nullspace-QR tentative run on
+        the device too (torch-orchestrated over the device aggregation /
+        spgemm engines — host twins: pointwise_matrix, expand_strong,
+        tentative_nullspace, filtered_smoother_matrix in core.cpp);
+        AMGCL_HOST_NULLSPACE=1 restores the host fallback."""
+        import os
+
         from ..backend import hip_setup
         from ..profiler import prof
 
-        if int(self.prm["block_size"]) > 1 or self.B is not None:
-            raise OverflowError("block/nullspace coarsening runs on the host")
+        bsize = int(self.prm["block_size"])
+        if (bsize > 1 or self.B is not None) and os.environ.get(
+                "AMGCL_HOST_NULLSPACE"):
+            raise OverflowError("block/nullspace coarsening forced to the host")
 
-        with prof.scope("aggregates(dev)"):
-            naggr, ids, strong = hip_setup.aggregates(A, self.eps_strong)
+        if bsize > 1 and A.nrows % bsize == 0:
+            with prof.scope("aggregates(dev)"):
+                naggr, ids, strong = self._pointwise_aggregates_device(A, bsize)
+        else:
+            with prof.scope("aggregates(dev)"):
+                naggr, ids, strong = hip_setup.aggregates(A, self.eps_strong)
         self.eps_strong *= 0.5
         omega = float(self.prm["relax"])
         if self.prm["estimate_spectral_radius"]:
@@ -161,10 +171,127 @@ class SmoothedAggregation:
         else:
             omega *= 2.0 / 3.0
         with prof.scope("smooth_P(dev)"):
-            P = hip_setup.smoothed_prolongation(A, strong, ids, naggr, omega)
+            if self.B is None:
+                P = hip_setup.smoothed_prolongation(A, strong, ids, naggr, omega)
+            else:
+                P = self._nullspace_prolongation_device(A, strong, ids, naggr,
+                                                        omega)
         with prof.scope("transpose_R(dev)"):
             R = hip_setup.transpose(P)
         return P, R
+
+    def _pointwise_aggregates_device(self, A, bsize):
+        """Pointwise (block) aggregation on the device: condense the block
+        rows to a scalar matrix of Frobenius norms (torch sparse coalesce),
+        run the device MIS on it, expand ids and the strong mask back to
+        scalar unknowns (host twins: pointwise_matrix + expand_strong)."""
+        import torch
+
+        from ..backend import hip_setup
+        from ..backend.hip import DeviceCSR
+
+        n = A.nrows
+        dev = A.val.device
+        t64 = torch.int64
+        lens = (A.ptr[1:] - A.ptr[:-1]).to(t64)
+        row_of = torch.repeat_interleave(
+            torch.arange(n, device=dev, dtype=t64), lens)
+        col64 = A.col.to(t64)
+        np_ = n // bsize
+        rpt = row_of // bsize
+        cpt = col64 // bsize
+        pt = torch.sparse_coo_tensor(torch.stack([rpt, cpt]), A.val * A.val,
+                                     (np_, np_)).coalesce()
+        prow, pcol = pt.indices()
+        pval = pt.values().sqrt()
+        pptr = torch.zeros(np_ + 1, dtype=t64, device=dev)
+        torch.cumsum(torch.bincount(prow, minlength=np_), 0, out=pptr[1:])
+        Apt = DeviceCSR.from_tensors(
+            np_, np_, pptr.to(torch.int32).contiguous(),
+            pcol.to(torch.int32).contiguous(), pval.contiguous())
+        naggr, ids_pt, S_pt = hip_setup.aggregates(Apt, self.eps_strong)
+        ids = ids_pt.repeat_interleave(bsize).contiguous()
+        # strong mask per scalar entry = the pointwise entry's flag
+        # (diagonal-point entries are never strong)
+        key = prow * np_ + pcol  # ascending (coalesce ordering)
+        ekey = rpt * np_ + cpt
+        pos = torch.searchsorted(key, ekey)
+        S = S_pt[pos]
+        S = torch.where(rpt == cpt, torch.zeros_like(S), S).contiguous()
+        self._keep_pt = (Apt,)  # keep alive until setup finishes
+        return naggr, ids, S
+
+    def _nullspace_prolongation_device(self, A, strong, ids, naggr, omega):
+        """Nullspace tentative prolongation (per-aggregate batched QR) +
+        filtered-Jacobi smoothing, device-resident (host twins:
+        tentative_nullspace + filtered_smoother_matrix).  Assumes every row
+        carries a diagonal entry (true for the supported generators)."""
+        import torch
+
+        from ..backend import hip_setup
+        from ..backend.hip import DeviceCSR
+
+        n = A.nrows
+        dev = A.val.device
+        t64 = torch.int64
+        k = self.B.shape[1]
+        B_d = torch.from_numpy(np.ascontiguousarray(self.B)).to(dev)
+
+        ids64 = ids.to(t64)
+        assigned = ids64 >= 0
+        rows_assigned = torch.nonzero(assigned).ravel()
+        agg_of = ids64[rows_assigned]
+        order = torch.argsort(agg_of, stable=True)
+        members = rows_assigned[order]            # grouped by aggregate
+        magg = agg_of[order]
+        counts = torch.bincount(magg, minlength=naggr)
+        maxd = int(counts.max().item())
+        offs = torch.zeros(naggr + 1, dtype=t64, device=dev)
+        torch.cumsum(counts, 0, out=offs[1:])
+        slot = (torch.arange(members.numel(), device=dev, dtype=t64)
+                - offs[:-1].repeat_interleave(counts))
+        batch = torch.zeros(naggr, maxd, k, dtype=torch.float64, device=dev)
+        batch[magg, slot] = B_d[members]
+        Q, Rf = torch.linalg.qr(batch, mode="reduced")
+        # coarse-level nullspace = per-aggregate R blocks (naggr*k x k)
+        self.B = Rf.reshape(naggr * k, k).cpu().numpy()
+        self.prm["block_size"] = k if k > 1 else 1
+
+        # P_tent rows in row order: row i gets Q[id[i], slot_of(i), :]
+        slot_of = torch.zeros(n, dtype=t64, device=dev)
+        slot_of[members] = slot
+        vals = Q[ids64[rows_assigned], slot_of[rows_assigned]]  # (nass, k)
+        pptr = torch.zeros(n + 1, dtype=t64, device=dev)
+        torch.cumsum(assigned.to(t64) * k, 0, out=pptr[1:])
+        pcols = (ids64[rows_assigned].unsqueeze(1) * k
+                 + torch.arange(k, device=dev, dtype=t64)).reshape(-1)
+        P_tent = DeviceCSR.from_tensors(
+            n, naggr * k, pptr.to(torch.int32).contiguous(),
+            pcols.to(torch.int32).contiguous(), vals.reshape(-1).contiguous())
+
+        # S_F = I - omega Df^-1 Af (weak off-diagonals folded into Df)
+        lens = (A.ptr[1:] - A.ptr[:-1]).to(t64)
+        row_of = torch.repeat_interleave(
+            torch.arange(n, device=dev, dtype=t64), lens)
+        col64 = A.col.to(t64)
+        dia_mask = row_of == col64
+        Sb = strong.to(torch.bool)
+        contrib = torch.where(dia_mask | ~Sb, A.val, torch.zeros_like(A.val))
+        dia = torch.zeros(n, dtype=torch.float64, device=dev)
+        dia.scatter_add_(0, row_of, contrib)
+        w = torch.where(dia != 0, -omega / dia,
+                        torch.zeros_like(dia))
+        keep = dia_mask | Sb
+        kr = row_of[keep]
+        kv = torch.where(dia_mask[keep],
+                         torch.full_like(A.val[keep], 1.0 - omega),
+                         w[kr] * A.val[keep])
+        sptr = torch.zeros(n + 1, dtype=t64, device=dev)
+        torch.cumsum(torch.bincount(kr, minlength=n), 0, out=sptr[1:])
+        S_F = DeviceCSR.from_tensors(
+            n, n, sptr.to(torch.int32).contiguous(),
+            A.col[keep].contiguous(), kv.contiguous())
+        return hip_setup.spgemm(S_F, P_tent, sort=True)
 
     def coarse_operator(self, A, P, R):
         return galerkin(R, A, P)

@@ -64,7 +64,8 @@ def main():
     Ah, bh, coords = elasticity3d(n)
     B = rigid_body_modes(coords)
     bd = hip.from_host(bh)
-    run(f"elasticity {n}^3 nodes CG+SA(esr)+RBM+Chebyshev+BSR(3)", Ah, bd,
+    Ad = hip.matrix(Ah)  # device input -> device block/nullspace setup
+    run(f"elasticity {n}^3 nodes CG+SA(esr)+RBM+Chebyshev+BSR(3)", Ad, bd,
         {"precond": {"class": "amg", "block_value": 3, "keep_host_matrices": True,
                      "relax": {"type": "chebyshev"},
                      "coarsening": {"type": "smoothed_aggregation",
