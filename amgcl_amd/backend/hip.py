@@ -178,6 +178,44 @@ class DeviceBSR:
     def nnz(self):
         return self.val.numel()
 
+    @classmethod
+    def from_device(cls, dcsr, bsize, device):
+        """Convert a device CSR to BSR entirely on the GPU (torch ops): the
+        block pattern is the coalesced pointwise pattern, values scatter by
+        (block index, r, c).  Host twin: _core.csr_to_bsr."""
+        import torch
+
+        b = int(bsize)
+        n = dcsr.nrows
+        nb = n // b
+        dev = dcsr.val.device
+        t64 = torch.int64
+        lens = (dcsr.ptr[1:] - dcsr.ptr[:-1]).to(t64)
+        row_of = torch.repeat_interleave(torch.arange(n, device=dev, dtype=t64),
+                                         lens)
+        col64 = dcsr.col.to(t64)
+        rpt = row_of // b
+        cpt = col64 // b
+        key = rpt * nb + cpt
+        ukey = torch.unique(key, sorted=True)  # block pattern, row-major
+        blk = torch.searchsorted(ukey, key)    # block index per scalar entry
+        nblocks = ukey.numel()
+        bptr = torch.zeros(nb + 1, dtype=t64, device=dev)
+        torch.cumsum(torch.bincount(ukey // nb, minlength=nb), 0, out=bptr[1:])
+        bcol = (ukey % nb).to(torch.int32).contiguous()
+        bval = torch.zeros(nblocks * b * b, dtype=dcsr.val.dtype, device=dev)
+        dest = blk * (b * b) + (row_of % b) * b + (col64 % b)
+        bval[dest] = dcsr.val
+        self = cls.__new__(cls)
+        self.nbrows = nb
+        self.bsize = b
+        self.nrows = n
+        self.ncols = dcsr.ncols
+        self.ptr = bptr.to(torch.int32).contiguous()
+        self.col = bcol
+        self.val = bval
+        return self
+
 
 @register("hip")
 class HipBackend:

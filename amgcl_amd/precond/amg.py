@@ -288,10 +288,12 @@ class AMG:
             A = lvl.A
             if A.nrows % bsize:
                 continue
+            if isinstance(A, DeviceCSR):
+                # device-resident conversion, no D2H round-trip
+                lvl.A = DeviceBSR.from_device(A, bsize, self.backend.device)
+                continue
             host = lvl.A_host
             if host is None or not isinstance(host, CSR):
-                host = hip_setup.download(A) if isinstance(A, DeviceCSR) else host
-            if host is None:
                 continue
             lvl.A = DeviceBSR(host, bsize, self.backend.device)
 

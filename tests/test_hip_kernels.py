@@ -427,3 +427,32 @@ def test_sell_sigma_sorted_matches(hip):
     hip.relax_diag(Ad, md, bd, x2, td)
     np.testing.assert_allclose(hip.to_host(x2), x + m * (b - ref),
                                rtol=1e-12, atol=1e-10)
+
+
+def test_bsr_from_device_matches_host(hip):
+    """Device CSR->BSR conversion (torch ops) produces the same BSR arrays
+    as the host converter (_core.csr_to_bsr)."""
+    from amgcl_amd.backend.hip import DeviceBSR
+
+    rng = np.random.default_rng(23)
+    nb, B = 120, 3
+    a = sp.random(nb, nb, density=0.08, random_state=rng, format="csr")
+    a = (a + sp.identity(nb)).tocsr()
+    a.data[:] = 1.0
+    blocks = rng.standard_normal((a.nnz, B, B))
+    # zero a few entries inside blocks so scalar CSR has partial blocks
+    blocks[rng.random(a.nnz) < 0.3, 0, 1] = 0.0
+    m = sp.bsr_matrix((blocks, a.indices, a.indptr),
+                      shape=(nb * B, nb * B)).tocsr()
+    m.eliminate_zeros()
+    m.sort_indices()
+    A = CSR(nb * B, nb * B, m.indptr, m.indices, m.data)
+    d_host = DeviceBSR(A, B, hip.device)
+    d_dev = DeviceBSR.from_device(hip.matrix(A), B, hip.device)
+    assert d_host.nbrows == d_dev.nbrows
+    np.testing.assert_array_equal(d_host.ptr.cpu().numpy(),
+                                  d_dev.ptr.cpu().numpy())
+    np.testing.assert_array_equal(d_host.col.cpu().numpy(),
+                                  d_dev.col.cpu().numpy())
+    np.testing.assert_allclose(d_host.val.cpu().numpy(),
+                               d_dev.val.cpu().numpy(), atol=0, rtol=0)
