@@ -252,7 +252,24 @@ class SmoothedAggregation:
                 - offs[:-1].repeat_interleave(counts))
         batch = torch.zeros(naggr, maxd, k, dtype=torch.float64, device=dev)
         batch[magg, slot] = B_d[members]
-        Q, Rf = torch.linalg.qr(batch, mode="reduced")
+        # vectorized MGS with one re-orthogonalization (the host engine's
+        # exact algorithm, batched over aggregates; the padded zero rows
+        # contribute nothing to the inner products).  Measured faster than
+        # MAGMA's batched small-matrix QR and keeps host parity.
+        Q = batch
+        Rf = torch.zeros(naggr, k, k, dtype=torch.float64, device=dev)
+        for c in range(k):
+            qc = Q[:, :, c]
+            for _pass in range(2):
+                for p in range(c):
+                    qp = Q[:, :, p]
+                    h = (qp * qc).sum(dim=1)
+                    qc -= h.unsqueeze(1) * qp
+                    Rf[:, p, c] += h
+            nrm = qc.square().sum(dim=1).sqrt()
+            Rf[:, c, c] = nrm
+            safe = nrm > 1e-300
+            qc /= torch.where(safe, nrm, torch.ones_like(nrm)).unsqueeze(1)
         # coarse-level nullspace = per-aggregate R blocks (naggr*k x k)
         self.B = Rf.reshape(naggr * k, k).cpu().numpy()
         self.prm["block_size"] = k if k > 1 else 1
