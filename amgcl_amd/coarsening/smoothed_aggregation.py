@@ -190,18 +190,21 @@ class SmoothedAggregation:
         from ..backend import hip_setup
         from ..backend.hip import DeviceCSR
 
+        from ..profiler import prof
+
         n = A.nrows
         dev = A.val.device
         t64 = torch.int64
-        lens = (A.ptr[1:] - A.ptr[:-1]).to(t64)
-        row_of = torch.repeat_interleave(
-            torch.arange(n, device=dev, dtype=t64), lens)
-        col64 = A.col.to(t64)
-        np_ = n // bsize
-        rpt = row_of // bsize
-        cpt = col64 // bsize
-        pt = torch.sparse_coo_tensor(torch.stack([rpt, cpt]), A.val * A.val,
-                                     (np_, np_)).coalesce()
+        with prof.scope("pt_build"):
+            lens = (A.ptr[1:] - A.ptr[:-1]).to(t64)
+            row_of = torch.repeat_interleave(
+                torch.arange(n, device=dev, dtype=t64), lens)
+            col64 = A.col.to(t64)
+            np_ = n // bsize
+            rpt = row_of // bsize
+            cpt = col64 // bsize
+            pt = torch.sparse_coo_tensor(torch.stack([rpt, cpt]), A.val * A.val,
+                                         (np_, np_)).coalesce()
         prow, pcol = pt.indices()
         pval = pt.values().sqrt()
         pptr = torch.zeros(np_ + 1, dtype=t64, device=dev)
@@ -209,15 +212,17 @@ class SmoothedAggregation:
         Apt = DeviceCSR.from_tensors(
             np_, np_, pptr.to(torch.int32).contiguous(),
             pcol.to(torch.int32).contiguous(), pval.contiguous())
-        naggr, ids_pt, S_pt = hip_setup.aggregates(Apt, self.eps_strong)
-        ids = ids_pt.repeat_interleave(bsize).contiguous()
-        # strong mask per scalar entry = the pointwise entry's flag
-        # (diagonal-point entries are never strong)
-        key = prow * np_ + pcol  # ascending (coalesce ordering)
-        ekey = rpt * np_ + cpt
-        pos = torch.searchsorted(key, ekey)
-        S = S_pt[pos]
-        S = torch.where(rpt == cpt, torch.zeros_like(S), S).contiguous()
+        with prof.scope("pt_mis"):
+            naggr, ids_pt, S_pt = hip_setup.aggregates(Apt, self.eps_strong)
+        with prof.scope("pt_expand"):
+            ids = ids_pt.repeat_interleave(bsize).contiguous()
+            # strong mask per scalar entry = the pointwise entry's flag
+            # (diagonal-point entries are never strong)
+            key = prow * np_ + pcol  # ascending (coalesce ordering)
+            ekey = rpt * np_ + cpt
+            pos = torch.searchsorted(key, ekey)
+            S = S_pt[pos]
+            S = torch.where(rpt == cpt, torch.zeros_like(S), S).contiguous()
         self._keep_pt = (Apt,)  # keep alive until setup finishes
         return naggr, ids, S
 
@@ -237,6 +242,8 @@ class SmoothedAggregation:
         k = self.B.shape[1]
         B_d = torch.from_numpy(np.ascontiguousarray(self.B)).to(dev)
 
+        from ..profiler import prof
+
         ids64 = ids.to(t64)
         assigned = ids64 >= 0
         rows_assigned = torch.nonzero(assigned).ravel()
@@ -250,6 +257,7 @@ class SmoothedAggregation:
         torch.cumsum(counts, 0, out=offs[1:])
         slot = (torch.arange(members.numel(), device=dev, dtype=t64)
                 - offs[:-1].repeat_interleave(counts))
+        prof.tic("ns_qr")
         batch = torch.zeros(naggr, maxd, k, dtype=torch.float64, device=dev)
         batch[magg, slot] = B_d[members]
         # vectorized MGS with one re-orthogonalization (the host engine's
@@ -274,6 +282,7 @@ class SmoothedAggregation:
         self.B = Rf.reshape(naggr * k, k).cpu().numpy()
         self.prm["block_size"] = k if k > 1 else 1
 
+        prof.toc("ns_qr")
         # P_tent rows in row order: row i gets Q[id[i], slot_of(i), :]
         slot_of = torch.zeros(n, dtype=t64, device=dev)
         slot_of[members] = slot
@@ -308,7 +317,8 @@ class SmoothedAggregation:
         S_F = DeviceCSR.from_tensors(
             n, n, sptr.to(torch.int32).contiguous(),
             A.col[keep].contiguous(), kv.contiguous())
-        return hip_setup.spgemm(S_F, P_tent, sort=True)
+        with prof.scope("ns_spgemm"):
+            return hip_setup.spgemm(S_F, P_tent, sort=True)
 
     def coarse_operator(self, A, P, R):
         return galerkin(R, A, P)
