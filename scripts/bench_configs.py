@@ -67,7 +67,6 @@ def main():
     Ad = hip.matrix(Ah)  # device input -> device block/nullspace setup
     run(f"elasticity {n}^3 nodes CG+SA(esr)+RBM+Chebyshev+BSR(3)", Ad, bd,
         {"precond": {"class": "amg", "block_value": 3,
-                     "coarse_enough": 1000,  # reference: 3000/block_rows
                      "relax": {"type": "chebyshev"},
                      "coarsening": {"type": "smoothed_aggregation",
                                     "nullspace_raw": B, "block_size": 3,

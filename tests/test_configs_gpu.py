@@ -37,7 +37,6 @@ def test_config3_elasticity_bsr_nullspace(hip):
     s = am.make_solver(
         Ad,
         {"precond": {"class": "amg", "block_value": 3,
-                     "coarse_enough": 1000,  # reference: 3000/block_rows
                      "relax": {"type": "chebyshev"},
                      "coarsening": {"type": "smoothed_aggregation",
                                     "nullspace_raw": B, "block_size": 3,
