@@ -58,6 +58,19 @@ class LevelDescC(ctypes.Structure):
         ("bptr", ctypes.c_void_p),
         ("bcol", ctypes.c_void_p),
         ("bval", ctypes.c_void_p),
+        ("ilu_iters", ctypes.c_int),
+        ("ilu_damping", ctypes.c_double),
+        ("ilu_jdamping", ctypes.c_double),
+        ("lptr", ctypes.c_void_p),
+        ("lcol", ctypes.c_void_p),
+        ("lval", ctypes.c_void_p),
+        ("uptr", ctypes.c_void_p),
+        ("ucol", ctypes.c_void_p),
+        ("uval", ctypes.c_void_p),
+        ("ilu_dinv", ctypes.c_void_p),
+        ("ilu_y", ctypes.c_void_p),
+        ("ilu_s", ctypes.c_void_p),
+        ("ilu_b", ctypes.c_void_p),
     ]
 
 
@@ -107,6 +120,7 @@ class NativeDriver:
         import torch
 
         from ..relaxation.chebyshev import Chebyshev
+        from ..relaxation.ilu0 import ILU0
         from ..relaxation.spai0 import DiagonalSmootherBase
         from .hip import DeviceBSR, DeviceCSR, DeviceDenseSolver
 
@@ -121,18 +135,24 @@ class NativeDriver:
         if not all(isinstance(l.A, (DeviceCSR, DeviceBSR)) for l in levels):
             raise TypeError("native driver needs device-resident levels")
         self._mixed = bool(getattr(amg, "_mixed", False))
+        def relax_ok(r):
+            if isinstance(r, (DiagonalSmootherBase, Chebyshev)):
+                return True
+            # ILU0 with the iterated-Jacobi solve (graph-capturable); the
+            # exact cooperative sptrsv stays on the generic path
+            return (isinstance(r, ILU0) and not getattr(r, "_serial", True)
+                    and not getattr(r, "_exact", False))
+
         for l in levels[:-1]:
-            if not isinstance(l.relax, (DiagonalSmootherBase, Chebyshev)):
+            if not relax_ok(l.relax):
                 raise TypeError(
-                    "native driver supports diagonal/Chebyshev smoothers")
+                    "native driver supports diagonal/Chebyshev/ILU0(jacobi)")
         if amg.coarse_solve is not None and not isinstance(
             amg.coarse_solve, DeviceDenseSolver
         ):
             raise TypeError("native driver needs the dense coarse solver")
-        if amg.coarse_solve is None and not isinstance(
-            levels[-1].relax, (DiagonalSmootherBase, Chebyshev)
-        ):
-            raise TypeError("native coarsest smoother must be diagonal/Chebyshev")
+        if amg.coarse_solve is None and not relax_ok(levels[-1].relax):
+            raise TypeError("native coarsest smoother unsupported")
 
         self._keep = []  # tensor refs
         descs = (LevelDescC * len(levels))()
@@ -174,7 +194,23 @@ class NativeDriver:
                 d.rsrows = _ptr(l.R.srows)
                 self._keep.extend([l.R.soff, l.R.scol, l.R.sval, l.R.srows])
             relax = l.relax
-            if isinstance(relax, Chebyshev):
+            if isinstance(relax, ILU0):
+                d.ilu_iters = relax.solve_iters
+                d.ilu_damping = relax.damping
+                d.ilu_jdamping = relax.solve_damping
+                d.lptr, d.lcol, d.lval = (_ptr(relax.L.ptr), _ptr(relax.L.col),
+                                          _ptr(relax.L.val))
+                d.uptr, d.ucol, d.uval = (_ptr(relax.U.ptr), _ptr(relax.U.col),
+                                          _ptr(relax.U.val))
+                d.ilu_dinv = _ptr(relax.Dinv)
+                work = [torch.empty(A.nrows, dtype=torch.float64,
+                                    device=backend.device) for _ in range(3)]
+                d.ilu_y, d.ilu_s, d.ilu_b = (_ptr(work[0]), _ptr(work[1]),
+                                             _ptr(work[2]))
+                self._keep.extend([relax.L.ptr, relax.L.col, relax.L.val,
+                                   relax.U.ptr, relax.U.col, relax.U.val,
+                                   relax.Dinv] + work)
+            elif isinstance(relax, Chebyshev):
                 d.cheb_degree = relax.degree
                 d.cheb_theta = relax.theta
                 d.cheb_delta = relax.delta
@@ -194,7 +230,7 @@ class NativeDriver:
             self._keep.extend([A.ptr, A.col, A.val, l.f, l.u, l.t])
             if l.P is not None:
                 self._keep.extend([l.P.ptr, l.P.col, l.P.val, l.R.ptr, l.R.col, l.R.val])
-            if relax is not None and not isinstance(relax, Chebyshev):
+            if relax is not None and not isinstance(relax, (Chebyshev, ILU0)):
                 self._keep.append(relax.M)
 
         inv = amg.coarse_solve.inv if amg.coarse_solve is not None else None

@@ -101,4 +101,21 @@ struct LevelDesc {
     const int *bptr;
     const int *bcol;
     const double *bval;
+    // ILU(0) smoothing via damped-Jacobi iterated triangular solves
+    // (ilu_iters > 0; the reference's GPU-native ilu_solve.hpp route).
+    // L strictly lower (unit diag implied), U strictly upper; ilu_dinv is
+    // the inverted diagonal of U.
+    int ilu_iters;
+    double ilu_damping;       // outer relaxation damping
+    double ilu_jdamping;      // inner Jacobi damping (0.72)
+    const int *lptr;
+    const int *lcol;
+    const double *lval;
+    const int *uptr;
+    const int *ucol;
+    const double *uval;
+    const double *ilu_dinv;
+    double *ilu_y;            // work vectors (level-sized)
+    double *ilu_s;
+    double *ilu_b;
 };

@@ -311,8 +311,49 @@ static int cheb_apply(Driver *D, const LevelDesc &L, const T *rhs, T *x, T *r) {
     return 0;
 }
 
+// ILU(0) smoothing step, in place on x (driver twin of relaxation/ilu0.py
+// _solve_jacobi; parity: amgcl/relaxation/detail/ilu_solve.hpp:44-124):
+//   t = rhs - A x;  t <- (LU)^{-1} t by damped-Jacobi iterated triangular
+//   solves;  x += damping * t.   fp64 only (ILU + mixed is not offered).
+template <typename T>
+static int ilu_apply(Driver *D, const LevelDesc &L, const T *rhs, T *x, T *t) {
+    return (int)hipErrorInvalidValue;  // specialized for double below
+}
+
+template <>
+int ilu_apply<double>(Driver *D, const LevelDesc &L, const double *rhs, double *x,
+                      double *t) {
+    const int64_t n = L.nrows;
+    hipStream_t st = D->stream;
+    double *y = L.ilu_y, *s = L.ilu_s, *b = L.ilu_b;
+    const double om = L.ilu_jdamping;
+    CHK(level_residual<double>(D, L, rhs, x, t));
+    // lower: (I + L) y = t
+    CHK(hipMemcpyAsync(y, t, n * sizeof(double), hipMemcpyDeviceToDevice, st));
+    for (int it = 0; it < L.ilu_iters; ++it) {
+        CHK(amg_spmv_f64(n, 0, L.lptr, L.lcol, L.lval, y, -1.0, 0.0, s, 2, st));
+        CHK(amg_axpby_f64(n, 1.0, t, 1.0, s, st));
+        CHK(amg_axpby_f64(n, om, s, 1.0 - om, y, st));
+    }
+    // upper: (D + U) z = y  ->  z = Dinv (y - U z), z kept in y
+    CHK(hipMemcpyAsync(b, y, n * sizeof(double), hipMemcpyDeviceToDevice, st));
+    CHK(amg_vmul_f64(n, 1.0, L.ilu_dinv, b, 0.0, y, st));
+    for (int it = 0; it < L.ilu_iters; ++it) {
+        CHK(amg_spmv_f64(n, 0, L.uptr, L.ucol, L.uval, y, -1.0, 0.0, s, 2, st));
+        CHK(amg_axpby_f64(n, 1.0, b, 1.0, s, st));
+        CHK(amg_vmul_f64(n, 1.0, L.ilu_dinv, s, 0.0, s, st));
+        CHK(amg_axpby_f64(n, om, s, 1.0 - om, y, st));
+    }
+    CHK(amg_axpby_f64(n, L.ilu_damping, y, 1.0, x, st));
+    return 0;
+}
+
 template <typename T>
 static int relax_swap(Driver *D, const LevelDesc &L, const T *rhs, T **x, T **xn) {
+    if (L.ilu_iters) {  // in place, no pointer swap
+        CHK(ilu_apply<T>(D, L, rhs, *x, *xn));
+        return 0;
+    }
     if (L.cheb_degree) {  // in place, no pointer swap
         CHK(cheb_apply<T>(D, L, rhs, *x, *xn));
         return 0;
@@ -365,7 +406,7 @@ static int cycle(Driver *D, int li, const T *f, T **u_io, T **scratch, bool u_is
     // relax_zero writes into the swap buffer and swaps, so zero-guess and
     // general smooths have identical pointer parity (no copy-back needed)
     auto relax_zero = [&](LevelDesc &LL, const T *ff, T **u2, T **sc) -> int {
-        if (LL.cheb_degree || LL.bsize) {
+        if (LL.cheb_degree || LL.bsize || LL.ilu_iters) {
             // no fused zero-guess form for these smoothers: clear + general
             CHK(ops<T>::fill(LL.nrows, *u2, D->stream));
             return relax_swap<T>(D, LL, ff, u2, sc);

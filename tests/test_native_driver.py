@@ -52,12 +52,12 @@ def test_native_w_cycle(hip):
 
 
 def test_native_rejects_unsupported(hip):
-    # ILU smoothing is outside the driver's smoother set -> generic path
+    # multicolor Gauss-Seidel is outside the driver's smoother set
     A, b = am.poisson3d(16)
     s = am.make_solver(
         A,
         {
-            "precond": {"class": "amg", "relax": {"type": "ilu0"}},
+            "precond": {"class": "amg", "relax": {"type": "gauss_seidel"}},
             "solver": {"type": "cg", "tol": 1e-8},
         },
         backend=hip,
@@ -65,6 +65,38 @@ def test_native_rejects_unsupported(hip):
     assert s._native is None
     x, iters, resid = s(b)
     assert resid < 1e-8
+
+
+def test_native_driver_ilu0_jacobi(hip):
+    """ILU(0) with iterated-Jacobi triangular solves runs inside the native
+    driver (ilu_solve.hpp semantics); the exact-sptrsv variant stays on the
+    generic path (cooperative kernels are not graph-capturable)."""
+    A, b = am.poisson3d(24, rhs="random", anisotropy=50.0)
+    prm = {"precond": {"class": "amg", "coarse_enough": 300,
+                       "relax": {"type": "ilu0"}},
+           "solver": {"type": "cg", "tol": 1e-8, "maxiter": 200}}
+    s = am.make_solver(A, prm, backend=hip)
+    assert s._native is not None
+    x1, it1, r1 = s(b)
+    assert r1 < 1e-8
+
+    import copy
+
+    prm2 = copy.deepcopy(prm)
+    prm2["solver"]["verbose"] = True
+    s2 = am.make_solver(A, prm2, backend=hip)
+    assert s2._native is None
+    x2, it2, r2 = s2(b)
+    assert abs(it1 - it2) <= 1, (it1, it2)
+    xa, xb = hip.to_host(x1), hip.to_host(x2)
+    assert np.linalg.norm(xa - xb) / np.linalg.norm(xa) < 1e-8
+
+    prm3 = copy.deepcopy(prm)
+    prm3["precond"]["relax"]["solve"] = "exact"
+    s3 = am.make_solver(A, prm3, backend=hip)
+    assert s3._native is None  # exact sptrsv -> generic path
+    x3, it3, r3 = s3(b)
+    assert r3 < 1e-8
 
 
 @pytest.mark.parametrize("solver", ["cg", "bicgstab"])
