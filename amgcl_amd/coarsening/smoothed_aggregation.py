@@ -203,10 +203,19 @@ class SmoothedAggregation:
             np_ = n // bsize
             rpt = row_of // bsize
             cpt = col64 // bsize
-            pt = torch.sparse_coo_tensor(torch.stack([rpt, cpt]), A.val * A.val,
-                                         (np_, np_)).coalesce()
-        prow, pcol = pt.indices()
-        pval = pt.values().sqrt()
+            # one sort + segment-sum (sparse coalesce measured ~4x slower),
+            # and the inverse map doubles as the scalar->pointwise entry
+            # lookup for the strong-mask expansion below
+            key = rpt * np_ + cpt
+            sk, perm = torch.sort(key)
+            uk, inv = torch.unique_consecutive(sk, return_inverse=True)
+            pv = torch.zeros(uk.numel(), dtype=torch.float64, device=dev)
+            pv.scatter_add_(0, inv, (A.val * A.val)[perm])
+            prow = uk // np_
+            pcol = uk % np_
+            pval = pv.sqrt()
+            pos = torch.empty(key.numel(), dtype=t64, device=dev)
+            pos[perm] = inv
         pptr = torch.zeros(np_ + 1, dtype=t64, device=dev)
         torch.cumsum(torch.bincount(prow, minlength=np_), 0, out=pptr[1:])
         Apt = DeviceCSR.from_tensors(
@@ -217,10 +226,7 @@ class SmoothedAggregation:
         with prof.scope("pt_expand"):
             ids = ids_pt.repeat_interleave(bsize).contiguous()
             # strong mask per scalar entry = the pointwise entry's flag
-            # (diagonal-point entries are never strong)
-            key = prow * np_ + pcol  # ascending (coalesce ordering)
-            ekey = rpt * np_ + cpt
-            pos = torch.searchsorted(key, ekey)
+            # (diagonal-point entries are never strong); pos computed above
             S = S_pt[pos]
             S = torch.where(rpt == cpt, torch.zeros_like(S), S).contiguous()
         self._keep_pt = (Apt,)  # keep alive until setup finishes
