@@ -242,3 +242,21 @@ print("TORCHFREE_OK", it.value, res.value)
     out = subprocess.check_output([sys.executable, "-c", prog], text=True,
                                   stderr=subprocess.STDOUT, timeout=300)
     assert "TORCHFREE_OK" in out
+
+
+@pytest.mark.gpu
+def test_gpu_capi_from_c(tmp_path):
+    """Compile examples/gpu_capi.c with hipcc and run it — a complete GPU
+    solve from plain C with no Python/torch in the process."""
+    import shutil
+    import subprocess
+
+    hipcc = shutil.which("hipcc") or "/opt/rocm/bin/hipcc"
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    libdir = os.path.join(root, "amgcl_amd", "_hip")
+    exe = str(tmp_path / "gpu_capi")
+    subprocess.check_call(
+        [hipcc, os.path.join(root, "examples", "gpu_capi.c"), "-o", exe,
+         f"-L{libdir}", "-lamghip", f"-Wl,-rpath,{libdir}"])
+    out = subprocess.check_output([exe], text=True, timeout=240)
+    assert "rc=0" in out
