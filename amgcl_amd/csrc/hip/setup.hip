@@ -808,20 +808,28 @@ __global__ void spgemm_ub_k(int64_t an, const int *__restrict__ aptr,
     }
 }
 
+// templated on the LDS table size, tiered by ub: the dominant fine-level
+// A*P rows have <= 24 candidates, where a 32-slot table quarters the
+// init/extract traffic of the 128-slot one (measured: the init dominates
+// 5:1 for 24-candidate rows)
+template <int SLOTS>
 __global__ void spgemm_count_small_k(int64_t an, const int *__restrict__ aptr,
                                      const int *__restrict__ acol,
                                      const int *__restrict__ bptr,
                                      const int *__restrict__ bcol,
-                                     const int *__restrict__ ub, int *__restrict__ cnt) {
-    __shared__ int keys[32][SSLOTS];  // 32 groups of 8 lanes (256 threads)
+                                     const int *__restrict__ ub, int *__restrict__ cnt,
+                                     int ub_lo, int ub_hi) {
+    __shared__ int keys[32][SLOTS];  // 32 groups of 8 lanes (256 threads)
+    constexpr int MASK = SLOTS - 1;
     int gid = threadIdx.x / SGRP;
     int lane = threadIdx.x & (SGRP - 1);
     int64_t row = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / SGRP;
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) / SGRP;
     int *tk = keys[gid];
     for (; row < an; row += stride) {
-        if (ub[row] > SGSMALL) continue;
-        for (int t = lane; t < SSLOTS; t += SGRP) tk[t] = -1;
+        int u = ub[row];
+        if (u <= ub_lo || u > ub_hi) continue;
+        for (int t = lane; t < SLOTS; t += SGRP) tk[t] = -1;
         __builtin_amdgcn_s_waitcnt(0);
         __builtin_amdgcn_wave_barrier();
         int inserted = 0;
@@ -829,12 +837,12 @@ __global__ void spgemm_count_small_k(int64_t an, const int *__restrict__ aptr,
             int ca = acol[ja];
             for (int jb = bptr[ca] + lane; jb < bptr[ca + 1]; jb += SGRP) {
                 int key = bcol[jb];
-                uint32_t h = ((uint32_t)key * 2654435761u) & SSMASK;
+                uint32_t h = ((uint32_t)key * 2654435761u) & MASK;
                 while (true) {
                     int old = atomicCAS(&tk[h], -1, key);
                     if (old == -1) { ++inserted; break; }
                     if (old == key) break;
-                    h = (h + 1) & SSMASK;
+                    h = (h + 1) & MASK;
                 }
             }
         }
@@ -1244,7 +1252,10 @@ extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol, co
     int rc = scan_i32_device(flags, an, s);
     if (rc) return rc;
     spg_bigscatter_k<<<nblk(an), 256, 0, s>>>(an, ub, flags, biglist, nbig);
-    spgemm_count_small_k<<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, bptr, bcol, ub, cnt);
+    spgemm_count_small_k<32><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, bptr,
+                                                             bcol, ub, cnt, 0, 24);
+    spgemm_count_small_k<SSLOTS><<<nblk(an * SGRP), 256, 0, s>>>(
+        an, aptr, acol, bptr, bcol, ub, cnt, 24, SGSMALL);
     spgemm_count_k<HSLOTS><<<nblk(an * WAVE), 256, 0, s>>>(
         an, biglist, nbig, aptr, acol, bptr, bcol, ub, cnt, overflow, SGSMALL, 1 << 30);
     return (int)hipGetLastError();
@@ -1256,9 +1267,12 @@ extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol, con
                                hipStream_t s) {
     const int *biglist = bigscratch + an;
     const int *nbig = bigscratch + 2 * an;
+    spgemm_fill_small_k<32><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr,
+                                                            bcol, bval, ub, cptr_scanned,
+                                                            ccol, cval, do_sort, 0, 20);
     spgemm_fill_small_k<64><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr,
                                                             bcol, bval, ub, cptr_scanned,
-                                                            ccol, cval, do_sort, 0, 48);
+                                                            ccol, cval, do_sort, 20, 48);
     spgemm_fill_small_k<SSLOTS><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, aval, bptr,
                                                                 bcol, bval, ub, cptr_scanned,
                                                                 ccol, cval, do_sort, 48,
