@@ -1252,10 +1252,11 @@ extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol, co
     int rc = scan_i32_device(flags, an, s);
     if (rc) return rc;
     spg_bigscatter_k<<<nblk(an), 256, 0, s>>>(an, ub, flags, biglist, nbig);
-    spgemm_count_small_k<32><<<nblk(an * SGRP), 256, 0, s>>>(an, aptr, acol, bptr,
-                                                             bcol, ub, cnt, 0, 24);
+    // single count pass: a 32-slot count tier measured WORSE (the count is
+    // candidate-probing-bound, not table-init-bound, and the second sweep
+    // costs a full pass) — the small table only pays in the FILL
     spgemm_count_small_k<SSLOTS><<<nblk(an * SGRP), 256, 0, s>>>(
-        an, aptr, acol, bptr, bcol, ub, cnt, 24, SGSMALL);
+        an, aptr, acol, bptr, bcol, ub, cnt, 0, SGSMALL);
     spgemm_count_k<HSLOTS><<<nblk(an * WAVE), 256, 0, s>>>(
         an, biglist, nbig, aptr, acol, bptr, bcol, ub, cnt, overflow, SGSMALL, 1 << 30);
     return (int)hipGetLastError();
