@@ -23,10 +23,13 @@ from amgcl_amd.backend import make_backend
 
 
 def run(name, A, b, prm, hip):
+    import copy
     import math
 
+    am.make_solver(A, copy.deepcopy(prm), backend=hip)  # warmup (cold allocs)
+    hip.synchronize()
     t0 = time.perf_counter()
-    solve = am.make_solver(A, prm, backend=hip)
+    solve = am.make_solver(A, copy.deepcopy(prm), backend=hip)
     hip.synchronize()
     t1 = time.perf_counter()
     solve(b)  # warm
