@@ -65,6 +65,27 @@ _SIGS = {
     "amg_sptrsv_f64": [ctypes.c_int64] + [ctypes.c_void_p] * 7
                       + [ctypes.c_int, ctypes.c_void_p],
     "amg_coop_supported": [],
+    # --- complex128 solve kernels (host complex setup + device solve) ---
+    "amg_spmv_c128": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+                     + [ctypes.c_double] * 4 + [ctypes.c_void_p, ctypes.c_int,
+                                                ctypes.c_void_p],
+    "amg_residual_c128": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 6
+                         + [ctypes.c_int, ctypes.c_void_p],
+    "amg_relax_diag_c128": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 7
+                           + [ctypes.c_int, ctypes.c_void_p],
+    "amg_axpby_c128": [ctypes.c_int64] + [ctypes.c_double] * 2 + [ctypes.c_void_p]
+                      + [ctypes.c_double] * 2 + [ctypes.c_void_p] * 2,
+    "amg_axpbypcz_c128": [ctypes.c_int64] + [ctypes.c_double] * 2
+                         + [ctypes.c_void_p] + [ctypes.c_double] * 2
+                         + [ctypes.c_void_p] + [ctypes.c_double] * 2
+                         + [ctypes.c_void_p] * 2,
+    "amg_vmul_c128": [ctypes.c_int64] + [ctypes.c_double] * 2
+                     + [ctypes.c_void_p] * 2 + [ctypes.c_double] * 2
+                     + [ctypes.c_void_p] * 2,
+    "amg_dot_c128": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
+    "amg_gemv_c128": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
+    "amg_fill_c128": [ctypes.c_int64] + [ctypes.c_double] * 2
+                     + [ctypes.c_void_p] * 2,
     # --- SELL-64 (wave-native sliced-ELL) solve kernels ---
     "amg_sell_spmv_f64": [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 5
                          + [ctypes.c_double, ctypes.c_double]

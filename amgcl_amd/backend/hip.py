@@ -221,13 +221,19 @@ class DeviceBSR:
 class HipBackend:
     name = "hip"
 
-    def __init__(self, device=None):
+    def __init__(self, device=None, dtype=None):
         import torch
 
         if not torch.cuda.is_available():
             raise RuntimeError("hip backend requires a GPU (torch.cuda unavailable)")
         self.torch = torch
         self.device = torch.device(device or "cuda")
+        # default value type for vectors; complex128 selects the native
+        # complex kernels (parity: amgcl/value_type/complex.hpp on the
+        # device backends)
+        self.dtype = dtype or torch.float64
+        if self.dtype not in (torch.float64, torch.complex128):
+            raise ValueError("hip backend supports float64 and complex128")
         lib()  # fail loudly now if the kernel library is missing
         self._dotbuf = torch.zeros(2, dtype=torch.float64, device=self.device)
         self._dothost = torch.zeros(2, dtype=torch.float64, pin_memory=True)
@@ -239,10 +245,12 @@ class HipBackend:
         return DeviceCSR(csr, self.device)
 
     def vector(self, n, dtype=None):
-        return self.torch.zeros(n, dtype=dtype or self.torch.float64, device=self.device)
+        return self.torch.zeros(n, dtype=dtype or self.dtype, device=self.device)
 
     def from_host(self, a):
-        t = self.torch.from_numpy(np.ascontiguousarray(a, dtype=np.float64))
+        dt = np.complex128 if (np.iscomplexobj(a)
+                               or self.dtype == self.torch.complex128) else np.float64
+        t = self.torch.from_numpy(np.ascontiguousarray(a, dtype=dt))
         return t.to(self.device)
 
     def to_host(self, v):
@@ -253,7 +261,8 @@ class HipBackend:
     def _fn(name, t):
         import torch
 
-        sfx = "_f32" if t.dtype == torch.float32 else "_f64"
+        sfx = {torch.float32: "_f32", torch.float64: "_f64",
+               torch.complex128: "_c128"}[t.dtype]
         return getattr(lib(), "amg_" + name + sfx)
 
     def spmv(self, alpha, A, x, beta, y):
@@ -262,6 +271,13 @@ class HipBackend:
                                          A.col.data_ptr(), A.val.data_ptr(),
                                          x.data_ptr(), alpha, beta, y.data_ptr(),
                                          _stream()), "bsr_spmv")
+            return
+        if A.val.is_complex():
+            a, b = complex(alpha), complex(beta)
+            check(lib().amg_spmv_c128(A.nrows, A.nnz, A.ptr.data_ptr(),
+                                      A.col.data_ptr(), A.val.data_ptr(),
+                                      x.data_ptr(), a.real, a.imag, b.real, b.imag,
+                                      y.data_ptr(), A.subw, _stream()), "cspmv")
             return
         if getattr(A, "nslice", 0):
             check(self._fn("sell_spmv", A.sval)(A.nrows, A.nslice, A.soff.data_ptr(),
@@ -280,6 +296,12 @@ class HipBackend:
                                              A.col.data_ptr(), A.val.data_ptr(),
                                              b.data_ptr(), x.data_ptr(), r.data_ptr(),
                                              _stream()), "bsr_residual")
+            return
+        if A.val.is_complex():
+            check(lib().amg_residual_c128(A.nrows, A.nnz, A.ptr.data_ptr(),
+                                          A.col.data_ptr(), A.val.data_ptr(),
+                                          b.data_ptr(), x.data_ptr(), r.data_ptr(),
+                                          A.subw, _stream()), "cresidual")
             return
         if getattr(A, "nslice", 0):
             check(self._fn("sell_residual", A.sval)(A.nrows, A.nslice,
@@ -304,6 +326,14 @@ class HipBackend:
                                           t.data_ptr(), _stream()), "bsr_relax")
             self.axpby(1.0, t, 1.0, x)
             return
+        if A.val.is_complex():
+            check(lib().amg_relax_diag_c128(A.nrows, A.nnz, A.ptr.data_ptr(),
+                                            A.col.data_ptr(), A.val.data_ptr(),
+                                            M.data_ptr(), rhs.data_ptr(),
+                                            x.data_ptr(), t.data_ptr(), A.subw,
+                                            _stream()), "crelax")
+            self.axpby(1.0, t, 1.0, x)
+            return
         if getattr(A, "nslice", 0):
             # SELL relax writes x_new = x + M(rhs - Ax) into t, then copy back
             check(self._fn("sell_relax", A.sval)(A.nrows, A.nslice, A.soff.data_ptr(),
@@ -322,20 +352,43 @@ class HipBackend:
         self.axpby(1.0, t, 1.0, x)
 
     def clear(self, x):
+        if x.is_complex():
+            check(lib().amg_fill_c128(x.numel(), 0.0, 0.0, x.data_ptr(),
+                                      _stream()), "cfill")
+            return
         check(self._fn("fill", x)(x.numel(), 0.0, x.data_ptr(), _stream()), "fill")
 
     def copy(self, x, y):
         y.copy_(x)
 
     def axpby(self, a, x, b, y):
+        if x.is_complex():
+            a, b = complex(a), complex(b)
+            check(lib().amg_axpby_c128(x.numel(), a.real, a.imag, x.data_ptr(),
+                                       b.real, b.imag, y.data_ptr(), _stream()),
+                  "caxpby")
+            return
         check(self._fn("axpby", x)(x.numel(), a, x.data_ptr(), b, y.data_ptr(),
                                    _stream()), "axpby")
 
     def axpbypcz(self, a, x, b, y, c, z):
+        if x.is_complex():
+            a, b, c = complex(a), complex(b), complex(c)
+            check(lib().amg_axpbypcz_c128(x.numel(), a.real, a.imag, x.data_ptr(),
+                                          b.real, b.imag, y.data_ptr(), c.real,
+                                          c.imag, z.data_ptr(), _stream()),
+                  "caxpbypcz")
+            return
         check(self._fn("axpbypcz", x)(x.numel(), a, x.data_ptr(), b, y.data_ptr(), c,
                                       z.data_ptr(), _stream()), "axpbypcz")
 
     def vmul(self, a, m, x, b, z):
+        if x.is_complex():
+            a, b = complex(a), complex(b)
+            check(lib().amg_vmul_c128(x.numel(), a.real, a.imag, m.data_ptr(),
+                                      x.data_ptr(), b.real, b.imag, z.data_ptr(),
+                                      _stream()), "cvmul")
+            return
         check(self._fn("vmul", x)(x.numel(), a, m.data_ptr(), x.data_ptr(), b,
                                   z.data_ptr(), _stream()), "vmul")
 
@@ -352,11 +405,19 @@ class HipBackend:
                                      _stream()), "cast")
 
     def dot(self, x, y):
+        if x.is_complex():
+            self._dotbuf.zero_()
+            check(lib().amg_dot_c128(x.numel(), x.data_ptr(), y.data_ptr(),
+                                     self._dotbuf.data_ptr(), _stream()), "cdot")
+            self._dothost.copy_(self._dotbuf, non_blocking=False)
+            return complex(float(self._dothost[0]), float(self._dothost[1]))
         check(self._fn("dot", x)(x.numel(), x.data_ptr(), y.data_ptr(),
                                  self._dotbuf.data_ptr(), _stream()), "dot")
         return float(self._dotbuf[0].item())
 
     def dot2(self, x1, y1, x2, y2):
+        if x1.is_complex():
+            return self.dot(x1, y1), self.dot(x2, y2)
         check(lib().amg_dot2_f64(x1.numel(), x1.data_ptr(), y1.data_ptr(),
                                  x2.data_ptr(), y2.data_ptr(),
                                  self._dotbuf.data_ptr(), _stream()), "dot2")
@@ -370,10 +431,16 @@ class HipBackend:
                                          y.data_ptr(), _stream()), "blkdiag_vmul")
 
     def gather(self, x, idx, buf):
+        if x.is_complex():
+            self.torch.index_select(x, 0, idx.to(self.torch.int64), out=buf)
+            return
         check(self._fn("gather", x)(idx.numel(), x.data_ptr(), idx.data_ptr(),
                                     buf.data_ptr(), _stream()), "gather")
 
     def scatter(self, buf, idx, x):
+        if x.is_complex():
+            x[idx.to(self.torch.int64)] = buf
+            return
         check(self._fn("scatter", x)(idx.numel(), buf.data_ptr(), idx.data_ptr(),
                                      x.data_ptr(), _stream()), "scatter")
 
@@ -479,6 +546,10 @@ class DeviceDenseSolver:
     def __call__(self, f, u):
         import torch
 
+        if self.inv.is_complex():
+            check(lib().amg_gemv_c128(self.n, self.inv.data_ptr(), f.data_ptr(),
+                                      u.data_ptr(), _stream()), "cgemv")
+            return
         fn = lib().amg_gemv_f32 if self.inv.dtype == torch.float32 else lib().amg_gemv_f64
         check(fn(self.n, self.inv.data_ptr(), f.data_ptr(), u.data_ptr(), _stream()),
               "gemv")

@@ -311,6 +311,8 @@ def try_native(make_solver_obj):
     P, S, backend = make_solver_obj.P, make_solver_obj.S, make_solver_obj.backend
     if getattr(backend, "name", "") != "hip":
         return None
+    if "complex" in str(getattr(backend, "dtype", "")):
+        return None  # complex solves use the generic per-kernel path
     if not isinstance(P, AMG):
         return None
     if isinstance(S, CG):

@@ -104,12 +104,17 @@ def make_solver(A, prm=None, backend="cpu", **backend_kwargs):
     if isinstance(backend, str):
         from .backend import make_backend
 
-        if (backend == "cpu" and isinstance(A, CSR) and A.is_complex
+        if (backend in ("cpu", "hip") and isinstance(A, CSR) and A.is_complex
                 and "dtype" not in backend_kwargs):
             # native complex solve (parity: amgcl/value_type/complex.hpp;
             # the reference instantiates the same templates over
             # std::complex).  The HIP backend keeps the 2x2-real adapter
             # route (adapter.complex_to_real).
-            backend_kwargs["dtype"] = np.complex128
+            if backend == "hip":
+                import torch
+
+                backend_kwargs["dtype"] = torch.complex128
+            else:
+                backend_kwargs["dtype"] = np.complex128
         backend = make_backend(backend, **backend_kwargs)
     return MakeSolver(A, prm, backend)

@@ -107,6 +107,8 @@ class AMG:
         def eligible(M):
             if not isinstance(M, DeviceCSR) or M.nrows < min_rows:
                 return False
+            if M.val.is_complex():
+                return False  # SELL kernels are real-valued (f64/f32)
             return M.nnz / max(M.nrows, 1) >= min_mean
 
         for lvl in self.levels:

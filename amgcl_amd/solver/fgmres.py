@@ -25,6 +25,8 @@ class FGMRES(SolverBase):
         self.v = [b.vector(n) for _ in range(m + 1)]
         self.z = [b.vector(n) for _ in range(m)]
         dt = getattr(backend, "dtype", np.float64)
+        dt = (np.complex128 if "complex" in str(dt)
+              else (np.float32 if "float32" in str(dt) else np.float64))
         self._cplx = np.issubdtype(np.dtype(dt), np.complexfloating)
         self.H = np.zeros((m + 1, m), dtype=dt)
         self.cs = np.zeros(m + 1)  # real also in the complex rotation

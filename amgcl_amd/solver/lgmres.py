@@ -53,6 +53,8 @@ class LGMRES(SolverBase):
             naug = min(len(self.aug), self.k)
             msize = self.m + naug
             dt = getattr(self.backend, "dtype", np.float64)
+            dt = (np.complex128 if "complex" in str(dt)
+                  else (np.float32 if "float32" in str(dt) else np.float64))
             cplx = np.issubdtype(np.dtype(dt), np.complexfloating)
             H = np.zeros((msize + 1, msize), dtype=dt)
             cs = np.zeros(msize + 1)
