@@ -20,6 +20,9 @@ class RugeStuben:
         self.prm = merge_params(self.defaults(), prm)
 
     def transfer_operators(self, A):
+        if getattr(A, "is_complex", False):
+            raise ValueError("ruge_stuben is real-valued; complex systems use "
+                             "smoothed_aggregation (or the 2x2-real adapter)")
         if not isinstance(A, CSR):
             # host-only algorithm: trigger make_solver/AMG's host fallback
             raise OverflowError("ruge_stuben runs on the host")

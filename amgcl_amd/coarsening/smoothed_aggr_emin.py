@@ -43,6 +43,9 @@ class SmoothedAggrEMin:
         return Af, dia
 
     def transfer_operators(self, A):
+        if getattr(A, "is_complex", False):
+            raise ValueError("smoothed_aggr_emin is real-valued; complex "
+                             "systems use smoothed_aggregation")
         if not isinstance(A, CSR):
             raise OverflowError("smoothed_aggr_emin runs on the host")
         import scipy.sparse as sp

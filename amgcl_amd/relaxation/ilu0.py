@@ -40,6 +40,10 @@ class ILU0:
             from ..backend import hip_setup
 
             A = hip_setup.download(A)  # ILU factorizations are host-side
+        if A.is_complex:
+            raise ValueError("ILU smoothers are real-valued; complex systems "
+                             "use spai0/damped_jacobi/chebyshev (or the "
+                             "2x2-real adapter)")
         p = merge_params(self.defaults(), prm)
         self.damping = float(p["damping"])
         self.backend = backend

@@ -139,3 +139,32 @@ def test_native_complex_on_hip(solver):
             "solver": {"type": solver, "tol": 1e-8, "maxiter": 200}})
     x2, it2, r2 = s2(b)
     assert abs(iters - it2) <= 2, (iters, it2)
+
+
+def test_complex_unsupported_components_fail_loudly():
+    """Real-only components reject complex systems with clear errors
+    instead of silently discarding imaginary parts."""
+    A, m = helmholtz(14)
+    b = np.ones(A.nrows, dtype=complex)
+    for prm, err in (
+        ({"coarsening": {"type": "ruge_stuben"}}, "ruge_stuben"),
+        ({"relax": {"type": "ilu0"}}, "ILU"),
+        ({"coarsening": {"type": "smoothed_aggr_emin"}}, "emin"),
+    ):
+        with pytest.raises((ValueError, TypeError)):
+            am.make_solver(A, {"precond": {"class": "amg", "coarse_enough": 300,
+                                           **prm},
+                               "solver": {"type": "bicgstab"}})
+
+
+def test_complex_chebyshev_smoother():
+    A, m = helmholtz(14)
+    rng = np.random.default_rng(5)
+    b = rng.standard_normal(A.nrows) + 1j * rng.standard_normal(A.nrows)
+    s = am.make_solver(A, {"precond": {"class": "amg", "coarse_enough": 300,
+                                       "relax": {"type": "chebyshev"}},
+                           "solver": {"type": "bicgstab", "tol": 1e-8,
+                                      "maxiter": 300}})
+    x, it, r = s(b)
+    assert r < 1e-8
+    assert np.linalg.norm(b - m @ x) / np.linalg.norm(b) < 1e-7
