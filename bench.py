@@ -153,22 +153,24 @@ def run_distributed(args, dist_ctx):
     solve.backend.synchronize()
     dist.barrier()
     t2 = time.perf_counter()
-    # true residual check against the full operator (rank 0)
-    xh = solve.gather_solution(x)
-    if args.weak:
-        bs = [None] * world
-        dist.all_gather_object(bs, np.asarray(b))
-    true_rel = -1.0
-    if xh is not None:
-        if args.weak:
-            from amgcl_amd.generators import poisson3d_box_strip
+    # True residual check, strip-local: every rank computes
+    # ||b_strip - A_strip x_full||^2 against its own strip (global columns)
+    # and the norms are all-reduced.  This is exactly ||b - A x|| without
+    # rebuilding the full operator anywhere (a host 512^3 assembly costs
+    # ~2.5 min of untimed wall on a quota-limited box, per run).
+    xs = [None] * world
+    xh = solve.backend.to_host(x)
+    dist.all_gather_object(xs, np.asarray(xh, dtype=np.float64))
+    x_full = np.concatenate(xs)
+    r = np.asarray(b, dtype=np.float64) - A_strip @ x_full
+    import torch
 
-            A_full, _, _, _ = poisson3d_box_strip(n, n, nz, 0, 1)
-            bg = np.concatenate(bs)
-        else:
-            A_full, _ = am.poisson3d(n)
-            bg = b_global
-        true_rel = float(np.linalg.norm(bg - A_full @ xh) / np.linalg.norm(bg))
+    nums = torch.tensor([float(r @ r), float(np.asarray(b) @ np.asarray(b))],
+                        dtype=torch.float64)
+    if dist.get_backend() == "nccl":
+        nums = nums.cuda()  # NCCL collectives need device tensors
+    dist.all_reduce(nums)
+    true_rel = float((nums[0] / nums[1]).sqrt())
     return t0, t1, t2, iters, resid, true_rel
 
 
