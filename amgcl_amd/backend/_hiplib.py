@@ -134,8 +134,8 @@ _SIGS = {
     "amg_spgemm_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 10 + [ctypes.c_int]
                        + [ctypes.c_void_p] * 2,
     "amg_gershgorin": [ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p],
-    "amg_poisson_cnt": [ctypes.c_int64] + [ctypes.c_void_p] * 2,
-    "amg_poisson_fill": [ctypes.c_int64] + [ctypes.c_void_p] * 4,
+    "amg_poisson_cnt": [ctypes.c_int64] * 4 + [ctypes.c_void_p] * 2,
+    "amg_poisson_fill": [ctypes.c_int64] * 4 + [ctypes.c_void_p] * 4,
 }
 
 

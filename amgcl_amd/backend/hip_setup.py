@@ -238,14 +238,65 @@ def poisson3d_device(n, device="cuda"):
             "unknowns, 2.1G nnz) — shard larger problems across ranks "
             "(bench.py --gpus N)")
     """7-point Poisson fixture generated directly on the GPU."""
+    return poisson3d_device_strip(n, 0, n**3, device)
+
+
+def poisson3d_device_strip(n, row_beg, row_end, device="cuda", nz=None):
+    """Row strip [row_beg, row_end) of the 7-point Poisson operator on an
+    n x n x nz box (nz defaults to n: the cube) with GLOBAL columns,
+    generated directly in device memory — the per-rank distributed fixture
+    (reference: examples/mpi/mpi_solver.cpp assemble_poisson3d).  Global
+    column ids are int32, so the GLOBAL problem may reach 2^31 unknowns
+    (~1290^3) across ranks."""
+    nz = n if nz is None else nz
+    ntot = n * n * nz
+    if ntot > 2**31:
+        raise ValueError(f"global column ids exceed int32 ({ntot/1e9:.2f}G unknowns)")
+    nloc = row_end - row_beg
+    if 7 * nloc >= 2**31:
+        raise ValueError("strip nonzeros exceed int32; use more ranks")
     t = _torch()
-    n3 = n**3
-    ptr = _new_ptr(n3, device)
-    check(lib().amg_poisson_cnt(n, ptr[1:].data_ptr(), _stream()), "poisson_cnt")
+    ptr = _new_ptr(nloc, device)
+    check(lib().amg_poisson_cnt(n, nz, row_beg, row_end, ptr[1:].data_ptr(),
+                                _stream()), "poisson_cnt")
     _scan_ptr(ptr)
     nnz = int(ptr[-1].item())
     col = t.empty(nnz, dtype=t.int32, device=device)
     val = t.empty(nnz, dtype=t.float64, device=device)
-    check(lib().amg_poisson_fill(n, ptr[1:].data_ptr(), col.data_ptr(), val.data_ptr(),
-                                 _stream()), "poisson_fill")
-    return device_csr(n3, n3, ptr, col, val)
+    check(lib().amg_poisson_fill(n, nz, row_beg, row_end, ptr[1:].data_ptr(),
+                                 col.data_ptr(), val.data_ptr(), _stream()),
+          "poisson_fill")
+    return device_csr(nloc, ntot, ptr, col, val)
+
+
+def split_strip_torch(ptr, col, val, col_beg, col_end):
+    """Local/remote split of a row strip with global columns, in torch ops
+    (device-agnostic: runs on the GPU for device strips; tested on CPU
+    tensors against the C++ _core.split_strip).  Returns
+    (lp, lc, lv, rp, rc, rv, ghost_global): local part with columns
+    renumbered to [0, col_end-col_beg), remote part with columns renumbered
+    into the sorted ghost_global id list (parity:
+    amgcl/mpi/distributed_matrix.hpp:370-430)."""
+    t = _torch()
+    nloc = ptr.numel() - 1
+    dev = col.device
+    lens = (ptr[1:] - ptr[:-1]).to(t.int64)
+    row_ids = t.repeat_interleave(t.arange(nloc, device=dev, dtype=t.int64), lens)
+    mask = (col >= col_beg) & (col < col_end)
+    lloc = t.bincount(row_ids[mask], minlength=nloc)
+    lp = t.zeros(nloc + 1, dtype=t.int64, device=dev)
+    t.cumsum(lloc, 0, out=lp[1:])
+    lc = (col[mask] - col_beg).contiguous()
+    lv = val[mask].contiguous()
+    maskr = ~mask
+    rcg = col[maskr]
+    rows_r = row_ids[maskr]
+    lrem = t.bincount(rows_r, minlength=nloc) if rcg.numel() else t.zeros(
+        nloc, dtype=t.int64, device=dev)
+    rp = t.zeros(nloc + 1, dtype=t.int64, device=dev)
+    t.cumsum(lrem, 0, out=rp[1:])
+    ghost_global = t.unique(rcg.to(t.int64))
+    rc = t.searchsorted(ghost_global, rcg.to(t.int64)).to(t.int32).contiguous()
+    rv = val[maskr].contiguous()
+    return (lp.to(t.int32).contiguous(), lc, lv,
+            rp.to(t.int32).contiguous(), rc, rv, ghost_global)

@@ -1200,41 +1200,51 @@ __global__ void spgemm_fill_k(int64_t an, const int *__restrict__ biglist,
 // 7-point Poisson fixture generated directly in device memory (same
 // semantics as the host generator / reference tests/sample_problem.hpp:11).
 // ---------------------------------------------------------------------------
-__global__ void poisson_cnt_k(int64_t n, int *__restrict__ cnt) {
-    int64_t n3 = n * n * n;
-    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+// Row-strip form: rows [row_beg, row_end) with GLOBAL columns — the
+// distributed fixture (one strip per rank, reference
+// examples/mpi/mpi_solver.cpp assemble_poisson3d shape).  Global column ids
+// stay int32, so the GLOBAL problem may reach 2^31 unknowns (~1290^3) even
+// though one GPU holds only its strip.
+__global__ void poisson_cnt_k(int64_t n, int64_t nz, int64_t row_beg, int64_t row_end,
+                              int *__restrict__ cnt) {
+    int64_t idx = row_beg + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; idx < n3; idx += stride) {
+    for (; idx < row_end; idx += stride) {
         int64_t i = idx % n, j = (idx / n) % n, k = idx / (n * n);
-        cnt[idx] = 1 + (k > 0) + (j > 0) + (i > 0) + (i + 1 < n) + (j + 1 < n) + (k + 1 < n);
+        cnt[idx - row_beg] =
+            1 + (k > 0) + (j > 0) + (i > 0) + (i + 1 < n) + (j + 1 < n) + (k + 1 < nz);
     }
 }
 
-__global__ void poisson_fill_k(int64_t n, const int *__restrict__ ptr_scanned,
+__global__ void poisson_fill_k(int64_t n, int64_t nz, int64_t row_beg, int64_t row_end,
+                               const int *__restrict__ ptr_scanned,
                                int *__restrict__ col, double *__restrict__ val) {
-    int64_t n3 = n * n * n;
-    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t idx = row_beg + (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; idx < n3; idx += stride) {
+    for (; idx < row_end; idx += stride) {
         int64_t i = idx % n, j = (idx / n) % n, k = idx / (n * n);
-        int h = (idx == 0) ? 0 : ptr_scanned[idx - 1];
-        if (k > 0)     { col[h] = (int)(idx - n * n); val[h] = -1.0; ++h; }
-        if (j > 0)     { col[h] = (int)(idx - n);     val[h] = -1.0; ++h; }
-        if (i > 0)     { col[h] = (int)(idx - 1);     val[h] = -1.0; ++h; }
+        int64_t loc = idx - row_beg;
+        int h = (loc == 0) ? 0 : ptr_scanned[loc - 1];
+        if (k > 0)      { col[h] = (int)(idx - n * n); val[h] = -1.0; ++h; }
+        if (j > 0)      { col[h] = (int)(idx - n);     val[h] = -1.0; ++h; }
+        if (i > 0)      { col[h] = (int)(idx - 1);     val[h] = -1.0; ++h; }
         col[h] = (int)idx; val[h] = 6.0; ++h;
-        if (i + 1 < n) { col[h] = (int)(idx + 1);     val[h] = -1.0; ++h; }
-        if (j + 1 < n) { col[h] = (int)(idx + n);     val[h] = -1.0; ++h; }
-        if (k + 1 < n) { col[h] = (int)(idx + n * n); val[h] = -1.0; ++h; }
+        if (i + 1 < n)  { col[h] = (int)(idx + 1);     val[h] = -1.0; ++h; }
+        if (j + 1 < n)  { col[h] = (int)(idx + n);     val[h] = -1.0; ++h; }
+        if (k + 1 < nz) { col[h] = (int)(idx + n * n); val[h] = -1.0; ++h; }
     }
 }
 
-extern "C" int amg_poisson_cnt(int64_t n, int *cnt, hipStream_t s) {
-    poisson_cnt_k<<<nblk(n * n * n), 256, 0, s>>>(n, cnt);
+extern "C" int amg_poisson_cnt(int64_t n, int64_t nz, int64_t row_beg, int64_t row_end,
+                               int *cnt, hipStream_t s) {
+    poisson_cnt_k<<<nblk(row_end - row_beg), 256, 0, s>>>(n, nz, row_beg, row_end, cnt);
     return (int)hipGetLastError();
 }
-extern "C" int amg_poisson_fill(int64_t n, const int *ptr_scanned, int *col, double *val,
+extern "C" int amg_poisson_fill(int64_t n, int64_t nz, int64_t row_beg, int64_t row_end,
+                                const int *ptr_scanned, int *col, double *val,
                                 hipStream_t s) {
-    poisson_fill_k<<<nblk(n * n * n), 256, 0, s>>>(n, ptr_scanned, col, val);
+    poisson_fill_k<<<nblk(row_end - row_beg), 256, 0, s>>>(n, nz, row_beg, row_end,
+                                                           ptr_scanned, col, val);
     return (int)hipGetLastError();
 }
 

@@ -54,22 +54,44 @@ class SubdomainDeflation:
         else:
             self.Z = np.ascontiguousarray(Z)
 
-        # --- AZ columns (distributed spmv per global deflation vector) -----
-        az_cols = []
+        # --- AZ columns -----------------------------------------------------
+        # One halo exchange per LOCAL deflation column (ndv total), not per
+        # GLOBAL one (ndv*world): exchanging the stacked vector
+        # W_j[i] = Z_owner(i)[i, j] delivers every neighbor's j-th column
+        # ghosts at once, and AZ[:, r*ndv+j] = A_rem @ (ghosts owned by r)
+        # separates the owners exactly (A couples only neighboring strips,
+        # so all other columns are exactly zero — the same zeros the
+        # ndv*world-spmv loop computed). O(1) exchanges in world size
+        # (parity: mpi/subdomain_deflation.hpp:242-453 builds AZ from the
+        # same single-exchange ghost rows).
+        az_cols = [base.vector(nloc) for _ in range(m)]
+        for c in az_cols:
+            base.clear(c)
         zcol = base.vector(nloc)
-        for gcol in range(m):
-            owner = gcol // self.ndv
-            local = gcol % self.ndv
-            if owner == rank:
-                if is_hip:
-                    zcol.copy_(self.Z[:, local])
-                else:
-                    np.copyto(zcol, Z[:, local])
+        ghost_masks = []
+        if dist_A.A_rem is not None:
+            for r in dist_A.recv_ranks:
+                msk = (dist_A.ghost_owner == r).astype(np.float64)
+                ghost_masks.append(
+                    torch.from_numpy(msk).to(base.device) if is_hip else msk)
+        for j in range(self.ndv):
+            if is_hip:
+                zcol.copy_(self.Z[:, j])
             else:
-                base.clear(zcol)
-            out = base.vector(nloc)
-            backend.spmv(1.0, dist_A, zcol, 0.0, out)
-            az_cols.append(out)
+                np.copyto(zcol, Z[:, j])
+            works = dist_A.start_exchange(zcol)
+            base.spmv(1.0, dist_A.A_loc, zcol, 0.0, az_cols[self.col0 + j])
+            dist_A.finish_exchange(works)
+            if dist_A.A_rem is not None:
+                xr = dist_A.x_rem
+                masked = base.vector(dist_A.n_ghost)
+                for r, msk in zip(dist_A.recv_ranks, ghost_masks):
+                    if is_hip:
+                        torch.mul(xr, msk, out=masked)
+                    else:
+                        np.multiply(xr, msk, out=masked)
+                    base.spmv(1.0, dist_A.A_rem, masked, 1.0,
+                              az_cols[r * self.ndv + j])
         if is_hip:
             self.AZ = torch.stack(az_cols, dim=1)  # nloc x m
         else:

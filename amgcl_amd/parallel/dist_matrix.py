@@ -49,25 +49,43 @@ class DistMatrix:
         self.n_loc_cols = col_end - col_beg
         self.n_global_cols = int(self.col_begs[-1])
 
-        # split into local + remote (ghost) parts (C++ engine;
-        # parity: distributed_matrix.hpp:370-430)
-        from .. import _core
+        # split into local + remote (ghost) parts
+        # (parity: distributed_matrix.hpp:370-430)
+        self._A_loc_host = self._A_rem_host = None
+        self._rem_host_done = False
+        if isinstance(strip, CSR):
+            # host strip: C++/OpenMP split, then upload
+            from .. import _core
 
-        lp, lc, lv, rp, rc, rv, ghost_global = _core.split_strip(
-            n_loc, col_beg, col_end, strip.ptr, strip.col, strip.val
-        )
-        ghost_global = np.asarray(ghost_global)
-        self.n_ghost = len(ghost_global)
+            lp, lc, lv, rp, rc, rv, ghost_global = _core.split_strip(
+                n_loc, col_beg, col_end, strip.ptr, strip.col, strip.val
+            )
+            ghost_global = np.asarray(ghost_global)
+            self.n_ghost = len(ghost_global)
+            self._A_loc_host = CSR(n_loc, self.n_loc_cols, lp, lc, lv)
+            self.A_loc = base.matrix(self._A_loc_host)
+            self._A_rem_host = (CSR(n_loc, self.n_ghost, rp, rc, rv)
+                                if self.n_ghost else None)
+            self._rem_host_done = True
+            self.A_rem = base.matrix(self._A_rem_host) if self.n_ghost else None
+        else:
+            # device strip (DeviceCSR with global columns): split stays on
+            # the GPU — no host round-trip inside the timed setup
+            from ..backend import hip_setup
+            from ..backend.hip import DeviceCSR
 
-        self.A_loc_host = CSR(n_loc, self.n_loc_cols, lp, lc, lv)
-        self.A_loc = base.matrix(self.A_loc_host)
-        self.A_rem_host = (CSR(n_loc, self.n_ghost, rp, rc, rv)
-                           if self.n_ghost else None)
-        self.A_rem = base.matrix(self.A_rem_host) if self.n_ghost else None
+            lp, lc, lv, rp, rc, rv, gg = hip_setup.split_strip_torch(
+                strip.ptr, strip.col, strip.val, col_beg, col_end)
+            self.n_ghost = int(gg.numel())
+            self.A_loc = DeviceCSR.from_tensors(n_loc, self.n_loc_cols, lp, lc, lv)
+            self.A_rem = (DeviceCSR.from_tensors(n_loc, self.n_ghost, rp, rc, rv)
+                          if self.n_ghost else None)
+            ghost_global = gg.cpu().numpy()
         self.ghost_global = ghost_global  # sorted global ids of ghost columns
 
         # --- comm pattern (who owns each ghost column; what must we send) ---
         owner = np.searchsorted(self.col_begs, ghost_global, side="right") - 1
+        self.ghost_owner = owner
         self.recv_ranks = []
         self.recv_counts = []
         need_from = [np.empty(0, dtype=np.int64)] * self.world
@@ -110,6 +128,27 @@ class DistMatrix:
         if isinstance(v, np.ndarray):
             return self._torch.from_numpy(v)
         return v
+
+    @property
+    def A_loc_host(self):
+        """Host CSR view of the local part (lazy download for device-split
+        strips; only the host-coupled components — dist_amg/pmis, schur,
+        cpr — pay for it)."""
+        if self._A_loc_host is None:
+            from ..backend import hip_setup
+
+            self._A_loc_host = hip_setup.download(self.A_loc)
+        return self._A_loc_host
+
+    @property
+    def A_rem_host(self):
+        if not self._rem_host_done:
+            from ..backend import hip_setup
+
+            self._A_rem_host = (hip_setup.download(self.A_rem)
+                                if self.A_rem is not None else None)
+            self._rem_host_done = True
+        return self._A_rem_host
 
     @property
     def nrows(self):
