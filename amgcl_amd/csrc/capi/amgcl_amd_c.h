@@ -69,9 +69,16 @@ int amgcl_amd_solver_report(amgcl_amd_handle solver, char *buf, int len);
 void amgcl_amd_solver_destroy(amgcl_amd_handle solver);
 
 /* Torch-free GPU solver (exported by libamghip.so, NOT libamgclamd_c.so):
- * host hierarchy assembly + raw hipMalloc upload + the native gfx950 solve
- * driver.  `config` is "key=value;key=value" with the same keys as the
- * Params object (solver.type/tol/maxiter, precond.coarse_enough, ...). */
+ * hierarchy assembly + the native gfx950 solve driver, no Python/torch in
+ * the process.  `config` is "key=value;key=value" with the same keys as the
+ * Params object (solver.type/tol/maxiter, precond.coarse_enough, ...).
+ * `precond.setup` selects where the hierarchy is built: "device" runs the
+ * whole SA setup (strong/MIS aggregation/smoothed P/Galerkin + SELL-64
+ * images) on the GPU via the setup.hip kernels, "host" uses the C++/OpenMP
+ * engine + upload, "auto" (default) picks device above 200k rows.  The
+ * device setup hands the tail below `precond.device_handoff` (20000) rows
+ * to the host engine, and falls back to the host path entirely if a row
+ * overflows its LDS buffers. */
 amgcl_amd_handle amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *col,
                                              const double *val, const char *config);
 int amgcl_amd_gpu_solver_solve(amgcl_amd_handle solver, const double *rhs,

@@ -13,6 +13,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstdio>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -27,6 +28,46 @@ extern "C" void *amg_driver_create(const LevelDesc *levels, int nlevels,
                                    const int *a64_col, const double *a64_val,
                                    int a64_subw, hipStream_t stream);
 extern "C" void amg_driver_destroy(void *h);
+// device setup engine (setup.hip)
+extern "C" int amg_setup_diag(int64_t n, const int *ptr, const int *col,
+                              const double *val, double *d, hipStream_t s);
+extern "C" int amg_setup_strong(int64_t n, const int *ptr, const int *col,
+                                const double *val, const double *d, double eps2,
+                                uint8_t *S, hipStream_t s);
+extern "C" int amg_setup_spai0(int64_t n, const int *ptr, const int *col,
+                               const double *val, double *m, hipStream_t s);
+extern "C" int amg_agg_init(int64_t n, const int *ptr, const uint8_t *S, int *id,
+                            hipStream_t s);
+extern "C" int amg_agg_run(int64_t n, const int *ptr, const int *col, const uint8_t *S,
+                           int *id, uint8_t *prov, uint64_t *m1, uint8_t *newroot,
+                           uint8_t *near, int *remaining, int sync_stride,
+                           int max_rounds, int *rounds_out, int *lists, hipStream_t s);
+extern "C" int amg_agg_renumber(int64_t n, int *id, int *mark, hipStream_t s);
+extern "C" int amg_psmooth_count(int64_t n, const int *ptr, const int *col,
+                                 const uint8_t *S, const int *id, int *cnt,
+                                 int *overflow, hipStream_t s);
+extern "C" int amg_psmooth_fill(int64_t n, const int *ptr, const int *col,
+                                const double *val, const uint8_t *S, const int *id,
+                                double omega, const int *pptr_scanned, int *pcol,
+                                double *pval, hipStream_t s);
+extern "C" int amg_transpose_count(int64_t nnz, const int *col, int *tcnt,
+                                   hipStream_t s);
+extern "C" int amg_transpose_scatter(int64_t n, const int *ptr, const int *col,
+                                     const double *val, int *cursor, int *tcol,
+                                     double *tval, hipStream_t s);
+extern "C" int amg_spgemm_count(int64_t an, const int *aptr, const int *acol,
+                                const int *bptr, const int *bcol, int *ub, int *cnt,
+                                int *overflow, int *bigscratch, hipStream_t s);
+extern "C" int amg_spgemm_fill(int64_t an, const int *aptr, const int *acol,
+                               const double *aval, const int *bptr, const int *bcol,
+                               const double *bval, const int *ub,
+                               const int *cptr_scanned, int *ccol, double *cval,
+                               int do_sort, const int *bigscratch, hipStream_t s);
+extern "C" int amg_scan_i32(int *a, int64_t n, hipStream_t s);
+extern "C" int amg_sell_fill_f64(int64_t nrows, int64_t nslice, const int *ptr,
+                                 const int *col, const double *val,
+                                 const int64_t *soff, const int *srows, int *scol,
+                                 double *sval, hipStream_t stream);
 extern "C" int amg_driver_cg(void *h, const double *rhs, double *x, double *r,
                              double *s, double *p, double *q, double *s_swap,
                              double tol, double abstol, int maxiter,
@@ -107,6 +148,306 @@ Params parse_config(const char *cfg) {
     return p;
 }
 
+// ---------------------------------------------------------------------------
+// Device-resident setup: the same SA pipeline the Python flagship runs
+// (strong -> MIS aggregation -> smoothed P -> R = P^T -> Galerkin), driven
+// from C++ over the setup.hip kernels with raw hipMalloc buffers — no torch,
+// no Python.  The tail below `precond.device_handoff` rows (default 20000)
+// is downloaded and finished by the host engine (same policy and rationale
+// as backend/hip_setup.py: tiny levels are launch-bound on the GPU and the
+// host engine supports every smoother/coarse solver).
+// ---------------------------------------------------------------------------
+
+__global__ void capi_dj_k(int64_t n, const double *__restrict__ d, double damping,
+                          double *__restrict__ m) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) m[i] = d[i] != 0.0 ? damping / d[i] : 0.0;
+}
+
+__global__ void capi_slicew_k(int64_t n, int64_t nslice, const int *__restrict__ ptr,
+                              int *__restrict__ w) {
+    int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; s < nslice; s += stride) {
+        int mx = 0;
+        int64_t r0 = s * 64, r1 = r0 + 64 < n ? r0 + 64 : n;
+        for (int64_t r = r0; r < r1; ++r) {
+            int l = ptr[r + 1] - ptr[r];
+            if (l > mx) mx = l;
+        }
+        w[s] = mx;
+    }
+}
+
+namespace {
+
+struct DevSetupFail {};  // -> host fallback
+
+// tracked hipMalloc: `keep` buffers live with the solver, `tmp` buffers are
+// freed when the Tracker goes out of scope (per-level scratch)
+struct Tracker {
+    std::vector<void *> bufs;
+    ~Tracker() {
+        for (void *p : bufs) (void)hipFree(p);
+    }
+    template <typename T>
+    T *alloc(size_t count, bool zero = false) {
+        void *p = nullptr;
+        if (hipMalloc(&p, count * sizeof(T)) != hipSuccess) throw DevSetupFail{};
+        if (zero) (void)hipMemset(p, 0, count * sizeof(T));
+        bufs.push_back(p);
+        return (T *)p;
+    }
+    void *release(void *p) {  // transfer ownership to the caller
+        for (size_t i = 0; i < bufs.size(); ++i)
+            if (bufs[i] == p) {
+                bufs.erase(bufs.begin() + i);
+                return p;
+            }
+        return p;
+    }
+};
+
+static void dev_check(int rc) {
+    if (rc != 0) throw DevSetupFail{};
+}
+
+struct DevCsr {
+    int64_t n = 0, ncols = 0, nnz = 0;
+    int *ptr = nullptr;
+    int *col = nullptr;
+    double *val = nullptr;
+};
+
+static int read_i32(const int *dev_p) {
+    int v = 0;
+    if (hipMemcpy(&v, dev_p, sizeof(int), hipMemcpyDeviceToHost) != hipSuccess)
+        throw DevSetupFail{};
+    return v;
+}
+
+// scan counts stored at p[1..n] in place, p[0] = 0; returns total
+static int scan_counts(int *p, int64_t n) {
+    dev_check(amg_scan_i32(p + 1, n, 0));
+    return read_i32(p + n);
+}
+
+// SELL-64 image built on the device (same layout as DeviceCSR.build_sell)
+static void build_sell_capi(GpuSolver *S, int64_t n, const int *ptr, const int *col,
+                            const double *val, int64_t &nslice_out,
+                            const int64_t *&soff_out, const int *&scol_out,
+                            const void *&sval_out) {
+    int64_t nslice = (n + 63) / 64;
+    Tracker tmp;
+    int *w = tmp.alloc<int>(nslice);
+    capi_slicew_k<<<amg_nblocks(nslice), 256, 0, 0>>>(n, nslice, ptr, w);
+    dev_check((int)hipGetLastError());
+    std::vector<int> wh(nslice);
+    if (hipMemcpy(wh.data(), w, nslice * sizeof(int), hipMemcpyDeviceToHost) !=
+        hipSuccess)
+        throw DevSetupFail{};
+    std::vector<int64_t> soff(nslice + 1, 0);
+    for (int64_t s = 0; s < nslice; ++s) soff[s + 1] = soff[s] + (int64_t)wh[s] * 64;
+    int64_t total = soff[nslice];
+    int64_t *soff_d = (int64_t *)S->keep(upload(soff));
+    if (!soff_d) throw DevSetupFail{};
+    Tracker out;
+    int *scol = out.alloc<int>(total);
+    double *sval = out.alloc<double>(total);
+    dev_check(amg_sell_fill_f64(n, nslice, ptr, col, val, soff_d, nullptr, scol,
+                                sval, 0));
+    S->keep(out.release(scol));
+    S->keep(out.release(sval));
+    nslice_out = nslice;
+    soff_out = soff_d;
+    scol_out = scol;
+    sval_out = sval;
+}
+
+// One SpGEMM product C = A*B on the device (ownership of C's arrays moves to
+// the caller's tracker)
+static DevCsr spgemm_dev(Tracker &own, const DevCsr &A, const DevCsr &B, int sort) {
+    Tracker tmp;
+    int *ub = tmp.alloc<int>(A.n);
+    int *cptr = own.alloc<int>(A.n + 1, true);
+    int *overflow = tmp.alloc<int>(1, true);
+    int *bigscratch = tmp.alloc<int>(2 * A.n + 1);
+    dev_check(amg_spgemm_count(A.n, A.ptr, A.col, B.ptr, B.col, ub, cptr + 1,
+                               overflow, bigscratch, 0));
+    int nnz = scan_counts(cptr, A.n);
+    if (read_i32(overflow) != 0) throw DevSetupFail{};
+    DevCsr C;
+    C.n = A.n;
+    C.ncols = B.ncols;
+    C.nnz = nnz;
+    C.ptr = cptr;
+    C.col = own.alloc<int>(nnz);
+    C.val = own.alloc<double>(nnz);
+    dev_check(amg_spgemm_fill(A.n, A.ptr, A.col, A.val, B.ptr, B.col, B.val, ub,
+                              cptr + 1, C.col, C.val, sort, bigscratch, 0));
+    return C;
+}
+
+// Build the device part of the hierarchy; returns the handoff matrix
+// downloaded to the host (tail continues in the host engine).
+static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *col,
+                               const double *val, const Params &p,
+                               std::vector<LevelDesc> &descs, double &eps_inout) {
+    const int handoff = p.geti("precond.device_handoff", 20000);
+    const int coarse_enough = p.geti("precond.coarse_enough", 1000);
+    const int max_levels = p.geti("precond.max_levels", 20);
+    const std::string relax = p.gets("precond.relax.type", "spai0");
+    const double damping = p.getf("precond.relax.damping", 0.72);
+    const int64_t sell_min = 500000;
+    if (relax != "spai0" && relax != "damped_jacobi") throw DevSetupFail{};
+
+    Tracker own;  // per-hierarchy ownership; transferred to S->keep on success
+    DevCsr A;
+    A.n = n;
+    A.ncols = n;
+    A.nnz = ptr[n];
+    A.ptr = own.alloc<int>(n + 1);
+    A.col = own.alloc<int>(A.nnz);
+    A.val = own.alloc<double>(A.nnz);
+    if (hipMemcpy(A.ptr, ptr, (n + 1) * sizeof(int), hipMemcpyHostToDevice) !=
+            hipSuccess ||
+        hipMemcpy(A.col, col, A.nnz * sizeof(int), hipMemcpyHostToDevice) !=
+            hipSuccess ||
+        hipMemcpy(A.val, val, A.nnz * sizeof(double), hipMemcpyHostToDevice) !=
+            hipSuccess)
+        throw DevSetupFail{};
+
+    double eps = eps_inout;
+    while (A.n > handoff && A.n > coarse_enough &&
+           (int)descs.size() + 1 < max_levels) {
+        Tracker tmp;
+        // smoother weights
+        double *M = own.alloc<double>(A.n);
+        if (relax == "spai0") {
+            dev_check(amg_setup_spai0(A.n, A.ptr, A.col, A.val, M, 0));
+        } else {
+            double *d0 = tmp.alloc<double>(A.n);
+            dev_check(amg_setup_diag(A.n, A.ptr, A.col, A.val, d0, 0));
+            capi_dj_k<<<amg_nblocks(A.n), 256, 0, 0>>>(A.n, d0, damping, M);
+            dev_check((int)hipGetLastError());
+        }
+        // strong connections + MIS aggregation
+        double *d = tmp.alloc<double>(A.n);
+        dev_check(amg_setup_diag(A.n, A.ptr, A.col, A.val, d, 0));
+        uint8_t *Sg = tmp.alloc<uint8_t>(A.nnz);
+        dev_check(amg_setup_strong(A.n, A.ptr, A.col, A.val, d, eps * eps, Sg, 0));
+        int *ids = tmp.alloc<int>(A.n);
+        dev_check(amg_agg_init(A.n, A.ptr, Sg, ids, 0));
+        uint8_t *prov = tmp.alloc<uint8_t>(A.n, true);
+        uint64_t *m1 = tmp.alloc<uint64_t>(A.n);
+        uint8_t *newroot = tmp.alloc<uint8_t>(A.n);
+        uint8_t *near = tmp.alloc<uint8_t>(A.n);
+        int *remaining = tmp.alloc<int>(1, true);
+        int *lists = tmp.alloc<int>(3 * (size_t)A.n);
+        int rounds = 0;
+        dev_check(amg_agg_run(A.n, A.ptr, A.col, Sg, ids, prov, m1, newroot, near,
+                              remaining, 2, 64, &rounds, lists, 0));
+        int *mark = tmp.alloc<int>(A.n);
+        dev_check(amg_agg_renumber(A.n, ids, mark, 0));
+        int naggr = read_i32(mark + A.n - 1);
+        if (naggr <= 0 || naggr >= A.n) break;  // no progress: host tail
+        eps *= 0.5;
+        // smoothed prolongation P = (I - omega D^-1 A_F) T, omega = 2/3
+        DevCsr P;
+        P.n = A.n;
+        P.ncols = naggr;
+        P.ptr = own.alloc<int>(A.n + 1, true);
+        int *pov = tmp.alloc<int>(1, true);
+        dev_check(amg_psmooth_count(A.n, A.ptr, A.col, Sg, ids, P.ptr + 1, pov, 0));
+        P.nnz = scan_counts(P.ptr, A.n);
+        if (read_i32(pov) != 0) throw DevSetupFail{};
+        P.col = own.alloc<int>(P.nnz);
+        P.val = own.alloc<double>(P.nnz);
+        dev_check(amg_psmooth_fill(A.n, A.ptr, A.col, A.val, Sg, ids, 2.0 / 3.0,
+                                   P.ptr + 1, P.col, P.val, 0));
+        // R = P^T
+        DevCsr R;
+        R.n = naggr;
+        R.ncols = A.n;
+        R.nnz = P.nnz;
+        R.ptr = own.alloc<int>(naggr + 1, true);
+        dev_check(amg_transpose_count(P.nnz, P.col, R.ptr + 1, 0));
+        (void)scan_counts(R.ptr, naggr);
+        int *cursor = tmp.alloc<int>(naggr);
+        if (hipMemcpy(cursor, R.ptr, naggr * sizeof(int),
+                      hipMemcpyDeviceToDevice) != hipSuccess)
+            throw DevSetupFail{};
+        R.col = own.alloc<int>(R.nnz);
+        R.val = own.alloc<double>(R.nnz);
+        dev_check(amg_transpose_scatter(A.n, P.ptr, P.col, P.val, cursor, R.col,
+                                        R.val, 0));
+        // Galerkin Ac = R*(A*P) (association measured in matrix.py galerkin)
+        DevCsr Ac;
+        {
+            Tracker mid;
+            DevCsr AP = spgemm_dev(mid, A, P, 0);
+            Ac = spgemm_dev(own, R, AP, 1);
+        }
+        // record this level
+        LevelDesc dsc;
+        std::memset(&dsc, 0, sizeof dsc);
+        dsc.nrows = A.n;
+        dsc.nnz = A.nnz;
+        dsc.ptr = A.ptr;
+        dsc.col = A.col;
+        dsc.val = A.val;
+        dsc.subw = pick_subw(A.n, A.nnz);
+        dsc.pnnz = P.nnz;
+        dsc.pptr = P.ptr;
+        dsc.pcol = P.col;
+        dsc.pval = P.val;
+        dsc.psubw = pick_subw(P.n, P.nnz);
+        dsc.rnnz = R.nnz;
+        dsc.rptr = R.ptr;
+        dsc.rcol = R.col;
+        dsc.rval = R.val;
+        dsc.rsubw = pick_subw(R.n, R.nnz);
+        dsc.M = M;
+        dsc.f = (double *)own.alloc<double>(A.n, true);
+        dsc.u = (double *)own.alloc<double>(A.n, true);
+        dsc.t = (double *)own.alloc<double>(A.n, true);
+        if (A.n >= sell_min)
+            build_sell_capi(S, A.n, A.ptr, A.col, A.val, dsc.nslice, dsc.soff,
+                            dsc.scol, dsc.sval);
+        if (P.n >= sell_min) {
+            build_sell_capi(S, P.n, P.ptr, P.col, P.val, dsc.pnslice, dsc.psoff,
+                            dsc.pscol, dsc.psval);
+            build_sell_capi(S, R.n, R.ptr, R.col, R.val, dsc.rnslice, dsc.rsoff,
+                            dsc.rscol, dsc.rsval);
+        }
+        descs.push_back(dsc);
+        A = Ac;
+    }
+    // download the handoff matrix for the host tail
+    std::vector<int> hp(A.n + 1), hc(A.nnz);
+    std::vector<double> hv(A.nnz);
+    if (hipMemcpy(hp.data(), A.ptr, (A.n + 1) * sizeof(int),
+                  hipMemcpyDeviceToHost) != hipSuccess ||
+        hipMemcpy(hc.data(), A.col, A.nnz * sizeof(int), hipMemcpyDeviceToHost) !=
+            hipSuccess ||
+        hipMemcpy(hv.data(), A.val, A.nnz * sizeof(double),
+                  hipMemcpyDeviceToHost) != hipSuccess)
+        throw DevSetupFail{};
+    Csr tail;
+    tail.n = tail.m = A.n;
+    tail.ptr = std::move(hp);
+    tail.col = std::move(hc);
+    tail.val = std::move(hv);
+    // success: everything still owned by `own` now belongs to the solver
+    for (void *b : own.bufs) S->keep(b);
+    own.bufs.clear();
+    eps_inout = eps;
+    return tail;
+}
+
+}  // namespace (inner)
+
 }  // namespace
 
 extern "C" void *amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *col,
@@ -118,18 +459,44 @@ extern "C" void *amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *c
     S->tol = p.getf("solver.tol", 1e-8);
     S->maxiter = p.geti("solver.maxiter", 200);
 
-    // host hierarchy (same engine as the CPU C API)
-    Precond P;
-    {
-        Csr A = amgclamd_host::make_csr(n, ptr, col, val, 0);
-        P.build(std::move(A), p);
+    // device levels first (precond.setup=device, or auto above 200k rows),
+    // then the host engine finishes the tail below precond.device_handoff
+    std::vector<LevelDesc> descs;
+    double eps = p.getf("precond.coarsening.eps_strong", 0.08);
+    const std::string setup = p.gets("precond.setup", "auto");
+    Csr tail;
+    bool have_tail = false;
+    if (setup == "device" || (setup == "auto" && n > 200000)) {
+        try {
+            tail = build_device_levels(S, n, ptr, col, val, p, descs, eps);
+            have_tail = true;
+        } catch (DevSetupFail &) {
+            descs.clear();  // partial buffers stay in S->blobs until destroy
+        }
     }
 
-    // upload levels
-    std::vector<LevelDesc> descs(P.lvl.size());
+    // host hierarchy (same engine as the CPU C API) — the whole thing, or
+    // just the tail the device setup handed off
+    Precond P;
+    {
+        Params p2 = p;
+        char ebuf[32];
+        std::snprintf(ebuf, sizeof ebuf, "%.17g", eps);
+        p2.kv["precond.coarsening.eps_strong"] = ebuf;
+        if (have_tail)
+            P.build(std::move(tail), p2);
+        else {
+            Csr A = amgclamd_host::make_csr(n, ptr, col, val, 0);
+            P.build(std::move(A), p2);
+        }
+    }
+
+    // upload the host-built levels, appended after the device-built ones
+    size_t base0 = descs.size();
+    descs.resize(base0 + P.lvl.size());
     for (size_t i = 0; i < P.lvl.size(); ++i) {
         auto &L = P.lvl[i];
-        LevelDesc &d = descs[i];
+        LevelDesc &d = descs[base0 + i];
         std::memset(&d, 0, sizeof d);
         d.nrows = L.A.n;
         d.nnz = L.A.nnz();
