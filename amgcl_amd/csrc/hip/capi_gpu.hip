@@ -189,6 +189,10 @@ struct DevSetupFail {};  // -> host fallback
 struct Tracker {
     std::vector<void *> bufs;
     ~Tracker() {
+        // kernels that read this scratch (SpGEMM fill, the AP intermediate)
+        // may still be in flight — hipFree here does NOT order against them
+        // (measured: async fault at 256^3, clean under AMD_SERIALIZE_KERNEL)
+        if (!bufs.empty()) (void)hipStreamSynchronize(0);
         for (void *p : bufs) (void)hipFree(p);
     }
     template <typename T>
@@ -211,6 +215,13 @@ struct Tracker {
 
 static void dev_check(int rc) {
     if (rc != 0) throw DevSetupFail{};
+}
+
+static void trace(const char *what, long long a = -1, long long b = -1) {
+    if (!getenv("AMGCL_CAPI_TRACE")) return;
+    fprintf(stderr, "[capi setup] %s %lld %lld\n", what, a, b);
+    (void)hipDeviceSynchronize();
+    fflush(stderr);
 }
 
 struct DevCsr {
@@ -332,6 +343,7 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
             capi_dj_k<<<amg_nblocks(A.n), 256, 0, 0>>>(A.n, d0, damping, M);
             dev_check((int)hipGetLastError());
         }
+        trace("smoother", A.n, A.nnz);
         // strong connections + MIS aggregation
         double *d = tmp.alloc<double>(A.n);
         dev_check(amg_setup_diag(A.n, A.ptr, A.col, A.val, d, 0));
@@ -348,9 +360,11 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
         int rounds = 0;
         dev_check(amg_agg_run(A.n, A.ptr, A.col, Sg, ids, prov, m1, newroot, near,
                               remaining, 2, 64, &rounds, lists, 0));
+        trace("agg_run");
         int *mark = tmp.alloc<int>(A.n);
         dev_check(amg_agg_renumber(A.n, ids, mark, 0));
         int naggr = read_i32(mark + A.n - 1);
+        trace("naggr", naggr);
         if (naggr <= 0 || naggr >= A.n) break;  // no progress: host tail
         eps *= 0.5;
         // smoothed prolongation P = (I - omega D^-1 A_F) T, omega = 2/3
@@ -366,6 +380,7 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
         P.val = own.alloc<double>(P.nnz);
         dev_check(amg_psmooth_fill(A.n, A.ptr, A.col, A.val, Sg, ids, 2.0 / 3.0,
                                    P.ptr + 1, P.col, P.val, 0));
+        trace("psmooth", P.nnz);
         // R = P^T
         DevCsr R;
         R.n = naggr;
@@ -382,6 +397,7 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
         R.val = own.alloc<double>(R.nnz);
         dev_check(amg_transpose_scatter(A.n, P.ptr, P.col, P.val, cursor, R.col,
                                         R.val, 0));
+        trace("transpose");
         // Galerkin Ac = R*(A*P) (association measured in matrix.py galerkin)
         DevCsr Ac;
         {
@@ -389,6 +405,7 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
             DevCsr AP = spgemm_dev(mid, A, P, 0);
             Ac = spgemm_dev(own, R, AP, 1);
         }
+        trace("galerkin", Ac.n, Ac.nnz);
         // record this level
         LevelDesc dsc;
         std::memset(&dsc, 0, sizeof dsc);
@@ -412,6 +429,7 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
         dsc.f = (double *)own.alloc<double>(A.n, true);
         dsc.u = (double *)own.alloc<double>(A.n, true);
         dsc.t = (double *)own.alloc<double>(A.n, true);
+        trace("record");
         if (A.n >= sell_min)
             build_sell_capi(S, A.n, A.ptr, A.col, A.val, dsc.nslice, dsc.soff,
                             dsc.scol, dsc.sval);
@@ -421,9 +439,11 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
             build_sell_capi(S, R.n, R.ptr, R.col, R.val, dsc.rnslice, dsc.rsoff,
                             dsc.rscol, dsc.rsval);
         }
+        trace("sell");
         descs.push_back(dsc);
         A = Ac;
     }
+    trace("loop done", A.n);
     // download the handoff matrix for the host tail
     std::vector<int> hp(A.n + 1), hc(A.nnz);
     std::vector<double> hv(A.nnz);
