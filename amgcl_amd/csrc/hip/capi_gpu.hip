@@ -430,6 +430,11 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
         dsc.u = (double *)own.alloc<double>(A.n, true);
         dsc.t = (double *)own.alloc<double>(A.n, true);
         trace("record");
+        if (getenv("AMGCL_CAPI_NO_SELL")) {
+            descs.push_back(dsc);
+            A = Ac;
+            continue;
+        }
         if (A.n >= sell_min)
             build_sell_capi(S, A.n, A.ptr, A.col, A.val, dsc.nslice, dsc.soff,
                             dsc.scol, dsc.sval);
@@ -569,6 +574,9 @@ extern "C" void *amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *c
     for (int i = 0; i < nwork; ++i) S->work.push_back((double *)S->keep(dalloc(n)));
     S->rhs_d = (double *)S->keep(dalloc(n));
     S->x_d = (double *)S->keep(dalloc(n));
+    // setup kernels (SELL fills etc.) may still be in flight; finish them so
+    // create/solve wall times split honestly for callers that time them
+    (void)hipDeviceSynchronize();
     return S;
 }
 
