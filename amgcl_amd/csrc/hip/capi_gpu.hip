@@ -189,17 +189,19 @@ struct DevSetupFail {};  // -> host fallback
 struct Tracker {
     std::vector<void *> bufs;
     ~Tracker() {
-        // kernels that read this scratch (SpGEMM fill, the AP intermediate)
-        // may still be in flight — hipFree here does NOT order against them
-        // (measured: async fault at 256^3, clean under AMD_SERIALIZE_KERNEL)
-        if (!bufs.empty()) (void)hipStreamSynchronize(0);
-        for (void *p : bufs) (void)hipFree(p);
+        // stream-ordered frees: ordered after the kernels that read this
+        // scratch (SpGEMM fill, the AP intermediate) with no device sync,
+        // and the pool recycles the pages instead of unmapping them (raw
+        // hipFree faulted async at 256^3 and its unmap churn showed up as
+        // ~90 ms launch stalls later in the process)
+        for (void *p : bufs) (void)hipFreeAsync(p, 0);
     }
     template <typename T>
     T *alloc(size_t count, bool zero = false) {
         void *p = nullptr;
-        if (hipMalloc(&p, count * sizeof(T)) != hipSuccess) throw DevSetupFail{};
-        if (zero) (void)hipMemset(p, 0, count * sizeof(T));
+        if (hipMallocAsync(&p, count * sizeof(T), 0) != hipSuccess)
+            throw DevSetupFail{};
+        if (zero) (void)hipMemsetAsync(p, 0, count * sizeof(T), 0);
         bufs.push_back(p);
         return (T *)p;
     }

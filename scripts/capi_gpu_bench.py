@@ -52,7 +52,6 @@ def main(m=256):
         cfg = (f"solver.type=cg;solver.tol=1e-6;precond.coarse_enough=1000;"
                f"precond.setup={mode}").encode()
         for rep in range(2):  # first = cold allocator, second = warm
-            x[:] = 0.0
             t0 = time.perf_counter()
             h = lib.amgcl_amd_gpu_solver_create(
                 n, ptr.ctypes.data, col.ctypes.data, val.ctypes.data, cfg)
@@ -60,14 +59,19 @@ def main(m=256):
             assert h, "create failed"
             it = ctypes.c_int(0)
             res = ctypes.c_double(0.0)
-            rc = lib.amgcl_amd_gpu_solver_solve(
-                h, b.ctypes.data, x.ctypes.data, ctypes.byref(it),
-                ctypes.byref(res))
-            t2 = time.perf_counter()
-            assert rc == 0 and res.value < 1e-6, (rc, res.value)
+            solves = []
+            for s in range(3):  # production pattern: many solves per handle
+                x[:] = 0.0
+                ts = time.perf_counter()
+                rc = lib.amgcl_amd_gpu_solver_solve(
+                    h, b.ctypes.data, x.ctypes.data, ctypes.byref(it),
+                    ctypes.byref(res))
+                solves.append(time.perf_counter() - ts)
+                assert rc == 0 and res.value < 1e-6, (rc, res.value)
             lib.amgcl_amd_gpu_solver_destroy(h)
+            sv = "/".join(f"{s:.3f}" for s in solves)
             print(f"setup={mode:6s} rep={rep} create={t1-t0:7.3f} s "
-                  f"solve={t2-t1:7.3f} s iters={it.value} resid={res.value:.2e}")
+                  f"solve={sv} s iters={it.value} resid={res.value:.2e}")
     assert "torch" not in sys.modules
     print("TORCHFREE_BENCH_OK")
 
