@@ -152,7 +152,8 @@ static int pick_subw(int64_t nrows, int64_t nnz) {
 
 struct GpuSolver {
     void *driver = nullptr;
-    std::vector<void *> blobs;  // everything hipMalloc'd
+    std::vector<void *> blobs;       // hipMalloc'd
+    std::vector<void *> pool_blobs;  // hipMallocAsync'd (device-setup levels)
     int64_t n = 0;
     std::string type = "cg";
     double tol = 1e-8;
@@ -163,10 +164,17 @@ struct GpuSolver {
     ~GpuSolver() {
         if (driver) amg_driver_destroy(driver);
         for (void *p : blobs) (void)hipFree(p);
+        for (void *p : pool_blobs) (void)hipFreeAsync(p, 0);
+        if (!pool_blobs.empty()) (void)hipStreamSynchronize(0);
     }
 
     void *keep(void *p) {
         if (p) blobs.push_back(p);
+        return p;
+    }
+
+    void *keep_pool(void *p) {
+        if (p) pool_blobs.push_back(p);
         return p;
     }
 };
@@ -310,8 +318,8 @@ static void build_sell_capi(GpuSolver *S, int64_t n, const int *ptr, const int *
     double *sval = out.alloc<double>(total);
     dev_check(amg_sell_fill_f64(n, nslice, ptr, col, val, soff_d, nullptr, scol,
                                 sval, 0));
-    S->keep(out.release(scol));
-    S->keep(out.release(sval));
+    S->keep_pool(out.release(scol));
+    S->keep_pool(out.release(sval));
     nslice_out = nslice;
     soff_out = soff_d;
     scol_out = scol;
@@ -504,7 +512,8 @@ static Csr build_device_levels(GpuSolver *S, int n, const int *ptr, const int *c
     tail.col = std::move(hc);
     tail.val = std::move(hv);
     // success: everything still owned by `own` now belongs to the solver
-    for (void *b : own.bufs) S->keep(b);
+    // (hipMallocAsync memory -> freed with hipFreeAsync at destroy)
+    for (void *b : own.bufs) S->keep_pool(b);
     own.bufs.clear();
     eps_inout = eps;
     return tail;
