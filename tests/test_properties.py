@@ -273,3 +273,26 @@ def test_aggregates_partition_validity(n, seed):
         valid = ids[ids >= 0]
         assert valid.size and valid.max() < naggr
         assert len(np.unique(valid)) == naggr  # every aggregate nonempty
+
+
+@settings(**COMMON)
+@given(n=st.integers(2, 60), nglob=st.integers(10, 300), seed=st.integers(0, 10**6),
+       frac=st.floats(0.05, 0.9))
+def test_split_strip_torch_matches_core_property(n, nglob, seed, frac):
+    """Device-path strip split == C++ split for ANY strip and any column
+    window (empty rows, all-ghost rows, window at either edge)."""
+    import torch
+
+    nglob = max(nglob, n)
+    a = rand_csr(n, nglob, 0.15, seed)
+    beg = int(frac * (nglob - 1))
+    end = min(nglob, beg + max(1, n))
+    ptr = a.indptr.astype(np.int32)
+    col = a.indices.astype(np.int32)
+    ref = _core.split_strip(n, beg, end, ptr, col, a.data)
+    from amgcl_amd.backend.hip_setup import split_strip_torch
+
+    got = split_strip_torch(torch.from_numpy(ptr), torch.from_numpy(col),
+                            torch.from_numpy(a.data), beg, end)
+    for r, g in zip(ref, got):
+        np.testing.assert_array_equal(np.asarray(r), g.numpy())
