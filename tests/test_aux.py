@@ -111,3 +111,19 @@ def test_examples_run(name, args):
         [sys.executable, os.path.join(root, "examples", f"{name}.py"), *args],
         capture_output=True, text=True, timeout=420, cwd=root)
     assert out.returncode == 0, out.stdout + out.stderr
+
+
+def test_distributed_example_runs():
+    """examples/distributed_solver.py under torchrun (2 ranks, gloo on CPU) —
+    the reference's mpi_solver example equivalent."""
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29641",
+         os.path.join(root, "examples", "distributed_solver.py"), "24"],
+        capture_output=True, text=True, timeout=420, cwd=root)
+    assert out.returncode == 0, out.stdout + out.stderr
