@@ -567,6 +567,7 @@ extern "C" void *amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *c
     // upload the host-built levels, appended after the device-built ones
     size_t base0 = descs.size();
     descs.resize(base0 + P.lvl.size());
+    bool oom = false;
     for (size_t i = 0; i < P.lvl.size(); ++i) {
         auto &L = P.lvl[i];
         LevelDesc &d = descs[base0 + i];
@@ -593,6 +594,16 @@ extern "C" void *amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *c
         d.f = (double *)S->keep(dalloc(L.A.n));
         d.u = (double *)S->keep(dalloc(L.A.n));
         d.t = (double *)S->keep(dalloc(L.A.n));
+        // hipMalloc failure shows up as a null field: fail the create
+        // cleanly instead of handing the driver dangling level pointers
+        oom |= !d.ptr || !d.col || !d.val || !d.f || !d.u || !d.t;
+        oom |= (L.P.n != 0) && (!d.pptr || !d.pcol || !d.pval ||
+                                !d.rptr || !d.rcol || !d.rval);
+        oom |= !L.M.empty() && !d.M;
+    }
+    if (oom) {
+        delete S;
+        return nullptr;
     }
 
     // coarsest: dense inverse on device (solve LU against identity columns)
