@@ -91,8 +91,14 @@ static amgcl_amd_handle solver_create(int n, const int *ptr, const int *col,
     Params dflt;
     const Params &p = prm ? *static_cast<Params *>(prm) : dflt;
     auto *S = new Solver;
-    S->A = make_csr(n, ptr, col, val, base);
     S->type = p.gets("solver.type", "cg");
+    if (S->type != "cg" && S->type != "bicgstab") {
+        // the compiled engine carries CG and BiCGStab; fail loudly instead
+        // of silently substituting (the Python API has the full Krylov set)
+        delete S;
+        return nullptr;
+    }
+    S->A = make_csr(n, ptr, col, val, base);
     S->tol = p.getf("solver.tol", 1e-8);
     S->maxiter = p.geti("solver.maxiter", 200);
     S->prec.build(S->A, p);

@@ -529,6 +529,12 @@ extern "C" void *amgcl_amd_gpu_solver_create(int n, const int *ptr, const int *c
     auto *S = new GpuSolver;
     S->n = n;
     S->type = p.gets("solver.type", "cg");
+    if (S->type != "cg" && S->type != "bicgstab") {
+        // the native driver carries CG and BiCGStab; fail loudly instead of
+        // silently substituting (the Python API has the full Krylov set)
+        delete S;
+        return nullptr;
+    }
     S->tol = p.getf("solver.tol", 1e-8);
     S->maxiter = p.geti("solver.maxiter", 200);
 

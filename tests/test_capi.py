@@ -325,3 +325,15 @@ print("DEVICE_SETUP_OK", it.value, it2.value, res.value)
     out = subprocess.check_output([sys.executable, "-c", prog], text=True,
                                   stderr=subprocess.STDOUT, timeout=600)
     assert "DEVICE_SETUP_OK" in out
+
+
+def test_capi_rejects_unknown_solver(capi):
+    """The compiled engine carries CG/BiCGStab; asking for anything else
+    must fail at create (NULL handle), never silently substitute."""
+    A, b = am.poisson3d(8)
+    ptr, col, val = _arrays(A)
+    prm = capi.amgcl_amd_params_create()
+    capi.amgcl_amd_params_sets(prm, b"solver.type", b"gmres")
+    s = capi.amgcl_amd_solver_create(A.nrows, _ptr(ptr), _ptr(col), _ptr(val), prm)
+    capi.amgcl_amd_params_destroy(prm)
+    assert not s
