@@ -108,6 +108,9 @@ class CpuBackend:
             return SpluCoarseSolver(csr, self)
         if kind == "skyline":
             return SkylineCoarseSolver(csr, self)
+        if kind != "dense":
+            raise ValueError(f"unknown direct_solver '{kind}' "
+                             "(dense, skyline, splu)")
         return DenseCoarseSolver(csr, self)
 
     def synchronize(self):

@@ -455,6 +455,9 @@ class HipBackend:
             # reference's coarse-solve shape (hip.hpp:73-96); only worth it
             # for very large, very sparse coarse levels
             return HostSpluSolver(csr, self)
+        if kind != "dense":
+            raise ValueError(f"unknown direct_solver '{kind}' "
+                             "(dense, skyline, splu)")
         if isinstance(csr, DeviceCSR):
             return DeviceDenseSolver.from_device(csr, self)
         return DeviceDenseSolver(csr, self)
