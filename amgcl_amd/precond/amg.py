@@ -77,6 +77,17 @@ class AMG:
             backend = make_backend("cpu")
         self.backend = backend
         self.prm = merge_params(self.defaults(), prm, opaque=("coarsening", "relax"))
+        p = self.prm
+        if p["precision"] not in ("fp64", "mixed"):
+            raise ValueError(f"precision must be fp64|mixed, got '{p['precision']}'")
+        if int(p["block_value"]) < 0:
+            raise ValueError("block_value must be >= 0")
+        if int(p["ncycle"]) < 1 or int(p["pre_cycles"]) < 1:
+            raise ValueError("ncycle and pre_cycles must be >= 1")
+        if int(p["npre"]) < 0 or int(p["npost"]) < 0:
+            raise ValueError("npre/npost must be >= 0")
+        if int(p["max_levels"]) < 1:
+            raise ValueError("max_levels must be >= 1")
         self.levels = []
         self.coarse_solve = None
         self._coarsening = None

@@ -92,6 +92,9 @@ class ILU0:
             self._t0 = backend.vector(self.n)
             self._t1 = backend.vector(self.n)
             self._t2 = backend.vector(self.n)
+            if str(p["solve"]) not in ("jacobi", "exact"):
+                raise ValueError(f"ilu solve must be jacobi|exact, "
+                                 f"got '{p['solve']}'")
             self._exact = str(p["solve"]) == "exact"
             if self._exact:
                 import torch

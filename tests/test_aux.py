@@ -127,3 +127,19 @@ def test_distributed_example_runs():
          os.path.join(root, "examples", "distributed_solver.py"), "24"],
         capture_output=True, text=True, timeout=420, cwd=root)
     assert out.returncode == 0, out.stdout + out.stderr
+
+
+@pytest.mark.parametrize("prm,msg", [
+    ({"precond": {"class": "amg", "precision": "bogus"}}, "precision"),
+    ({"precond": {"class": "amg", "block_value": -3}}, "block_value"),
+    ({"precond": {"class": "amg", "ncycle": 0}}, "ncycle"),
+    ({"precond": {"class": "amg", "npre": -1}}, "npre"),
+    ({"precond": {"class": "amg", "max_levels": 0}}, "max_levels"),
+    ({"precond": {"class": "amg", "direct_solver": "nope"}}, "direct_solver"),
+])
+def test_degenerate_params_rejected(prm, msg):
+    """Degenerate parameter values fail loudly at construction instead of
+    producing a silently misconfigured hierarchy."""
+    A, b = am.poisson3d(8)
+    with pytest.raises(ValueError, match=msg):
+        am.make_solver(A, prm)
