@@ -22,6 +22,8 @@ class Chebyshev:
 
         p = merge_params(self.defaults(), prm)
         self.degree = int(p["degree"])
+        if self.degree < 1:
+            raise ValueError("chebyshev degree must be >= 1")
         self.backend = backend
         self.scale = bool(p["scale"])
 

@@ -20,6 +20,8 @@ class GMRES(SolverBase):
         self._init_common(n, prm, backend, inner_product)
         b = self.backend
         m = int(self.prm["M"])
+        if m < 1:
+            raise ValueError("gmres restart M must be >= 1")
         self.m = m
         self.r = b.vector(n)
         self.w = b.vector(n)

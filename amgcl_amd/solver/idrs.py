@@ -18,6 +18,8 @@ class IDRS(SolverBase):
         self._init_common(n, prm, backend, inner_product)
         b = self.backend
         s = int(self.prm["s"])
+        if s < 1:
+            raise ValueError("idrs shadow-space size s must be >= 1")
         self.s = s
         rng = np.random.default_rng(71)
         # random shadow space, orthonormalized with backend ops (modified

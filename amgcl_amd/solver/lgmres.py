@@ -21,6 +21,8 @@ class LGMRES(SolverBase):
         self._init_common(n, prm, backend, inner_product)
         b = self.backend
         m = int(self.prm["M"])
+        if m < 1:
+            raise ValueError("gmres restart M must be >= 1")
         k = int(self.prm["K"])
         self.m, self.k = m, k
         self.r = b.vector(n)

@@ -143,3 +143,20 @@ def test_degenerate_params_rejected(prm, msg):
     A, b = am.poisson3d(8)
     with pytest.raises(ValueError, match=msg):
         am.make_solver(A, prm)
+
+
+@pytest.mark.parametrize("prm,msg", [
+    ({"solver": {"type": "gmres", "M": 0}}, "restart M"),
+    ({"solver": {"type": "fgmres", "M": 0}}, "restart M"),
+    ({"solver": {"type": "lgmres", "M": 0}}, "restart M"),
+    ({"solver": {"type": "idrs", "s": 0}}, "shadow-space"),
+    ({"precond": {"class": "amg", "coarse_enough": 100,
+                  "relax": {"type": "chebyshev", "degree": 0}}},
+     "degree"),  # coarse_enough < n so a smoothed level actually exists
+])
+def test_degenerate_solver_params_rejected(prm, msg):
+    """gmres(M=0) previously looped forever (zero Krylov steps per restart);
+    these all fail loudly now."""
+    A, b = am.poisson3d(8)
+    with pytest.raises(ValueError, match=msg):
+        am.make_solver(A, prm)
