@@ -39,6 +39,8 @@ class CPR:
         self.backend = backend
         p = merge_params(self.defaults(), prm, opaque=("pprecond", "sprecond"))
         b = int(p["block_size"])
+        if b < 1:
+            raise ValueError("cpr block_size must be >= 1")
         self.bsize = b
 
         if not isinstance(A, CSR):

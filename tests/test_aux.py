@@ -160,3 +160,19 @@ def test_degenerate_solver_params_rejected(prm, msg):
     A, b = am.poisson3d(8)
     with pytest.raises(ValueError, match=msg):
         am.make_solver(A, prm)
+
+
+def test_truncated_mm_rejected(tmp_path):
+    p = str(tmp_path / "t.mtx")
+    with open(p, "w") as f:
+        f.write("%%MatrixMarket matrix coordinate real general\n3 3 5\n1 1 2.0\n")
+    from amgcl_amd import io
+
+    with pytest.raises(ValueError, match="truncated"):
+        io.mm_read(p)
+
+
+def test_cpr_zero_block_size_rejected():
+    A, _ = am.poisson3d(8)
+    with pytest.raises(ValueError, match="block_size"):
+        am.make_solver(A, {"precond": {"class": "cpr", "block_size": 0}})
