@@ -32,6 +32,10 @@ class MakeSolver:
         for key in prm:
             if key not in ("precond", "solver"):
                 raise ValueError(f"unknown parameter '{key}'")
+        nr = getattr(A, "nrows", None)
+        nc = getattr(A, "ncols", None)
+        if nr is not None and nc is not None and nr != nc:
+            raise ValueError(f"system matrix must be square, got {nr}x{nc}")
         self.A_host = A
         self.P = make_preconditioner(A, prm.get("precond"), backend)
         self.S = make_solver_component(

@@ -176,3 +176,13 @@ def test_cpr_zero_block_size_rejected():
     A, _ = am.poisson3d(8)
     with pytest.raises(ValueError, match="block_size"):
         am.make_solver(A, {"precond": {"class": "cpr", "block_size": 0}})
+
+
+def test_non_square_rejected():
+    import scipy.sparse as sp
+
+    from amgcl_amd.matrix import CSR
+
+    ns = sp.random(6, 4, density=0.5, format="csr")
+    with pytest.raises(ValueError, match="square"):
+        am.make_solver(CSR.from_scipy(ns), {})
