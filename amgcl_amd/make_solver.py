@@ -104,7 +104,14 @@ def make_solver(A, prm=None, backend="cpu", **backend_kwargs):
                 n = len(ptr) - 1
             A = CSR(n, n, ptr, col, val)
         else:
-            A = CSR.from_scipy(A)
+            import numpy as _np
+
+            if isinstance(A, _np.ndarray):
+                import scipy.sparse as _sp
+
+                A = CSR.from_scipy(_sp.csr_matrix(_np.atleast_2d(A)))
+            else:
+                A = CSR.from_scipy(A)
     if isinstance(backend, str):
         from .backend import make_backend
 
