@@ -117,10 +117,11 @@ def make_solver(A, prm=None, backend="cpu", **backend_kwargs):
 
         if (backend in ("cpu", "hip") and isinstance(A, CSR) and A.is_complex
                 and "dtype" not in backend_kwargs):
-            # native complex solve (parity: amgcl/value_type/complex.hpp;
-            # the reference instantiates the same templates over
-            # std::complex).  The HIP backend keeps the 2x2-real adapter
-            # route (adapter.complex_to_real).
+            # native complex solve on BOTH backends (parity:
+            # amgcl/value_type/complex.hpp — the reference instantiates the
+            # same templates over std::complex; HIP runs hand-written
+            # double2 kernels).  adapter.complex_to_real remains as the
+            # 2x2-real alternative route.
             if backend == "hip":
                 import torch
 
