@@ -51,6 +51,9 @@ class MakeSolver:
     def __call__(self, rhs, x=None, A=None):
         """Solve A x = rhs. Returns (x, iters, relative residual)."""
         b = self.backend
+        if len(rhs) != self.A_host.nrows:
+            raise ValueError(f"rhs has {len(rhs)} entries, system has "
+                             f"{self.A_host.nrows} rows")
         rhs_dev = rhs if not isinstance(rhs, np.ndarray) or b.name == "cpu" else b.from_host(rhs)
         if x is None:
             x_dev = b.vector(self.A_host.nrows)

@@ -186,3 +186,11 @@ def test_non_square_rejected():
     ns = sp.random(6, 4, density=0.5, format="csr")
     with pytest.raises(ValueError, match="square"):
         am.make_solver(CSR.from_scipy(ns), {})
+
+
+def test_wrong_size_rhs_rejected():
+    """A mismatched rhs previously returned (iters=0, resid=nan) silently."""
+    A, _ = am.poisson3d(8)
+    s = am.make_solver(A, {})
+    with pytest.raises(ValueError, match="rhs"):
+        s(np.ones(100))
