@@ -56,9 +56,13 @@ def read_crs(path, row_beg=None, row_end=None):
             return _read_crs_reference(f, row_beg, row_end)
         nrows, ncols, nnz = np.fromfile(f, dtype=np.int64, count=3)
         ptr = np.fromfile(f, dtype=np.int64, count=nrows + 1)
+        if len(ptr) != nrows + 1:
+            raise ValueError(f"truncated binary matrix file '{path}'")
         if row_beg is None:
             col = np.fromfile(f, dtype=np.int32, count=nnz)
             val = np.fromfile(f, dtype=np.float64, count=nnz)
+            if len(col) != nnz or len(val) != nnz:
+                raise ValueError(f"truncated binary matrix file '{path}'")
             return CSR(nrows, ncols, ptr, col, val)
         # strip read
         lo, hi = int(ptr[row_beg]), int(ptr[row_end])

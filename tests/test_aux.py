@@ -194,3 +194,16 @@ def test_wrong_size_rhs_rejected():
     s = am.make_solver(A, {})
     with pytest.raises(ValueError, match="rhs"):
         s(np.ones(100))
+
+
+def test_truncated_binary_rejected(tmp_path):
+    from amgcl_amd import io
+
+    A, _ = am.poisson3d(6)
+    p = str(tmp_path / "a.bin")
+    io.write_crs(p, A)
+    data = open(p, "rb").read()
+    q = str(tmp_path / "trunc.bin")
+    open(q, "wb").write(data[: len(data) // 2])
+    with pytest.raises(ValueError, match="truncated"):
+        io.read_crs(q)
