@@ -61,6 +61,9 @@ class DistSolver:
 
     def __call__(self, rhs, x=None):
         b = self.backend
+        if len(rhs) != self.A.n_loc:
+            raise ValueError(f"rhs has {len(rhs)} entries, this rank's strip "
+                             f"has {self.A.n_loc} rows")
         if isinstance(rhs, np.ndarray) and b.base.name != "cpu":
             rhs = b.from_host(rhs)
         if x is None:
