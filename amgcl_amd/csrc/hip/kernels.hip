@@ -519,7 +519,7 @@ __global__ void dot2_k(int64_t n, const T *__restrict__ x1, const T *__restrict_
 extern "C" int amg_dot_f64(int64_t n, const double *x, const double *y, double *out,
                            hipStream_t stream) {
     fill_k<double><<<1, 64, 0, stream>>>(1, 0.0, out);
-    dot_k<double><<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x, y, out);
+    dot_k<double><<<nblocks(n, 256, 2048), 256, 0, stream>>>(n, x, y, out);
     return (int)hipGetLastError();
 }
 
@@ -527,7 +527,7 @@ extern "C" int amg_dot2_f64(int64_t n, const double *x1, const double *y1,
                             const double *x2, const double *y2, double *out,
                             hipStream_t stream) {
     fill_k<double><<<1, 64, 0, stream>>>(2, 0.0, out);
-    dot2_k<double><<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x1, y1, x2, y2, out);
+    dot2_k<double><<<nblocks(n, 256, 2048), 256, 0, stream>>>(n, x1, y1, x2, y2, out);
     return (int)hipGetLastError();
 }
 
@@ -676,7 +676,7 @@ extern "C" int amg_fill_f32(int64_t n, double v, float *x, hipStream_t stream) {
 extern "C" int amg_dot_f32(int64_t n, const float *x, const float *y, double *out,
                            hipStream_t stream) {
     fill_k<double><<<1, 64, 0, stream>>>(1, 0.0, out);
-    dot_k<float><<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, x, y, out);
+    dot_k<float><<<nblocks(n, 256, 2048), 256, 0, stream>>>(n, x, y, out);
     return (int)hipGetLastError();
 }
 extern "C" int amg_gather_f32(int64_t n, const float *x, const int *idx, float *buf,
@@ -776,7 +776,7 @@ __global__ void cg_tail_k(int64_t n, double alpha, const double *__restrict__ p,
 extern "C" int amg_cg_tail_f64(int64_t n, double alpha, const double *p, const double *q,
                                double *x, double *r, double *out, hipStream_t stream) {
     fill_k<double><<<1, 64, 0, stream>>>(1, 0.0, out);
-    cg_tail_k<<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, alpha, p, q, x, r, out);
+    cg_tail_k<<<nblocks(n, 256, 2048), 256, 0, stream>>>(n, alpha, p, q, x, r, out);
     return (int)hipGetLastError();
 }
 
@@ -946,7 +946,7 @@ __global__ void cdot_k(int64_t n, const c128 *__restrict__ x,
 
 extern "C" int amg_dot_c128(int64_t n, const void *x, const void *y, double *out,
                             hipStream_t stream) {
-    cdot_k<<<nblocks(n, 256, 1024), 256, 0, stream>>>(n, (const c128 *)x,
+    cdot_k<<<nblocks(n, 256, 2048), 256, 0, stream>>>(n, (const c128 *)x,
                                                       (const c128 *)y, out);
     return (int)hipGetLastError();
 }
