@@ -1607,6 +1607,142 @@ static void gauss_seidel(i64 nrows, arr<i32> ptr, arr<i32> col, arr<double> val,
     else         for (i64 i = nrows - 1; i >= 0; --i) sweep_row(i);
 }
 
+// Block-valued ILU(0): IKJ factorization over bxb blocks (parity:
+// amgcl/relaxation/ilu0.hpp:51 instantiated over static_matrix<double,B,B>
+// via value_type/static_matrix.hpp — the reference's block-valued route,
+// tutorial CoupCons3D).  Input is BSR (ptr/col over block rows, val
+// flattened bxb row-major per block, sorted rows).  Diagonal blocks are
+// stored INVERTED (Gauss-Jordan with partial pivoting, the reference's
+// detail/inverse.hpp:45) so the sweeps multiply instead of solving.
+// Returns (lu_blocks, dia_idx).
+static void gj_invert(double *m, double *inv, int b) {
+    // inv = m^-1 by Gauss-Jordan with partial pivoting (b <= 8)
+    double a[64 * 2];
+    for (int r = 0; r < b; ++r) {
+        for (int c = 0; c < b; ++c) {
+            a[r * 2 * b + c] = m[r * b + c];
+            a[r * 2 * b + b + c] = (r == c) ? 1.0 : 0.0;
+        }
+    }
+    for (int k = 0; k < b; ++k) {
+        int p = k;
+        for (int r = k + 1; r < b; ++r)
+            if (std::fabs(a[r * 2 * b + k]) > std::fabs(a[p * 2 * b + k])) p = r;
+        if (p != k)
+            for (int c = 0; c < 2 * b; ++c) std::swap(a[k * 2 * b + c], a[p * 2 * b + c]);
+        double d = a[k * 2 * b + k];
+        if (d == 0.0) throw std::runtime_error("block_ilu0: singular diagonal block");
+        for (int c = 0; c < 2 * b; ++c) a[k * 2 * b + c] /= d;
+        for (int r = 0; r < b; ++r) {
+            if (r == k) continue;
+            double f = a[r * 2 * b + k];
+            if (f == 0.0) continue;
+            for (int c = 0; c < 2 * b; ++c) a[r * 2 * b + c] -= f * a[k * 2 * b + c];
+        }
+    }
+    for (int r = 0; r < b; ++r)
+        for (int c = 0; c < b; ++c) inv[r * b + c] = a[r * 2 * b + b + c];
+}
+
+static py::tuple block_ilu0_factor(i64 nb, i64 bs, arr<i32> ptr_a, arr<i32> col_a,
+                                   arr<double> val_a) {
+    const int b = (int)bs;
+    if (b < 1 || b > 8) throw std::runtime_error("block_ilu0: 1 <= b <= 8");
+    const i32 *ptr = ptr_a.data();
+    const i32 *col = col_a.data();
+    const i64 bb = (i64)b * b;
+    const i64 nnzb = (i64)col_a.size();
+    arr<double> lu_a(nnzb * bb);
+    double *LU = lu_a.mutable_data();
+    std::memcpy(LU, val_a.data(), (size_t)nnzb * bb * sizeof(double));
+    arr<i32> dia_a(nb);
+    i32 *dia = dia_a.mutable_data();
+    for (i64 i = 0; i < nb; ++i) {
+        dia[i] = -1;
+        for (i32 j = ptr[i]; j < ptr[i + 1]; ++j)
+            if (col[j] == (i32)i) { dia[i] = j; break; }
+        if (dia[i] < 0) throw std::runtime_error("block_ilu0: missing diagonal block");
+    }
+    std::vector<i32> work(nb, -1);
+    double lik[64], t[64];
+    for (i64 i = 0; i < nb; ++i) {
+        i32 rb = ptr[i], re = ptr[i + 1];
+        for (i32 j = rb; j < re; ++j) work[col[j]] = j;
+        for (i32 j = rb; j < re && col[j] < (i32)i; ++j) {
+            i32 k = col[j];
+            // lik = LU[j] * Dinv[k] (diagonal of row k already inverted)
+            const double *A_ = LU + (i64)j * bb, *B_ = LU + (i64)dia[k] * bb;
+            for (int r = 0; r < b; ++r)
+                for (int c = 0; c < b; ++c) {
+                    double s = 0.0;
+                    for (int q = 0; q < b; ++q) s += A_[r * b + q] * B_[q * b + c];
+                    lik[r * b + c] = s;
+                }
+            std::memcpy(LU + (i64)j * bb, lik, (size_t)bb * sizeof(double));
+            for (i32 jk = dia[k] + 1; jk < ptr[k + 1]; ++jk) {
+                i32 w = work[col[jk]];
+                if (w < 0) continue;
+                const double *U_ = LU + (i64)jk * bb;
+                double *W_ = LU + (i64)w * bb;
+                for (int r = 0; r < b; ++r)
+                    for (int c = 0; c < b; ++c) {
+                        double s = 0.0;
+                        for (int q = 0; q < b; ++q) s += lik[r * b + q] * U_[q * b + c];
+                        W_[r * b + c] -= s;
+                    }
+            }
+        }
+        gj_invert(LU + (i64)dia[i] * bb, t, b);
+        std::memcpy(LU + (i64)dia[i] * bb, t, (size_t)bb * sizeof(double));
+        for (i32 j = rb; j < re; ++j) work[col[j]] = -1;
+    }
+    return py::make_tuple(lu_a, dia_a);
+}
+
+// Serial block forward/backward sweeps over the combined factor
+// (z := M^-1 z; unit block diagonal in L, inverted diagonal blocks in U).
+static void block_ilu0_solve(i64 nb, i64 bs, arr<i32> ptr_a, arr<i32> col_a,
+                             arr<double> lu_a, arr<i32> dia_a,
+                             py::array_t<double> z_a) {
+    const int b = (int)bs;
+    const i64 bb = (i64)b * b;
+    const i32 *ptr = ptr_a.data(), *col = col_a.data(), *dia = dia_a.data();
+    const double *LU = lu_a.data();
+    double *Z = z_a.mutable_data();
+    double acc[8];
+    for (i64 i = 0; i < nb; ++i) {  // forward: z_i -= sum L_ik z_k
+        double *zi = Z + i * b;
+        for (i32 j = ptr[i]; j < dia[i]; ++j) {
+            const double *M = LU + (i64)j * bb;
+            const double *zk = Z + (i64)col[j] * b;
+            for (int r = 0; r < b; ++r) {
+                double s = 0.0;
+                for (int c = 0; c < b; ++c) s += M[r * b + c] * zk[c];
+                zi[r] -= s;
+            }
+        }
+    }
+    for (i64 i = nb - 1; i >= 0; --i) {  // backward: z_i = Dinv (z_i - sum U z_k)
+        double *zi = Z + i * b;
+        for (i32 j = dia[i] + 1; j < ptr[i + 1]; ++j) {
+            const double *M = LU + (i64)j * bb;
+            const double *zk = Z + (i64)col[j] * b;
+            for (int r = 0; r < b; ++r) {
+                double s = 0.0;
+                for (int c = 0; c < b; ++c) s += M[r * b + c] * zk[c];
+                zi[r] -= s;
+            }
+        }
+        const double *D = LU + (i64)dia[i] * bb;
+        for (int r = 0; r < b; ++r) {
+            double s = 0.0;
+            for (int c = 0; c < b; ++c) s += D[r * b + c] * zi[c];
+            acc[r] = s;
+        }
+        for (int r = 0; r < b; ++r) zi[r] = acc[r];
+    }
+}
+
 // ILU(0) factorization, in-place IKJ (parity: amgcl/relaxation/ilu0.hpp:51).
 // Requires sorted rows. Returns (luval, dia_idx) where dia_idx[i] points at
 // the diagonal entry of row i inside the CSR arrays; U's diagonal is stored
@@ -2279,6 +2415,8 @@ PYBIND11_MODULE(_core, m) {
     m.def("residual", &residual);
     m.def("gauss_seidel", &gauss_seidel);
     m.def("ilu0_factor", &ilu0_factor);
+    m.def("block_ilu0_factor", &block_ilu0_factor);
+    m.def("block_ilu0_solve", &block_ilu0_solve);
     m.def("iluk_factor", &iluk_factor);
     m.def("ilut_factor", &ilut_factor);
     m.def("ilu0_chow_patel", &ilu0_chow_patel);

@@ -11,6 +11,7 @@ from .damped_jacobi import DampedJacobi
 from .chebyshev import Chebyshev
 from .gauss_seidel import GaussSeidel
 from .ilu0 import ILU0, ILU0ChowPatel, ILUK, ILUP, ILUT
+from .block_ilu0 import BlockILU0
 from .as_block import AsBlock
 
 REGISTRY = {
@@ -20,6 +21,7 @@ REGISTRY = {
     "chebyshev": Chebyshev,
     "gauss_seidel": GaussSeidel,
     "ilu0": ILU0,
+    "block_ilu0": BlockILU0,
     "iluk": ILUK,
     "ilup": ILUP,
     "ilut": ILUT,
