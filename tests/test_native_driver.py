@@ -248,3 +248,15 @@ def test_block_ilu0_on_hip():
     s2 = am.make_solver(A, prm)  # CPU twin: exact sweeps vs iterated-Jacobi
     x2, it2, r2 = s2(b)
     assert iters <= it2 + 10, (iters, it2)
+
+    # the full CoupCons3D-class configuration: BSR level storage AND the
+    # block-valued ILU smoother together (generic path)
+    prm3 = {"precond": {"class": "amg", "coarse_enough": 500,
+                        "block_value": 3,
+                        "relax": {"type": "block_ilu0", "block_size": 3}},
+            "solver": {"type": "cg", "tol": 1e-8, "maxiter": 200}}
+    s3 = am.make_solver(A, prm3, backend="hip")
+    x3, it3, r3 = s3(b)
+    assert r3 < 1e-8
+    xh3 = s3.backend.to_host(x3)
+    assert np.linalg.norm(b - A.to_scipy() @ xh3) / np.linalg.norm(b) < 1e-7
